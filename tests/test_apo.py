@@ -1,0 +1,272 @@
+"""APO service: patterns, suggestions, rule packing, segment lifecycle, beam."""
+
+import json
+
+import pytest
+
+from senweaver_amd.apo import (
+    APOService,
+    BeamSearchEngine,
+    LocalGradientEngine,
+    StubBackend,
+    inject_rules,
+    pack_rules,
+)
+from senweaver_amd.storage import MemoryStorage, APO_SEGMENTS_KEY, APO_BEAM_STATE_KEY
+from senweaver_amd.trace import TraceCollector
+
+
+def make_services(fixed_clock, seq_uuid, optimizer=None):
+    storage = MemoryStorage()
+    tc = TraceCollector(storage=storage, clock=fixed_clock, uuid_fn=seq_uuid)
+    apo = APOService(tc, storage=storage, clock=fixed_clock, uuid_fn=seq_uuid, optimizer=optimizer)
+    return tc, apo, storage
+
+
+def seed_bad_traces(tc, n_bad=5, n_good=2, errors=True, tool_fail=True, tokens=12000,
+                    llm_calls=3, user_msgs=4, tool_dur=20000):
+    for i in range(n_bad):
+        th = f"bad{i}"
+        tid = tc.start_trace(th, {"chatMode": "normal"})
+        for m in range(user_msgs):
+            tc.record_user_message(th, m * 2, f"question {m}")
+            tc.record_assistant_message(th, m * 2 + 1, f"answer {m}")
+        for _ in range(llm_calls):
+            tc.record_llm_call(th, 1, input_tokens=tokens // llm_calls, output_tokens=0)
+        if tool_fail:
+            tc.record_tool_call(th, 1, tool_name="run_command", tool_success=False,
+                                tool_result="exit 1", duration=tool_dur)
+        if errors:
+            tc.record_error(th, 1, "boom")
+        tc.end_trace(tid)
+        tc.record_user_feedback(th, 1, "bad")
+    for i in range(n_good):
+        th = f"good{i}"
+        tid = tc.start_trace(th, {"chatMode": "normal"})
+        tc.record_user_message(th, 0, "hi")
+        tc.record_assistant_message(th, 1, "done")
+        tc.end_trace(tid)
+        tc.record_user_feedback(th, 1, "good")
+
+
+def test_six_patterns_all_fire(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    seed_bad_traces(tc, n_bad=5)
+    report = apo.analyze_prompt_effectiveness()
+    descs = [p.description for p in report.patterns]
+    assert "Users give negative feedback after errors occur in conversations" in descs
+    assert "Tool call failures lead to user dissatisfaction" in descs
+    assert "User feedback is poor in conversations with high token consumption" in descs
+    assert "Users still dissatisfied after multiple LLM calls (possible retries)" in descs
+    assert "Long conversations with many turns still result in user dissatisfaction" in descs
+    assert "Slow tool execution (>15s total) correlates with user dissatisfaction" in descs
+    # severity rules: 5 error traces -> high; 5 tool-fail -> high; token always medium
+    by_desc = {p.description: p for p in report.patterns}
+    assert by_desc["Users give negative feedback after errors occur in conversations"].severity == "high"
+    assert by_desc["User feedback is poor in conversations with high token consumption"].severity == "medium"
+    assert by_desc["Users still dissatisfied after multiple LLM calls (possible retries)"].severity == "high"
+    # examples capped at 3
+    for p in report.patterns:
+        assert len(p.examples) <= 3
+
+
+def test_patterns_below_threshold_do_not_fire(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    seed_bad_traces(tc, n_bad=1)  # every pattern needs >=2 (tokens >=3)
+    report = apo.analyze_prompt_effectiveness()
+    trace_pattern_descs = [p.description for p in report.patterns if "dimension reward" not in p.description]
+    assert trace_pattern_descs == []
+
+
+def test_good_rate_and_by_mode(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    seed_bad_traces(tc, n_bad=6, n_good=2)
+    report = apo.analyze_prompt_effectiveness()
+    assert report.good_feedback_count == 2
+    assert report.bad_feedback_count == 6
+    assert report.good_rate == pytest.approx(2 / 8)
+    assert report.by_mode["normal"]["total"] == 8
+    # low good rate triggers the core_behavior suggestion
+    assert any("Overall approval rate is only" in s.description for s in report.suggestions)
+
+
+def test_rule_packing_budget():
+    rules = ["a" * 900, "b" * 900, "c" * 900]
+    content, included = pack_rules(rules)
+    assert included == 2
+    assert content == "a" * 900 + "\n" + "b" * 900
+    msg = inject_rules("SYSTEM", rules)
+    assert "# APO Optimized Rules (2/3 rules, budget limited)" in msg
+    assert msg.startswith("SYSTEM\n\n# APO Optimized Rules")
+    # all fit -> no truncation note
+    msg2 = inject_rules("SYSTEM", ["short rule", "another"])
+    assert "# APO Optimized Rules\n" in msg2
+    assert "budget limited" not in msg2
+    # greedy stop: a rule that doesn't fit stops packing even if later ones would
+    content3, inc3 = pack_rules(["a" * 1500, "b" * 1000, "c" * 10])
+    assert inc3 == 1 and content3 == "a" * 1500
+    # empty rules -> unchanged
+    assert inject_rules("SYSTEM", []) == "SYSTEM"
+
+
+def test_rule_packing_boundary_exact():
+    # exactly 2000 chars fits
+    content, inc = pack_rules(["x" * 2000])
+    assert inc == 1
+    content, inc = pack_rules(["x" * 2001])
+    assert inc == 0 and content == ""
+
+
+def test_suggestion_lifecycle_apply_revert(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    seg = apo.add_segment("core_behavior", "original rule")
+    from senweaver_amd.apo import PromptOptimizationSuggestion
+    sug = PromptOptimizationSuggestion(
+        id="sug-1", target_category="core_behavior", type="modify", priority="high",
+        description="d", reasoning="r", estimated_impact="e",
+        suggested_content="improved rule", target_segment_id=seg.id,
+    )
+    apo._suggestions.append(sug)
+    apo.apply_suggestion("sug-1")
+    assert seg.content == "improved rule"
+    assert seg.original_content == "original rule"
+    assert seg.is_optimized and seg.version == 2
+    assert apo.get_optimized_rules() == ["improved rule"]
+    apo.revert_suggestion("sug-1")
+    assert seg.content == "original rule"
+    assert seg.original_content is None
+    assert not seg.is_optimized and seg.version == 3
+    assert sug.status == "reverted"
+
+
+def test_suggestion_add_and_revert_removes(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    from senweaver_amd.apo import PromptOptimizationSuggestion
+    sug = PromptOptimizationSuggestion(
+        id="sug-2", target_category="tool_usage", type="add", priority="medium",
+        description="d", reasoning="r", estimated_impact="e", suggested_content="new tool rule",
+    )
+    apo._suggestions.append(sug)
+    apo.apply_suggestion("sug-2")
+    assert apo.get_optimized_rules() == ["new tool rule"]
+    apo.revert_suggestion("sug-2")
+    assert apo.get_optimized_rules() == []
+
+
+def test_reject_suggestion(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    from senweaver_amd.apo import PromptOptimizationSuggestion
+    sug = PromptOptimizationSuggestion(
+        id="sug-3", target_category="core_behavior", type="modify", priority="low",
+        description="d", reasoning="r", estimated_impact="e",
+    )
+    apo._suggestions.append(sug)
+    apo.reject_suggestion("sug-3")
+    assert sug.status == "rejected"
+    apo.apply_suggestion("sug-3")  # no-op on non-pending
+    assert sug.status == "rejected"
+
+
+def test_textual_gradient_local(fixed_clock, seq_uuid):
+    backend = StubBackend()
+    engine = LocalGradientEngine(backend)
+    tc, apo, _ = make_services(fixed_clock, seq_uuid, optimizer=engine)
+    seed_bad_traces(tc, n_bad=4)
+    tg = apo.request_textual_gradient()
+    assert tg is not None
+    assert tg.prompt_version == "v0"
+    assert "Based on 4 rollouts" in tg.rollout_summary
+    # gradient prompt fed to backend contains the reference template markers
+    gp = backend.generate_calls[0]
+    assert "You are an expert prompt engineer" in gp
+    assert "--- Experiment 1 ---" in gp
+    assert "Less than 350 words." in gp
+    # apply-edit decode produced a pending suggestion
+    pending = apo.get_pending_suggestions()
+    assert any(s.description.startswith("Textual Gradient:") for s in pending)
+
+
+def test_gradient_requires_two_rollouts(fixed_clock, seq_uuid):
+    backend = StubBackend()
+    tc, apo, _ = make_services(fixed_clock, seq_uuid, optimizer=LocalGradientEngine(backend))
+    seed_bad_traces(tc, n_bad=1, n_good=0)
+    assert apo.request_textual_gradient() is None
+
+
+def test_beam_search_rounds_and_topk(fixed_clock, seq_uuid):
+    backend = StubBackend()
+    tc, apo, storage = make_services(fixed_clock, seq_uuid)
+    seed_bad_traces(tc, n_bad=4)
+    engine = BeamSearchEngine(backend)
+    state = engine.run_search(apo, rounds=3)
+    cfg = apo.get_config()
+    assert state.current_round == 3
+    assert len(state.beam) == cfg["beamWidth"]
+    assert state.history_best_prompt is not None
+    assert state.history_best_score == max(b.score for b in state.beam)
+    # beam is sorted best-first and scores are deterministic under the stub
+    scores = [b.score for b in state.beam]
+    assert scores == sorted(scores, reverse=True)
+    # best prompt rules were applied as segments
+    assert len(apo.get_optimized_rules()) > 0
+    # beam state persisted under the reference storage key
+    apo.flush()
+    raw = json.loads(storage.get(APO_BEAM_STATE_KEY))
+    assert raw["currentRound"] == 3
+    assert len(raw["beam"]) == cfg["beamWidth"]
+    assert raw["historyBestPrompt"]["version"].startswith("v")
+
+
+def test_beam_state_resume(fixed_clock, seq_uuid):
+    backend = StubBackend()
+    tc, apo, storage = make_services(fixed_clock, seq_uuid)
+    seed_bad_traces(tc, n_bad=4)
+    engine = BeamSearchEngine(backend)
+    engine.run_round(apo)
+    apo.flush()
+    # Simulate crash + restart: new service over the same storage
+    apo2 = APOService(tc, storage=storage, clock=fixed_clock, uuid_fn=seq_uuid)
+    st = apo2.get_beam_state()
+    assert st is not None and st.current_round == 1
+    # continue the search from restored state
+    engine.run_round(apo2)
+    assert apo2.get_beam_state().current_round == 2
+
+
+def test_segments_persist_roundtrip(fixed_clock, seq_uuid):
+    tc, apo, storage = make_services(fixed_clock, seq_uuid)
+    apo.add_segment("tool_usage", "verify output", is_optimized=True)
+    apo.flush()
+    raw = json.loads(storage.get(APO_SEGMENTS_KEY))
+    assert raw[0]["category"] == "tool_usage"
+    assert raw[0]["isActive"] is True and raw[0]["isOptimized"] is True
+    apo2 = APOService(tc, storage=storage, clock=fixed_clock, uuid_fn=seq_uuid)
+    assert apo2.get_optimized_rules() == ["verify output"]
+
+
+def test_auto_analyze_gates(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    assert not apo.should_auto_analyze()  # no traces
+    seed_bad_traces(tc, n_bad=12, n_good=3)  # 15 traces < 20 min
+    assert not apo.should_auto_analyze()
+    seed_bad_traces(tc, n_bad=5, n_good=2)  # >20 traces, >10 feedbacks
+    assert apo.should_auto_analyze()
+    report = apo.try_auto_analyze()
+    assert report is not None
+    # immediately after a report, the interval gate blocks
+    assert not apo.should_auto_analyze()
+
+
+def test_stats_shape(fixed_clock, seq_uuid):
+    tc, apo, _ = make_services(fixed_clock, seq_uuid)
+    seed_bad_traces(tc, n_bad=3)
+    apo.analyze_prompt_effectiveness()
+    stats = apo.get_stats()
+    for key in ("totalReports", "totalSuggestions", "appliedSuggestions", "rejectedSuggestions",
+                "activeSegments", "optimizedSegments", "lastAnalysisTime", "currentGoodRate",
+                "beamSearchActive", "beamCurrentRound", "beamBestScore", "totalTextualGradients",
+                "avgFinalReward"):
+        assert key in stats
+    assert stats["totalReports"] == 1
+    assert stats["beamSearchActive"] is False
+    assert stats["beamBestScore"] is None
