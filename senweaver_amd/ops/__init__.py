@@ -1,0 +1,141 @@
+"""Op dispatch: hand-written gfx950 HIP kernels on GPU, fp32 torch on CPU.
+
+On a GPU box the HIP extension is REQUIRED — ops raise if a CUDA(=HIP) tensor
+arrives and the extension is missing, so nothing silently falls back to eager
+PyTorch on the hardware the kernels target.  CPU tensors use the fp32
+reference implementations (tests / CPU-only engine mode).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from . import reference as ref
+
+_ext = None
+_ext_err: Optional[str] = None
+
+
+def _load_ext():
+    global _ext, _ext_err
+    if _ext is not None or _ext_err is not None:
+        return _ext
+    from . import build as _build
+
+    try:
+        _ext = _build.load_prebuilt()
+        if _ext is None:
+            _ext = _build.build()
+    except Exception as e:  # pragma: no cover
+        _ext_err = str(e)
+        _ext = None
+    return _ext
+
+
+def hip_ext():
+    """The loaded HIP extension module, or raise with the build error."""
+    ext = _load_ext()
+    if ext is None:
+        raise RuntimeError(
+            "senweaver_amd HIP extension is not available on this GPU host "
+            "(refusing silent eager fallback). Build error: " + str(_ext_err)
+        )
+    return ext
+
+
+def extension_loaded() -> bool:
+    return _load_ext() is not None
+
+
+def _on_gpu(*tensors) -> bool:
+    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+
+
+# ---------------- public ops ----------------
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if _on_gpu(x):
+        return hip_ext().rmsnorm(x.contiguous(), w.contiguous(), eps)
+    return ref.rmsnorm_ref(x, w, eps)
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor, w: torch.Tensor, eps: float = 1e-5):
+    """Returns y; updates `residual` in place to x + residual."""
+    if _on_gpu(x):
+        return hip_ext().fused_add_rmsnorm(x.contiguous(), residual, w.contiguous(), eps)
+    y, r = ref.fused_add_rmsnorm_ref(x, residual, w, eps)
+    residual.copy_(r)
+    return y
+
+
+def rope_inplace(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor, positions: torch.Tensor):
+    """q: [T,Hq,D], k: [T,Hk,D] modified in place."""
+    if _on_gpu(q):
+        hip_ext().rope_inplace(q, k, cos_sin, positions.to(torch.int32))
+        return
+    q.copy_(ref.rope_ref(q, cos_sin, positions))
+    k.copy_(ref.rope_ref(k, cos_sin, positions))
+
+
+def swiglu(gateup: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(gateup):
+        return hip_ext().swiglu(gateup.contiguous())
+    return ref.swiglu_ref(gateup)
+
+
+def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """C[M,N] = A[M,K] @ B[N,K]^T.  Pads M to the kernel tile on GPU."""
+    if not _on_gpu(a):
+        return ref.gemm_bt_ref(a, b)
+    ext = hip_ext()
+    M = a.shape[0]
+    if M <= 16 and b.shape[1] % 1024 == 0 and b.shape[0] % 64 == 0:
+        return ext.gemm_bt(a.contiguous(), b.contiguous())
+    pad = (-M) % 128
+    if pad:
+        a = torch.nn.functional.pad(a, (0, 0, 0, pad))
+    c = ext.gemm_bt(a.contiguous(), b.contiguous())
+    return c[:M] if pad else c
+
+
+def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional[float] = None,
+             vt: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Causal GQA attention. q [B,H,S,D], k/v [B,Hk,S,D] (S % 64 == 0 on GPU).
+
+    ``vt`` may be passed pre-transposed ([B,Hk,D,S]) to skip the transpose.
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        if vt is None:
+            vt = v.transpose(-1, -2).contiguous()
+        return hip_ext().attn_fwd(q.contiguous(), k.contiguous(), vt, scale)
+    return ref.attn_fwd_ref(q, k, v, scale, causal=True)
+
+
+def paged_decode_attn(q, kcache, vcache, block_table, ctx_lens, scale: Optional[float] = None):
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        return hip_ext().paged_decode_attn(
+            q.contiguous(), kcache, vcache,
+            block_table.to(torch.int32), ctx_lens.to(torch.int32), scale)
+    return ref.paged_decode_attn_ref(q, kcache, vcache, block_table, ctx_lens, scale)
+
+
+def argmax_rows(logits: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(logits):
+        return hip_ext().argmax_rows(logits.contiguous())
+    return ref.argmax_rows_ref(logits)
+
+
+def target_logprob(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(logits):
+        return hip_ext().target_logprob(logits.contiguous(), targets.to(torch.int32))
+    return ref.target_logprob_ref(logits, targets)
+
+
+rope_tables = ref.rope_tables

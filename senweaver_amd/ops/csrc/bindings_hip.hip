@@ -1,0 +1,233 @@
+// PyTorch bindings for the senweaver_amd gfx950 kernel library.
+// HIP-native throughout (c10::hip stream accessors) — no CUDA shims.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+// ---- kernel prototypes (defined in the sibling .hip translation units) ----
+extern "C" __global__ void rmsnorm_fwd_kernel(const ushort*, const ushort*, ushort*, ushort*, int, float);
+extern "C" __global__ void rope_fwd_kernel(ushort*, ushort*, const float*, const int*, int, int, int);
+extern "C" __global__ void swiglu_fwd_kernel(const ushort*, ushort*, long long, int);
+extern "C" __global__ void add_bf16_kernel(const ushort*, const ushort*, ushort*, long long);
+extern "C" __global__ void argmax_rows_kernel(const ushort*, int*, int);
+extern "C" __global__ void target_logprob_kernel(const ushort*, const int*, float*, int);
+extern "C" __global__ void gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_kernel(const ushort*, const ushort*, float*, int, int, int, int);
+extern "C" __global__ void gemv_reduce_kernel(const float*, ushort*, int, int, int);
+extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
+extern "C" __global__ void paged_decode_attn_kernel(const ushort*, const ushort*, const ushort*, ushort*,
+                                                    const int*, const int*, int, int, int, float);
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+inline const ushort* bf16_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const ushort*>(t.data_ptr());
+}
+inline ushort* bf16_mut(torch::Tensor& t) {
+  return reinterpret_cast<ushort*>(t.data_ptr());
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+}  // namespace
+
+// ---------------- RMSNorm ----------------
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const int hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden % 8 != 0");
+  const long long rows = x.numel() / hidden;
+  auto y = torch::empty_like(x);
+ hipLaunchKernelGGL(( rmsnorm_fwd_kernel), dim3(dim3((unsigned)rows)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(x), bf16_ptr(w), bf16_mut(y), nullptr, hidden, (float)eps);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+// y = rmsnorm(x + residual); residual <- x + residual (in place)
+torch::Tensor fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                                torch::Tensor w, double eps) {
+  check_bf16(x, "x");
+  check_bf16(residual, "residual");
+  check_bf16(w, "w");
+  const int hidden = x.size(-1);
+  const long long rows = x.numel() / hidden;
+  auto y = torch::empty_like(x);
+ hipLaunchKernelGGL(( rmsnorm_fwd_kernel), dim3(dim3((unsigned)rows)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(x), bf16_ptr(w), bf16_mut(y), bf16_mut(residual), hidden, (float)eps);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+// ---------------- RoPE (in place over q and k) ----------------
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
+                  torch::Tensor positions) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32 && cos_sin.is_contiguous());
+  TORCH_CHECK(positions.scalar_type() == torch::kInt32 && positions.is_contiguous());
+  const int T = q.size(0);
+  const int Hq = q.size(1), Hk = k.size(1), D = q.size(2);
+  TORCH_CHECK(k.size(0) == T && k.size(2) == D);
+  TORCH_CHECK(cos_sin.size(1) == D);
+ hipLaunchKernelGGL(( rope_fwd_kernel), dim3(dim3(T, Hq + Hk)), dim3(dim3(64)), 0, cur_stream(), 
+      bf16_mut(q), bf16_mut(k), cos_sin.data_ptr<float>(),
+      positions.data_ptr<int>(), Hq, Hk, D);
+  HIP_CHECK_KERNEL();
+}
+
+// ---------------- SwiGLU ----------------
+torch::Tensor swiglu(torch::Tensor gateup) {
+  check_bf16(gateup, "gateup");
+  const int inter2 = gateup.size(-1);
+  TORCH_CHECK(inter2 % 16 == 0);
+  const int inter = inter2 / 2;
+  const long long rows = gateup.numel() / inter2;
+  auto sizes = gateup.sizes().vec();
+  sizes.back() = inter;
+  auto y = torch::empty(sizes, gateup.options());
+  const long long total = rows * (inter / 8);
+  const int blocks = (int)std::min<long long>((total + 255) / 256, 2048);
+ hipLaunchKernelGGL(( swiglu_fwd_kernel), dim3(dim3(blocks)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(gateup), bf16_mut(y), rows, inter);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() % 8 == 0);
+  auto y = torch::empty_like(a);
+  const long long n8 = a.numel() / 8;
+  const int blocks = (int)std::min<long long>((n8 + 255) / 256, 2048);
+ hipLaunchKernelGGL(( add_bf16_kernel), dim3(dim3(blocks)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(a), bf16_ptr(b), bf16_mut(y), n8);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+// ---------------- GEMM: C[M,N] = A[M,K] @ B[N,K]^T ----------------
+torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  const int M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(B.size(1) == K, "K mismatch");
+  auto C = torch::empty({M, N}, A.options());
+  if (M <= 16) {
+    TORCH_CHECK(N % 64 == 0 && K % 1024 == 0, "gemv path needs N%64==0, K%1024==0");
+    // split-K sized so blocks ~= CU count (microarch: wall ~= one block's
+    // latency until blocks exceed CUs)
+    int splitk = 1;
+    while ((long long)(N / 64) * splitk * 2 <= 256 && (K / (splitk * 2)) % 256 == 0 && splitk < 8)
+      splitk *= 2;
+    auto part = torch::empty({splitk, M, N}, A.options().dtype(torch::kFloat32));
+   hipLaunchKernelGGL(( gemv_bt_bf16_kernel), dim3(dim3(N / 64, splitk)), dim3(dim3(256)), 0, cur_stream(), 
+        bf16_ptr(A), bf16_ptr(B), part.data_ptr<float>(), M, N, K, splitk);
+    HIP_CHECK_KERNEL();
+    const long long total = (long long)M * N;
+    const int blocks = (int)std::min<long long>((total + 255) / 256, 2048);
+   hipLaunchKernelGGL(( gemv_reduce_kernel), dim3(dim3(blocks)), dim3(dim3(256)), 0, cur_stream(), 
+        part.data_ptr<float>(), bf16_mut(C), M, N, splitk);
+    HIP_CHECK_KERNEL();
+    return C;
+  }
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
+              "gemm_bt needs M,N % 128 == 0 and K % 64 == 0 (pad host-side); got ",
+              M, "x", N, "x", K);
+  const int nwg = (M / 128) * (N / 128);
+ hipLaunchKernelGGL(( gemm_bt_bf16_kernel), dim3(dim3(nwg)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+  HIP_CHECK_KERNEL();
+  return C;
+}
+
+// ---------------- Flash attention prefill ----------------
+torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
+                       double scale) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(vt, "vt");
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hk = k.size(1);
+  TORCH_CHECK(D == 128, "attn_fwd supports D=128");
+  TORCH_CHECK(S % 64 == 0, "S must be padded to 64");
+  TORCH_CHECK(vt.size(1) == Hk && vt.size(2) == D && vt.size(3) == S, "vt must be [B,Hk,D,S]");
+  TORCH_CHECK(H % Hk == 0);
+  auto o = torch::empty_like(q);
+ hipLaunchKernelGGL(( attn_fwd_bf16_kernel), dim3(dim3(S / 64, H, B)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(q), bf16_ptr(k), bf16_ptr(vt), bf16_mut(o), B, H, Hk, S, (float)scale);
+  HIP_CHECK_KERNEL();
+  return o;
+}
+
+// ---------------- Paged decode attention ----------------
+torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor kcache,
+                                torch::Tensor vcache, torch::Tensor block_table,
+                                torch::Tensor ctx_lens, double scale) {
+  check_bf16(q, "q");
+  check_bf16(kcache, "kcache");
+  check_bf16(vcache, "vcache");
+  TORCH_CHECK(block_table.scalar_type() == torch::kInt32 && block_table.is_contiguous());
+  TORCH_CHECK(ctx_lens.scalar_type() == torch::kInt32 && ctx_lens.is_contiguous());
+  const int B = q.size(0), H = q.size(1), D = q.size(2);
+  const int Hk = kcache.size(2);
+  TORCH_CHECK(D == 128 && kcache.size(3) == D);
+  TORCH_CHECK(kcache.size(1) == 16, "PAGE_SIZE=16");
+  const int max_pages = block_table.size(1);
+  auto o = torch::empty_like(q);
+ hipLaunchKernelGGL(( paged_decode_attn_kernel), dim3(dim3(H, B)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(q), bf16_ptr(kcache), bf16_ptr(vcache), bf16_mut(o),
+      block_table.data_ptr<int>(), ctx_lens.data_ptr<int>(), H, Hk, max_pages,
+      (float)scale);
+  HIP_CHECK_KERNEL();
+  return o;
+}
+
+// ---------------- Logit post-processing ----------------
+torch::Tensor argmax_rows(torch::Tensor logits) {
+  check_bf16(logits, "logits");
+  const int rows = logits.size(0), vocab = logits.size(1);
+  auto out = torch::empty({rows}, logits.options().dtype(torch::kInt32));
+ hipLaunchKernelGGL(( argmax_rows_kernel), dim3(dim3(rows)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(logits), out.data_ptr<int>(), vocab);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+torch::Tensor target_logprob(torch::Tensor logits, torch::Tensor targets) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(targets.scalar_type() == torch::kInt32 && targets.is_contiguous());
+  const int rows = logits.size(0), vocab = logits.size(1);
+  TORCH_CHECK(targets.size(0) == rows);
+  auto out = torch::empty({rows}, logits.options().dtype(torch::kFloat32));
+ hipLaunchKernelGGL(( target_logprob_kernel), dim3(dim3(rows)), dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(logits), targets.data_ptr<int>(), out.data_ptr<float>(), vocab);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm fwd (bf16)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual += x; y = rmsnorm(residual)");
+  m.def("rope_inplace", &rope_inplace, "RoPE in place over q,k");
+  m.def("swiglu", &swiglu, "silu(gate)*up from fused gateup");
+  m.def("add_bf16", &add_bf16, "a + b (bf16)");
+  m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");
+  m.def("attn_fwd", &attn_fwd, "causal flash attention fwd (D=128, GQA)");
+  m.def("paged_decode_attn", &paged_decode_attn, "paged decode attention");
+  m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");
+  m.def("target_logprob", &target_logprob, "fused log_softmax gather");
+}

@@ -1,0 +1,146 @@
+#include "hip/hip_runtime.h"
+// Fused elementwise / normalization kernels for the Llama backbone (gfx950).
+//
+// All memory-bound: target the HBM roofline (~6.3 TB/s achievable on MI355X),
+// so every bf16 access is vectorized as 16 B/lane (bf16x8) per the CDNA4
+// guideline (scalar bf16 loads cost ~2x).  The reference framework has no GPU
+// math at all (its LLM is a remote HTTPS provider); these
+// kernels are the MI355X-native compute path that replaces it.
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// RMSNorm (optionally fused with residual add).
+//   residual != nullptr:  r = x + residual  (written back to residual buffer,
+//                         becoming the next layer's residual stream)
+//                         y = r * rsqrt(mean(r^2) + eps) * w
+// One workgroup (256 threads) per row; f32 accumulation.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+rmsnorm_fwd_kernel(const ushort* __restrict__ x, const ushort* __restrict__ w,
+                   ushort* __restrict__ y, ushort* __restrict__ residual,
+                   int hidden, float eps) {
+  const int row = blockIdx.x;
+  const long long base = (long long)row * hidden;
+  const int nvec = hidden / 8;  // hidden % 8 == 0 enforced host-side
+  const bool fuse_res = residual != nullptr;
+
+  // Pass 1: sum of squares (with fused residual add written back; pass 2
+  // re-reads from L1/L2 — a register cache here would be runtime-indexed and
+  // spill to scratch, which is slower than the cache hit)
+  float ss = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + base + i * 8);
+    if (fuse_res) {
+      bf16x8 r = *reinterpret_cast<const bf16x8*>(residual + base + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v.v[j] = f2bf(bf2f(v.v[j]) + bf2f(r.v[j]));
+      *reinterpret_cast<bf16x8*>(residual + base + i * 8) = v;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(v.v[j]);
+      ss += f * f;
+    }
+  }
+
+  // Block reduction
+  __shared__ float smem[8];
+  float wsum = wave_reduce_sum(ss);
+  const int wid = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) smem[wid] = wsum;
+  __syncthreads();
+  float total = 0.f;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) total += smem[i];
+  const float inv_rms = rsqrtf(total / (float)hidden + eps);
+
+  const ushort* src = fuse_res ? residual : x;
+  // Pass 2: scale
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(src + base + i * 8);
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + i * 8);
+    bf16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      out.v[j] = f2bf(bf2f(v.v[j]) * inv_rms * bf2f(wv.v[j]));
+    *reinterpret_cast<bf16x8*>(y + base + i * 8) = out;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (Llama / NeoX rotate-half style), fused over Q and K.
+//   q: [T, Hq, D], k: [T, Hk, D] bf16, modified in place.
+//   cos_sin: [max_pos, D] f32 laid out [cos(0..D/2) | sin(0..D/2)] per row
+//   positions: [T] int32
+// Precomputed host-side trig table (on-device sinf/cosf turns this
+// memory-bound op VALU-bound).  Grid: (T, Hq+Hk).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(64)
+rope_fwd_kernel(ushort* __restrict__ q, ushort* __restrict__ k,
+                const float* __restrict__ cos_sin,
+                const int* __restrict__ positions, int Hq, int Hk, int D) {
+  const int t = blockIdx.x;
+  const int h = blockIdx.y;
+  const int half = D / 2;
+  ushort* ptr;
+  if (h < Hq) {
+    ptr = q + ((long long)t * Hq + h) * D;
+  } else {
+    ptr = k + ((long long)t * Hk + (h - Hq)) * D;
+  }
+  const float* cs = cos_sin + (long long)positions[t] * D;
+  // D = 128: 64 lanes each handle one rotation pair (d, d+half)
+  for (int d = threadIdx.x; d < half; d += blockDim.x) {
+    float c = cs[d];
+    float s = cs[half + d];
+    float x1 = bf2f(ptr[d]);
+    float x2 = bf2f(ptr[d + half]);
+    ptr[d] = f2bf(x1 * c - x2 * s);
+    ptr[d + half] = f2bf(x2 * c + x1 * s);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU activation: y = silu(gate) * up, bf16, vectorized.
+//   gateup: [T, 2*I] (fused gate|up GEMM output), y: [T, I]
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+swiglu_fwd_kernel(const ushort* __restrict__ gateup, ushort* __restrict__ y,
+                  long long rows, int inter) {
+  const int nvec = inter / 8;
+  const long long total = rows * nvec;
+  for (long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       idx < total; idx += (long long)gridDim.x * blockDim.x) {
+    const long long r = idx / nvec;
+    const int i = (int)(idx % nvec);
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(gateup + r * 2 * inter + i * 8);
+    bf16x8 u = *reinterpret_cast<const bf16x8*>(gateup + r * 2 * inter + inter + i * 8);
+    bf16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g.v[j]);
+      float uf = bf2f(u.v[j]);
+      float silu = gf / (1.f + __expf(-gf));
+      out.v[j] = f2bf(silu * uf);
+    }
+    *reinterpret_cast<bf16x8*>(y + r * inter + i * 8) = out;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Residual add (bf16): y = a + b  (used where the fused rmsnorm path isn't)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+add_bf16_kernel(const ushort* __restrict__ a, const ushort* __restrict__ b,
+                ushort* __restrict__ y, long long n8) {
+  for (long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       idx < n8; idx += (long long)gridDim.x * blockDim.x) {
+    bf16x8 va = *reinterpret_cast<const bf16x8*>(a + idx * 8);
+    bf16x8 vb = *reinterpret_cast<const bf16x8*>(b + idx * 8);
+    bf16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out.v[j] = f2bf(bf2f(va.v[j]) + bf2f(vb.v[j]));
+    *reinterpret_cast<bf16x8*>(y + idx * 8) = out;
+  }
+}

@@ -1,0 +1,216 @@
+// bf16 MFMA GEMM for gfx950 (MI355X): C[M,N] = A[M,K] @ B[N,K]^T
+//
+// This is the backbone projection GEMM (Llama weights are stored
+// [out_features, in_features], so both operands are K-contiguous — the
+// "B^T input" form whose fragments are plain 16-byte row loads).
+//
+// Structure (the CDNA4 guide's "step-3" LDS-staged design):
+//   - 128x128 output tile per 256-thread block (4 waves, 2x2; each wave owns
+//     a 64x64 quadrant = 4x4 fragments of v_mfma_f32_16x16x32_bf16)
+//   - K-loop in BK=64 steps; A-tile and B-tile (16 KiB each) staged by
+//     global_load_lds_dwordx4 (direct HBM->LDS DMA, 16 B/lane), double
+//     buffered; one __syncthreads per K-step (its vmcnt(0) drains the DMA)
+//   - LDS XOR swizzle (chunk ^= row&7) applied on the *source* address and
+//     the ds_read address (glds writes lane-linear, so the swizzle cannot go
+//     on the LDS destination) — keeps ds_read_b128 bank conflicts <=2-way
+//   - XCD-aware bijective blockIdx remap so neighboring tiles (sharing A/B
+//     panels) land on the same XCD's L2 (8 XCDs with private 4 MiB L2s)
+//
+// f32 accumulation in AGPRs; bf16 output.  M,N % 128 == 0, K % 64 == 0
+// (host wrapper pads M; Llama-3 dims satisfy N,K natively).
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define WARPS 4
+
+// chunk = 16 bytes = 8 bf16.  A row of BK=64 bf16 is 8 chunks.
+#define ROW_CHUNKS 8
+
+__device__ __forceinline__ void stage_tile_glds(
+    const ushort* __restrict__ src,  // tile origin: &X[row0*K + k0]
+    long long ldK,                   // leading dimension (elements)
+    ushort* lds_tile,                // 128*64 bf16, lane-linear destination
+    int tid) {
+  // 128 rows x 8 chunks = 1024 chunks; 256 threads -> 4 glds each.
+  // Linear slot s holds global chunk (row = s/8, c = (s%8) ^ (row&7)):
+  // inverse-swizzled source + swizzled read = same involution (rule 21).
+  // glds lane destination = wave-uniform base + lane*16, so the base must
+  // include this wave's 64-chunk sub-block, not just the iteration offset.
+  const int wave_chunk = tid & ~63;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = i * 256 + tid;
+    const int row = s >> 3;
+    const int c = (s & 7) ^ (row & 7);
+    const ushort* g = src + (long long)row * ldK + c * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(lds_tile + (long long)(i * 256 + wave_chunk) * 8),
+        16, 0, 0);
+  }
+}
+
+// ds_read_b128 of logical (row, chunk c) from a swizzled tile
+__device__ __forceinline__ short8 read_frag(const ushort* lds_tile, int row, int c) {
+  const int slot = row * ROW_CHUNKS + (c ^ (row & 7));
+  return *reinterpret_cast<const short8*>(lds_tile + slot * 8);
+}
+
+extern "C" __global__ void __launch_bounds__(256, 2)
+gemm_bt_bf16_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                    ushort* __restrict__ C, int M, int N, int K) {
+  // ---- XCD-aware bijective remap of the flat workgroup id (T1) ----
+  const int nwg = (M / BM) * (N / BN);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / BN;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;      // wave 0..3
+  const int lane = tid & 63;
+  const int wm = wid >> 1;       // 2x2 wave grid: quadrant row
+  const int wn = wid & 1;        // quadrant col
+
+  __shared__ __attribute__((aligned(16))) ushort lds[2][2][BM * BK];  // [buf][A/B]
+
+  const ushort* Atile = A + (long long)tile_m * BM * K;
+  const ushort* Btile = B + (long long)tile_n * BN * K;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int m_base = wm * 64;  // within tile
+  const int n_base = wn * 64;
+  const int frag_row = lane & 15;      // m or n within 16
+  const int frag_kgrp = lane >> 4;     // k-group 0..3 (8 bf16 each)
+
+  const int ntiles = K / BK;
+  // Prologue: stage tile 0 into buffer 0
+  stage_tile_glds(Atile, K, lds[0][0], tid);
+  stage_tile_glds(Btile, K, lds[0][1], tid);
+  __syncthreads();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      stage_tile_glds(Atile + (long long)(t + 1) * BK, K, lds[buf ^ 1][0], tid);
+      stage_tile_glds(Btile + (long long)(t + 1) * BK, K, lds[buf ^ 1][1], tid);
+    }
+    const ushort* Al = lds[buf][0];
+    const ushort* Bl = lds[buf][1];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {  // two K=32 MFMA steps per BK=64
+      short8 af[4], bf[4];
+      const int c = kk * 4 + frag_kgrp;  // logical chunk of this lane's 8 bf16
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = read_frag(Al, m_base + mi * 16 + frag_row, c);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bf[ni] = read_frag(Bl, n_base + ni * 16 + frag_row, c);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();  // drains the in-flight glds (vmcnt 0) + barrier
+    buf ^= 1;
+  }
+
+  // ---- Epilogue: C[m][n], C-fragment map row=(lane>>4)*4+e, col=lane&15 ----
+  const long long c_row0 = (long long)tile_m * BM + m_base + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * BN + n_base + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Split-K GEMV-ish path for skinny M (decode): M <= 16.
+// C[M,N] = A[M,K] @ B[N,K]^T with one workgroup per 64 N-rows x K-slice;
+// weights stream straight to VGPRs (LDS round-trip is pure overhead at M<=16
+// since B rows are used once), f32 atomics-free two-pass reduction.
+// Grid: (N/64, SPLITK); partials [SPLITK, M, N] f32 then reduced by
+// gemv_reduce_kernel (launch-boundary reduce — cheaper than in-launch
+// combine at this size per the microarch price list).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+gemv_bt_bf16_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                    float* __restrict__ Cpart, int M, int N, int K, int splitk) {
+  const int n = blockIdx.x * 64 + (threadIdx.x & 63);  // one N row per lane
+  const int slice = blockIdx.y;
+  const int wid = threadIdx.x >> 6;  // 4 waves split K within the slice
+  const int kper = K / splitk;
+  const int k0 = slice * kper + wid * (kper / 4);
+  const int k1 = k0 + kper / 4;
+
+  const ushort* brow = B + (long long)n * K;
+  // M <= 16 accumulators per lane
+  float acc[16];
+#pragma unroll
+  for (int m = 0; m < 16; ++m) acc[m] = 0.f;
+
+  for (int k = k0; k < k1; k += 8) {
+    bf16x8 bv = *reinterpret_cast<const bf16x8*>(brow + k);
+    float bfv[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) bfv[j] = bf2f(bv.v[j]);
+    for (int m = 0; m < M; ++m) {
+      bf16x8 av = *reinterpret_cast<const bf16x8*>(A + (long long)m * K + k);
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s += bf2f(av.v[j]) * bfv[j];
+      acc[m] += s;
+    }
+  }
+  // combine the 4 waves' partials via LDS, then write one partial per slice
+  __shared__ float part[4][16][64];
+  const int lane = threadIdx.x & 63;
+  for (int m = 0; m < M; ++m) part[wid][m][lane] = acc[m];
+  __syncthreads();
+  if (wid == 0) {
+    for (int m = 0; m < M; ++m) {
+      float v = part[0][m][lane] + part[1][m][lane] + part[2][m][lane] + part[3][m][lane];
+      Cpart[((long long)slice * M + m) * N + n] = v;
+    }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+gemv_reduce_kernel(const float* __restrict__ Cpart, ushort* __restrict__ C,
+                   int M, int N, int splitk) {
+  const long long total = (long long)M * N;
+  for (long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       idx < total; idx += (long long)gridDim.x * blockDim.x) {
+    float v = 0.f;
+    for (int s = 0; s < splitk; ++s) v += Cpart[(long long)s * total + idx];
+    C[idx] = f2bf(v);
+  }
+}
