@@ -34,12 +34,20 @@ class BeamSearchEngine:
         backend: PromptOptimizerBackend,
         score_fn: Optional[Callable[[List[str], List[RolloutResult]], List[float]]] = None,
         clock: Optional[Callable[[], int]] = None,
+        max_critique_tokens: int = 350,
+        max_edit_tokens: int = 512,
+        expand_fn: Optional[Callable[..., List[str]]] = None,
     ) -> None:
         """``score_fn`` scores a *batch* of candidate prompts (hook for the
-        RCCL candidate-parallel path); defaults to serial backend.score."""
+        RCCL candidate-parallel path); defaults to serial backend.score.
+        ``expand_fn(parent_contents, rollouts, branch_factor) -> contents``
+        overrides candidate generation (e.g. rank-0-generate + broadcast)."""
         self._backend = backend
         self._score_fn = score_fn
         self._clock = clock
+        self._max_critique_tokens = max_critique_tokens
+        self._max_edit_tokens = max_edit_tokens
+        self._expand_fn = expand_fn
 
     # -- candidate expansion --
 
@@ -63,16 +71,23 @@ class BeamSearchEngine:
                 created_at=self._now(svc),
             )]
             state.beam = parents
+        if self._expand_fn is not None:
+            contents = self._expand_fn([p.content for p in parents], rollouts, branch_factor)
+            return [VersionedPromptTemplate(
+                version=self._next_version(state), content=c, score=None,
+                parent_version=parents[i // branch_factor].version if i // branch_factor < len(parents) else None,
+                created_at=self._now(svc),
+            ) for i, c in enumerate(contents)]
         children: List[VersionedPromptTemplate] = []
         for parent in parents:
             parent_rules = [ln for ln in parent.content.split("\n") if ln.strip()]
             gradient_prompt = build_textual_gradient_prompt(parent_rules, rollouts)
-            critique = self._backend.generate(gradient_prompt, max_new_tokens=350)
+            critique = self._backend.generate(gradient_prompt, max_new_tokens=self._max_critique_tokens)
             for b in range(branch_factor):
                 # Vary the decode per branch by conditioning on the branch index,
                 # so branches explore different edits of the same parent.
                 edit_prompt = build_apply_edit_prompt(parent_rules, f"{critique}\n(Variant {b + 1}: emphasize a different single issue.)")
-                content = self._backend.generate(edit_prompt, max_new_tokens=512)
+                content = self._backend.generate(edit_prompt, max_new_tokens=self._max_edit_tokens)
                 children.append(VersionedPromptTemplate(
                     version=self._next_version(state),
                     content=content,

@@ -1,0 +1,172 @@
+"""LlamaBackend — the local optimizer backend (generation + candidate scoring).
+
+Implements ``senweaver_amd.apo.optimizer.PromptOptimizerBackend`` on the
+MI355X Llama backbone: greedy decode for textual-gradient critiques and
+apply-edit rewrites, and reward-weighted teacher-forced log-likelihood for
+beam-candidate scoring (the work the reference shipped to its backend via
+POST /api/apo/gradient and /api/apo/optimize).
+
+Scoring definition (see optimizer.PromptOptimizerBackend.score):
+    score(c) = sum_r w_r * avg_logprob(assistant tokens | c + context_r)
+               / sum_r |w_r|
+with w_r = finalReward (or +-1 from rollout status).  Prompts that raise the
+likelihood of high-reward conversations and lower the likelihood of
+negative-reward ones score higher.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+from .. import ops
+from ..apo.optimizer import rollout_weight
+from ..apo.schema import RolloutResult
+from ..models.config import ModelConfig, get_config
+from ..models.llama import LlamaModel
+from .kvcache import PAGE_SIZE, PagedKVCache
+from . import tokenizer as tok
+
+
+def _pad64(n: int) -> int:
+    return (n + 63) & ~63
+
+
+class LlamaBackend:
+    def __init__(self, config: ModelConfig | str = "llama-3-8b", device: Optional[str] = None,
+                 seed: int = 0, max_seq: int = 2048, micro_batch: int = 8) -> None:
+        if isinstance(config, str):
+            config = get_config(config)
+        if device is None:
+            device = "cuda:0" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        self.config = config
+        self.model = LlamaModel(config, device=self.device, seed=seed)
+        self.tokenizer = tok.HashTokenizer(config.vocab_size)
+        self.max_seq = min(max_seq, config.max_position)
+        self.micro_batch = micro_batch
+
+    # ------------------------------------------------------------------
+    # Sequence building
+    # ------------------------------------------------------------------
+    _ROLE_IDS = {"user": tok.ROLE_USER, "assistant": tok.ROLE_ASSISTANT,
+                 "tool": tok.ROLE_TOOL}
+
+    def build_scored_sequence(self, candidate_prompt: str, rollout: RolloutResult
+                              ) -> Tuple[List[int], List[bool]]:
+        """Token ids + mask marking assistant-content tokens (the scored ones)."""
+        ids = [tok.BOS, tok.ROLE_SYSTEM]
+        mask = [False, False]
+        ids.extend(self.tokenizer.encode(candidate_prompt, max_tokens=600))
+        mask.extend([False] * (len(ids) - len(mask)))
+        for m in rollout.messages:
+            role_id = self._ROLE_IDS.get(m.role, tok.ROLE_USER)
+            ids.append(role_id)
+            mask.append(False)
+            content = self.tokenizer.encode(m.content, max_tokens=400)
+            ids.extend(content)
+            mask.extend([m.role == "assistant"] * len(content))
+            ids.append(tok.EOS)
+            mask.append(m.role == "assistant")
+        ids = ids[: self.max_seq]
+        mask = mask[: self.max_seq]
+        return ids, mask
+
+    # ------------------------------------------------------------------
+    # Teacher-forced log-prob of masked positions for a batch of sequences
+    # ------------------------------------------------------------------
+    def sequence_logprobs(self, sequences: Sequence[Tuple[List[int], List[bool]]]
+                          ) -> List[float]:
+        """Returns the mean per-token logprob of masked tokens per sequence."""
+        results: List[float] = []
+        mb = self.micro_batch
+        for i in range(0, len(sequences), mb):
+            chunk = sequences[i: i + mb]
+            results.extend(self._logprobs_chunk(chunk))
+        return results
+
+    def _logprobs_chunk(self, chunk) -> List[float]:
+        B = len(chunk)
+        S = _pad64(max(len(ids) for ids, _ in chunk))
+        tokens = torch.zeros(B, S, dtype=torch.long)
+        for b, (ids, _) in enumerate(chunk):
+            tokens[b, : len(ids)] = torch.tensor(ids, dtype=torch.long)
+        tokens = tokens.to(self.device)
+        hidden = self.model.prefill(tokens)  # [B, S, H]
+
+        # positions p>=1 whose token is scored: predicted from hidden[p-1]
+        gather_rows: List[int] = []
+        targets: List[int] = []
+        seq_of: List[int] = []
+        for b, (ids, mask) in enumerate(chunk):
+            for p in range(1, len(ids)):
+                if mask[p]:
+                    gather_rows.append(b * S + p - 1)
+                    targets.append(ids[p])
+                    seq_of.append(b)
+        if not gather_rows:
+            return [0.0] * B
+        flat = hidden.reshape(B * S, -1)
+        rows = flat[torch.tensor(gather_rows, device=self.device)]
+        logits = self.model.logits(rows)
+        lp = ops.target_logprob(logits, torch.tensor(targets, dtype=torch.int32,
+                                                     device=self.device))
+        lp = lp.float().cpu()
+        sums = [0.0] * B
+        counts = [0] * B
+        for j, b in enumerate(seq_of):
+            sums[b] += float(lp[j])
+            counts[b] += 1
+        return [sums[b] / counts[b] if counts[b] else 0.0 for b in range(B)]
+
+    # ------------------------------------------------------------------
+    # PromptOptimizerBackend API
+    # ------------------------------------------------------------------
+    def score_batch(self, candidate_prompts: Sequence[str],
+                    rollouts: Sequence[RolloutResult]) -> List[float]:
+        """Score each candidate against all rollouts (one forward per pair)."""
+        seqs = []
+        for c in candidate_prompts:
+            for r in rollouts:
+                seqs.append(self.build_scored_sequence(c, r))
+        lps = self.sequence_logprobs(seqs)
+        weights = [rollout_weight(r) for r in rollouts]
+        wnorm = sum(abs(w) for w in weights) or 1.0
+        out = []
+        R = len(rollouts)
+        for ci in range(len(candidate_prompts)):
+            s = sum(weights[ri] * lps[ci * R + ri] for ri in range(R))
+            out.append(s / wnorm)
+        return out
+
+    def score(self, candidate_prompt: str, rollouts: Sequence[RolloutResult]) -> float:
+        return self.score_batch([candidate_prompt], rollouts)[0]
+
+    @torch.no_grad()
+    def generate(self, prompt: str, max_new_tokens: int = 256) -> str:
+        ids = [tok.BOS] + self.tokenizer.encode(prompt, max_tokens=self.max_seq - max_new_tokens - 8)
+        ids.append(tok.ROLE_ASSISTANT)
+        real = len(ids)
+        S = _pad64(real)
+        total = real + max_new_tokens + 1
+        pages_needed = (total + PAGE_SIZE - 1) // PAGE_SIZE + 2
+        cache = PagedKVCache(self.config, pages_needed, self.device)
+        seq = cache.new_seq()
+        tokens = torch.zeros(1, S, dtype=torch.long)
+        tokens[0, :real] = torch.tensor(ids, dtype=torch.long)
+        hidden = self.model.prefill(tokens.to(self.device), cache=cache, seqs=[seq],
+                                    real_lens=[real])
+        last_hidden = hidden[0, real - 1]
+        out_ids: List[int] = []
+        for step in range(max_new_tokens):
+            logits = self.model.logits(last_hidden.reshape(1, -1))
+            nxt = int(ops.argmax_rows(logits)[0])
+            if nxt == tok.EOS:
+                break
+            out_ids.append(nxt)
+            pos = torch.tensor([real + step], device=self.device)
+            last_hidden = self.model.decode_step(
+                torch.tensor([nxt], device=self.device), pos, cache, [seq])[0]
+        return self.tokenizer.decode(out_ids)

@@ -1,0 +1,5 @@
+from .config import ModelConfig, get_config, llama3_8b, llama3_70b, mixtral_8x7b, tiny_debug
+from .llama import LlamaModel
+
+__all__ = ["ModelConfig", "get_config", "llama3_8b", "llama3_70b", "mixtral_8x7b",
+           "tiny_debug", "LlamaModel"]

@@ -1,0 +1,153 @@
+"""Llama backbone on the senweaver_amd gfx950 kernel library.
+
+Replaces the reference's remote LLM providers (sendLLMMessage.impl.ts's 20
+HTTPS backends) with a local model whose hot path is entirely hand-written
+HIP: MFMA GEMM for every projection, fused residual+RMSNorm, table-driven
+RoPE, flash prefill attention, paged decode attention, SwiGLU, and fused
+logit reductions.  PyTorch supplies tensor storage, reshapes/transposes and
+the embedding gather only.
+
+Weights are random-init bf16 (no network access for checkpoints — bench data
+is declared synthetic).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+
+from .. import ops
+from ..engine.kvcache import PagedKVCache
+from .config import ModelConfig
+
+
+class LlamaModel:
+    def __init__(self, config: ModelConfig, device="cpu", dtype=torch.bfloat16, seed: int = 0):
+        self.config = config
+        self.device = torch.device(device)
+        self.dtype = dtype
+        c = config
+        if self.device.type == "cuda":
+            # Device-side init: seconds for 8B instead of minutes of CPU RNG +
+            # PCIe transfer.  Same seed + same call order on every rank =>
+            # identical weights across the DP group (Philox is device-count
+            # independent), which candidate-parallel scoring requires.
+            gen = torch.Generator(device=self.device).manual_seed(seed)
+
+            def W(*shape, std=0.02):
+                t = torch.empty(shape, dtype=torch.float32, device=self.device)
+                t.normal_(0.0, std, generator=gen)
+                return t.to(dtype)
+        else:
+            g = torch.Generator(device="cpu").manual_seed(seed)
+
+            def W(*shape, std=0.02):
+                t = torch.empty(shape, dtype=torch.float32)
+                t.normal_(0.0, std, generator=g)
+                return t.to(dtype).to(self.device)
+
+        self.embed = W(c.vocab_size, c.hidden_size)
+        self.layers = []
+        for _ in range(c.num_layers):
+            self.layers.append({
+                "input_norm": torch.ones(c.hidden_size, dtype=dtype, device=self.device),
+                "qkv": W(c.q_size + 2 * c.kv_size, c.hidden_size),
+                "o": W(c.hidden_size, c.q_size),
+                "post_norm": torch.ones(c.hidden_size, dtype=dtype, device=self.device),
+                "gateup": W(2 * c.intermediate_size, c.hidden_size),
+                "down": W(c.hidden_size, c.intermediate_size),
+            })
+        self.final_norm = torch.ones(c.hidden_size, dtype=dtype, device=self.device)
+        self.lm_head = self.embed if c.tie_embeddings else W(c.vocab_size, c.hidden_size)
+        self.cos_sin = ops.rope_tables(c.max_position, c.head_dim, c.rope_theta).to(self.device)
+        self.scale = 1.0 / math.sqrt(c.head_dim)
+
+    # ------------------------------------------------------------------
+    # Prefill: tokens [B, S] -> final hidden states [B, S, H].
+    # If `cache` is given, K/V of every position are appended (seqs maps
+    # batch row -> cache sequence id).
+    # ------------------------------------------------------------------
+    def prefill(self, tokens: torch.Tensor, cache: Optional[PagedKVCache] = None,
+                seqs: Optional[List[int]] = None,
+                real_lens: Optional[List[int]] = None) -> torch.Tensor:
+        c = self.config
+        B, S = tokens.shape
+        T = B * S
+        flat = tokens.reshape(T)
+        positions = torch.arange(S, device=self.device, dtype=torch.int32).repeat(B)
+
+        hidden = self.embed[flat.long()]  # [T, H]
+        residual = None
+        for li, L in enumerate(self.layers):
+            if residual is None:
+                residual = hidden.clone()
+                h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
+            else:
+                h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
+            qkv = ops.gemm_bt(h, L["qkv"])  # [T, q+2kv]
+            q = qkv[:, : c.q_size].reshape(T, c.num_heads, c.head_dim).contiguous()
+            k = qkv[:, c.q_size: c.q_size + c.kv_size].reshape(T, c.num_kv_heads, c.head_dim).contiguous()
+            v = qkv[:, c.q_size + c.kv_size:].reshape(T, c.num_kv_heads, c.head_dim).contiguous()
+            ops.rope_inplace(q, k, self.cos_sin, positions)
+            if cache is not None:
+                for b in range(B):
+                    n = real_lens[b] if real_lens is not None else S
+                    cache.append(li, seqs[b], k[b * S: b * S + n], v[b * S: b * S + n],
+                                 advance_len=(li == c.num_layers - 1))
+            qb = q.reshape(B, S, c.num_heads, c.head_dim).transpose(1, 2).contiguous()
+            kb = k.reshape(B, S, c.num_kv_heads, c.head_dim).transpose(1, 2).contiguous()
+            vb = v.reshape(B, S, c.num_kv_heads, c.head_dim).transpose(1, 2).contiguous()
+            attn = ops.attn_fwd(qb, kb, vb, self.scale)  # [B,Hq,S,D]
+            attn = attn.transpose(1, 2).reshape(T, c.q_size).contiguous()
+            o = ops.gemm_bt(attn, L["o"])
+            h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
+            gateup = ops.gemm_bt(h, L["gateup"])
+            act = ops.swiglu(gateup)
+            hidden = ops.gemm_bt(act, L["down"])
+        h = ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)
+        return h.reshape(B, S, c.hidden_size)
+
+    # ------------------------------------------------------------------
+    # Decode: one new token per sequence.  tokens [B], positions [B].
+    # ------------------------------------------------------------------
+    def decode_step(self, tokens: torch.Tensor, positions: torch.Tensor,
+                    cache: PagedKVCache, seqs: List[int]) -> torch.Tensor:
+        c = self.config
+        B = tokens.shape[0]
+        hidden = self.embed[tokens.long()]
+        residual = None
+        pos32 = positions.to(torch.int32)
+        for li, L in enumerate(self.layers):
+            if residual is None:
+                residual = hidden.clone()
+                h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
+            else:
+                h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
+            qkv = ops.gemm_bt(h, L["qkv"])
+            q = qkv[:, : c.q_size].reshape(B, c.num_heads, c.head_dim).contiguous()
+            k = qkv[:, c.q_size: c.q_size + c.kv_size].reshape(B, c.num_kv_heads, c.head_dim).contiguous()
+            v = qkv[:, c.q_size + c.kv_size:].reshape(B, c.num_kv_heads, c.head_dim).contiguous()
+            ops.rope_inplace(q, k, self.cos_sin, pos32)
+            for b in range(B):
+                cache.append(li, seqs[b], k[b: b + 1], v[b: b + 1],
+                             advance_len=(li == c.num_layers - 1))
+            bt = cache.block_table_tensor(seqs)
+            # context includes the just-appended token for every layer:
+            ctx = torch.tensor([cache.seq_lens[s] + (1 if li < c.num_layers - 1 else 0)
+                                for s in seqs], dtype=torch.int32, device=self.device)
+            attn = ops.paged_decode_attn(q, cache.k[li], cache.v[li], bt, ctx, self.scale)
+            attn = attn.reshape(B, c.q_size)
+            o = ops.gemm_bt(attn, L["o"])
+            h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
+            gateup = ops.gemm_bt(h, L["gateup"])
+            act = ops.swiglu(gateup)
+            hidden = ops.gemm_bt(act, L["down"])
+        h = ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)
+        return h  # [B, hidden]
+
+    def logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        """hidden [*, H] -> logits [*, vocab] through the lm_head GEMM."""
+        flat = hidden.reshape(-1, self.config.hidden_size)
+        return ops.gemm_bt(flat, self.lm_head)

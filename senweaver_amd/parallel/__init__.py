@@ -1,0 +1,10 @@
+from .dist import (
+    CandidateParallelScorer,
+    broadcast_strings,
+    dp_scores_allreduce,
+    init_from_env,
+    rank_world,
+)
+
+__all__ = ["CandidateParallelScorer", "broadcast_strings", "dp_scores_allreduce",
+           "init_from_env", "rank_world"]

@@ -1,0 +1,104 @@
+"""Multi-process (gloo, world_size=2) tests for the candidate-parallel path.
+
+These run on CPU here; the identical code path runs over RCCL ("nccl"
+backend) on the 8-GPU xGMI node.
+"""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+
+def _worker(rank, world, port, fn_name, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        result = globals()[fn_name](rank, world)
+        q.put((rank, result))
+    finally:
+        dist.destroy_process_group()
+
+
+def _run_dist(fn_name, world=2):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = 29511 + (os.getpid() % 500)
+    procs = [ctx.Process(target=_worker, args=(r, world, port, fn_name, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    results = {}
+    while not q.empty():
+        rank, res = q.get()
+        results[rank] = res
+    for p in procs:
+        assert p.exitcode == 0, f"worker exited {p.exitcode}"
+    assert len(results) == world
+    return results
+
+
+# --- payload fns (module-level so spawn can pickle by name) ---
+
+def _payload_allreduce_scores(rank, world):
+    from senweaver_amd.parallel import dp_scores_allreduce
+    my_idx = list(range(rank, 5, world))
+    my_scores = [float(10 * i + 1) for i in my_idx]
+    return dp_scores_allreduce(5, my_idx, my_scores, torch.device("cpu"))
+
+
+def _payload_broadcast(rank, world):
+    from senweaver_amd.parallel import broadcast_strings
+    strings = ["alpha", "beta", "gamma"] if rank == 0 else None
+    return broadcast_strings(strings, src=0)
+
+
+def _payload_candidate_parallel(rank, world):
+    from senweaver_amd.apo.optimizer import StubBackend
+    from senweaver_amd.apo.schema import RolloutMessage, RolloutResult
+    from senweaver_amd.parallel import CandidateParallelScorer
+
+    class CPUStub(StubBackend):
+        device = torch.device("cpu")
+
+        def score_batch(self, prompts, rollouts):
+            return [self.score(p, rollouts) for p in prompts]
+
+    r = RolloutResult(
+        trace_id="t", thread_id="th", status="succeeded", final_reward=1.0,
+        reward_dimensions=[], chat_mode="normal",
+        messages=[RolloutMessage("assistant", "x")],
+        tool_call_stats={"totalCalls": 0, "succeeded": 0, "failed": 0,
+                         "successRate": None, "byToolName": {}, "totalDurationMs": 0},
+        llm_stats={"totalCalls": 1, "totalTokens": 1},
+    )
+    backend = CPUStub(seed=7)
+    scorer = CandidateParallelScorer(backend)
+    prompts = [f"- rule {i}" for i in range(6)]
+    scores = scorer(prompts, [r])
+    serial = [StubBackend(seed=7).score(p, [r]) for p in prompts]
+    return (scores, serial)
+
+
+def test_dp_scores_allreduce_world2():
+    results = _run_dist("_payload_allreduce_scores")
+    for rank, vec in results.items():
+        assert vec == [1.0, 11.0, 21.0, 31.0, 41.0]
+
+
+def test_broadcast_strings_world2():
+    results = _run_dist("_payload_broadcast")
+    assert results[0] == results[1] == ["alpha", "beta", "gamma"]
+
+
+def test_candidate_parallel_scorer_world2():
+    results = _run_dist("_payload_candidate_parallel")
+    for rank, (scores, serial) in results.items():
+        assert scores == pytest.approx(serial, abs=1e-6)  # f32 collective rounding
+    # both ranks computed the identical full vector
+    assert results[0][0] == pytest.approx(results[1][0], abs=0)
