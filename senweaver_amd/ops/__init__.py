@@ -94,7 +94,10 @@ def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     M = a.shape[0]
     if M <= 16 and b.shape[1] % 1024 == 0 and b.shape[0] % 64 == 0:
         return ext.gemm_bt(a.contiguous(), b.contiguous())
-    pad = (-M) % 128
+    # pad M to the 256-tile when N allows it and the grid fills the chip
+    N = b.shape[0]
+    tile = 256 if (N % 256 == 0 and ((M + 255) // 256) * (N // 256) >= 160) else 128
+    pad = (-M) % tile
     if pad:
         a = torch.nn.functional.pad(a, (0, 0, 0, pad))
     c = ext.gemm_bt(a.contiguous(), b.contiguous())

@@ -15,6 +15,7 @@ extern "C" __global__ void add_bf16_kernel(const ushort*, const ushort*, ushort*
 extern "C" __global__ void argmax_rows_kernel(const ushort*, int*, int);
 extern "C" __global__ void target_logprob_kernel(const ushort*, const int*, float*, int);
 extern "C" __global__ void gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_kernel(const ushort*, const ushort*, float*, int, int, int, int);
 extern "C" __global__ void gemv_reduce_kernel(const float*, ushort*, int, int, int);
 extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
@@ -144,12 +145,21 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
     HIP_CHECK_KERNEL();
     return C;
   }
-  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
-              "gemm_bt needs M,N % 128 == 0 and K % 64 == 0 (pad host-side); got ",
-              M, "x", N, "x", K);
-  const int nwg = (M / 128) * (N / 128);
-  gemm_bt_bf16_kernel<<<dim3(nwg), dim3(256), 0, cur_stream()>>>(
-      bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+  TORCH_CHECK(K % 64 == 0, "gemm_bt needs K % 64 == 0; got K=", K);
+  // 256-tile kernel runs 1 block/CU (128 KiB LDS): it needs >=~160 blocks to
+  // fill 256 CUs; below that the 2-block/CU 128-tile kernel wins.
+  if (M % 256 == 0 && N % 256 == 0 && (M / 256) * (N / 256) >= 160) {
+    const int nwg = (M / 256) * (N / 256);
+    gemm_bt_bf16_256_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
+        bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+  } else {
+    TORCH_CHECK(M % 128 == 0 && N % 128 == 0,
+                "gemm_bt needs M,N % 128 == 0 (pad host-side); got ",
+                M, "x", N, "x", K);
+    const int nwg = (M / 128) * (N / 128);
+    gemm_bt_bf16_kernel<<<dim3(nwg), dim3(256), 0, cur_stream()>>>(
+        bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+  }
   HIP_CHECK_KERNEL();
   return C;
 }
