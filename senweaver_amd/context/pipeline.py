@@ -1,0 +1,112 @@
+"""System-message assembly + prepared-request pipeline.
+
+Rebuilds the reference's ConvertToLLMMessageService surface
+(browser/convertToLLMMessageService.ts): system-message generation with the
+multi-agent section (:788-830) and the APO 2000-char rule injection
+(:832-853, via senweaver_amd.apo.rules.inject_rules), key+timestamp caching
+(:661-663), and prepare_llm_chat_messages combining history compression with
+the 4-phase fitting.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import List, Optional, Tuple
+
+from ..agents.registry import get_agent_composition, get_agent_definition
+from ..apo.rules import inject_rules
+from .compress import CompressibleMessage, compress_old_messages
+from .fitting import Msg, prepare_messages
+
+SYSTEM_CACHE_TTL_MS = 30_000
+
+
+def base_system_message(chat_mode: str, workspace_overview: str = "",
+                        system_info: str = "") -> str:
+    """Compact analog of chat_systemMessage (prompts.ts:806-1237): mode
+    header + system info + files overview.  Tool XML definitions are
+    appended by the transport's tool registry when tools are enabled."""
+    headers = {
+        "normal": "You are SenWeaver-AMD, an expert coding assistant. Answer questions "
+                  "about the user's codebase; you may read files but not modify them.",
+        "agent": "You are SenWeaver-AMD in agent mode: complete the user's task end to "
+                 "end using the available tools. Verify your work before finishing.",
+        "designer": "You are SenWeaver-AMD in designer mode: build and refine UI "
+                    "components and their backing APIs.",
+        "gather": "You are SenWeaver-AMD in gather mode: collect and summarize relevant "
+                  "context from the codebase. Read-only.",
+    }
+    parts = [headers.get(chat_mode, headers["normal"])]
+    if system_info:
+        parts.append(f"<system_info>\n{system_info}\n</system_info>")
+    if workspace_overview:
+        parts.append(f"<files_overview>\n{workspace_overview}\n</files_overview>")
+    return "\n\n".join(parts)
+
+
+def multi_agent_section(chat_mode: str) -> str:
+    """The multi-agent system section (:788-830)."""
+    comp = get_agent_composition(chat_mode)
+    if not comp.available_sub_agents:
+        return ""
+    parts = [f"\n## Primary Agent\n{get_agent_definition(comp.primary_agent).name}"]
+    subs = []
+    for a in comp.available_sub_agents:
+        d = get_agent_definition(a)
+        if d:
+            subs.append(f"- {d.name}: {d.description}")
+    if subs:
+        parts.append("\n## Available Sub-Agents\n" + "\n".join(subs))
+    if comp.enable_parallel:
+        parts.append(
+            f"\n## Parallel Execution Capability\nYou support up to {comp.max_parallel} "
+            "sub-tasks running in parallel. For sub-tasks that can be completed "
+            "independently, prefer parallel processing to improve efficiency.")
+    return "# Multi-Agent System" + "".join(parts)
+
+
+class ConvertToLLMMessages:
+    def __init__(self, apo_service=None, clock=None) -> None:
+        self._apo = apo_service
+        self._clock = clock or (lambda: int(time.time() * 1000))
+        self._cached_msg: Optional[str] = None
+        self._cached_key: Optional[str] = None
+        self._cached_at = 0
+
+    def invalidate_cache(self) -> None:
+        """Directory-change invalidation hook (reference :698-700, 2s debounce)."""
+        self._cached_msg = None
+
+    def generate_system_message(self, chat_mode: str, workspace_overview: str = "",
+                                system_info: str = "") -> str:
+        key = f"{chat_mode}:{hash(workspace_overview)}:{hash(system_info)}"
+        now = self._clock()
+        if (self._cached_msg is not None and self._cached_key == key
+                and now - self._cached_at < SYSTEM_CACHE_TTL_MS):
+            return self._cached_msg
+        msg = base_system_message(chat_mode, workspace_overview, system_info)
+        ma = multi_agent_section(chat_mode)
+        if ma:
+            msg = msg + "\n\n" + ma
+        if self._apo is not None:
+            try:
+                msg = inject_rules(msg, self._apo.get_optimized_rules())
+            except Exception:
+                pass  # APO failure never breaks message assembly
+        self._cached_msg = msg
+        self._cached_key = key
+        self._cached_at = now
+        return msg
+
+    def prepare_llm_chat_messages(self, history: List[CompressibleMessage],
+                                  chat_mode: str, context_window: int,
+                                  workspace_overview: str = "",
+                                  reserved_output: Optional[int] = None
+                                  ) -> Tuple[str, List[Msg]]:
+        """History compression -> 4-phase fit -> (system_message, messages)."""
+        sys_msg = self.generate_system_message(chat_mode, workspace_overview)
+        compressed = compress_old_messages(history)
+        msgs = [Msg(m.role if m.role != "tool" else "assistant",
+                    m.content if m.role != "tool" else f"[tool {m.tool_name} result]\n{m.content}")
+                for m in compressed]
+        return prepare_messages(msgs, sys_msg, context_window, reserved_output)

@@ -145,6 +145,42 @@ class LlamaBackend:
         return self.score_batch([candidate_prompt], rollouts)[0]
 
     @torch.no_grad()
+    def stream_generate(self, prompt: str, max_new_tokens: int,
+                        should_stop, on_chunk) -> str:
+        """Incremental greedy decode for the transport layer: calls
+        on_chunk(cumulative_text) per token, honors should_stop between
+        tokens.  Same decode path as generate()."""
+        ids = [tok.BOS] + self.tokenizer.encode(prompt, max_tokens=self.max_seq - max_new_tokens - 8)
+        ids.append(tok.ROLE_ASSISTANT)
+        real = len(ids)
+        S = _pad64(real)
+        total = real + max_new_tokens + 1
+        pages_needed = (total + PAGE_SIZE - 1) // PAGE_SIZE + 2
+        cache = PagedKVCache(self.config, pages_needed, self.device)
+        seq = cache.new_seq()
+        tokens = torch.zeros(1, S, dtype=torch.long)
+        tokens[0, :real] = torch.tensor(ids, dtype=torch.long)
+        hidden = self.model.prefill(tokens.to(self.device), cache=cache, seqs=[seq],
+                                    real_lens=[real])
+        last_hidden = hidden[0, real - 1]
+        out_ids = []
+        text = ""
+        for step in range(max_new_tokens):
+            if should_stop():
+                break
+            logits = self.model.logits(last_hidden.reshape(1, -1))
+            nxt = int(ops.argmax_rows(logits)[0])
+            if nxt == tok.EOS:
+                break
+            out_ids.append(nxt)
+            text = self.tokenizer.decode(out_ids)
+            on_chunk(text)
+            pos = torch.tensor([real + step], device=self.device)
+            last_hidden = self.model.decode_step(
+                torch.tensor([nxt], device=self.device), pos, cache, [seq])[0]
+        return text
+
+    @torch.no_grad()
     def generate(self, prompt: str, max_new_tokens: int = 256) -> str:
         ids = [tok.BOS] + self.tokenizer.encode(prompt, max_tokens=self.max_seq - max_new_tokens - 8)
         ids.append(tok.ROLE_ASSISTANT)

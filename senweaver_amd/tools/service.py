@@ -1,0 +1,364 @@
+"""ToolsService — validation + execution of the built-in tools.
+
+Capability-compatible with the reference's ToolsService
+(browser/toolsService.ts: callTool map :1693, stringOfResult map :3265,
+param validation).  The filesystem/search/terminal tools are implemented
+locally (sandboxed under a workspace root); web/vision/document tools —
+which the reference served via localhost sidecar HTTP servers
+(browser/start*.cjs, ports 3000-3008) — raise a structured offline error in
+this no-network environment.
+"""
+
+from __future__ import annotations
+
+import fnmatch
+import os
+import re
+import shutil
+import subprocess
+import threading
+import time
+import uuid
+from dataclasses import dataclass
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+from .registry import APPROVAL_TYPE_OF_TOOL, BUILTIN_TOOLS, is_builtin_tool
+
+MAX_TERMINAL_CHARS = 100_000            # prompts.ts:29
+MAX_TERMINAL_INACTIVE_TIME_S = 8        # prompts.ts:30
+MAX_TERMINAL_BG_COMMAND_TIME_S = 5      # prompts.ts:31
+MAX_FILE_CHARS = 200_000
+PAGE_SIZE_RESULTS = 50
+
+# search/replace block format — prompts.ts:38-40
+SR_ORIGINAL = "<<<<<<< ORIGINAL"
+SR_DIVIDER = "======="
+SR_UPDATED = ">>>>>>> UPDATED"
+
+
+class ToolError(Exception):
+    pass
+
+
+@dataclass
+class ToolResult:
+    tool_name: str
+    result: Any
+    text: str  # stringOfResult
+
+
+def parse_search_replace_blocks(blocks: str) -> List[Tuple[str, str]]:
+    """Parse '<<<<<<< ORIGINAL / ======= / >>>>>>> UPDATED' blocks."""
+    out: List[Tuple[str, str]] = []
+    lines = blocks.split("\n")
+    i = 0
+    while i < len(lines):
+        if lines[i].strip() == SR_ORIGINAL:
+            orig: List[str] = []
+            updated: List[str] = []
+            i += 1
+            while i < len(lines) and lines[i].strip() != SR_DIVIDER:
+                orig.append(lines[i])
+                i += 1
+            i += 1  # divider
+            while i < len(lines) and lines[i].strip() != SR_UPDATED:
+                updated.append(lines[i])
+                i += 1
+            out.append(("\n".join(orig), "\n".join(updated)))
+        i += 1
+    if not out:
+        raise ToolError("No ORIGINAL/UPDATED search-replace blocks found")
+    return out
+
+
+class PersistentTerminal:
+    def __init__(self, cwd: str) -> None:
+        self.id = str(uuid.uuid4())[:8]
+        self.cwd = cwd
+        self.proc = subprocess.Popen(
+            ["/bin/bash"], stdin=subprocess.PIPE, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT, cwd=cwd, text=True, bufsize=1)
+        self._buffer: List[str] = []
+        self._lock = threading.Lock()
+        self._reader = threading.Thread(target=self._read, daemon=True)
+        self._reader.start()
+
+    def _read(self) -> None:
+        try:
+            for line in self.proc.stdout:
+                with self._lock:
+                    self._buffer.append(line)
+        except Exception:
+            pass
+
+    def run(self, command: str, wait_s: float = MAX_TERMINAL_BG_COMMAND_TIME_S) -> str:
+        with self._lock:
+            self._buffer.clear()
+        marker = f"__DONE_{uuid.uuid4().hex[:8]}__"
+        self.proc.stdin.write(command + f"\necho {marker}\n")
+        self.proc.stdin.flush()
+        deadline = time.time() + wait_s
+        while time.time() < deadline:
+            with self._lock:
+                text = "".join(self._buffer)
+            if marker in text:
+                return text.split(marker)[0][:MAX_TERMINAL_CHARS]
+            time.sleep(0.05)
+        with self._lock:
+            return ("".join(self._buffer))[:MAX_TERMINAL_CHARS] + "\n[command still running in background]"
+
+    def kill(self) -> None:
+        try:
+            self.proc.kill()
+        except Exception:
+            pass
+
+
+class ToolsService:
+    def __init__(self, workspace_root: str, subagent_runner=None) -> None:
+        self.root = os.path.abspath(workspace_root)
+        self.terminals: Dict[str, PersistentTerminal] = {}
+        self._subagents = subagent_runner
+
+    # ---- path safety ----
+    def _resolve(self, uri: Optional[str]) -> str:
+        p = uri or ""
+        if not os.path.isabs(p):
+            p = os.path.join(self.root, p)
+        p = os.path.abspath(p)
+        if not (p == self.root or p.startswith(self.root + os.sep)):
+            raise ToolError(f"path {uri!r} escapes the workspace root")
+        return p
+
+    # ---- validation (reference validateParams semantics) ----
+    def validate_params(self, tool_name: str, raw: Dict[str, str]) -> Dict[str, Any]:
+        if not is_builtin_tool(tool_name):
+            raise ToolError(f"unknown tool {tool_name!r}")
+        allowed = set(BUILTIN_TOOLS[tool_name])
+        params = {k: v for k, v in raw.items() if k in allowed}
+        required = {
+            "read_file": ["uri"], "get_dir_tree": ["uri"], "search_pathnames_only": ["query"],
+            "search_for_files": ["query"], "search_in_file": ["uri", "query"],
+            "read_lint_errors": ["uri"], "create_file_or_folder": ["uri"],
+            "delete_file_or_folder": ["uri"], "edit_file": ["uri", "search_replace_blocks"],
+            "rewrite_file": ["uri", "new_content"], "run_command": ["command"],
+            "run_persistent_command": ["command", "persistent_terminal_id"],
+            "kill_persistent_terminal": ["persistent_terminal_id"],
+            "spawn_subagent": ["label", "task_prompt"], "skill": ["name"],
+        }.get(tool_name, [])
+        missing = [r for r in required if not params.get(r)]
+        if missing:
+            raise ToolError(f"missing required param(s) {missing} for {tool_name}")
+        for bkey in ("is_regex", "is_recursive", "headless", "crawl_links"):
+            if bkey in params:
+                params[bkey] = str(params[bkey]).strip().lower() in ("true", "1", "yes")
+        for ikey in ("start_line", "end_line", "page_number", "timeout_ms", "max_results"):
+            if ikey in params and params[ikey] not in (None, ""):
+                try:
+                    params[ikey] = int(str(params[ikey]).strip())
+                except ValueError:
+                    raise ToolError(f"param {ikey} must be an integer")
+        return params
+
+    def approval_type(self, tool_name: str) -> Optional[str]:
+        return APPROVAL_TYPE_OF_TOOL.get(tool_name)
+
+    # ---- execution ----
+    def call_tool(self, tool_name: str, params: Dict[str, Any]) -> ToolResult:
+        fn = getattr(self, f"_tool_{tool_name}", None)
+        if fn is None:
+            raise ToolError(f"tool {tool_name!r} is not available in this environment "
+                            "(reference serves it via a localhost sidecar; offline here)")
+        return fn(params)
+
+    # --- context-gathering ---
+    def _tool_read_file(self, p):
+        path = self._resolve(p["uri"])
+        with open(path, "r", encoding="utf-8", errors="replace") as f:
+            lines = f.readlines()
+        s = p.get("start_line") or 1
+        e = p.get("end_line") or len(lines)
+        text = "".join(lines[s - 1: e])[:MAX_FILE_CHARS]
+        return ToolResult("read_file", {"uri": p["uri"], "lines": len(lines)}, text)
+
+    def _tool_ls_dir(self, p):
+        path = self._resolve(p.get("uri") or "")
+        entries = sorted(os.listdir(path))
+        page = (p.get("page_number") or 1) - 1
+        chunk = entries[page * PAGE_SIZE_RESULTS: (page + 1) * PAGE_SIZE_RESULTS]
+        text = "\n".join((e + "/" if os.path.isdir(os.path.join(path, e)) else e) for e in chunk)
+        return ToolResult("ls_dir", {"entries": chunk}, text)
+
+    def _tool_get_dir_tree(self, p):
+        # caps from directoryStrService.ts: MAX_FILES_TOTAL=1000, depth 3
+        root = self._resolve(p.get("uri") or "")
+        out: List[str] = []
+        count = 0
+        for dirpath, dirnames, filenames in os.walk(root):
+            depth = dirpath[len(root):].count(os.sep)
+            if depth >= 3:
+                dirnames[:] = []
+                continue
+            dirnames[:] = [d for d in sorted(dirnames) if not d.startswith(".") and d != "__pycache__"]
+            indent = "  " * depth
+            out.append(f"{indent}{os.path.basename(dirpath) or '.'}/")
+            for fn in sorted(filenames):
+                if count >= 1000:
+                    out.append(f"{indent}  ...[truncated at 1000 files]")
+                    return ToolResult("get_dir_tree", {}, "\n".join(out))
+                out.append(f"{indent}  {fn}")
+                count += 1
+        return ToolResult("get_dir_tree", {}, "\n".join(out))
+
+    def _tool_search_pathnames_only(self, p):
+        q = p["query"]
+        include = p.get("include_pattern")
+        hits = []
+        for dirpath, dirnames, filenames in os.walk(self.root):
+            dirnames[:] = [d for d in dirnames if not d.startswith(".") and d not in ("__pycache__", "node_modules")]
+            for fn in filenames:
+                rel = os.path.relpath(os.path.join(dirpath, fn), self.root)
+                if q.lower() in rel.lower() and (not include or fnmatch.fnmatch(rel, include)):
+                    hits.append(rel)
+        page = (p.get("page_number") or 1) - 1
+        chunk = hits[page * PAGE_SIZE_RESULTS: (page + 1) * PAGE_SIZE_RESULTS]
+        return ToolResult("search_pathnames_only", {"hits": chunk}, "\n".join(chunk) or "(no matches)")
+
+    def _tool_search_for_files(self, p):
+        q = p["query"]
+        rx = re.compile(q) if p.get("is_regex") else None
+        base = self._resolve(p.get("search_in_folder") or "")
+        hits = []
+        for dirpath, dirnames, filenames in os.walk(base):
+            dirnames[:] = [d for d in dirnames if not d.startswith(".") and d not in ("__pycache__", "node_modules")]
+            for fn in filenames:
+                fp = os.path.join(dirpath, fn)
+                try:
+                    if os.path.getsize(fp) > 2_000_000:
+                        continue
+                    with open(fp, "r", encoding="utf-8", errors="ignore") as f:
+                        content = f.read()
+                except OSError:
+                    continue
+                if (rx.search(content) if rx else q in content):
+                    hits.append(os.path.relpath(fp, self.root))
+                if len(hits) >= PAGE_SIZE_RESULTS:
+                    break
+        return ToolResult("search_for_files", {"hits": hits}, "\n".join(hits) or "(no matches)")
+
+    def _tool_search_in_file(self, p):
+        path = self._resolve(p["uri"])
+        q = p["query"]
+        rx = re.compile(q) if p.get("is_regex") else None
+        lines = open(path, "r", encoding="utf-8", errors="replace").readlines()
+        nums = [i + 1 for i, ln in enumerate(lines) if (rx.search(ln) if rx else q in ln)]
+        return ToolResult("search_in_file", {"lines": nums}, ", ".join(map(str, nums)) or "(no matches)")
+
+    def _tool_read_lint_errors(self, p):
+        # local lint: python syntax check for .py; otherwise none available
+        path = self._resolve(p["uri"])
+        errors: List[str] = []
+        if path.endswith(".py"):
+            import ast
+            try:
+                ast.parse(open(path, encoding="utf-8", errors="replace").read())
+            except SyntaxError as e:
+                errors.append(f"{e.lineno}: {e.msg}")
+        text = "\n".join(errors) if errors else "No lint errors found."
+        return ToolResult("read_lint_errors", {"errors": errors}, text)
+
+    # --- editing ---
+    def _tool_create_file_or_folder(self, p):
+        uri = p["uri"]
+        path = self._resolve(uri)
+        if uri.endswith("/"):
+            os.makedirs(path, exist_ok=True)
+        else:
+            os.makedirs(os.path.dirname(path), exist_ok=True)
+            if not os.path.exists(path):
+                open(path, "w").close()
+        return ToolResult("create_file_or_folder", {"uri": uri}, f"Created {uri}")
+
+    def _tool_delete_file_or_folder(self, p):
+        path = self._resolve(p["uri"])
+        if os.path.isdir(path):
+            if p.get("is_recursive"):
+                shutil.rmtree(path)
+            else:
+                os.rmdir(path)
+        elif os.path.exists(path):
+            os.unlink(path)
+        return ToolResult("delete_file_or_folder", {}, f"Deleted {p['uri']}")
+
+    def _tool_rewrite_file(self, p):
+        path = self._resolve(p["uri"])
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        with open(path, "w", encoding="utf-8") as f:
+            f.write(p["new_content"])
+        return ToolResult("rewrite_file", {}, f"Rewrote {p['uri']}")
+
+    def _tool_edit_file(self, p):
+        path = self._resolve(p["uri"])
+        content = open(path, "r", encoding="utf-8", errors="replace").read()
+        for orig, updated in parse_search_replace_blocks(p["search_replace_blocks"]):
+            if orig not in content:
+                raise ToolError(f"ORIGINAL block not found in {p['uri']}: {orig[:80]!r}")
+            content = content.replace(orig, updated, 1)
+        with open(path, "w", encoding="utf-8") as f:
+            f.write(content)
+        return ToolResult("edit_file", {}, f"Applied edits to {p['uri']}")
+
+    # --- terminal ---
+    def _tool_run_command(self, p):
+        cwd = self._resolve(p.get("cwd") or "")
+        try:
+            r = subprocess.run(["/bin/bash", "-c", p["command"]], cwd=cwd,
+                               capture_output=True, text=True,
+                               timeout=MAX_TERMINAL_INACTIVE_TIME_S * 4)
+            out = (r.stdout + r.stderr)[:MAX_TERMINAL_CHARS]
+            text = out + (f"\n[exit code {r.returncode}]" if r.returncode else "")
+        except subprocess.TimeoutExpired as e:
+            text = ((e.stdout or "") + (e.stderr or ""))[:MAX_TERMINAL_CHARS] + "\n[command timed out]"
+        return ToolResult("run_command", {}, text)
+
+    def _tool_open_persistent_terminal(self, p):
+        term = PersistentTerminal(self._resolve(p.get("cwd") or ""))
+        self.terminals[term.id] = term
+        return ToolResult("open_persistent_terminal", {"persistent_terminal_id": term.id},
+                          f"Opened persistent terminal {term.id}")
+
+    def _tool_run_persistent_command(self, p):
+        term = self.terminals.get(p["persistent_terminal_id"])
+        if not term:
+            raise ToolError(f"no persistent terminal {p['persistent_terminal_id']!r}")
+        return ToolResult("run_persistent_command", {}, term.run(p["command"]))
+
+    def _tool_kill_persistent_terminal(self, p):
+        term = self.terminals.pop(p["persistent_terminal_id"], None)
+        if term:
+            term.kill()
+        return ToolResult("kill_persistent_terminal", {}, "Terminal killed")
+
+    # --- agents ---
+    def _tool_spawn_subagent(self, p):
+        if self._subagents is None:
+            raise ToolError("subagent runner not configured")
+        from ..agents.subagents import SubagentInput
+        inp = SubagentInput(
+            label=p["label"], task_prompt=p["task_prompt"],
+            summary_prompt=p.get("summary_prompt") or "Summarize your findings concisely.",
+            timeout_ms=p.get("timeout_ms") or 300000,
+            allowed_tools=(p.get("allowed_tools") or "").split(",") if p.get("allowed_tools") else None,
+        )
+        res = self._subagents.spawn(inp)
+        if not res.success:
+            raise ToolError(res.error or "subagent failed")
+        return ToolResult("spawn_subagent", {"taskId": res.task_id}, res.summary)
+
+    def _tool_skill(self, p):
+        from .skills import SkillService
+        svc = SkillService(self.root)
+        skill = svc.get_skill(p["name"])
+        if skill is None:
+            raise ToolError(f"skill {p['name']!r} not found")
+        return ToolResult("skill", {"name": skill.name}, skill.content)

@@ -1,0 +1,189 @@
+"""Agent-loop tests: tool-use while-loop, approval gate, retry ladder, tracing."""
+
+import pytest
+
+from senweaver_amd.chat import (
+    ChatThreadService,
+    GlobalSettings,
+    TPMRateLimiter,
+    get_retry_delay_ms,
+    is_context_length_error,
+    is_rate_limit_error,
+)
+from senweaver_amd.storage import MemoryStorage
+from senweaver_amd.tools import ToolsService
+from senweaver_amd.trace import TraceCollector
+from senweaver_amd.transport import LLMMessageService
+
+
+class ScriptedBackend:
+    """Streams scripted responses, one per send (token-chunked)."""
+
+    def __init__(self, responses):
+        self.responses = list(responses)
+        self.calls = 0
+
+    def stream_generate(self, prompt, max_new_tokens, should_stop, on_chunk):
+        self.calls += 1
+        if not self.responses:
+            return ""
+        item = self.responses.pop(0)
+        if isinstance(item, Exception):
+            raise item
+        acc = ""
+        for i in range(0, len(item), 7):
+            if should_stop():
+                return acc
+            acc = item[: i + 7]
+            on_chunk(acc)
+        return item
+
+
+def make_service(tmp_path, responses, auto_approve=None, sleeps=None):
+    (tmp_path / "hello.txt").write_text("salutations world\n")
+    backend = ScriptedBackend(responses)
+    llm = LLMMessageService(backend)
+    tools = ToolsService(str(tmp_path))
+    tc = TraceCollector(storage=MemoryStorage())
+    recorded_sleeps = sleeps if sleeps is not None else []
+    svc = ChatThreadService(llm, tools, tc,
+                            settings=GlobalSettings(auto_approve=auto_approve or {}),
+                            sleep=lambda s: recorded_sleeps.append(s))
+    return svc, backend, tc
+
+
+def test_tool_loop_end_to_end(tmp_path):
+    responses = [
+        "Let me check the file <read_file><uri>hello.txt</uri></read_file>",
+        "The file says: salutations world. Done.",
+    ]
+    svc, backend, tc = make_service(tmp_path, responses)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "what does hello.txt say?")
+    roles = [m.role for m in thread.messages]
+    assert roles == ["checkpoint", "user", "assistant", "tool", "assistant", "checkpoint"]
+    tool_msg = thread.messages[3]
+    assert tool_msg.tool_success is True
+    assert "salutations world" in tool_msg.content
+    assert backend.calls == 2
+    # trace recorded spans + reward
+    trace = tc.get_all_traces()[0]
+    types = [s.type for s in trace.spans]
+    assert "user_message" in types and "tool_call" in types and "assistant_message" in types
+    assert trace.summary.total_tool_calls == 1
+    assert trace.summary.tool_calls_succeeded == 1
+    assert trace.summary.final_reward is not None
+
+
+def test_approval_gate_pauses_and_resumes(tmp_path):
+    responses = [
+        "<run_command><command>echo approved-$((40+2))</command></run_command>",
+        "The command printed approved-42.",
+    ]
+    svc, backend, tc = make_service(tmp_path, responses)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "run the thing")
+    # paused awaiting approval (terminal class not auto-approved)
+    assert thread.pending_tool is not None
+    assert thread.messages[-2].role == "tool_request"  # before the final checkpoint
+    svc.approve_latest_tool_request(thread.id)
+    tool_msgs = [m for m in thread.messages if m.role == "tool"]
+    assert tool_msgs and "approved-42" in tool_msgs[-1].content
+    assert backend.calls == 2
+
+
+def test_auto_approve_skips_gate(tmp_path):
+    responses = [
+        "<run_command><command>echo fast-$((1+1))</command></run_command>",
+        "done",
+    ]
+    svc, backend, tc = make_service(tmp_path, responses, auto_approve={"terminal": True})
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "run it")
+    assert thread.pending_tool is None
+    tool_msgs = [m for m in thread.messages if m.role == "tool"]
+    assert "fast-2" in tool_msgs[0].content
+
+
+def test_invalid_params_feed_back(tmp_path):
+    responses = [
+        "<read_file></read_file>",   # missing uri
+        "Sorry, retrying without the tool.",
+    ]
+    svc, backend, tc = make_service(tmp_path, responses)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "read something")
+    tool_msgs = [m for m in thread.messages if m.role == "tool"]
+    assert tool_msgs[0].tool_success is False
+    assert "Invalid parameters" in tool_msgs[0].content
+    assert backend.calls == 2  # loop continued after invalid params
+
+
+def test_tool_failure_becomes_tool_error(tmp_path):
+    responses = [
+        "<read_file><uri>missing.txt</uri></read_file>",
+        "The file does not exist.",
+    ]
+    svc, backend, tc = make_service(tmp_path, responses)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "read missing file")
+    tool_msgs = [m for m in thread.messages if m.role == "tool"]
+    assert tool_msgs[0].tool_success is False
+    trace = tc.get_all_traces()[0]
+    assert trace.summary.tool_calls_failed == 1
+
+
+def test_retry_ladder_on_transient_errors(tmp_path):
+    sleeps = []
+    responses = [
+        RuntimeError("connection reset"),
+        RuntimeError("connection reset again"),
+        "All good now.",
+    ]
+    svc, backend, tc = make_service(tmp_path, responses, sleeps=sleeps)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "hello")
+    assert thread.messages[-2].content == "All good now."
+    assert len(sleeps) == 2  # two backoff sleeps
+    assert sleeps[0] == pytest.approx(get_retry_delay_ms(1, False) / 1000)
+
+
+def test_retry_exhaustion_records_error(tmp_path):
+    responses = [RuntimeError(f"boom {i}") for i in range(10)]
+    svc, backend, tc = make_service(tmp_path, responses)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "hello")
+    assert thread.messages[-2].content.startswith("[error]")
+    trace = tc.get_all_traces()[0]
+    assert trace.summary.has_errors
+
+
+def test_rate_limit_cooldown_and_recovery(tmp_path):
+    sleeps = []
+    responses = [RuntimeError("429 rate limit, retry-after: 2"), "recovered"]
+    svc, backend, tc = make_service(tmp_path, responses, sleeps=sleeps)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "hello")
+    assert thread.messages[-2].content == "recovered"
+    assert any(abs(s - 2.0) < 0.01 for s in sleeps)  # retry-after honored
+
+
+def test_error_string_classification():
+    assert is_context_length_error("Error 400: maximum context length exceeded")
+    assert is_context_length_error("input is too long for this model")
+    assert not is_context_length_error("connection refused")
+    assert is_rate_limit_error("HTTP 429 Too Many Requests")
+    assert not is_rate_limit_error("500 internal error")
+
+
+def test_tpm_rate_limiter_reactive():
+    t = {"now": 0.0}
+    limiter = TPMRateLimiter(clock=lambda: t["now"])
+    assert limiter.get_wait_time_ms("default") == 0
+    cd = limiter.handle_rate_limit_error("default", "429 retry-after: 5")
+    assert cd == 5000
+    assert limiter.get_wait_time_ms("default") == 5000
+    t["now"] = 3000
+    assert limiter.get_wait_time_ms("default") == 2000
+    limiter.record_success("default")
+    assert limiter.get_wait_time_ms("default") == 0
