@@ -74,11 +74,22 @@ def tiny_debug(vocab: int = 512) -> ModelConfig:
     )
 
 
+def tiny_moe(vocab: int = 512) -> ModelConfig:
+    """CPU-testable MoE model (grouped-GEMM path)."""
+    return ModelConfig(
+        name="tiny-moe", hidden_size=256, intermediate_size=512,
+        num_layers=2, num_heads=2, num_kv_heads=1, head_dim=128,
+        vocab_size=vocab, max_position=2048, num_experts=4,
+        num_experts_per_tok=2,
+    )
+
+
 PRESETS = {
     "llama-3-8b": llama3_8b,
     "llama-3-70b": llama3_70b,
     "mixtral-8x7b": mixtral_8x7b,
     "tiny-debug": tiny_debug,
+    "tiny-moe": tiny_moe,
 }
 
 

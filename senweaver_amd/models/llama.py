@@ -19,8 +19,10 @@ from typing import List, Optional
 import torch
 
 from .. import ops
-from ..engine.kvcache import PagedKVCache
 from .config import ModelConfig
+
+if False:  # type-checking only; avoids the engine<->models import cycle
+    from ..engine.kvcache import PagedKVCache
 
 
 class LlamaModel:
@@ -51,14 +53,21 @@ class LlamaModel:
         self.embed = W(c.vocab_size, c.hidden_size)
         self.layers = []
         for _ in range(c.num_layers):
-            self.layers.append({
+            layer = {
                 "input_norm": torch.ones(c.hidden_size, dtype=dtype, device=self.device),
                 "qkv": W(c.q_size + 2 * c.kv_size, c.hidden_size),
                 "o": W(c.hidden_size, c.q_size),
                 "post_norm": torch.ones(c.hidden_size, dtype=dtype, device=self.device),
-                "gateup": W(2 * c.intermediate_size, c.hidden_size),
-                "down": W(c.hidden_size, c.intermediate_size),
-            })
+            }
+            if c.num_experts > 0:
+                # Mixtral-style MoE: fused gate|up per expert + down, router
+                layer["router"] = W(c.num_experts, c.hidden_size)
+                layer["w13"] = W(c.num_experts, 2 * c.intermediate_size, c.hidden_size)
+                layer["w2"] = W(c.num_experts, c.hidden_size, c.intermediate_size)
+            else:
+                layer["gateup"] = W(2 * c.intermediate_size, c.hidden_size)
+                layer["down"] = W(c.hidden_size, c.intermediate_size)
+            self.layers.append(layer)
         self.final_norm = torch.ones(c.hidden_size, dtype=dtype, device=self.device)
         self.lm_head = self.embed if c.tie_embeddings else W(c.vocab_size, c.hidden_size)
         self.cos_sin = ops.rope_tables(c.max_position, c.head_dim, c.rope_theta).to(self.device)
@@ -69,7 +78,7 @@ class LlamaModel:
     # If `cache` is given, K/V of every position are appended (seqs maps
     # batch row -> cache sequence id).
     # ------------------------------------------------------------------
-    def prefill(self, tokens: torch.Tensor, cache: Optional[PagedKVCache] = None,
+    def prefill(self, tokens: torch.Tensor, cache=None,
                 seqs: Optional[List[int]] = None,
                 real_lens: Optional[List[int]] = None) -> torch.Tensor:
         c = self.config
@@ -103,17 +112,61 @@ class LlamaModel:
             attn = attn.transpose(1, 2).reshape(T, c.q_size).contiguous()
             o = ops.gemm_bt(attn, L["o"])
             h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
-            gateup = ops.gemm_bt(h, L["gateup"])
-            act = ops.swiglu(gateup)
-            hidden = ops.gemm_bt(act, L["down"])
+            hidden = self._ffn(h, L)
         h = ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)
         return h.reshape(B, S, c.hidden_size)
+
+    def _ffn(self, h: torch.Tensor, L: dict) -> torch.Tensor:
+        c = self.config
+        if c.num_experts == 0:
+            gateup = ops.gemm_bt(h, L["gateup"])
+            act = ops.swiglu(gateup)
+            return ops.gemm_bt(act, L["down"])
+        return self._moe_ffn(h, L)
+
+    def _moe_ffn(self, h: torch.Tensor, L: dict) -> torch.Tensor:
+        """Mixtral top-k routed FFN on the grouped-GEMM kernel.
+
+        Routing/sort/combine are small tensor ops; the expert math runs as
+        ONE grouped GEMM per projection (tokens sorted by expert, tile map
+        built host-side so router imbalance costs no idle blocks).
+        """
+        c = self.config
+        T = h.shape[0]
+        k = c.num_experts_per_tok
+        # router is [E=8, H]: far below the MFMA tile (N=8) — a plain skinny
+        # library matmul, not a hot op
+        router_logits = h @ L["router"].t()  # [T, E]
+        probs = torch.softmax(router_logits.float(), dim=-1)
+        topw, topi = probs.topk(k, dim=-1)           # [T, k]
+        topw = topw / topw.sum(dim=-1, keepdim=True)  # renormalize (Mixtral)
+        flat_expert = topi.reshape(-1)                # [T*k]
+        flat_token = torch.arange(T, device=h.device).repeat_interleave(k)
+        order = torch.argsort(flat_expert, stable=True)
+        sorted_token = flat_token[order]
+        sorted_weight = topw.reshape(-1)[order]
+        counts = torch.bincount(flat_expert, minlength=c.num_experts)
+        seg_starts = [0]
+        for e in range(c.num_experts):
+            seg_starts.append(seg_starts[-1] + int(counts[e]))
+        Tk = T * k
+        pad = 128  # last-tile overread margin for the grouped kernel
+        a_sorted = torch.zeros(Tk + pad, c.hidden_size, dtype=h.dtype, device=h.device)
+        a_sorted[:Tk] = h[sorted_token]
+        gateup = ops.grouped_gemm_bt(a_sorted, L["w13"], seg_starts)
+        act = ops.swiglu(gateup[:Tk])
+        act_p = torch.zeros(Tk + pad, c.intermediate_size, dtype=h.dtype, device=h.device)
+        act_p[:Tk] = act
+        down = ops.grouped_gemm_bt(act_p, L["w2"], seg_starts)[:Tk]
+        out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=h.device)
+        out.index_add_(0, sorted_token, down.float() * sorted_weight.unsqueeze(1))
+        return out.to(h.dtype)
 
     # ------------------------------------------------------------------
     # Decode: one new token per sequence.  tokens [B], positions [B].
     # ------------------------------------------------------------------
     def decode_step(self, tokens: torch.Tensor, positions: torch.Tensor,
-                    cache: PagedKVCache, seqs: List[int]) -> torch.Tensor:
+                    cache, seqs: List[int]) -> torch.Tensor:
         c = self.config
         B = tokens.shape[0]
         hidden = self.embed[tokens.long()]
@@ -141,9 +194,7 @@ class LlamaModel:
             attn = attn.reshape(B, c.q_size)
             o = ops.gemm_bt(attn, L["o"])
             h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
-            gateup = ops.gemm_bt(h, L["gateup"])
-            act = ops.swiglu(gateup)
-            hidden = ops.gemm_bt(act, L["down"])
+            hidden = self._ffn(h, L)
         h = ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)
         return h  # [B, hidden]
 

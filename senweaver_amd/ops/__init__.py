@@ -104,6 +104,36 @@ def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return c[:M] if pad else c
 
 
+def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
+                    ) -> torch.Tensor:
+    """Segment-grouped C = A_seg @ W[e]^T for MoE.
+
+    ``a_sorted``: [T_pad, K] tokens sorted by expert (padded by >=128 rows);
+    ``w``: [E, N, K]; ``seg_starts_cpu``: list/1D tensor of E+1 cumulative
+    row offsets (host side).  Rows outside segments are left unwritten.
+    """
+    E = w.shape[0]
+    starts = [int(x) for x in seg_starts_cpu]
+    if not _on_gpu(a_sorted):
+        out = torch.zeros(a_sorted.shape[0], w.shape[1], dtype=a_sorted.dtype)
+        for e in range(E):
+            s, t = starts[e], starts[e + 1]
+            if t > s:
+                out[s:t] = ref.gemm_bt_ref(a_sorted[s:t], w[e])
+        return out
+    tile_expert, tile_m0 = [], []
+    for e in range(E):
+        s, t = starts[e], starts[e + 1]
+        for m0 in range(s, t, 128):
+            tile_expert.append(e)
+            tile_m0.append(m0)
+    dev = a_sorted.device
+    te = torch.tensor(tile_expert, dtype=torch.int32, device=dev)
+    tm = torch.tensor(tile_m0, dtype=torch.int32, device=dev)
+    ends = torch.tensor(starts[1:], dtype=torch.int32, device=dev)
+    return hip_ext().grouped_gemm_bt(a_sorted.contiguous(), w.contiguous(), te, tm, ends)
+
+
 def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional[float] = None,
              vt: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Causal GQA attention. q [B,H,S,D], k/v [B,Hk,S,D] (S % 64 == 0 on GPU).

@@ -16,6 +16,7 @@ extern "C" __global__ void argmax_rows_kernel(const ushort*, int*, int);
 extern "C" __global__ void target_logprob_kernel(const ushort*, const int*, float*, int);
 extern "C" __global__ void gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_kernel(const ushort*, const ushort*, float*, int, int, int, int);
 extern "C" __global__ void gemv_reduce_kernel(const float*, ushort*, int, int, int);
 extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
@@ -164,6 +165,30 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+// ---------------- Grouped GEMM (MoE) ----------------
+torch::Tensor grouped_gemm_bt(torch::Tensor A, torch::Tensor W,
+                              torch::Tensor tile_expert, torch::Tensor tile_m0,
+                              torch::Tensor seg_ends) {
+  check_bf16(A, "A");
+  check_bf16(W, "W");
+  TORCH_CHECK(tile_expert.scalar_type() == torch::kInt32 && tile_expert.is_contiguous());
+  TORCH_CHECK(tile_m0.scalar_type() == torch::kInt32 && tile_m0.is_contiguous());
+  TORCH_CHECK(seg_ends.scalar_type() == torch::kInt32 && seg_ends.is_contiguous());
+  const int K = A.size(1);
+  const int N = W.size(1);
+  TORCH_CHECK(W.size(2) == K && N % 128 == 0 && K % 64 == 0);
+  const int n_mtiles = tile_expert.size(0);
+  const int tiles_n = N / 128;
+  auto C = torch::empty({A.size(0), (long)N}, A.options());
+  if (n_mtiles > 0) {
+    grouped_gemm_bt_bf16_kernel<<<dim3(n_mtiles * tiles_n), dim3(256), 0, cur_stream()>>>(
+        bf16_ptr(A), bf16_ptr(W), bf16_mut(C), tile_expert.data_ptr<int>(),
+        tile_m0.data_ptr<int>(), seg_ends.data_ptr<int>(), N, K, tiles_n);
+    HIP_CHECK_KERNEL();
+  }
+  return C;
+}
+
 // ---------------- Flash attention prefill ----------------
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
                        double scale) {
@@ -236,6 +261,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu", &swiglu, "silu(gate)*up from fused gateup");
   m.def("add_bf16", &add_bf16, "a + b (bf16)");
   m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");
+  m.def("grouped_gemm_bt", &grouped_gemm_bt, "segment-grouped C = A @ W[e]^T (MoE)");
   m.def("attn_fwd", &attn_fwd, "causal flash attention fwd (D=128, GQA)");
   m.def("paged_decode_attn", &paged_decode_attn, "paged decode attention");
   m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");
