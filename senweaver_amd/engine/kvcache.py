@@ -16,11 +16,13 @@ PAGE_SIZE = 16
 
 
 class PagedKVCache:
-    def __init__(self, config, num_pages: int, device, dtype=torch.bfloat16) -> None:
+    def __init__(self, config, num_pages: int, device, dtype=torch.bfloat16,
+                 num_kv_heads=None) -> None:
         self.config = config
         self.num_pages = num_pages
         self.device = device
-        shape = (num_pages, PAGE_SIZE, config.num_kv_heads, config.head_dim)
+        self.num_kv_heads = num_kv_heads if num_kv_heads is not None else config.num_kv_heads
+        shape = (num_pages, PAGE_SIZE, self.num_kv_heads, config.head_dim)
         self.k = [torch.zeros(shape, dtype=dtype, device=device) for _ in range(config.num_layers)]
         self.v = [torch.zeros(shape, dtype=dtype, device=device) for _ in range(config.num_layers)]
         self._free = list(range(num_pages - 1, -1, -1))
@@ -64,8 +66,8 @@ class PagedKVCache:
         self._ensure_capacity(seq, start + T)
         slots = self.slot_ids(seq, start, T)
         cfg = self.config
-        kflat = self.k[layer].view(self.num_pages * PAGE_SIZE, cfg.num_kv_heads, cfg.head_dim)
-        vflat = self.v[layer].view(self.num_pages * PAGE_SIZE, cfg.num_kv_heads, cfg.head_dim)
+        kflat = self.k[layer].view(self.num_pages * PAGE_SIZE, self.num_kv_heads, cfg.head_dim)
+        vflat = self.v[layer].view(self.num_pages * PAGE_SIZE, self.num_kv_heads, cfg.head_dim)
         kflat.index_copy_(0, slots, k)
         vflat.index_copy_(0, slots, v)
         if advance_len:  # the last layer advances the logical length

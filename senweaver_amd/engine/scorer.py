@@ -36,14 +36,14 @@ def _pad64(n: int) -> int:
 
 class LlamaBackend:
     def __init__(self, config: ModelConfig | str = "llama-3-8b", device: Optional[str] = None,
-                 seed: int = 0, max_seq: int = 2048, micro_batch: int = 8) -> None:
+                 seed: int = 0, max_seq: int = 2048, micro_batch: int = 8, tp=None) -> None:
         if isinstance(config, str):
             config = get_config(config)
         if device is None:
             device = "cuda:0" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
         self.config = config
-        self.model = LlamaModel(config, device=self.device, seed=seed)
+        self.model = LlamaModel(config, device=self.device, seed=seed, tp=tp)
         self.tokenizer = tok.HashTokenizer(config.vocab_size)
         self.max_seq = min(max_seq, config.max_position)
         self.micro_batch = micro_batch
@@ -156,7 +156,8 @@ class LlamaBackend:
         S = _pad64(real)
         total = real + max_new_tokens + 1
         pages_needed = (total + PAGE_SIZE - 1) // PAGE_SIZE + 2
-        cache = PagedKVCache(self.config, pages_needed, self.device)
+        cache = PagedKVCache(self.config, pages_needed, self.device,
+                             num_kv_heads=self.model.local_kv_heads)
         seq = cache.new_seq()
         tokens = torch.zeros(1, S, dtype=torch.long)
         tokens[0, :real] = torch.tensor(ids, dtype=torch.long)
@@ -188,7 +189,8 @@ class LlamaBackend:
         S = _pad64(real)
         total = real + max_new_tokens + 1
         pages_needed = (total + PAGE_SIZE - 1) // PAGE_SIZE + 2
-        cache = PagedKVCache(self.config, pages_needed, self.device)
+        cache = PagedKVCache(self.config, pages_needed, self.device,
+                             num_kv_heads=self.model.local_kv_heads)
         seq = cache.new_seq()
         tokens = torch.zeros(1, S, dtype=torch.long)
         tokens[0, :real] = torch.tensor(ids, dtype=torch.long)

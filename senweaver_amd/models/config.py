@@ -84,12 +84,22 @@ def tiny_moe(vocab: int = 512) -> ModelConfig:
     )
 
 
+def tiny_tp(vocab: int = 512) -> ModelConfig:
+    """CPU-testable TP model (heads divisible by 2)."""
+    return ModelConfig(
+        name="tiny-tp", hidden_size=512, intermediate_size=512,
+        num_layers=2, num_heads=4, num_kv_heads=2, head_dim=128,
+        vocab_size=vocab, max_position=2048,
+    )
+
+
 PRESETS = {
     "llama-3-8b": llama3_8b,
     "llama-3-70b": llama3_70b,
     "mixtral-8x7b": mixtral_8x7b,
     "tiny-debug": tiny_debug,
     "tiny-moe": tiny_moe,
+    "tiny-tp": tiny_tp,
 }
 
 

@@ -26,11 +26,25 @@ if False:  # type-checking only; avoids the engine<->models import cycle
 
 
 class LlamaModel:
-    def __init__(self, config: ModelConfig, device="cpu", dtype=torch.bfloat16, seed: int = 0):
+    def __init__(self, config: ModelConfig, device="cpu", dtype=torch.bfloat16, seed: int = 0,
+                 tp=None):
+        from ..parallel.tp import TPContext, shard_gateup, shard_qkv, shard_rows
         self.config = config
         self.device = torch.device(device)
         self.dtype = dtype
+        self.tp = tp if tp is not None else TPContext()
         c = config
+        tpw = self.tp.world
+        if tpw > 1:
+            assert c.num_heads % tpw == 0 and c.num_kv_heads % tpw == 0, \
+                f"TP={tpw} must divide heads ({c.num_heads}/{c.num_kv_heads})"
+            assert c.intermediate_size % tpw == 0
+            assert c.num_experts == 0, "MoE+TP combo not supported yet"
+        self.local_heads = c.num_heads // tpw
+        self.local_kv_heads = c.num_kv_heads // tpw
+        self.local_q_size = self.local_heads * c.head_dim
+        self.local_kv_size = self.local_kv_heads * c.head_dim
+        self.local_inter = c.intermediate_size // tpw
         if self.device.type == "cuda":
             # Device-side init: seconds for 8B instead of minutes of CPU RNG +
             # PCIe transfer.  Same seed + same call order on every rank =>
@@ -53,20 +67,31 @@ class LlamaModel:
         self.embed = W(c.vocab_size, c.hidden_size)
         self.layers = []
         for _ in range(c.num_layers):
+            # full weights are generated with the shared seed, then sliced to
+            # this rank's shard: every TP degree sees the SAME model, and all
+            # ranks agree by construction
+            qkv_full = W(c.q_size + 2 * c.kv_size, c.hidden_size)
+            o_full = W(c.hidden_size, c.q_size)
             layer = {
                 "input_norm": torch.ones(c.hidden_size, dtype=dtype, device=self.device),
-                "qkv": W(c.q_size + 2 * c.kv_size, c.hidden_size),
-                "o": W(c.hidden_size, c.q_size),
+                "qkv": shard_qkv(qkv_full, c.num_heads, c.num_kv_heads, c.head_dim, self.tp)
+                        if tpw > 1 else qkv_full,
+                "o": shard_rows(o_full, self.tp, dim=1) if tpw > 1 else o_full,
                 "post_norm": torch.ones(c.hidden_size, dtype=dtype, device=self.device),
             }
+            del qkv_full, o_full
             if c.num_experts > 0:
                 # Mixtral-style MoE: fused gate|up per expert + down, router
                 layer["router"] = W(c.num_experts, c.hidden_size)
                 layer["w13"] = W(c.num_experts, 2 * c.intermediate_size, c.hidden_size)
                 layer["w2"] = W(c.num_experts, c.hidden_size, c.intermediate_size)
             else:
-                layer["gateup"] = W(2 * c.intermediate_size, c.hidden_size)
-                layer["down"] = W(c.hidden_size, c.intermediate_size)
+                gateup_full = W(2 * c.intermediate_size, c.hidden_size)
+                down_full = W(c.hidden_size, c.intermediate_size)
+                layer["gateup"] = (shard_gateup(gateup_full, c.intermediate_size, self.tp)
+                                   if tpw > 1 else gateup_full)
+                layer["down"] = shard_rows(down_full, self.tp, dim=1) if tpw > 1 else down_full
+                del gateup_full, down_full
             self.layers.append(layer)
         self.final_norm = torch.ones(c.hidden_size, dtype=dtype, device=self.device)
         self.lm_head = self.embed if c.tie_embeddings else W(c.vocab_size, c.hidden_size)
@@ -95,22 +120,24 @@ class LlamaModel:
                 h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
             else:
                 h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
-            qkv = ops.gemm_bt(h, L["qkv"])  # [T, q+2kv]
-            q = qkv[:, : c.q_size].reshape(T, c.num_heads, c.head_dim).contiguous()
-            k = qkv[:, c.q_size: c.q_size + c.kv_size].reshape(T, c.num_kv_heads, c.head_dim).contiguous()
-            v = qkv[:, c.q_size + c.kv_size:].reshape(T, c.num_kv_heads, c.head_dim).contiguous()
+            qkv = ops.gemm_bt(h, L["qkv"])  # [T, local q+2kv]
+            qs, kvs = self.local_q_size, self.local_kv_size
+            q = qkv[:, :qs].reshape(T, self.local_heads, c.head_dim).contiguous()
+            k = qkv[:, qs: qs + kvs].reshape(T, self.local_kv_heads, c.head_dim).contiguous()
+            v = qkv[:, qs + kvs:].reshape(T, self.local_kv_heads, c.head_dim).contiguous()
             ops.rope_inplace(q, k, self.cos_sin, positions)
             if cache is not None:
                 for b in range(B):
                     n = real_lens[b] if real_lens is not None else S
                     cache.append(li, seqs[b], k[b * S: b * S + n], v[b * S: b * S + n],
                                  advance_len=(li == c.num_layers - 1))
-            qb = q.reshape(B, S, c.num_heads, c.head_dim).transpose(1, 2).contiguous()
-            kb = k.reshape(B, S, c.num_kv_heads, c.head_dim).transpose(1, 2).contiguous()
-            vb = v.reshape(B, S, c.num_kv_heads, c.head_dim).transpose(1, 2).contiguous()
-            attn = ops.attn_fwd(qb, kb, vb, self.scale)  # [B,Hq,S,D]
-            attn = attn.transpose(1, 2).reshape(T, c.q_size).contiguous()
+            qb = q.reshape(B, S, self.local_heads, c.head_dim).transpose(1, 2).contiguous()
+            kb = k.reshape(B, S, self.local_kv_heads, c.head_dim).transpose(1, 2).contiguous()
+            vb = v.reshape(B, S, self.local_kv_heads, c.head_dim).transpose(1, 2).contiguous()
+            attn = ops.attn_fwd(qb, kb, vb, self.scale)  # [B,Hq_local,S,D]
+            attn = attn.transpose(1, 2).reshape(T, self.local_q_size).contiguous()
             o = ops.gemm_bt(attn, L["o"])
+            self.tp.all_reduce_(o)  # row-parallel o_proj partial sum
             h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
             hidden = self._ffn(h, L)
         h = ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)
@@ -121,7 +148,9 @@ class LlamaModel:
         if c.num_experts == 0:
             gateup = ops.gemm_bt(h, L["gateup"])
             act = ops.swiglu(gateup)
-            return ops.gemm_bt(act, L["down"])
+            out = ops.gemm_bt(act, L["down"])
+            self.tp.all_reduce_(out)  # row-parallel down_proj partial sum
+            return out
         return self._moe_ffn(h, L)
 
     def _moe_ffn(self, h: torch.Tensor, L: dict) -> torch.Tensor:
@@ -179,9 +208,10 @@ class LlamaModel:
             else:
                 h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
             qkv = ops.gemm_bt(h, L["qkv"])
-            q = qkv[:, : c.q_size].reshape(B, c.num_heads, c.head_dim).contiguous()
-            k = qkv[:, c.q_size: c.q_size + c.kv_size].reshape(B, c.num_kv_heads, c.head_dim).contiguous()
-            v = qkv[:, c.q_size + c.kv_size:].reshape(B, c.num_kv_heads, c.head_dim).contiguous()
+            qs, kvs = self.local_q_size, self.local_kv_size
+            q = qkv[:, :qs].reshape(B, self.local_heads, c.head_dim).contiguous()
+            k = qkv[:, qs: qs + kvs].reshape(B, self.local_kv_heads, c.head_dim).contiguous()
+            v = qkv[:, qs + kvs:].reshape(B, self.local_kv_heads, c.head_dim).contiguous()
             ops.rope_inplace(q, k, self.cos_sin, pos32)
             for b in range(B):
                 cache.append(li, seqs[b], k[b: b + 1], v[b: b + 1],
@@ -191,8 +221,9 @@ class LlamaModel:
             ctx = torch.tensor([cache.seq_lens[s] + (1 if li < c.num_layers - 1 else 0)
                                 for s in seqs], dtype=torch.int32, device=self.device)
             attn = ops.paged_decode_attn(q, cache.k[li], cache.v[li], bt, ctx, self.scale)
-            attn = attn.reshape(B, c.q_size)
+            attn = attn.reshape(B, self.local_q_size)
             o = ops.gemm_bt(attn, L["o"])
+            self.tp.all_reduce_(o)
             h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
             hidden = self._ffn(h, L)
         h = ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)

@@ -102,3 +102,32 @@ def test_candidate_parallel_scorer_world2():
         assert scores == pytest.approx(serial, abs=1e-6)  # f32 collective rounding
     # both ranks computed the identical full vector
     assert results[0][0] == pytest.approx(results[1][0], abs=0)
+
+
+def _payload_tp_prefill(rank, world):
+    import torch
+    from senweaver_amd.models import tiny_tp
+    from senweaver_amd.models.llama import LlamaModel
+    from senweaver_amd.parallel.tp import TPContext
+
+    tp = TPContext.from_default_group()
+    model = LlamaModel(tiny_tp(), device="cpu", seed=5, tp=tp)
+    tokens = torch.randint(0, 512, (1, 64), generator=torch.Generator().manual_seed(9))
+    hidden = model.prefill(tokens)
+    return hidden.float().sum(-1).squeeze(0).tolist()[:8]
+
+
+def test_tp2_matches_tp1():
+    """TP=2 over gloo reproduces the single-rank model (same seed/full weights)."""
+    import torch
+    from senweaver_amd.models import tiny_tp
+    from senweaver_amd.models.llama import LlamaModel
+
+    ref_model = LlamaModel(tiny_tp(), device="cpu", seed=5)
+    tokens = torch.randint(0, 512, (1, 64), generator=torch.Generator().manual_seed(9))
+    ref = ref_model.prefill(tokens).float().sum(-1).squeeze(0).tolist()[:8]
+
+    results = _run_dist("_payload_tp_prefill")
+    for rank, vals in results.items():
+        assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
+    assert results[0] == pytest.approx(results[1], abs=1e-5)
