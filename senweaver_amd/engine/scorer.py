@@ -36,14 +36,14 @@ def _pad64(n: int) -> int:
 
 class LlamaBackend:
     def __init__(self, config: ModelConfig | str = "llama-3-8b", device: Optional[str] = None,
-                 seed: int = 0, max_seq: int = 2048, micro_batch: int = 8, tp=None) -> None:
+                 seed: int = 0, max_seq: int = 2048, micro_batch: int = 8, tp=None, quant: str = "bf16") -> None:
         if isinstance(config, str):
             config = get_config(config)
         if device is None:
             device = "cuda:0" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
         self.config = config
-        self.model = LlamaModel(config, device=self.device, seed=seed, tp=tp)
+        self.model = LlamaModel(config, device=self.device, seed=seed, tp=tp, quant=quant)
         self.tokenizer = tok.HashTokenizer(config.vocab_size)
         self.max_seq = min(max_seq, config.max_position)
         self.micro_batch = micro_batch

@@ -27,7 +27,7 @@ if False:  # type-checking only; avoids the engine<->models import cycle
 
 class LlamaModel:
     def __init__(self, config: ModelConfig, device="cpu", dtype=torch.bfloat16, seed: int = 0,
-                 tp=None):
+                 tp=None, quant: str = "bf16"):
         from ..parallel.tp import TPContext, shard_gateup, shard_qkv, shard_rows
         self.config = config
         self.device = torch.device(device)
@@ -45,6 +45,10 @@ class LlamaModel:
         self.local_q_size = self.local_heads * c.head_dim
         self.local_kv_size = self.local_kv_heads * c.head_dim
         self.local_inter = c.intermediate_size // tpw
+        assert quant in ("bf16", "fp8")
+        if quant == "fp8":
+            assert c.num_experts == 0, "fp8 MoE not supported yet"
+        self.quant = quant
         if self.device.type == "cuda":
             # Device-side init: seconds for 8B instead of minutes of CPU RNG +
             # PCIe transfer.  Same seed + same call order on every rank =>
@@ -95,8 +99,24 @@ class LlamaModel:
             self.layers.append(layer)
         self.final_norm = torch.ones(c.hidden_size, dtype=dtype, device=self.device)
         self.lm_head = self.embed if c.tie_embeddings else W(c.vocab_size, c.hidden_size)
+        if self.quant == "fp8":
+            # pre-quantize every projection weight row-wise to OCP e4m3 —
+            # halves weight bytes; activations quantize per token on the fly
+            for L in self.layers:
+                for name in ("qkv", "o", "gateup", "down"):
+                    q, s = ops.quant_fp8(L[name])
+                    L[name + "_q"], L[name + "_s"] = q, s
+                    del L[name]
+            self.lm_head_q, self.lm_head_s = ops.quant_fp8(self.lm_head)
         self.cos_sin = ops.rope_tables(c.max_position, c.head_dim, c.rope_theta).to(self.device)
         self.scale = 1.0 / math.sqrt(c.head_dim)
+
+    def _linear(self, x: torch.Tensor, L: dict, name: str) -> torch.Tensor:
+        """Projection through bf16 MFMA or the fp8 MFMA path."""
+        if self.quant == "fp8":
+            xq, xs = ops.quant_fp8(x)
+            return ops.gemm_bt_fp8(xq, xs, L[name + "_q"], L[name + "_s"])
+        return ops.gemm_bt(x, L[name])
 
     # ------------------------------------------------------------------
     # Prefill: tokens [B, S] -> final hidden states [B, S, H].
@@ -120,7 +140,7 @@ class LlamaModel:
                 h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
             else:
                 h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
-            qkv = ops.gemm_bt(h, L["qkv"])  # [T, local q+2kv]
+            qkv = self._linear(h, L, "qkv")  # [T, local q+2kv]
             qs, kvs = self.local_q_size, self.local_kv_size
             q = qkv[:, :qs].reshape(T, self.local_heads, c.head_dim).contiguous()
             k = qkv[:, qs: qs + kvs].reshape(T, self.local_kv_heads, c.head_dim).contiguous()
@@ -136,7 +156,7 @@ class LlamaModel:
             vb = v.reshape(B, S, self.local_kv_heads, c.head_dim).transpose(1, 2).contiguous()
             attn = ops.attn_fwd(qb, kb, vb, self.scale)  # [B,Hq_local,S,D]
             attn = attn.transpose(1, 2).reshape(T, self.local_q_size).contiguous()
-            o = ops.gemm_bt(attn, L["o"])
+            o = self._linear(attn, L, "o")
             self.tp.all_reduce_(o)  # row-parallel o_proj partial sum
             h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
             hidden = self._ffn(h, L)
@@ -146,9 +166,9 @@ class LlamaModel:
     def _ffn(self, h: torch.Tensor, L: dict) -> torch.Tensor:
         c = self.config
         if c.num_experts == 0:
-            gateup = ops.gemm_bt(h, L["gateup"])
+            gateup = self._linear(h, L, "gateup")
             act = ops.swiglu(gateup)
-            out = ops.gemm_bt(act, L["down"])
+            out = self._linear(act, L, "down")
             self.tp.all_reduce_(out)  # row-parallel down_proj partial sum
             return out
         return self._moe_ffn(h, L)
@@ -207,7 +227,7 @@ class LlamaModel:
                 h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
             else:
                 h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
-            qkv = ops.gemm_bt(h, L["qkv"])
+            qkv = self._linear(h, L, "qkv")
             qs, kvs = self.local_q_size, self.local_kv_size
             q = qkv[:, :qs].reshape(B, self.local_heads, c.head_dim).contiguous()
             k = qkv[:, qs: qs + kvs].reshape(B, self.local_kv_heads, c.head_dim).contiguous()
@@ -222,7 +242,7 @@ class LlamaModel:
                                 for s in seqs], dtype=torch.int32, device=self.device)
             attn = ops.paged_decode_attn(q, cache.k[li], cache.v[li], bt, ctx, self.scale)
             attn = attn.reshape(B, self.local_q_size)
-            o = ops.gemm_bt(attn, L["o"])
+            o = self._linear(attn, L, "o")
             self.tp.all_reduce_(o)
             h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
             hidden = self._ffn(h, L)
@@ -232,4 +252,7 @@ class LlamaModel:
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
         """hidden [*, H] -> logits [*, vocab] through the lm_head GEMM."""
         flat = hidden.reshape(-1, self.config.hidden_size)
+        if self.quant == "fp8":
+            xq, xs = ops.quant_fp8(flat)
+            return ops.gemm_bt_fp8(xq, xs, self.lm_head_q, self.lm_head_s)
         return ops.gemm_bt(flat, self.lm_head)

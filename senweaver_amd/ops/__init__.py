@@ -104,6 +104,28 @@ def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return c[:M] if pad else c
 
 
+def quant_fp8(x: torch.Tensor):
+    """Row-wise e4m3 quantization: returns (q_uint8, scale_f32[rows])."""
+    if _on_gpu(x):
+        q, s = hip_ext().quant_fp8(x.contiguous().reshape(-1, x.shape[-1]))
+        return q, s
+    return ref.quant_fp8_ref(x)
+
+
+def gemm_bt_fp8(a_q, a_s, b_q, b_s):
+    """C = (A_q @ B_q^T) * a_s[m] * b_s[n] on fp8 MFMA, bf16 out."""
+    if _on_gpu(a_q):
+        M = a_q.shape[0]
+        pad = (-M) % 128
+        if pad:
+            a_q = torch.nn.functional.pad(a_q, (0, 0, 0, pad))
+            a_s = torch.nn.functional.pad(a_s, (0, pad), value=1.0)
+        c = hip_ext().gemm_bt_fp8(a_q.contiguous(), a_s.contiguous(),
+                                  b_q.contiguous(), b_s.contiguous())
+        return c[:M] if pad else c
+    return ref.gemm_bt_fp8_ref(a_q, a_s, b_q, b_s)
+
+
 def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
                     ) -> torch.Tensor:
     """Segment-grouped C = A_seg @ W[e]^T for MoE.

@@ -17,6 +17,8 @@ extern "C" __global__ void target_logprob_kernel(const ushort*, const int*, floa
 extern "C" __global__ void gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
+extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
+extern "C" __global__ void gemm_bt_fp8_kernel(const unsigned char*, const float*, const unsigned char*, const float*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_kernel(const ushort*, const ushort*, float*, int, int, int, int);
 extern "C" __global__ void gemv_reduce_kernel(const float*, ushort*, int, int, int);
 extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
@@ -189,6 +191,37 @@ torch::Tensor grouped_gemm_bt(torch::Tensor A, torch::Tensor W,
   return C;
 }
 
+// ---------------- fp8 path ----------------
+std::vector<torch::Tensor> quant_fp8(torch::Tensor x) {
+  check_bf16(x, "x");
+  const int K = x.size(-1);
+  TORCH_CHECK(K % 8 == 0);
+  const long long rows = x.numel() / K;
+  auto q = torch::empty({rows, (long)K}, x.options().dtype(torch::kUInt8));
+  auto s = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  quant_fp8_rowwise_kernel<<<dim3((unsigned)rows), dim3(256), 0, cur_stream()>>>(
+      bf16_ptr(x), q.data_ptr<unsigned char>(), s.data_ptr<float>(), K);
+  HIP_CHECK_KERNEL();
+  return {q, s};
+}
+
+torch::Tensor gemm_bt_fp8(torch::Tensor Aq, torch::Tensor As,
+                          torch::Tensor Bq, torch::Tensor Bs) {
+  TORCH_CHECK(Aq.scalar_type() == torch::kUInt8 && Bq.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(As.scalar_type() == torch::kFloat32 && Bs.scalar_type() == torch::kFloat32);
+  const int M = Aq.size(0), K = Aq.size(1), N = Bq.size(0);
+  TORCH_CHECK(Bq.size(1) == K && M % 128 == 0 && N % 128 == 0 && K % 128 == 0,
+              "fp8 gemm needs M,N%128, K%128; got ", M, "x", N, "x", K);
+  auto C = torch::empty({M, N}, Aq.options().dtype(torch::kBFloat16));
+  const int nwg = (M / 128) * (N / 128);
+  gemm_bt_fp8_kernel<<<dim3(nwg), dim3(256), 0, cur_stream()>>>(
+      Aq.data_ptr<unsigned char>(), As.data_ptr<float>(),
+      Bq.data_ptr<unsigned char>(), Bs.data_ptr<float>(),
+      bf16_mut(C), M, N, K);
+  HIP_CHECK_KERNEL();
+  return C;
+}
+
 // ---------------- Flash attention prefill ----------------
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
                        double scale) {
@@ -262,6 +295,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_bf16", &add_bf16, "a + b (bf16)");
   m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");
   m.def("grouped_gemm_bt", &grouped_gemm_bt, "segment-grouped C = A @ W[e]^T (MoE)");
+  m.def("quant_fp8", &quant_fp8, "row-wise bf16 -> e4m3 + scale");
+  m.def("gemm_bt_fp8", &gemm_bt_fp8, "fp8 MFMA GEMM with row/col rescale");
   m.def("attn_fwd", &attn_fwd, "causal flash attention fwd (D=128, GQA)");
   m.def("paged_decode_attn", &paged_decode_attn, "paged decode attention");
   m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");

@@ -1,0 +1,174 @@
+// fp8 (OCP e4m3) path for the 70B scorer (BASELINE config 5).
+//
+// gfx950 uses OCP e4m3fn (max 448), NOT MI300X's fnuz.  Weights are
+// quantized once per-output-row, activations per-token-row on the fly;
+// the GEMM runs v_mfma_f32_16x16x32_fp8_fp8 over a 128x128 tile with
+// BK=128 (fp8 halves the staged bytes per K element, so the same 16 KiB
+// LDS tile covers twice the K depth of the bf16 kernel) and the epilogue
+// rescales by a_scale[m] * b_scale[n].
+//
+// Note on rates: non-MX-scaled fp8 MFMA runs at the bf16 rate on CDNA4
+// (the 2x peak needs the MX block-scaled K=128 instructions — a follow-up);
+// what this path buys today is halved weight/activation traffic and the
+// fp8 numerics contract for ranking-stability testing.
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define FBK 128  // fp8 K-tile depth (bytes == elements)
+
+// ---------------------------------------------------------------------------
+// Row-wise quantization: bf16 [rows, K] -> fp8 bytes + f32 scale per row.
+//   scale = amax/448; q = fp8(x/scale)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+quant_fp8_rowwise_kernel(const ushort* __restrict__ x, unsigned char* __restrict__ q,
+                         float* __restrict__ scales, int K) {
+  const long long base = (long long)blockIdx.x * K;
+  float amax = 0.f;
+  const int nvec = K / 8;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + base + i * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) amax = fmaxf(amax, fabsf(bf2f(v.v[j])));
+  }
+  __shared__ float sm[4];
+  float wmax = wave_reduce_max(amax);
+  if ((threadIdx.x & 63) == 0) sm[threadIdx.x / 64] = wmax;
+  __syncthreads();
+  const float total = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+  const float scale = (total > 0.f) ? total / 448.f : 1.f;
+  const float inv = 1.f / scale;
+  if (threadIdx.x == 0) scales[blockIdx.x] = scale;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + base + i * 8);
+    unsigned int lo = 0, hi = 0;
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v.v[0]) * inv, bf2f(v.v[1]) * inv, lo, false);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v.v[2]) * inv, bf2f(v.v[3]) * inv, lo, true);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v.v[4]) * inv, bf2f(v.v[5]) * inv, hi, false);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f(v.v[6]) * inv, bf2f(v.v[7]) * inv, hi, true);
+    uint2 out = {lo, hi};
+    *reinterpret_cast<uint2*>(q + base + i * 8) = out;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fp8 GEMM: C[M,N] = (A_q[M,K] @ B_q[N,K]^T) * a_s[m] * b_s[n], bf16 out.
+// 128x128 tile, 4 waves (2x2), glds double buffer, one barrier per K-tile.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void stage_fp8_tile(
+    const unsigned char* __restrict__ src, long long ldK, unsigned char* lds_tile,
+    int tid) {
+  // 128 rows x 8 chunks(16B=16 fp8) = 1024 chunks / 256 threads = 4 glds
+  const int wave_chunk = tid & ~63;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = i * 256 + tid;
+    const int row = s >> 3;
+    const int c = (s & 7) ^ (row & 7);
+    const unsigned char* g = src + (long long)row * ldK + c * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(lds_tile + (long long)(i * 256 + wave_chunk) * 16),
+        16, 0, 0);
+  }
+}
+
+__device__ __forceinline__ long read_fp8_frag(const unsigned char* lds_tile, int row,
+                                              int byte_off) {
+  const int c = byte_off >> 4;
+  const int phys = (c ^ (row & 7)) * 16 + (byte_off & 15);
+  return *reinterpret_cast<const long*>(lds_tile + row * FBK + phys);
+}
+
+extern "C" __global__ void __launch_bounds__(256, 2)
+gemm_bt_fp8_kernel(const unsigned char* __restrict__ A, const float* __restrict__ a_s,
+                   const unsigned char* __restrict__ B, const float* __restrict__ b_s,
+                   ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / 128) * (N / 128);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / 128;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1;
+  const int wn = wid & 1;
+
+  __shared__ __attribute__((aligned(16))) unsigned char lds[2][2][128 * FBK];
+
+  const unsigned char* Atile = A + (long long)tile_m * 128 * K;
+  const unsigned char* Btile = B + (long long)tile_n * 128 * K;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int m_base = wm * 64;
+  const int n_base = wn * 64;
+  const int frag_row = lane & 15;
+  const int frag_kgrp = lane >> 4;
+
+  const int ntiles = K / FBK;
+  stage_fp8_tile(Atile, K, lds[0][0], tid);
+  stage_fp8_tile(Btile, K, lds[0][1], tid);
+  __syncthreads();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      stage_fp8_tile(Atile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][0], tid);
+      stage_fp8_tile(Btile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][1], tid);
+    }
+    const unsigned char* Al = lds[buf][0];
+    const unsigned char* Bl = lds[buf][1];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {  // FBK=128 in 4 MFMA steps of K=32
+      long af[4], bf[4];
+      const int boff = kk * 32 + frag_kgrp * 8;
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = read_fp8_frag(Al, m_base + mi * 16 + frag_row, boff);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bf[ni] = read_fp8_frag(Bl, n_base + ni * 16 + frag_row, boff);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const long long c_row0 = (long long)tile_m * 128 + m_base + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * 128 + n_base + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+      const float as = a_s[row];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const long long col = c_col0 + ni * 16;
+        crow[col] = f2bf(acc[mi][ni][e] * as * b_s[col]);
+      }
+    }
+  }
+}

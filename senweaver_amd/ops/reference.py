@@ -108,3 +108,22 @@ def argmax_rows_ref(logits: torch.Tensor) -> torch.Tensor:
 def target_logprob_ref(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
     lp = torch.log_softmax(logits.float(), dim=-1)
     return lp.gather(1, targets.long().unsqueeze(1)).squeeze(1)
+
+
+FP8_MAX = 448.0  # OCP e4m3fn
+
+
+def quant_fp8_ref(x: torch.Tensor):
+    """CPU row-wise e4m3 quantization via torch.float8_e4m3fn."""
+    x2 = x.reshape(-1, x.shape[-1]).float()
+    amax = x2.abs().amax(dim=-1)
+    scale = torch.where(amax > 0, amax / FP8_MAX, torch.ones_like(amax))
+    q = (x2 / scale.unsqueeze(1)).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scale
+
+
+def gemm_bt_fp8_ref(a_q, a_s, b_q, b_s):
+    af = a_q.view(torch.float8_e4m3fn).float()
+    bf = b_q.view(torch.float8_e4m3fn).float()
+    c = (af @ bf.t()) * a_s.unsqueeze(1) * b_s.unsqueeze(0)
+    return c.to(torch.bfloat16)

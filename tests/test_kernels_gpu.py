@@ -193,3 +193,34 @@ def test_mixtral_tiny_forward_gpu(dev):
     tokens = torch.randint(0, 512, (2, 64), device=dev)
     h = model.prefill(tokens)
     assert torch.isfinite(h.float()).all()
+
+
+def test_quant_fp8_gpu(dev):
+    x = torch.randn(64, 512, dtype=torch.bfloat16, device=dev) * 2
+    q, s = ops.quant_fp8(x)
+    qr, sr = ref.quant_fp8_ref(x.cpu())
+    torch.testing.assert_close(s.cpu(), sr, atol=1e-3, rtol=1e-3)
+    back = q.cpu().view(torch.float8_e4m3fn).float() * s.cpu().unsqueeze(1)
+    back_ref = qr.view(torch.float8_e4m3fn).float() * sr.unsqueeze(1)
+    torch.testing.assert_close(back, back_ref, atol=0.05, rtol=0.05)
+
+
+def test_gemm_fp8_gpu(dev):
+    torch.manual_seed(5)
+    M, N, K = 256, 1024, 4096
+    a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    b = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+    aq, asc = ops.quant_fp8(a)
+    bq, bsc = ops.quant_fp8(b)
+    c = ops.gemm_bt_fp8(aq, asc, bq, bsc)
+    c_ref = ref.gemm_bt_fp8_ref(aq.cpu(), asc.cpu(), bq.cpu(), bsc.cpu())
+    torch.testing.assert_close(c.float().cpu(), c_ref.float(), atol=1.0, rtol=5e-2)
+
+
+def test_fp8_model_forward_gpu(dev):
+    from senweaver_amd.models import tiny_debug
+    from senweaver_amd.models.llama import LlamaModel
+    m = LlamaModel(tiny_debug(), device=dev, quant="fp8")
+    tokens = torch.randint(0, 512, (1, 128), device=dev)
+    h = m.prefill(tokens)
+    assert torch.isfinite(h.float()).all()
