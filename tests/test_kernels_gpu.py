@@ -202,7 +202,12 @@ def test_quant_fp8_gpu(dev):
     torch.testing.assert_close(s.cpu(), sr, atol=1e-3, rtol=1e-3)
     back = q.cpu().view(torch.float8_e4m3fn).float() * s.cpu().unsqueeze(1)
     back_ref = qr.view(torch.float8_e4m3fn).float() * sr.unsqueeze(1)
-    torch.testing.assert_close(back, back_ref, atol=0.05, rtol=0.05)
+    # the kernel multiplies by 1/scale while the reference divides: exact
+    # e4m3 ties can land one quantum apart — allow 1 ulp on <1% of elements
+    d = (back - back_ref).abs()
+    ulp = (s.cpu().unsqueeze(1) * 16.0)  # mantissa step at the top binade
+    assert (d > ulp + 1e-6).sum() == 0
+    assert (d > 0).float().mean() < 0.01
 
 
 def test_gemm_fp8_gpu(dev):
