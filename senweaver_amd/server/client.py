@@ -1,0 +1,110 @@
+"""Python client for the native daemon (the renderer-proxy analog).
+
+Mirrors sendLLMMessageService.ts: request-id-keyed hook registry; send
+returns the requestId which is the abort token; events dispatched to the
+registered callbacks.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import socket
+import subprocess
+import threading
+import time
+import uuid
+from typing import Any, Callable, Dict, List, Optional
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+DAEMON_BIN = os.path.join(_HERE, "_bin", "senweaver_daemon")
+
+
+def build_daemon(force: bool = False) -> str:
+    """Compile the C++ daemon with g++ (no GPU/torch dependency)."""
+    src = os.path.join(_HERE, "daemon.cpp")
+    os.makedirs(os.path.dirname(DAEMON_BIN), exist_ok=True)
+    if force or (not os.path.exists(DAEMON_BIN)
+                 or os.path.getmtime(DAEMON_BIN) < os.path.getmtime(src)):
+        subprocess.run(["g++", "-O2", "-std=c++17", "-o", DAEMON_BIN, src], check=True)
+    return DAEMON_BIN
+
+
+class DaemonClient:
+    def __init__(self, socket_path: str) -> None:
+        self._sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        self._sock.connect(socket_path)
+        self._hooks: Dict[str, Dict[str, Callable]] = {}
+        self._lock = threading.Lock()
+        self._closed = False
+        self.events: List[dict] = []
+        self._reader = threading.Thread(target=self._read_loop, daemon=True)
+        self._reader.start()
+
+    def _read_loop(self) -> None:
+        buf = b""
+        while not self._closed:
+            try:
+                chunk = self._sock.recv(65536)
+            except OSError:
+                break
+            if not chunk:
+                break
+            buf += chunk
+            while b"\n" in buf:
+                line, buf = buf.split(b"\n", 1)
+                if not line:
+                    continue
+                try:
+                    msg = json.loads(line)
+                except ValueError:
+                    continue
+                self.events.append(msg)
+                rid = msg.get("requestId", "")
+                with self._lock:
+                    hooks = self._hooks.get(rid, {})
+                ev = msg.get("event", "")
+                fn = hooks.get(ev)
+                if fn:
+                    try:
+                        fn(msg)
+                    except Exception:
+                        pass
+
+    def _send(self, obj: dict) -> None:
+        self._sock.sendall((json.dumps(obj) + "\n").encode())
+
+    def send_llm_message(self, messages: List[dict], on_text=None, on_final=None,
+                         on_error=None, chat_mode: Optional[str] = None,
+                         max_new_tokens: int = 64) -> str:
+        rid = str(uuid.uuid4())
+        with self._lock:
+            self._hooks[rid] = {"onText": on_text or (lambda m: None),
+                                "onFinalMessage": on_final or (lambda m: None),
+                                "onError": on_error or (lambda m: None),
+                                "onAbort": lambda m: None}
+        self._send({"method": "sendLLMMessage", "requestId": rid,
+                    "messages": messages, "chatMode": chat_mode,
+                    "maxNewTokens": max_new_tokens})
+        return rid
+
+    def abort(self, request_id: str) -> None:
+        self._send({"method": "abort", "requestId": request_id})
+
+    def ping(self, timeout: float = 30.0) -> bool:
+        got = threading.Event()
+        n0 = len(self.events)
+        self._send({"method": "ping"})
+        t0 = time.time()
+        while time.time() - t0 < timeout:
+            if any(e.get("event") == "pong" for e in self.events[n0:]):
+                return True
+            time.sleep(0.05)
+        return False
+
+    def close(self) -> None:
+        self._closed = True
+        try:
+            self._sock.close()
+        except OSError:
+            pass

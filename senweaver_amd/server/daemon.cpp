@@ -1,0 +1,308 @@
+// senweaver_daemon — native JSON-RPC streaming control plane.
+//
+// The reference splits LLM transport across processes: the renderer posts
+// requests over the IPC channel 'senweaver-channel-llmMessage' and the
+// Electron main process streams request-id-tagged events back
+// (electron-main/sendLLMMessageChannel.ts: listen onText_/onFinalMessage_/
+// onError_, call sendLLMMessage/abort/list; per-request abortRef map).
+//
+// This daemon is the MI355X-native equivalent: clients connect over a Unix
+// domain socket and speak newline-delimited JSON:
+//   -> {"method":"sendLLMMessage","requestId":"r1", ...payload}
+//   -> {"method":"abort","requestId":"r1"}
+//   -> {"method":"list"}
+//   <- {"event":"onText","requestId":"r1","fullText":...}    (streamed)
+//   <- {"event":"onFinalMessage","requestId":"r1",...}
+//   <- {"event":"onError","requestId":"r1","message":...}
+//
+// The daemon owns connection routing, the request-id -> client map, the
+// instant client-side abort fast path (an aborted id suppresses further
+// events immediately, before the engine has even seen the abort — the same
+// no-round-trip semantics as sendLLMMessageService.ts:142-146), and engine
+// lifecycle; the Python engine worker (engine_worker.py, spawned as a child
+// on the same box, GPU-side) does the model work.  Payloads transit the
+// daemon opaquely — only "method"/"requestId"/"event" are inspected.
+
+#include <arpa/inet.h>
+#include <errno.h>
+#include <fcntl.h>
+#include <poll.h>
+#include <signal.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <sys/socket.h>
+#include <sys/un.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
+#include <map>
+#include <set>
+#include <string>
+#include <vector>
+
+namespace {
+
+// Minimal extraction of a top-level string field from a JSON object line.
+// Handles escaped quotes; payloads are otherwise passed through opaquely.
+std::string json_field(const std::string& line, const std::string& key) {
+  const std::string pat = "\"" + key + "\"";
+  size_t p = line.find(pat);
+  if (p == std::string::npos) return "";
+  p = line.find(':', p + pat.size());
+  if (p == std::string::npos) return "";
+  ++p;
+  while (p < line.size() && (line[p] == ' ' || line[p] == '\t')) ++p;
+  if (p >= line.size() || line[p] != '"') return "";
+  ++p;
+  std::string out;
+  while (p < line.size()) {
+    char c = line[p];
+    if (c == '\\' && p + 1 < line.size()) {
+      out.push_back(line[p + 1]);
+      p += 2;
+      continue;
+    }
+    if (c == '"') break;
+    out.push_back(c);
+    ++p;
+  }
+  return out;
+}
+
+struct Client {
+  int fd;
+  std::string inbuf;
+};
+
+class Daemon {
+ public:
+  Daemon(std::string socket_path, std::vector<std::string> worker_argv)
+      : socket_path_(std::move(socket_path)), worker_argv_(std::move(worker_argv)) {}
+
+  int run() {
+    signal(SIGPIPE, SIG_IGN);
+    if (!spawn_worker()) return 1;
+    listen_fd_ = make_listen_socket();
+    if (listen_fd_ < 0) return 1;
+    fprintf(stderr, "[daemon] listening on %s\n", socket_path_.c_str());
+    loop();
+    return 0;
+  }
+
+ private:
+  std::string socket_path_;
+  std::vector<std::string> worker_argv_;
+  int listen_fd_ = -1;
+  int worker_in_ = -1;   // write requests to worker stdin
+  int worker_out_ = -1;  // read events from worker stdout
+  pid_t worker_pid_ = -1;
+  std::string worker_buf_;
+  std::map<int, Client> clients_;
+  std::map<std::string, int> request_client_;  // requestId -> client fd
+  std::set<std::string> aborted_;
+
+  bool spawn_worker() {
+    int in_pipe[2], out_pipe[2];
+    if (pipe(in_pipe) || pipe(out_pipe)) return false;
+    worker_pid_ = fork();
+    if (worker_pid_ < 0) return false;
+    if (worker_pid_ == 0) {
+      dup2(in_pipe[0], 0);
+      dup2(out_pipe[1], 1);
+      close(in_pipe[1]);
+      close(out_pipe[0]);
+      std::vector<char*> argv;
+      for (auto& a : worker_argv_) argv.push_back(const_cast<char*>(a.c_str()));
+      argv.push_back(nullptr);
+      execvp(argv[0], argv.data());
+      perror("execvp engine worker");
+      _exit(127);
+    }
+    close(in_pipe[0]);
+    close(out_pipe[1]);
+    worker_in_ = in_pipe[1];
+    worker_out_ = out_pipe[0];
+    return true;
+  }
+
+  int make_listen_socket() {
+    int fd = socket(AF_UNIX, SOCK_STREAM, 0);
+    if (fd < 0) return -1;
+    sockaddr_un addr{};
+    addr.sun_family = AF_UNIX;
+    unlink(socket_path_.c_str());
+    snprintf(addr.sun_path, sizeof(addr.sun_path), "%s", socket_path_.c_str());
+    if (bind(fd, (sockaddr*)&addr, sizeof(addr)) || listen(fd, 16)) {
+      perror("bind/listen");
+      close(fd);
+      return -1;
+    }
+    return fd;
+  }
+
+  static bool write_all(int fd, const std::string& s) {
+    size_t off = 0;
+    while (off < s.size()) {
+      ssize_t n = write(fd, s.data() + off, s.size() - off);
+      if (n <= 0) return false;
+      off += (size_t)n;
+    }
+    return true;
+  }
+
+  void handle_client_line(int fd, const std::string& line) {
+    const std::string method = json_field(line, "method");
+    const std::string rid = json_field(line, "requestId");
+    if (method == "sendLLMMessage") {
+      if (rid.empty()) {
+        write_all(fd, "{\"event\":\"onError\",\"message\":\"missing requestId\"}\n");
+        return;
+      }
+      request_client_[rid] = fd;
+      aborted_.erase(rid);
+      write_all(worker_in_, line + "\n");
+    } else if (method == "abort") {
+      // instant client-side abort: suppress events immediately, then tell
+      // the engine so it stops decoding
+      aborted_.insert(rid);
+      request_client_.erase(rid);
+      write_all(worker_in_, line + "\n");
+      write_all(fd, "{\"event\":\"onAbort\",\"requestId\":\"" + rid + "\"}\n");
+    } else if (method == "list" || method == "ping") {
+      write_all(worker_in_, line + "\n");
+      request_client_[method] = fd;  // single in-flight list/ping per method
+    } else if (method == "shutdown") {
+      write_all(fd, "{\"event\":\"shuttingDown\"}\n");
+      shutdown_all();
+      exit(0);
+    } else {
+      write_all(fd, "{\"event\":\"onError\",\"message\":\"unknown method\"}\n");
+    }
+  }
+
+  void handle_worker_line(const std::string& line) {
+    const std::string rid = json_field(line, "requestId");
+    const std::string ev = json_field(line, "event");
+    std::string key = rid;
+    if (key.empty()) key = ev == "listResult" ? "list" : ev == "pong" ? "ping" : "";
+    auto it = request_client_.find(key);
+    if (it == request_client_.end()) return;           // client gone
+    if (aborted_.count(rid)) return;                   // aborted: drop events
+    if (!write_all(it->second, line + "\n")) {
+      request_client_.erase(it);
+      return;
+    }
+    if (ev == "onFinalMessage" || ev == "onError" || ev == "listResult" || ev == "pong")
+      request_client_.erase(key);
+  }
+
+  void drop_client(int fd) {
+    close(fd);
+    clients_.erase(fd);
+    for (auto it = request_client_.begin(); it != request_client_.end();) {
+      if (it->second == fd) {
+        aborted_.insert(it->first);
+        it = request_client_.erase(it);
+      } else {
+        ++it;
+      }
+    }
+  }
+
+  void shutdown_all() {
+    if (worker_pid_ > 0) {
+      kill(worker_pid_, SIGTERM);
+      int status = 0;
+      waitpid(worker_pid_, &status, 0);
+    }
+    for (auto& [fd, c] : clients_) close(fd);
+    if (listen_fd_ >= 0) close(listen_fd_);
+    unlink(socket_path_.c_str());
+  }
+
+  void loop() {
+    std::string pending;
+    for (;;) {
+      std::vector<pollfd> fds;
+      fds.push_back({listen_fd_, POLLIN, 0});
+      fds.push_back({worker_out_, POLLIN, 0});
+      for (auto& [fd, c] : clients_) fds.push_back({fd, POLLIN, 0});
+      if (poll(fds.data(), fds.size(), 1000) < 0) {
+        if (errno == EINTR) continue;
+        break;
+      }
+      // worker died?
+      int status;
+      if (worker_pid_ > 0 && waitpid(worker_pid_, &status, WNOHANG) == worker_pid_) {
+        fprintf(stderr, "[daemon] engine worker exited (%d); restarting\n", status);
+        close(worker_in_);
+        close(worker_out_);
+        worker_pid_ = -1;
+        if (!spawn_worker()) break;
+        continue;
+      }
+      if (fds[0].revents & POLLIN) {
+        int cfd = accept(listen_fd_, nullptr, nullptr);
+        if (cfd >= 0) clients_[cfd] = Client{cfd, ""};
+      }
+      if (fds[1].revents & (POLLIN | POLLHUP)) {
+        char buf[65536];
+        ssize_t n = read(worker_out_, buf, sizeof(buf));
+        if (n > 0) {
+          worker_buf_.append(buf, (size_t)n);
+          size_t pos;
+          while ((pos = worker_buf_.find('\n')) != std::string::npos) {
+            handle_worker_line(worker_buf_.substr(0, pos));
+            worker_buf_.erase(0, pos + 1);
+          }
+        }
+      }
+      for (size_t i = 2; i < fds.size(); ++i) {
+        if (!(fds[i].revents & (POLLIN | POLLHUP | POLLERR))) continue;
+        int fd = fds[i].fd;
+        char buf[65536];
+        ssize_t n = read(fd, buf, sizeof(buf));
+        if (n <= 0) {
+          drop_client(fd);
+          continue;
+        }
+        auto& client = clients_[fd];
+        client.inbuf.append(buf, (size_t)n);
+        size_t pos;
+        while ((pos = client.inbuf.find('\n')) != std::string::npos) {
+          std::string line = client.inbuf.substr(0, pos);
+          client.inbuf.erase(0, pos + 1);
+          if (!line.empty()) handle_client_line(fd, line);
+        }
+      }
+    }
+    shutdown_all();
+  }
+};
+
+}  // namespace
+
+int main(int argc, char** argv) {
+  std::string socket_path = "/tmp/senweaver_amd.sock";
+  std::vector<std::string> worker{"python3", "-m", "senweaver_amd.server.engine_worker"};
+  for (int i = 1; i < argc; ++i) {
+    std::string a = argv[i];
+    if (a == "--socket" && i + 1 < argc) {
+      socket_path = argv[++i];
+    } else if (a == "--worker" && i + 1 < argc) {
+      worker.clear();
+      // split on spaces
+      std::string w = argv[++i];
+      size_t start = 0;
+      while (start < w.size()) {
+        size_t sp = w.find(' ', start);
+        if (sp == std::string::npos) sp = w.size();
+        if (sp > start) worker.push_back(w.substr(start, sp - start));
+        start = sp + 1;
+      }
+    }
+  }
+  Daemon d(socket_path, worker);
+  return d.run();
+}

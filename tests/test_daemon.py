@@ -1,0 +1,122 @@
+"""Native daemon end-to-end: C++ JSON-RPC control plane + Python engine worker."""
+
+import os
+import subprocess
+import threading
+import time
+import uuid
+
+import pytest
+
+from senweaver_amd.server.client import DaemonClient, build_daemon
+from senweaver_amd.utils.cache import LRUCache, MultiLayerCache
+from senweaver_amd.utils.observability import (
+    MetricsService,
+    PerformanceMonitor,
+    TokenUsageTracker,
+)
+
+
+@pytest.fixture(scope="module")
+def daemon():
+    bin_path = build_daemon()
+    sock = f"/tmp/senweaver_test_{os.getpid()}.sock"
+    env = dict(os.environ, SENWEAVER_MODEL="tiny-debug", SENWEAVER_MAX_SEQ="128")
+    proc = subprocess.Popen([bin_path, "--socket", sock], env=env,
+                            stderr=subprocess.PIPE)
+    # wait for the socket
+    for _ in range(200):
+        if os.path.exists(sock):
+            break
+        time.sleep(0.05)
+    else:
+        proc.kill()
+        pytest.fail("daemon socket never appeared")
+    yield sock
+    proc.terminate()
+    proc.wait(timeout=10)
+
+
+def test_daemon_ping(daemon):
+    c = DaemonClient(daemon)
+    assert c.ping(timeout=120)  # first ping waits for engine import/ready
+    c.close()
+
+
+def test_daemon_stream_and_final(daemon):
+    c = DaemonClient(daemon)
+    assert c.ping(timeout=120)
+    texts = []
+    done = threading.Event()
+
+    rid = c.send_llm_message(
+        [{"role": "user", "content": "hello engine"}],
+        on_text=lambda m: texts.append(m["fullText"]),
+        on_final=lambda m: (texts.append(m["fullText"]), done.set()),
+        on_error=lambda m: done.set(),
+        max_new_tokens=6)
+    assert done.wait(timeout=180), "no final message"
+    # cumulative streaming: each onText extends the previous
+    for a, b in zip(texts, texts[1:]):
+        assert b.startswith(a[: len(a)])
+    assert texts[-1]  # non-empty final
+    c.close()
+
+
+def test_daemon_abort(daemon):
+    c = DaemonClient(daemon)
+    assert c.ping(timeout=120)
+    rid = c.send_llm_message([{"role": "user", "content": "long task"}],
+                             max_new_tokens=64)
+    c.abort(rid)
+    time.sleep(1.0)
+    aborts = [e for e in c.events if e.get("event") == "onAbort" and e.get("requestId") == rid]
+    assert aborts, "abort ack not received"
+    # no final message may arrive for the aborted id after the abort ack
+    finals = [e for e in c.events if e.get("event") == "onFinalMessage" and e.get("requestId") == rid]
+    assert not finals
+    c.close()
+
+
+# ---- aux utils (cache / observability) ----
+
+def test_lru_cache():
+    c = LRUCache(capacity=2)
+    c.put("a", 1)
+    c.put("b", 2)
+    assert c.get("a") == 1
+    c.put("c", 3)  # evicts b (a was touched)
+    assert c.get("b") is None
+    assert c.get("a") == 1 and c.get("c") == 3
+    assert c.hit_rate is not None
+
+
+def test_multilayer_cache():
+    m = MultiLayerCache(hot_capacity=1, warm_capacity=4)
+    m.put("x", 10)
+    m.put("y", 20)  # x falls out of hot, stays warm
+    assert m.get("x") == 10  # warm hit promotes back
+
+
+def test_metrics_and_usage():
+    ms = MetricsService()
+    ms.capture("llm_send", {"model": "local"})
+    ms.capture("llm_send", {"model": "local"})
+    assert ms.debug_info()["byEvent"]["llm_send"] == 2
+    t = TokenUsageTracker()
+    t.record("r1", "llama-3-8b", 100, 50)
+    t.record("r2", "llama-3-8b", 10, 5)
+    s = t.stats()
+    assert s["totalInputTokens"] == 110
+    assert s["byModel"]["llama-3-8b"]["requests"] == 2
+
+
+def test_performance_monitor_slo():
+    pm = PerformanceMonitor(enabled=True)
+    pm.record("messageTrimming", 500)  # over the 200ms SLO
+    pm.record("messageTrimming", 10)
+    assert pm.violations and pm.violations[0]["metric"] == "messageTrimming"
+    assert pm.summary()["messageTrimming"]["count"] == 2
+    with pm.timer("systemMessageGeneration"):
+        pass
+    assert pm.summary()["systemMessageGeneration"]["count"] == 1
