@@ -118,13 +118,24 @@ def main() -> int:
 
     comm_dev = torch.device(device) if use_cuda else torch.device("cpu")
 
+    # tokenized (candidate x rollout) sequences as device tensors, sharded
+    my_cands = list(range(rank, n_candidates, world))
+    my_tok, my_mask = [], []
+    for ci in my_cands:
+        for ids, mask in sequences[ci * args.rollouts:(ci + 1) * args.rollouts]:
+            my_tok.append(torch.tensor(ids, dtype=torch.long))
+            my_mask.append(torch.tensor(mask, dtype=torch.bool))
+    dev_t = torch.device(device)
+    my_tok = torch.stack(my_tok).to(dev_t) if my_tok else torch.zeros(0, args.seq_len, dtype=torch.long, device=dev_t)
+    my_mask = torch.stack(my_mask).to(dev_t) if my_cands else None
+
     def one_step(step_idx: int):
-        # shard candidates over ranks (strong scaling)
-        my_cands = list(range(rank, n_candidates, world))
-        my_seqs = []
-        for ci in my_cands:
-            my_seqs.extend(sequences[ci * args.rollouts:(ci + 1) * args.rollouts])
-        lps = backend.sequence_logprobs(my_seqs)
+        # teacher-forced scoring of this rank's candidate shard (micro-batched)
+        mb = args.micro_batch
+        lps = []
+        for i in range(0, my_tok.shape[0], mb):
+            lps.append(backend.score_token_batch(my_tok[i:i + mb], my_mask[i:i + mb]))
+        lps = torch.cat(lps).tolist() if lps else []
         my_scores = []
         for j, ci in enumerate(my_cands):
             chunk = lps[j * args.rollouts:(j + 1) * args.rollouts]

@@ -91,35 +91,39 @@ class LlamaBackend:
         B = len(chunk)
         S = _pad64(max(len(ids) for ids, _ in chunk))
         tokens = torch.zeros(B, S, dtype=torch.long)
-        for b, (ids, _) in enumerate(chunk):
+        mask = torch.zeros(B, S, dtype=torch.bool)
+        for b, (ids, m) in enumerate(chunk):
             tokens[b, : len(ids)] = torch.tensor(ids, dtype=torch.long)
-        tokens = tokens.to(self.device)
-        hidden = self.model.prefill(tokens)  # [B, S, H]
+            mask[b, : len(m)] = torch.tensor(m, dtype=torch.bool)
+        return self.score_token_batch(tokens.to(self.device), mask.to(self.device)).tolist()
 
-        # positions p>=1 whose token is scored: predicted from hidden[p-1]
-        gather_rows: List[int] = []
-        targets: List[int] = []
-        seq_of: List[int] = []
-        for b, (ids, mask) in enumerate(chunk):
-            for p in range(1, len(ids)):
-                if mask[p]:
-                    gather_rows.append(b * S + p - 1)
-                    targets.append(ids[p])
-                    seq_of.append(b)
-        if not gather_rows:
-            return [0.0] * B
-        flat = hidden.reshape(B * S, -1)
-        rows = flat[torch.tensor(gather_rows, device=self.device)]
+    def score_token_batch(self, tokens: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
+        """Mean per-token logprob of masked positions per sequence.
+
+        tokens/mask: [B, S] on the engine device (S % 64 == 0).  Fully
+        vectorized: only assistant-position hidden rows go through the
+        lm_head GEMM (a ~4x lm_head saving at the bench's 25% mask).
+        Returns f32 [B] on CPU.
+        """
+        B, S = tokens.shape
+        hidden = self.model.prefill(tokens)  # [B, S, H]
+        m = mask.clone()
+        m[:, 0] = False  # position 0 has no predictor
+        flat_pos = m.reshape(-1).nonzero(as_tuple=False).squeeze(1)  # b*S + p
+        if flat_pos.numel() == 0:
+            return torch.zeros(B, dtype=torch.float32)
+        # p >= 1 within every sequence (position 0 masked off above), so
+        # flat_pos - 1 stays inside the same sequence's rows
+        rows = hidden.reshape(B * S, -1)[flat_pos - 1]
         logits = self.model.logits(rows)
-        lp = ops.target_logprob(logits, torch.tensor(targets, dtype=torch.int32,
-                                                     device=self.device))
-        lp = lp.float().cpu()
-        sums = [0.0] * B
-        counts = [0] * B
-        for j, b in enumerate(seq_of):
-            sums[b] += float(lp[j])
-            counts[b] += 1
-        return [sums[b] / counts[b] if counts[b] else 0.0 for b in range(B)]
+        targets = tokens.reshape(-1)[flat_pos].to(torch.int32)
+        lp = ops.target_logprob(logits, targets)
+        seq_of = torch.div(flat_pos, S, rounding_mode="floor")
+        sums = torch.zeros(B, dtype=torch.float32, device=lp.device)
+        counts = torch.zeros(B, dtype=torch.float32, device=lp.device)
+        sums.index_add_(0, seq_of, lp.float())
+        counts.index_add_(0, seq_of, torch.ones_like(lp, dtype=torch.float32))
+        return (sums / counts.clamp(min=1)).cpu()
 
     # ------------------------------------------------------------------
     # PromptOptimizerBackend API

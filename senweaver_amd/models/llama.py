@@ -145,16 +145,17 @@ class LlamaModel:
             q = qkv[:, :qs].reshape(T, self.local_heads, c.head_dim).contiguous()
             k = qkv[:, qs: qs + kvs].reshape(T, self.local_kv_heads, c.head_dim).contiguous()
             v = qkv[:, qs + kvs:].reshape(T, self.local_kv_heads, c.head_dim).contiguous()
-            ops.rope_inplace(q, k, self.cos_sin, positions)
+            # fused rope + [B,H,S,D] scatter (replaces rope + 2 transposes)
+            qb, kb = ops.rope_scatter(q, k, self.cos_sin, positions, B, S)
             if cache is not None:
+                kflat = kb.permute(0, 2, 1, 3).reshape(T, self.local_kv_heads, c.head_dim)
                 for b in range(B):
                     n = real_lens[b] if real_lens is not None else S
-                    cache.append(li, seqs[b], k[b * S: b * S + n], v[b * S: b * S + n],
+                    cache.append(li, seqs[b], kflat[b * S: b * S + n], v[b * S: b * S + n],
                                  advance_len=(li == c.num_layers - 1))
-            qb = q.reshape(B, S, self.local_heads, c.head_dim).transpose(1, 2).contiguous()
-            kb = k.reshape(B, S, self.local_kv_heads, c.head_dim).transpose(1, 2).contiguous()
-            vb = v.reshape(B, S, self.local_kv_heads, c.head_dim).transpose(1, 2).contiguous()
-            attn = ops.attn_fwd(qb, kb, vb, self.scale)  # [B,Hq_local,S,D]
+            # V^T computed directly: the attention kernel consumes [B,Hk,D,S]
+            vt = v.reshape(B, S, self.local_kv_heads, c.head_dim).permute(0, 2, 3, 1).contiguous()
+            attn = ops.attn_fwd(qb, kb, None, self.scale, vt=vt)  # [B,Hq_local,S,D]
             attn = attn.transpose(1, 2).reshape(T, self.local_q_size).contiguous()
             o = self._linear(attn, L, "o")
             self.tp.all_reduce_(o)  # row-parallel o_proj partial sum

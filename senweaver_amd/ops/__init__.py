@@ -80,6 +80,19 @@ def rope_inplace(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor, positi
     k.copy_(ref.rope_ref(k, cos_sin, positions))
 
 
+def rope_scatter(q: torch.Tensor, k: torch.Tensor, cos_sin, positions, B: int, S: int):
+    """q/k [T,H,D] -> rope -> ([B,Hq,S,D], [B,Hk,S,D]) in one pass."""
+    if _on_gpu(q):
+        qo, ko = hip_ext().rope_scatter(q.contiguous(), k.contiguous(), cos_sin,
+                                        positions.to(torch.int32), B, S)
+        return qo, ko
+    qr = ref.rope_ref(q, cos_sin, positions)
+    kr = ref.rope_ref(k, cos_sin, positions)
+    Hq, Hk, D = q.shape[1], k.shape[1], q.shape[2]
+    return (qr.reshape(B, S, Hq, D).permute(0, 2, 1, 3).contiguous(),
+            kr.reshape(B, S, Hk, D).permute(0, 2, 1, 3).contiguous())
+
+
 def swiglu(gateup: torch.Tensor) -> torch.Tensor:
     if _on_gpu(gateup):
         return hip_ext().swiglu(gateup.contiguous())
@@ -168,6 +181,8 @@ def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional[
         if vt is None:
             vt = v.transpose(-1, -2).contiguous()
         return hip_ext().attn_fwd(q.contiguous(), k.contiguous(), vt, scale)
+    if v is None:
+        v = vt.transpose(-1, -2).contiguous()
     return ref.attn_fwd_ref(q, k, v, scale, causal=True)
 
 

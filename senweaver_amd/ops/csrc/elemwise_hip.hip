@@ -102,6 +102,44 @@ rope_fwd_kernel(ushort* __restrict__ q, ushort* __restrict__ k,
 }
 
 // ---------------------------------------------------------------------------
+// Fused RoPE + head-major scatter: reads q/k in token-major [T, H, D] (the
+// qkv GEMM's natural layout), applies rotate-half RoPE, and writes the
+// attention kernel's [B, H, S, D] layout directly — replacing a rope pass
+// plus two transpose copies (one full q/k read+write each) with one pass.
+// Grid: (T, Hq+Hk), 64 lanes (D=128: lane d pairs with d+64).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(64)
+rope_scatter_kernel(const ushort* __restrict__ q_in, const ushort* __restrict__ k_in,
+                    ushort* __restrict__ q_out, ushort* __restrict__ k_out,
+                    const float* __restrict__ cos_sin,
+                    const int* __restrict__ positions, int Hq, int Hk, int D,
+                    int S) {
+  const int t = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = t / S;
+  const int s = t % S;
+  const int half = D / 2;
+  const ushort* src;
+  ushort* dst;
+  if (h < Hq) {
+    src = q_in + ((long long)t * Hq + h) * D;
+    dst = q_out + (((long long)b * Hq + h) * S + s) * D;
+  } else {
+    src = k_in + ((long long)t * Hk + (h - Hq)) * D;
+    dst = k_out + (((long long)b * Hk + (h - Hq)) * S + s) * D;
+  }
+  const float* cs = cos_sin + (long long)positions[t] * D;
+  for (int d = threadIdx.x; d < half; d += blockDim.x) {
+    const float c = cs[d];
+    const float sn = cs[half + d];
+    const float x1 = bf2f(src[d]);
+    const float x2 = bf2f(src[d + half]);
+    dst[d] = f2bf(x1 * c - x2 * sn);
+    dst[d + half] = f2bf(x2 * c + x1 * sn);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // SwiGLU activation: y = silu(gate) * up, bf16, vectorized.
 //   gateup: [T, 2*I] (fused gate|up GEMM output), y: [T, I]
 // ---------------------------------------------------------------------------

@@ -231,3 +231,17 @@ def test_fp8_model_forward_gpu(dev):
     tokens = torch.randint(0, 512, (1, 128), device=dev)
     h = m.prefill(tokens)
     assert torch.isfinite(h.float()).all()
+
+
+def test_rope_scatter_gpu(dev):
+    B, S, Hq, Hk, D = 2, 128, 4, 2, 128
+    T = B * S
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=dev)
+    cos_sin = ops.rope_tables(4096, D).to(dev)
+    pos = torch.arange(S, dtype=torch.int32, device=dev).repeat(B)
+    qo, ko = ops.rope_scatter(q, k, cos_sin, pos, B, S)
+    q_ref = ref.rope_ref(q, cos_sin, pos).reshape(B, S, Hq, D).permute(0, 2, 1, 3)
+    k_ref = ref.rope_ref(k, cos_sin, pos).reshape(B, S, Hk, D).permute(0, 2, 1, 3)
+    torch.testing.assert_close(qo.float(), q_ref.float(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(ko.float(), k_ref.float(), atol=2e-2, rtol=2e-2)
