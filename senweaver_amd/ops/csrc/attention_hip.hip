@@ -58,7 +58,10 @@ extern "C" __global__ void __launch_bounds__(256, 2)
 attn_fwd_bf16_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
                      const ushort* __restrict__ VT, ushort* __restrict__ O,
                      int B, int H, int Hk, int S, float scale) {
-  const int qb = blockIdx.x;           // Q block of 64 rows
+  // heaviest-first dispatch: causal work grows with the Q block index
+  // (block qb touches qb+1 KV tiles), so reverse the launch order to avoid
+  // a heavy straggler tail in the final dispatch rounds
+  const int qb = gridDim.x - 1 - blockIdx.x;  // Q block of 64 rows
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int kvh = h / (H / Hk);
@@ -74,9 +77,14 @@ attn_fwd_bf16_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
   const int lane = tid & 63;
   const int q0 = qb * QBLK + wid * 16;  // this wave's first Q row
 
-  __shared__ __attribute__((aligned(16))) ushort k_lds[KVBLK * 128];
-  __shared__ __attribute__((aligned(16))) ushort vt_lds[128 * KVBLK];
-  __shared__ __attribute__((aligned(16))) ushort p_lds[4][16 * KVBLK];
+  // double-buffered K/VT tiles (32 KiB per buffer) + wave-private P tiles:
+  // ONE __shared__ object (multiple objects make hipcc drain vmcnt before
+  // every ds_read of a glds pipeline — CDNA4 guide trap (a))
+  __shared__ __attribute__((aligned(16))) ushort smem[2 * 2 * KVBLK * 128 + 4 * 16 * KVBLK];
+  // layout: [buf][K | VT] tiles then the per-wave P region; buffer bases are
+  // computed by index (an initialized pointer array of LDS addresses fails
+  // to compile: addrspacecast in static initializer)
+  ushort* p_lds = smem + 4 * KVBLK * 128;  // [wave][16 * KVBLK]
 
   // Q fragments: lane holds Q[q0 + lane%16][kk*32 + (lane>>4)*8 .. +7]
   short8 qf[4];
@@ -100,11 +108,27 @@ attn_fwd_bf16_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
   for (int nt = 0; nt < 8; ++nt) o_acc[nt] = {0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = min(S, qb * QBLK + QBLK);  // causal upper bound
+  // prologue: stage tile 0 into buffer 0
+  stage_swz(Kh, D, 16, smem, KVBLK * 16, tid);
+  stage_swz(VTh, S, 8, smem + KVBLK * 128, 128 * 8, tid);
+  __syncthreads();
+  int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
-    // ---- stage K[kv0..+64][0..128] and VT[0..128][kv0..+64] ----
-    stage_swz(Kh + (long long)kv0 * D, D, 16, k_lds, KVBLK * 16, tid);
-    stage_swz(VTh + kv0, S, 8, vt_lds, 128 * 8, tid);
-    __syncthreads();  // drains glds + barrier
+    // Pipelined staging with COUNTED vmcnt across raw barriers: the next
+    // tile's 8 glds stay in flight through the whole compute phase
+    // (PMC showed 55% of wave-cycles parked on the __syncthreads drain).
+    // vmcnt(8) = previous tile landed, the 8 just issued may still fly.
+    if (kv0 + KVBLK < kv_end) {
+      ushort* nk = smem + (cur ^ 1) * (2 * KVBLK * 128);
+      stage_swz(Kh + (long long)(kv0 + KVBLK) * D, D, 16, nk, KVBLK * 16, tid);
+      stage_swz(VTh + kv0 + KVBLK, S, 8, nk + KVBLK * 128, 128 * 8, tid);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    const ushort* k_lds = smem + cur * (2 * KVBLK * 128);
+    const ushort* vt_lds = k_lds + KVBLK * 128;
 
     // ---- QK^T: S_tile[16 q][64 kv] = Q[16,128] @ K_tile^T ----
     f32x4 s_frag[4];
@@ -160,7 +184,7 @@ attn_fwd_bf16_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
         const int row = (lane >> 4) * 4 + e;
         const int col = (lane & 15) + kt * 16;
         const int phys = ((col >> 3) ^ (row & 7)) * 8 + (col & 7);
-        p_lds[wid][row * KVBLK + phys] = f2bf(p[kt][e]);
+        p_lds[wid * 16 * KVBLK + row * KVBLK + phys] = f2bf(p[kt][e]);
       }
     }
     // wave-private tile: in-wave ds ordering is handled by the compiler's
@@ -170,14 +194,16 @@ attn_fwd_bf16_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {  // kv 64 in 2 steps of 32
       short8 pfrag = *reinterpret_cast<const short8*>(
-          p_lds[wid] + ((lane & 15) * 8 + (((kk * 4 + kg) & 7) ^ ((lane & 15) & 7))) * 8);
+          p_lds + wid * 16 * KVBLK + ((lane & 15) * 8 + (((kk * 4 + kg) & 7) ^ ((lane & 15) & 7))) * 8);
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         short8 vfrag = read_swz(vt_lds, nt * 16 + (lane & 15), kk * 4 + kg, 8);
         o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag, o_acc[nt], 0, 0, 0);
       }
     }
-    __syncthreads();  // tile consumed; safe to restage
+    __builtin_amdgcn_s_barrier();  // all reads of this tile done (every
+    // ds_read's data was consumed by an MFMA, so lgkm already waited)
+    cur ^= 1;
   }
 
   // ---- epilogue: O /= l, write bf16 ----
