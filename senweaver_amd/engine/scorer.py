@@ -148,6 +148,28 @@ class LlamaBackend:
     def score(self, candidate_prompt: str, rollouts: Sequence[RolloutResult]) -> float:
         return self.score_batch([candidate_prompt], rollouts)[0]
 
+    # ------------------------------------------------------------------
+    # Decode (persistent cache + hipGraph-captured step on GPU)
+    # ------------------------------------------------------------------
+    def _decode_state(self):
+        """Persistent (cache, graph) reused across generate() calls."""
+        if getattr(self, "_decode_cache", None) is None:
+            pages = (self.max_seq + PAGE_SIZE - 1) // PAGE_SIZE + 2
+            self._decode_cache = PagedKVCache(self.config, pages, self.device,
+                                              num_kv_heads=self.model.local_kv_heads)
+            self._decode_graph = None
+            if self.device.type == "cuda":
+                from .graph import DecodeGraph
+                self._decode_graph = DecodeGraph(self.model, self._decode_cache, pages)
+        return self._decode_cache, self._decode_graph
+
+    def _decode_one(self, cache, graph, seq: int, token: int) -> torch.Tensor:
+        if graph is not None:
+            return graph.step(token, seq)[0]
+        pos = torch.tensor([cache.seq_lens[seq]], device=self.device)
+        return self.model.decode_step(torch.tensor([token], device=self.device),
+                                      pos, cache, [seq])[0]
+
     @torch.no_grad()
     def stream_generate(self, prompt: str, max_new_tokens: int,
                         should_stop, on_chunk) -> str:
@@ -158,10 +180,11 @@ class LlamaBackend:
         ids.append(tok.ROLE_ASSISTANT)
         real = len(ids)
         S = _pad64(real)
-        total = real + max_new_tokens + 1
-        pages_needed = (total + PAGE_SIZE - 1) // PAGE_SIZE + 2
-        cache = PagedKVCache(self.config, pages_needed, self.device,
-                             num_kv_heads=self.model.local_kv_heads)
+        cache, graph = self._decode_state()
+        if cache.seq_lens and any(cache.seq_lens):
+            for s in range(len(cache.seq_lens)):
+                cache.free_seq(s)
+            cache.block_tables, cache.seq_lens = [], []
         seq = cache.new_seq()
         tokens = torch.zeros(1, S, dtype=torch.long)
         tokens[0, :real] = torch.tensor(ids, dtype=torch.long)
@@ -180,35 +203,11 @@ class LlamaBackend:
             out_ids.append(nxt)
             text = self.tokenizer.decode(out_ids)
             on_chunk(text)
-            pos = torch.tensor([real + step], device=self.device)
-            last_hidden = self.model.decode_step(
-                torch.tensor([nxt], device=self.device), pos, cache, [seq])[0]
+            last_hidden = self._decode_one(cache, graph, seq, nxt)
         return text
 
     @torch.no_grad()
     def generate(self, prompt: str, max_new_tokens: int = 256) -> str:
-        ids = [tok.BOS] + self.tokenizer.encode(prompt, max_tokens=self.max_seq - max_new_tokens - 8)
-        ids.append(tok.ROLE_ASSISTANT)
-        real = len(ids)
-        S = _pad64(real)
-        total = real + max_new_tokens + 1
-        pages_needed = (total + PAGE_SIZE - 1) // PAGE_SIZE + 2
-        cache = PagedKVCache(self.config, pages_needed, self.device,
-                             num_kv_heads=self.model.local_kv_heads)
-        seq = cache.new_seq()
-        tokens = torch.zeros(1, S, dtype=torch.long)
-        tokens[0, :real] = torch.tensor(ids, dtype=torch.long)
-        hidden = self.model.prefill(tokens.to(self.device), cache=cache, seqs=[seq],
-                                    real_lens=[real])
-        last_hidden = hidden[0, real - 1]
-        out_ids: List[int] = []
-        for step in range(max_new_tokens):
-            logits = self.model.logits(last_hidden.reshape(1, -1))
-            nxt = int(ops.argmax_rows(logits)[0])
-            if nxt == tok.EOS:
-                break
-            out_ids.append(nxt)
-            pos = torch.tensor([real + step], device=self.device)
-            last_hidden = self.model.decode_step(
-                torch.tensor([nxt], device=self.device), pos, cache, [seq])[0]
-        return self.tokenizer.decode(out_ids)
+        return self.stream_generate(prompt, max_new_tokens,
+                                    should_stop=lambda: False,
+                                    on_chunk=lambda _t: None)

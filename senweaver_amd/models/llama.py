@@ -215,6 +215,43 @@ class LlamaModel:
     # ------------------------------------------------------------------
     # Decode: one new token per sequence.  tokens [B], positions [B].
     # ------------------------------------------------------------------
+    def decode_step_tensors(self, tokens: torch.Tensor, pos32: torch.Tensor,
+                            kcaches, vcaches, slot: torch.Tensor,
+                            bt: torch.Tensor, ctx: torch.Tensor) -> torch.Tensor:
+        """Graph-capturable decode step: every input is a device tensor with a
+        stable address (contents may change between replays); no host reads,
+        no allocations outside the caching allocator, no cache bookkeeping.
+
+        tokens/pos32/slot: [B]; bt: [B, max_pages] int32; ctx: [B] int32.
+        kcaches/vcaches: per-layer [P, 16, Hk_local, D] cache tensors.
+        """
+        c = self.config
+        B = tokens.shape[0]
+        hidden = self.embed[tokens]
+        residual = None
+        for li, L in enumerate(self.layers):
+            if residual is None:
+                residual = hidden.clone()
+                h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
+            else:
+                h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
+            qkv = self._linear(h, L, "qkv")
+            qs, kvs = self.local_q_size, self.local_kv_size
+            q = qkv[:, :qs].reshape(B, self.local_heads, c.head_dim).contiguous()
+            k = qkv[:, qs: qs + kvs].reshape(B, self.local_kv_heads, c.head_dim).contiguous()
+            v = qkv[:, qs + kvs:].reshape(B, self.local_kv_heads, c.head_dim).contiguous()
+            ops.rope_inplace(q, k, self.cos_sin, pos32)
+            P = kcaches[li].shape[0]
+            kcaches[li].view(P * 16, self.local_kv_heads, c.head_dim).index_copy_(0, slot, k)
+            vcaches[li].view(P * 16, self.local_kv_heads, c.head_dim).index_copy_(0, slot, v)
+            attn = ops.paged_decode_attn(q, kcaches[li], vcaches[li], bt, ctx, self.scale)
+            attn = attn.reshape(B, self.local_q_size)
+            o = self._linear(attn, L, "o")
+            self.tp.all_reduce_(o)
+            h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
+            hidden = self._ffn(h, L)
+        return ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)
+
     def decode_step(self, tokens: torch.Tensor, positions: torch.Tensor,
                     cache, seqs: List[int]) -> torch.Tensor:
         c = self.config
@@ -222,6 +259,9 @@ class LlamaModel:
         hidden = self.embed[tokens.long()]
         residual = None
         pos32 = positions.to(torch.int32)
+        # context length for this token, captured before any append
+        ctx_vals = [cache.seq_lens[s] + 1 for s in seqs]
+        bt = ctx = None
         for li, L in enumerate(self.layers):
             if residual is None:
                 residual = hidden.clone()
@@ -237,10 +277,11 @@ class LlamaModel:
             for b in range(B):
                 cache.append(li, seqs[b], k[b: b + 1], v[b: b + 1],
                              advance_len=(li == c.num_layers - 1))
-            bt = cache.block_table_tensor(seqs)
-            # context includes the just-appended token for every layer:
-            ctx = torch.tensor([cache.seq_lens[s] + (1 if li < c.num_layers - 1 else 0)
-                                for s in seqs], dtype=torch.int32, device=self.device)
+            if li == 0:
+                # block table / context lengths are layer-invariant for this
+                # token (context = pre-append len + 1): build them once
+                bt = cache.block_table_tensor(seqs)
+                ctx = torch.tensor(ctx_vals, dtype=torch.int32, device=self.device)
             attn = ops.paged_decode_attn(q, cache.k[li], cache.v[li], bt, ctx, self.scale)
             attn = attn.reshape(B, self.local_q_size)
             o = self._linear(attn, L, "o")

@@ -105,7 +105,7 @@ def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
         return ref.gemm_bt_ref(a, b)
     ext = hip_ext()
     M = a.shape[0]
-    if M <= 16 and b.shape[1] % 1024 == 0 and b.shape[0] % 64 == 0:
+    if M <= 16 and b.shape[1] % 512 == 0 and b.shape[0] % 4 == 0:
         return ext.gemm_bt(a.contiguous(), b.contiguous())
     # pad M to the 256-tile when N allows it and the grid fills the chip
     N = b.shape[0]

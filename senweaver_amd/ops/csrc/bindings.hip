@@ -20,8 +20,11 @@ extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*,
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
 extern "C" __global__ void gemm_bt_fp8_kernel(const unsigned char*, const float*, const unsigned char*, const float*, ushort*, int, int, int);
-extern "C" __global__ void gemv_bt_bf16_kernel(const ushort*, const ushort*, float*, int, int, int, int);
-extern "C" __global__ void gemv_reduce_kernel(const float*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_v2_m1(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_v2_m4(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_v2_m8(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_v2_m16(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
 extern "C" __global__ void paged_decode_attn_kernel(const ushort*, const ushort*, const ushort*, ushort*,
                                                     const int*, const int*, int, int, int, float);
@@ -149,20 +152,26 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
   TORCH_CHECK(B.size(1) == K, "K mismatch");
   auto C = torch::empty({M, N}, A.options());
   if (M <= 16) {
-    TORCH_CHECK(N % 64 == 0 && K % 1024 == 0, "gemv path needs N%64==0, K%1024==0");
-    // split-K sized so blocks ~= CU count (microarch: wall ~= one block's
-    // latency until blocks exceed CUs)
-    int splitk = 1;
-    while ((long long)(N / 64) * splitk * 2 <= 256 && (K / (splitk * 2)) % 256 == 0 && splitk < 8)
-      splitk *= 2;
-    auto part = torch::empty({splitk, M, N}, A.options().dtype(torch::kFloat32));
-    gemv_bt_bf16_kernel<<<dim3(N / 64, splitk), dim3(256), 0, cur_stream()>>>(
-        bf16_ptr(A), bf16_ptr(B), part.data_ptr<float>(), M, N, K, splitk);
-    HIP_CHECK_KERNEL();
-    const long long total = (long long)M * N;
-    const int blocks = (int)std::min<long long>((total + 255) / 256, 2048);
-    gemv_reduce_kernel<<<dim3(blocks), dim3(256), 0, cur_stream()>>>(
-        part.data_ptr<float>(), bf16_mut(C), M, N, splitk);
+    TORCH_CHECK(N % 4 == 0 && K % 512 == 0, "gemv path needs N%4==0, K%512==0");
+    // wave-per-row coalesced GEMV; A must be padded to the MM bucket rows
+    int MM = M <= 1 ? 1 : M <= 2 ? 2 : M <= 4 ? 4 : M <= 8 ? 8 : 16;
+    torch::Tensor Ap = A;
+    if (A.size(0) < MM) {
+      Ap = torch::zeros({MM, (long)K}, A.options());
+      Ap.narrow(0, 0, M).copy_(A);
+    }
+    const dim3 grid((N + 3) / 4);
+    auto launch = [&](auto kern) {
+      kern<<<grid, dim3(256), 0, cur_stream()>>>(bf16_ptr(Ap), bf16_ptr(B),
+                                                 bf16_mut(C), M, N, K);
+    };
+    switch (MM) {
+      case 1: launch(gemv_bt_bf16_v2_m1); break;
+      case 2: launch(gemv_bt_bf16_v2_m2); break;
+      case 4: launch(gemv_bt_bf16_v2_m4); break;
+      case 8: launch(gemv_bt_bf16_v2_m8); break;
+      default: launch(gemv_bt_bf16_v2_m16); break;
+    }
     HIP_CHECK_KERNEL();
     return C;
   }

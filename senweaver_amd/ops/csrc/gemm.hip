@@ -263,64 +263,44 @@ gemm_bt_bf16_256_kernel(const ushort* __restrict__ A, const ushort* __restrict__
 }
 
 // ---------------------------------------------------------------------------
-// Split-K GEMV-ish path for skinny M (decode): M <= 16.
-// C[M,N] = A[M,K] @ B[N,K]^T with one workgroup per 64 N-rows x K-slice;
-// weights stream straight to VGPRs (LDS round-trip is pure overhead at M<=16
-// since B rows are used once), f32 atomics-free two-pass reduction.
-// Grid: (N/64, SPLITK); partials [SPLITK, M, N] f32 then reduced by
-// gemv_reduce_kernel (launch-boundary reduce — cheaper than in-launch
-// combine at this size per the microarch price list).
+// Decode GEMV: M <= 16 (beam decode).  One WAVE per output row n, lanes
+// stride K (coalesced 16 B/lane row reads — the previous row-per-lane form
+// was ~4x off the bandwidth bound), f32 accumulation, one wave reduction
+// per (m, n) at the end.  N/4 blocks fill the chip without split-K, so the
+// partial-reduce pass is gone too.  A (M x K, <=128 KiB) streams from L2.
 // ---------------------------------------------------------------------------
-extern "C" __global__ void __launch_bounds__(256)
-gemv_bt_bf16_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
-                    float* __restrict__ Cpart, int M, int N, int K, int splitk) {
-  const int n = blockIdx.x * 64 + (threadIdx.x & 63);  // one N row per lane
-  const int slice = blockIdx.y;
-  const int wid = threadIdx.x >> 6;  // 4 waves split K within the slice
-  const int kper = K / splitk;
-  const int k0 = slice * kper + wid * (kper / 4);
-  const int k1 = k0 + kper / 4;
-
-  const ushort* brow = B + (long long)n * K;
-  // M <= 16 accumulators per lane
-  float acc[16];
-#pragma unroll
-  for (int m = 0; m < 16; ++m) acc[m] = 0.f;
-
-  for (int k = k0; k < k1; k += 8) {
-    bf16x8 bv = *reinterpret_cast<const bf16x8*>(brow + k);
-    float bfv[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) bfv[j] = bf2f(bv.v[j]);
-    for (int m = 0; m < M; ++m) {
-      bf16x8 av = *reinterpret_cast<const bf16x8*>(A + (long long)m * K + k);
-      float s = 0.f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) s += bf2f(av.v[j]) * bfv[j];
-      acc[m] += s;
-    }
+// concrete instantiations with C linkage for the bindings TU
+#define GEMV2_INST(MM)                                                        \
+  extern "C" __global__ void __launch_bounds__(256)                           \
+  gemv_bt_bf16_v2_m##MM(const ushort* A, const ushort* B, ushort* C, int M,   \
+                        int N, int K) {                                       \
+    const int wid = threadIdx.x >> 6;                                         \
+    const int lane = threadIdx.x & 63;                                        \
+    const int n = blockIdx.x * 4 + wid;                                       \
+    if (n >= N) return;                                                       \
+    const ushort* brow = B + (long long)n * K;                                \
+    float acc[MM];                                                            \
+    _Pragma("unroll") for (int m = 0; m < MM; ++m) acc[m] = 0.f;              \
+    for (int k = lane * 8; k < K; k += 64 * 8) {                              \
+      bf16x8 bv = *reinterpret_cast<const bf16x8*>(brow + k);                 \
+      float bfv[8];                                                           \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) bfv[j] = bf2f(bv.v[j]);   \
+      _Pragma("unroll") for (int m = 0; m < MM; ++m) {                        \
+        bf16x8 av = *reinterpret_cast<const bf16x8*>(A + (long long)m * K + k); \
+        float s = 0.f;                                                        \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j)                         \
+            s += bf2f(av.v[j]) * bfv[j];                                      \
+        acc[m] += s;                                                          \
+      }                                                                       \
+    }                                                                         \
+    _Pragma("unroll") for (int m = 0; m < MM; ++m) {                          \
+      float v = wave_reduce_sum(acc[m]);                                      \
+      if (lane == 0 && m < M) C[(long long)m * N + n] = f2bf(v);              \
+    }                                                                         \
   }
-  // combine the 4 waves' partials via LDS, then write one partial per slice
-  __shared__ float part[4][16][64];
-  const int lane = threadIdx.x & 63;
-  for (int m = 0; m < M; ++m) part[wid][m][lane] = acc[m];
-  __syncthreads();
-  if (wid == 0) {
-    for (int m = 0; m < M; ++m) {
-      float v = part[0][m][lane] + part[1][m][lane] + part[2][m][lane] + part[3][m][lane];
-      Cpart[((long long)slice * M + m) * N + n] = v;
-    }
-  }
-}
 
-extern "C" __global__ void __launch_bounds__(256)
-gemv_reduce_kernel(const float* __restrict__ Cpart, ushort* __restrict__ C,
-                   int M, int N, int splitk) {
-  const long long total = (long long)M * N;
-  for (long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-       idx < total; idx += (long long)gridDim.x * blockDim.x) {
-    float v = 0.f;
-    for (int s = 0; s < splitk; ++s) v += Cpart[(long long)s * total + idx];
-    C[idx] = f2bf(v);
-  }
-}
+GEMV2_INST(1)
+GEMV2_INST(2)
+GEMV2_INST(4)
+GEMV2_INST(8)
+GEMV2_INST(16)
