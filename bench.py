@@ -94,6 +94,10 @@ def main() -> int:
     ap.add_argument("--branch-factor", type=int, default=4)
     ap.add_argument("--rollouts", type=int, default=4)
     ap.add_argument("--micro-batch", type=int, default=8)
+    ap.add_argument("--quant", default="bf16", choices=["bf16", "fp8"])
+    ap.add_argument("--tp", action="store_true",
+                    help="tensor-parallel scoring over the whole world (config 5) "
+                         "instead of candidate-parallel DP")
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -107,8 +111,13 @@ def main() -> int:
     from senweaver_amd.apo.optimizer import rollout_weight  # noqa: F401
 
     cfg = get_config(args.model)
+    tp_ctx = None
+    if args.tp and world > 1:
+        from senweaver_amd.parallel.tp import TPContext
+        tp_ctx = TPContext.from_default_group()
     backend = LlamaBackend(cfg, device=device, seed=0, max_seq=args.seq_len,
-                           micro_batch=args.micro_batch)
+                           micro_batch=args.micro_batch, tp=tp_ctx,
+                           quant=args.quant)
 
     n_candidates = args.beam_width * args.branch_factor
     sequences = build_synthetic_workload(cfg, cfg.vocab_size, n_candidates,
@@ -119,7 +128,9 @@ def main() -> int:
     comm_dev = torch.device(device) if use_cuda else torch.device("cpu")
 
     # tokenized (candidate x rollout) sequences as device tensors, sharded
-    my_cands = list(range(rank, n_candidates, world))
+    # (TP mode: every rank holds a shard of the MODEL and scores ALL candidates)
+    my_cands = (list(range(n_candidates)) if args.tp
+                else list(range(rank, n_candidates, world)))
     my_tok, my_mask = [], []
     for ci in my_cands:
         for ids, mask in sequences[ci * args.rollouts:(ci + 1) * args.rollouts]:
@@ -140,7 +151,10 @@ def main() -> int:
         for j, ci in enumerate(my_cands):
             chunk = lps[j * args.rollouts:(j + 1) * args.rollouts]
             my_scores.append(sum(w * l for w, l in zip(weights, chunk)) / wnorm)
-        scores = P.dp_scores_allreduce(n_candidates, my_cands, my_scores, comm_dev)
+        if args.tp:
+            scores = my_scores  # identical on every rank (collective forward)
+        else:
+            scores = P.dp_scores_allreduce(n_candidates, my_cands, my_scores, comm_dev)
         # Top-K beam selection (deterministic, every rank)
         order = sorted(range(n_candidates), key=lambda i: (-scores[i], i))
         return [order[i] for i in range(args.beam_width)], scores
@@ -182,13 +196,13 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": args.quant,
             "data": "synthetic",
             "config": {
                 "model": args.model,
                 "global_batch": n_candidates * args.rollouts,
                 "seq_len": args.seq_len,
-                "parallelism": f"dp{n_gpus}",
+                "parallelism": (f"tp{n_gpus}" if args.tp else f"dp{n_gpus}"),
                 "beam_width": args.beam_width,
                 "branch_factor": args.branch_factor,
                 "gradient_batch_size": args.rollouts,
