@@ -56,7 +56,7 @@ def bench_attn():
         v = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
         vt = v.transpose(-1, -2).contiguous()
         scale = 1.0 / math.sqrt(D)
-        t = timeit(lambda: ops.attn_fwd(q, k, v, scale, vt=vt))
+        t = timeit(lambda: ops.attn_fwd_t(q, k, vt, scale))
         flops = 2 * 2 * B * H * S * S * D / 2  # causal half
         print(f"ATTN B{B} H{H} S{S}: {flops / t / 1e12:7.1f} TF/s  ({t * 1e3:.2f} ms)")
         tsdpa = timeit(lambda: torch.nn.functional.scaled_dot_product_attention(

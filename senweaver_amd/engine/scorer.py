@@ -31,7 +31,8 @@ from . import tokenizer as tok
 
 
 def _pad64(n: int) -> int:
-    return (n + 63) & ~63
+    # pad to the v2 attention kernel's 128-row Q-block granularity
+    return (n + 127) & ~127
 
 
 class LlamaBackend:
@@ -181,10 +182,9 @@ class LlamaBackend:
         real = len(ids)
         S = _pad64(real)
         cache, graph = self._decode_state()
-        if cache.seq_lens and any(cache.seq_lens):
-            for s in range(len(cache.seq_lens)):
-                cache.free_seq(s)
-            cache.block_tables, cache.seq_lens = [], []
+        # full reset (also recovers pages leaked by an aborted prefill)
+        cache._free = list(range(cache.num_pages - 1, -1, -1))
+        cache.block_tables, cache.seq_lens = [], []
         seq = cache.new_seq()
         tokens = torch.zeros(1, S, dtype=torch.long)
         tokens[0, :real] = torch.tensor(ids, dtype=torch.long)

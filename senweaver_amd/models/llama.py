@@ -155,8 +155,12 @@ class LlamaModel:
                                  advance_len=(li == c.num_layers - 1))
             # V^T computed directly: the attention kernel consumes [B,Hk,D,S]
             vt = v.reshape(B, S, self.local_kv_heads, c.head_dim).permute(0, 2, 3, 1).contiguous()
-            attn = ops.attn_fwd(qb, kb, None, self.scale, vt=vt)  # [B,Hq_local,S,D]
-            attn = attn.transpose(1, 2).reshape(T, self.local_q_size).contiguous()
+            if S % 128 == 0:
+                ot = ops.attn_fwd_t(qb, kb, vt, self.scale)  # O^T [B,Hq_local,D,S]
+                attn = ot.permute(0, 3, 1, 2).reshape(T, self.local_q_size).contiguous()
+            else:  # v1 kernel path for 64-granular sequences
+                ab = ops.attn_fwd(qb, kb, None, self.scale, vt=vt)
+                attn = ab.transpose(1, 2).reshape(T, self.local_q_size).contiguous()
             o = self._linear(attn, L, "o")
             self.tp.all_reduce_(o)  # row-parallel o_proj partial sum
             h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)

@@ -174,16 +174,36 @@ def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional[
     """Causal GQA attention. q [B,H,S,D], k/v [B,Hk,S,D] (S % 64 == 0 on GPU).
 
     ``vt`` may be passed pre-transposed ([B,Hk,D,S]) to skip the transpose.
+    Dispatches to the swapped-QK^T v2 kernel at S % 128 == 0 (faster), the
+    v1 kernel otherwise.
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
         if vt is None:
             vt = v.transpose(-1, -2).contiguous()
+        if q.shape[2] % 128 == 0:
+            ot = hip_ext().attn_fwd_v2(q.contiguous(), k.contiguous(), vt, scale)
+            return ot.transpose(-1, -2).contiguous()
         return hip_ext().attn_fwd(q.contiguous(), k.contiguous(), vt, scale)
     if v is None:
         v = vt.transpose(-1, -2).contiguous()
     return ref.attn_fwd_ref(q, k, v, scale, causal=True)
+
+
+def attn_fwd_t(q: torch.Tensor, k: torch.Tensor, vt: torch.Tensor,
+               scale: Optional[float] = None) -> torch.Tensor:
+    """Causal GQA attention returning O^T [B,H,D,S] (the v2 kernel's native
+    layout — saves one transpose copy when the caller wants token-major
+    activations next).  S % 128 == 0 on GPU."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        return hip_ext().attn_fwd_v2(q.contiguous(), k.contiguous(),
+                                     vt.contiguous(), scale)
+    v = vt.transpose(-1, -2).contiguous()
+    o = ref.attn_fwd_ref(q, k, v, scale, causal=True)
+    return o.transpose(-1, -2).contiguous()
 
 
 def paged_decode_attn(q, kcache, vcache, block_table, ctx_lens, scale: Optional[float] = None):

@@ -262,3 +262,30 @@ def test_decode_graph_matches_eager(dev):
     # repeated generate reuses the captured graph and stays deterministic
     out2 = b_graph.generate("check the graph decode path", max_new_tokens=10)
     assert out2 == out_graph
+
+
+@pytest.mark.parametrize("B,H,Hk,S", [(2, 8, 2, 256), (1, 32, 8, 1024)])
+def test_attn_fwd_v2(dev, B, H, Hk, S):
+    D = 128
+    q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+    vt = v.transpose(-1, -2).contiguous()
+    scale = 1.0 / math.sqrt(D)
+    ot = ops.attn_fwd_t(q, k, vt, scale)
+    o_ref = ref.attn_fwd_ref(q, k, v, scale, causal=True)
+    torch.testing.assert_close(ot.transpose(-1, -2).float(), o_ref.float(),
+                               atol=8e-2, rtol=8e-2)
+
+
+def test_attn_fwd_v2_spiked(dev):
+    B, H, Hk, S, D = 1, 2, 2, 512, 128
+    q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+    k[:, :, 100] *= 8
+    v = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+    scale = 1.0 / math.sqrt(D)
+    ot = ops.attn_fwd_t(q, k, v.transpose(-1, -2).contiguous(), scale)
+    o_ref = ref.attn_fwd_ref(q, k, v, scale, causal=True)
+    torch.testing.assert_close(ot.transpose(-1, -2).float(), o_ref.float(),
+                               atol=8e-2, rtol=8e-2)
