@@ -142,19 +142,21 @@ class LlamaModel:
                 h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
             qkv = self._linear(h, L, "qkv")  # [T, local q+2kv]
             qs, kvs = self.local_q_size, self.local_kv_size
-            q = qkv[:, :qs].reshape(T, self.local_heads, c.head_dim).contiguous()
-            k = qkv[:, qs: qs + kvs].reshape(T, self.local_kv_heads, c.head_dim).contiguous()
-            v = qkv[:, qs + kvs:].reshape(T, self.local_kv_heads, c.head_dim).contiguous()
-            # fused rope + [B,H,S,D] scatter (replaces rope + 2 transposes)
-            qb, kb = ops.rope_scatter(q, k, self.cos_sin, positions, B, S)
+            # fused rope + [B,H,S,D] scatter reading the qkv slices in place
+            qb, kb = ops.rope_scatter_qkv(qkv, self.cos_sin, positions,
+                                          self.local_heads, self.local_kv_heads,
+                                          c.head_dim, B, S)
+            vview = qkv[:, qs + kvs:].reshape(B, S, self.local_kv_heads, c.head_dim)
             if cache is not None:
                 kflat = kb.permute(0, 2, 1, 3).reshape(T, self.local_kv_heads, c.head_dim)
+                vflat = vview.reshape(T, self.local_kv_heads, c.head_dim)
                 for b in range(B):
                     n = real_lens[b] if real_lens is not None else S
-                    cache.append(li, seqs[b], kflat[b * S: b * S + n], v[b * S: b * S + n],
+                    cache.append(li, seqs[b], kflat[b * S: b * S + n],
+                                 vflat[b * S: b * S + n].contiguous(),
                                  advance_len=(li == c.num_layers - 1))
             # V^T computed directly: the attention kernel consumes [B,Hk,D,S]
-            vt = v.reshape(B, S, self.local_kv_heads, c.head_dim).permute(0, 2, 3, 1).contiguous()
+            vt = vview.permute(0, 2, 3, 1).contiguous()
             if S % 128 == 0:
                 ot = ops.attn_fwd_t(qb, kb, vt, self.scale)  # O^T [B,Hq_local,D,S]
                 attn = ot.permute(0, 3, 1, 2).reshape(T, self.local_q_size).contiguous()

@@ -80,15 +80,23 @@ def rope_inplace(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor, positi
     k.copy_(ref.rope_ref(k, cos_sin, positions))
 
 
-def rope_scatter(q: torch.Tensor, k: torch.Tensor, cos_sin, positions, B: int, S: int):
-    """q/k [T,H,D] -> rope -> ([B,Hq,S,D], [B,Hk,S,D]) in one pass."""
-    if _on_gpu(q):
-        qo, ko = hip_ext().rope_scatter(q.contiguous(), k.contiguous(), cos_sin,
-                                        positions.to(torch.int32), B, S)
+def rope_scatter_qkv(qkv: torch.Tensor, cos_sin, positions, Hq: int, Hk: int,
+                     D: int, B: int, S: int):
+    """Fused qkv [T, (Hq+2Hk)*D] -> rope(q,k) -> ([B,Hq,S,D], [B,Hk,S,D]).
+
+    Reads the q/k head slices straight from the fused projection output (no
+    contiguous slice copies).  The v slice is untouched (consumed separately
+    as V^T)."""
+    if _on_gpu(qkv):
+        qo, ko = hip_ext().rope_scatter_qkv(qkv.contiguous(), cos_sin,
+                                            positions.to(torch.int32),
+                                            Hq, Hk, D, B, S)
         return qo, ko
+    T = qkv.shape[0]
+    q = qkv[:, : Hq * D].reshape(T, Hq, D)
+    k = qkv[:, Hq * D: (Hq + Hk) * D].reshape(T, Hk, D)
     qr = ref.rope_ref(q, cos_sin, positions)
     kr = ref.rope_ref(k, cos_sin, positions)
-    Hq, Hk, D = q.shape[1], k.shape[1], q.shape[2]
     return (qr.reshape(B, S, Hq, D).permute(0, 2, 1, 3).contiguous(),
             kr.reshape(B, S, Hk, D).permute(0, 2, 1, 3).contiguous())
 

@@ -10,7 +10,7 @@
 // ---- kernel prototypes (defined in the sibling .hip translation units) ----
 extern "C" __global__ void rmsnorm_fwd_kernel(const ushort*, const ushort*, ushort*, ushort*, int, float);
 extern "C" __global__ void rope_fwd_kernel(ushort*, ushort*, const float*, const int*, int, int, int);
-extern "C" __global__ void rope_scatter_kernel(const ushort*, const ushort*, ushort*, ushort*, const float*, const int*, int, int, int, int);
+extern "C" __global__ void rope_scatter_kernel(const ushort*, long long, ushort*, ushort*, const float*, const int*, int, int, int, int);
 extern "C" __global__ void swiglu_fwd_kernel(const ushort*, ushort*, long long, int);
 extern "C" __global__ void add_bf16_kernel(const ushort*, const ushort*, ushort*, long long);
 extern "C" __global__ void argmax_rows_kernel(const ushort*, int*, int);
@@ -97,19 +97,21 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
   HIP_CHECK_KERNEL();
 }
 
-// q/k: [T, H, D] -> rope -> [B, H, S, D] (T = B*S)
-std::vector<torch::Tensor> rope_scatter(torch::Tensor q, torch::Tensor k,
-                                        torch::Tensor cos_sin, torch::Tensor positions,
-                                        int64_t B, int64_t S) {
-  check_bf16(q, "q");
-  check_bf16(k, "k");
-  const int T = q.size(0), Hq = q.size(1), Hk = k.size(1), D = q.size(2);
-  TORCH_CHECK(T == B * S && cos_sin.size(1) == D);
-  auto qo = torch::empty({B, (long)Hq, S, (long)D}, q.options());
-  auto ko = torch::empty({B, (long)Hk, S, (long)D}, k.options());
- hipLaunchKernelGGL(( rope_scatter_kernel), dim3(dim3(T, Hq + Hk)), dim3(dim3(64)), 0, cur_stream(), 
-      bf16_ptr(q), bf16_ptr(k), bf16_mut(qo), bf16_mut(ko),
-      cos_sin.data_ptr<float>(), positions.data_ptr<int>(), Hq, Hk, D, (int)S);
+// qkv: [T, Hq*D + 2*Hk*D] fused projection -> rope -> q [B,Hq,S,D], k [B,Hk,S,D]
+std::vector<torch::Tensor> rope_scatter_qkv(torch::Tensor qkv, torch::Tensor cos_sin,
+                                            torch::Tensor positions,
+                                            int64_t Hq, int64_t Hk, int64_t D,
+                                            int64_t B, int64_t S) {
+  check_bf16(qkv, "qkv");
+  const long long T = qkv.size(0);
+  const long long ld = qkv.size(1);
+  TORCH_CHECK(T == B * S && cos_sin.size(1) == D && ld >= (Hq + 2 * Hk) * D);
+  auto qo = torch::empty({B, Hq, S, D}, qkv.options());
+  auto ko = torch::empty({B, Hk, S, D}, qkv.options());
+ hipLaunchKernelGGL(( rope_scatter_kernel), dim3(dim3((unsigned)T, (unsigned)(Hq + Hk))), dim3(dim3(64)), 0, cur_stream(), 
+      bf16_ptr(qkv), ld, bf16_mut(qo), bf16_mut(ko),
+      cos_sin.data_ptr<float>(), positions.data_ptr<int>(), (int)Hq, (int)Hk,
+      (int)D, (int)S);
   HIP_CHECK_KERNEL();
   return {qo, ko};
 }
@@ -336,7 +338,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm fwd (bf16)");
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual += x; y = rmsnorm(residual)");
   m.def("rope_inplace", &rope_inplace, "RoPE in place over q,k");
-  m.def("rope_scatter", &rope_scatter, "fused RoPE + [B,H,S,D] scatter");
+  m.def("rope_scatter_qkv", &rope_scatter_qkv, "fused RoPE + [B,H,S,D] scatter from qkv");
   m.def("swiglu", &swiglu, "silu(gate)*up from fused gateup");
   m.def("add_bf16", &add_bf16, "a + b (bf16)");
   m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");

@@ -108,11 +108,13 @@ rope_fwd_kernel(ushort* __restrict__ q, ushort* __restrict__ k,
 // Grid: (T, Hq+Hk), 64 lanes (D=128: lane d pairs with d+64).
 // ---------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(64)
-rope_scatter_kernel(const ushort* __restrict__ q_in, const ushort* __restrict__ k_in,
+rope_scatter_kernel(const ushort* __restrict__ qkv, long long ld,
                     ushort* __restrict__ q_out, ushort* __restrict__ k_out,
                     const float* __restrict__ cos_sin,
                     const int* __restrict__ positions, int Hq, int Hk, int D,
                     int S) {
+  // reads q/k heads straight out of the fused qkv GEMM output (row stride
+  // ld = q_size + 2*kv_size) — no separate contiguous slice copies
   const int t = blockIdx.x;
   const int h = blockIdx.y;
   const int b = t / S;
@@ -121,10 +123,10 @@ rope_scatter_kernel(const ushort* __restrict__ q_in, const ushort* __restrict__ 
   const ushort* src;
   ushort* dst;
   if (h < Hq) {
-    src = q_in + ((long long)t * Hq + h) * D;
+    src = qkv + (long long)t * ld + (long long)h * D;
     dst = q_out + (((long long)b * Hq + h) * S + s) * D;
   } else {
-    src = k_in + ((long long)t * Hk + (h - Hq)) * D;
+    src = qkv + (long long)t * ld + (long long)Hq * D + (long long)(h - Hq) * D;
     dst = k_out + (((long long)b * Hk + (h - Hq)) * S + s) * D;
   }
   const float* cs = cos_sin + (long long)positions[t] * D;

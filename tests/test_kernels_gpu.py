@@ -236,11 +236,12 @@ def test_fp8_model_forward_gpu(dev):
 def test_rope_scatter_gpu(dev):
     B, S, Hq, Hk, D = 2, 128, 4, 2, 128
     T = B * S
-    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=dev)
-    k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=dev)
+    qkv = torch.randn(T, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=dev)
+    q = qkv[:, : Hq * D].reshape(T, Hq, D)
+    k = qkv[:, Hq * D: (Hq + Hk) * D].reshape(T, Hk, D)
     cos_sin = ops.rope_tables(4096, D).to(dev)
     pos = torch.arange(S, dtype=torch.int32, device=dev).repeat(B)
-    qo, ko = ops.rope_scatter(q, k, cos_sin, pos, B, S)
+    qo, ko = ops.rope_scatter_qkv(qkv, cos_sin, pos, Hq, Hk, D, B, S)
     q_ref = ref.rope_ref(q, cos_sin, pos).reshape(B, S, Hq, D).permute(0, 2, 1, 3)
     k_ref = ref.rope_ref(k, cos_sin, pos).reshape(B, S, Hk, D).permute(0, 2, 1, 3)
     torch.testing.assert_close(qo.float(), q_ref.float(), atol=2e-2, rtol=2e-2)
