@@ -115,10 +115,12 @@ class PersistentTerminal:
 
 
 class ToolsService:
-    def __init__(self, workspace_root: str, subagent_runner=None) -> None:
+    def __init__(self, workspace_root: str, subagent_runner=None,
+                 edit_backend=None) -> None:
         self.root = os.path.abspath(workspace_root)
         self.terminals: Dict[str, PersistentTerminal] = {}
         self._subagents = subagent_runner
+        self._edit_backend = edit_backend  # PromptOptimizerBackend-like .generate
 
     # ---- path safety ----
     def _resolve(self, uri: Optional[str]) -> str:
@@ -354,6 +356,32 @@ class ToolsService:
         if not res.success:
             raise ToolError(res.error or "subagent failed")
         return ToolResult("spawn_subagent", {"taskId": res.task_id}, res.summary)
+
+    def _tool_edit_agent(self, p):
+        """LLM-driven file edit (reference browser/editAgentService.ts:458):
+        modes 'create' | 'edit' | 'overwrite'; the model produces the new
+        content from the description (+ current content for edits)."""
+        if self._edit_backend is None:
+            raise ToolError("edit_agent backend not configured")
+        mode = (p.get("mode") or "edit").strip().lower()
+        uri = p.get("uri")
+        if not uri:
+            raise ToolError("edit_agent requires uri")
+        current = p.get("current_content")
+        if current is None and mode != "create":
+            path = self._resolve(uri)
+            current = (open(path, encoding="utf-8", errors="replace").read()
+                       if os.path.exists(path) else "")
+        prompt = (f"File: {uri}\nTask ({mode}): {p.get('description', '')}\n"
+                  + (f"Current content:\n{(current or '')[:8000]}\n" if mode != "create" else "")
+                  + "Return ONLY the complete new file content.")
+        new_content = self._edit_backend.generate(prompt, max_new_tokens=512)
+        path = self._resolve(uri)
+        os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+        with open(path, "w", encoding="utf-8") as f:
+            f.write(new_content)
+        return ToolResult("edit_agent", {"uri": uri, "mode": mode},
+                          f"{mode} applied to {uri} ({len(new_content)} chars)")
 
     def _tool_skill(self, p):
         from .skills import SkillService

@@ -291,3 +291,16 @@ def test_scheduler_parallel_execution():
     assert concurrency["max"] <= get_agent_composition("agent").max_parallel
     merged = sched.merge_sub_agent_results(results)
     assert "[OK]" in merged
+
+
+def test_edit_agent_tool(tmp_path):
+    from senweaver_amd.apo.optimizer import StubBackend
+    svc = ToolsService(str(tmp_path), edit_backend=StubBackend())
+    r = svc.call_tool("edit_agent", {"uri": "gen.md", "mode": "create",
+                                     "description": "write rules"})
+    assert "create applied" in r.text
+    assert (tmp_path / "gen.md").exists()
+    # edit mode reads current content into the prompt
+    (tmp_path / "x.txt").write_text("ORIGINAL-CONTENT")
+    svc.call_tool("edit_agent", {"uri": "x.txt", "mode": "edit", "description": "improve"})
+    assert (tmp_path / "x.txt").read_text() != "ORIGINAL-CONTENT"
