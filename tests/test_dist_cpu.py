@@ -131,3 +131,69 @@ def test_tp2_matches_tp1():
     for rank, vals in results.items():
         assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
     assert results[0] == pytest.approx(results[1], abs=1e-5)
+
+
+def _payload_distributed_beam(rank, world):
+    """Full distributed beam round: rank-0 expansion broadcast + sharded
+    scoring + all-reduced Top-K -> identical beam state on every rank."""
+    from senweaver_amd.apo import APOService, BeamSearchEngine
+    from senweaver_amd.apo.optimizer import StubBackend
+    from senweaver_amd.parallel import CandidateParallelScorer, broadcast_strings
+    from senweaver_amd.storage import MemoryStorage
+    from senweaver_amd.trace import TraceCollector
+    import torch
+
+    class CPUStub(StubBackend):
+        device = torch.device("cpu")
+
+        def score_batch(self, prompts, rollouts):
+            return [self.score(p, rollouts) for p in prompts]
+
+    backend = CPUStub(seed=11)
+    storage = MemoryStorage()
+    tc = TraceCollector(storage=storage, clock=lambda: 1700000000000,
+                        uuid_fn=lambda: f"u{id(object()) % 97}")
+    n = {"i": 0}
+
+    def uuid_fn():
+        n["i"] += 1
+        return f"uuid-{n['i']}"
+
+    tc._uuid = uuid_fn
+    for i in range(4):
+        th = f"th{i}"
+        tid = tc.start_trace(th, {"chatMode": "agent"})
+        tc.record_user_message(th, 0, f"task {i}")
+        tc.record_assistant_message(th, 1, f"done {i}")
+        tc.end_trace(tid)
+        tc.record_user_feedback(th, 1, "good" if i % 2 else "bad")
+    apo = APOService(tc, storage=storage, clock=lambda: 1700000000000, uuid_fn=uuid_fn)
+
+    def expand_fn(parent_contents, rollouts, branch_factor):
+        if rank == 0:
+            out = []
+            for p in parent_contents:
+                for b in range(branch_factor):
+                    out.append(backend.generate(f"{p}::variant{b}"))
+        else:
+            out = None
+        return broadcast_strings(out, src=0)
+
+    engine = BeamSearchEngine(backend, score_fn=CandidateParallelScorer(backend),
+                              expand_fn=expand_fn)
+    engine.run_round(apo)
+    engine.run_round(apo)
+    state = apo.get_beam_state()
+    return {
+        "round": state.current_round,
+        "beam": [(b.version, b.content[:40], round(b.score, 6)) for b in state.beam],
+        "best": state.history_best_prompt.version,
+        "best_score": round(state.history_best_score, 6),
+    }
+
+
+def test_distributed_beam_identical_state():
+    results = _run_dist("_payload_distributed_beam")
+    assert results[0] == results[1]
+    assert results[0]["round"] == 2
+    assert len(results[0]["beam"]) == 4
