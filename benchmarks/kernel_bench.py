@@ -47,6 +47,24 @@ def bench_gemm():
         print(f"GEMM {M}x{N}x{K}: ours {tf:7.1f} TF/s   hipblaslt {tf2:7.1f} TF/s")
 
 
+def bench_fp8():
+    dev = torch.device("cuda:0")
+    shapes = [(4096, 4096, 4096), (8192, 8192, 8192), (2048, 28672, 4096)]
+    for M, N, K in shapes:
+        a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+        b = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+        aq, asc = ops.quant_fp8(a)
+        bq, bsc = ops.quant_fp8(b)
+        t = timeit(lambda: ops.gemm_bt_fp8(aq, asc, bq, bsc))
+        amq, ams = ops.quant_mxfp8(a)
+        bmq, bms = ops.quant_mxfp8(b)
+        tm = timeit(lambda: ops.gemm_bt_mxfp8(amq, ams, bmq, bms))
+        tq = timeit(lambda: ops.quant_mxfp8(a))
+        fl = 2 * M * N * K / 1e12
+        print(f"FP8  {M}x{N}x{K}: rowwise {fl / t:7.1f} TF/s   mx-scaled {fl / tm:7.1f}"
+              f" TF/s   (quant_mx {M * K * 2 / tq / 1e9:5.0f} GB/s)")
+
+
 def bench_attn():
     dev = torch.device("cuda:0")
     for (B, H, Hk, S) in [(4, 32, 8, 2048), (1, 32, 8, 8192)]:
@@ -90,6 +108,8 @@ if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "gemm"):
         bench_gemm()
+    if which in ("all", "fp8"):
+        bench_fp8()
     if which in ("all", "attn"):
         bench_attn()
     if which in ("all", "mem"):

@@ -45,8 +45,8 @@ class LlamaModel:
         self.local_q_size = self.local_heads * c.head_dim
         self.local_kv_size = self.local_kv_heads * c.head_dim
         self.local_inter = c.intermediate_size // tpw
-        assert quant in ("bf16", "fp8")
-        if quant == "fp8":
+        assert quant in ("bf16", "fp8", "mxfp8")
+        if quant in ("fp8", "mxfp8"):
             assert c.num_experts == 0, "fp8 MoE not supported yet"
         self.quant = quant
         if self.device.type == "cuda":
@@ -99,15 +99,18 @@ class LlamaModel:
             self.layers.append(layer)
         self.final_norm = torch.ones(c.hidden_size, dtype=dtype, device=self.device)
         self.lm_head = self.embed if c.tie_embeddings else W(c.vocab_size, c.hidden_size)
-        if self.quant == "fp8":
-            # pre-quantize every projection weight row-wise to OCP e4m3 —
-            # halves weight bytes; activations quantize per token on the fly
+        if self.quant in ("fp8", "mxfp8"):
+            # pre-quantize every projection weight to OCP e4m3 — halves weight
+            # bytes.  "fp8": one f32 scale per output row (bf16 MFMA rate);
+            # "mxfp8": e8m0 scale per 32-element K block, run on the
+            # block-scaled 32x32x64 MFMA (2x fp8 rate, HW-fused dequant).
+            qfn = ops.quant_fp8 if self.quant == "fp8" else ops.quant_mxfp8
             for L in self.layers:
                 for name in ("qkv", "o", "gateup", "down"):
-                    q, s = ops.quant_fp8(L[name])
+                    q, s = qfn(L[name])
                     L[name + "_q"], L[name + "_s"] = q, s
                     del L[name]
-            self.lm_head_q, self.lm_head_s = ops.quant_fp8(self.lm_head)
+            self.lm_head_q, self.lm_head_s = qfn(self.lm_head)
         self.cos_sin = ops.rope_tables(c.max_position, c.head_dim, c.rope_theta).to(self.device)
         self.scale = 1.0 / math.sqrt(c.head_dim)
 
@@ -116,6 +119,9 @@ class LlamaModel:
         if self.quant == "fp8":
             xq, xs = ops.quant_fp8(x)
             return ops.gemm_bt_fp8(xq, xs, L[name + "_q"], L[name + "_s"])
+        if self.quant == "mxfp8":
+            xq, xs = ops.quant_mxfp8(x)
+            return ops.gemm_bt_mxfp8(xq, xs, L[name + "_q"], L[name + "_s"])
         return ops.gemm_bt(x, L[name])
 
     # ------------------------------------------------------------------
@@ -303,4 +309,7 @@ class LlamaModel:
         if self.quant == "fp8":
             xq, xs = ops.quant_fp8(flat)
             return ops.gemm_bt_fp8(xq, xs, self.lm_head_q, self.lm_head_s)
+        if self.quant == "mxfp8":
+            xq, xs = ops.quant_mxfp8(flat)
+            return ops.gemm_bt_mxfp8(xq, xs, self.lm_head_q, self.lm_head_s)
         return ops.gemm_bt(flat, self.lm_head)

@@ -147,6 +147,32 @@ def gemm_bt_fp8(a_q, a_s, b_q, b_s):
     return ref.gemm_bt_fp8_ref(a_q, a_s, b_q, b_s)
 
 
+def quant_mxfp8(x: torch.Tensor):
+    """OCP MX quant: (q_uint8[rows,K], e8m0 scales uint8[rows,K//32])."""
+    if _on_gpu(x):
+        q, s = hip_ext().quant_mxfp8(x.contiguous().reshape(-1, x.shape[-1]))
+        return q, s
+    return ref.quant_mxfp8_ref(x)
+
+
+def gemm_bt_mxfp8(a_q, a_s, b_q, b_s):
+    """C = dequant(A)@dequant(B)^T on the MX block-scaled fp8 MFMA (2x fp8 rate).
+
+    Scales are e8m0 bytes per 32-element K block; the 32x32x64 scaled MFMA
+    applies them in hardware, so no epilogue rescale.
+    """
+    if _on_gpu(a_q):
+        M = a_q.shape[0]
+        pad = (-M) % 128
+        if pad:
+            a_q = torch.nn.functional.pad(a_q, (0, 0, 0, pad))
+            a_s = torch.nn.functional.pad(a_s, (0, 0, 0, pad), value=127)
+        c = hip_ext().gemm_bt_mxfp8(a_q.contiguous(), a_s.contiguous(),
+                                    b_q.contiguous(), b_s.contiguous())
+        return c[:M] if pad else c
+    return ref.gemm_bt_mxfp8_ref(a_q, a_s, b_q, b_s)
+
+
 def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
                     ) -> torch.Tensor:
     """Segment-grouped C = A_seg @ W[e]^T for MoE.

@@ -20,6 +20,8 @@ extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*,
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
 extern "C" __global__ void gemm_bt_fp8_kernel(const unsigned char*, const float*, const unsigned char*, const float*, ushort*, int, int, int);
+extern "C" __global__ void quant_mxfp8_kernel(const ushort*, unsigned char*, unsigned char*, int);
+extern "C" __global__ void gemm_bt_mxfp8_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m4(const ushort*, const ushort*, ushort*, int, int, int);
@@ -252,6 +254,38 @@ torch::Tensor gemm_bt_fp8(torch::Tensor Aq, torch::Tensor As,
   return C;
 }
 
+// ---------------- MX block-scaled fp8 path ----------------
+std::vector<torch::Tensor> quant_mxfp8(torch::Tensor x) {
+  check_bf16(x, "x");
+  const int K = x.size(-1);
+  TORCH_CHECK(K % 32 == 0, "mxfp8 needs K%32; got ", K);
+  const long long rows = x.numel() / K;
+  auto q = torch::empty({rows, (long)K}, x.options().dtype(torch::kUInt8));
+  auto s = torch::empty({rows, (long)(K / 32)}, x.options().dtype(torch::kUInt8));
+  quant_mxfp8_kernel<<<dim3((unsigned)rows), dim3(256), 0, cur_stream()>>>(
+      bf16_ptr(x), q.data_ptr<unsigned char>(), s.data_ptr<unsigned char>(), K);
+  HIP_CHECK_KERNEL();
+  return {q, s};
+}
+
+torch::Tensor gemm_bt_mxfp8(torch::Tensor Aq, torch::Tensor As,
+                            torch::Tensor Bq, torch::Tensor Bs) {
+  TORCH_CHECK(Aq.scalar_type() == torch::kUInt8 && Bq.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(As.scalar_type() == torch::kUInt8 && Bs.scalar_type() == torch::kUInt8);
+  const int M = Aq.size(0), K = Aq.size(1), N = Bq.size(0);
+  TORCH_CHECK(Bq.size(1) == K && M % 128 == 0 && N % 128 == 0 && K % 128 == 0,
+              "mxfp8 gemm needs M,N%128, K%128; got ", M, "x", N, "x", K);
+  TORCH_CHECK(As.size(1) == K / 32 && Bs.size(1) == K / 32);
+  auto C = torch::empty({M, N}, Aq.options().dtype(torch::kBFloat16));
+  const int nwg = (M / 128) * (N / 128);
+  gemm_bt_mxfp8_kernel<<<dim3(nwg), dim3(256), 0, cur_stream()>>>(
+      Aq.data_ptr<unsigned char>(), As.data_ptr<unsigned char>(),
+      Bq.data_ptr<unsigned char>(), Bs.data_ptr<unsigned char>(),
+      bf16_mut(C), M, N, K);
+  HIP_CHECK_KERNEL();
+  return C;
+}
+
 // ---------------- Flash attention prefill ----------------
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
                        double scale) {
@@ -345,6 +379,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grouped_gemm_bt", &grouped_gemm_bt, "segment-grouped C = A @ W[e]^T (MoE)");
   m.def("quant_fp8", &quant_fp8, "row-wise bf16 -> e4m3 + scale");
   m.def("gemm_bt_fp8", &gemm_bt_fp8, "fp8 MFMA GEMM with row/col rescale");
+  m.def("quant_mxfp8", &quant_mxfp8, "OCP MX quant: bf16 -> e4m3 + e8m0 per-32 scales");
+  m.def("gemm_bt_mxfp8", &gemm_bt_mxfp8, "MX block-scaled fp8 MFMA GEMM (32x32x64)");
   m.def("attn_fwd", &attn_fwd, "causal flash attention fwd (D=128, GQA)");
   m.def("attn_fwd_v2", &attn_fwd_v2, "swapped-QK^T attention; O^T out [B,H,D,S]");
   m.def("paged_decode_attn", &paged_decode_attn, "paged decode attention");

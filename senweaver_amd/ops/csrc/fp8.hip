@@ -172,3 +172,172 @@ gemm_bt_fp8_kernel(const unsigned char* __restrict__ A, const float* __restrict_
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// MX block-scaled fp8 path — the CDNA4 5-PF-class fp8 rate.
+// v_mfma_scale_f32_32x32x64_f8f6f4 takes per-lane e8m0 scales covering each
+// lane's 32-element K block (HW-fused dequant: D += (A*2^sa)@(B*2^sb)), so
+// the epilogue needs NO rescale.  OCP MX quantization: per 32-element block,
+// scale exponent = floor(log2(amax)) - 8 (e4m3 emax), elements saturate at
+// +-448.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// bf16 [rows, K] -> fp8 q[rows, K] + e8m0 scales [rows, K/32]
+extern "C" __global__ void __launch_bounds__(256)
+quant_mxfp8_kernel(const ushort* __restrict__ x, unsigned char* __restrict__ q,
+                   unsigned char* __restrict__ scales, int K) {
+  const long long row = blockIdx.x;
+  const int nblocks = K / 32;
+  for (int blk = threadIdx.x; blk < nblocks; blk += blockDim.x) {
+    const ushort* src = x + row * K + blk * 32;
+    float v[32];
+    float amax = 0.f;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 w = *reinterpret_cast<const bf16x8*>(src + c * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        v[c * 8 + j] = bf2f(w.v[j]);
+        amax = fmaxf(amax, fabsf(v[c * 8 + j]));
+      }
+    }
+    int e;
+    if (amax > 0.f) {
+      int ex;
+      float m = frexpf(amax, &ex);  // amax = m * 2^ex, m in [0.5, 1)
+      e = (ex - 1) - 8;             // floor(log2(amax)) - emax(e4m3)
+      // non-saturating: if amax/2^e = m*512 > 448, step one finer
+      if (m > 0.875f) e += 1;
+    } else {
+      e = -127;
+    }
+    e = max(-127, min(127, e));
+    scales[row * nblocks + blk] = (unsigned char)(e + 127);
+    const float inv = exp2f((float)(-e));
+    unsigned char* dst = q + row * K + blk * 32;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      unsigned w = 0;
+      w = __builtin_amdgcn_cvt_pk_fp8_f32(v[c * 4 + 0] * inv, v[c * 4 + 1] * inv, w, false);
+      w = __builtin_amdgcn_cvt_pk_fp8_f32(v[c * 4 + 2] * inv, v[c * 4 + 3] * inv, w, true);
+      *reinterpret_cast<unsigned*>(dst + c * 4) = w;
+    }
+  }
+}
+
+// C[M,N] = dequant(A_q, A_s) @ dequant(B_q, B_s)^T, bf16 out.
+// 128x128 tile, BK=128 fp8 bytes, 4 waves (2x2, 64x64 per wave = 2x2 frags
+// of 32x32), glds double buffer + one __syncthreads per K-tile.
+__device__ __forceinline__ i32x8 read_mx_frag(const unsigned char* lds_tile,
+                                              int row, int c0) {
+  // two consecutive 16-B chunks (c0, c0+1), each XOR-swizzled
+  i32x8 out;
+  const int p0 = (c0 ^ (row & 7)) * 16;
+  const int p1 = ((c0 + 1) ^ (row & 7)) * 16;
+  *reinterpret_cast<int4*>(&out) = *reinterpret_cast<const int4*>(lds_tile + row * FBK + p0);
+  *(reinterpret_cast<int4*>(&out) + 1) = *reinterpret_cast<const int4*>(lds_tile + row * FBK + p1);
+  return out;
+}
+
+extern "C" __global__ void __launch_bounds__(256, 2)
+gemm_bt_mxfp8_kernel(const unsigned char* __restrict__ A, const unsigned char* __restrict__ As,
+                     const unsigned char* __restrict__ B, const unsigned char* __restrict__ Bs,
+                     ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / 128) * (N / 128);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / 128;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1;
+  const int wn = wid & 1;
+  const int l31 = lane & 31;
+  const int lhi = lane >> 5;
+
+  __shared__ __attribute__((aligned(16))) unsigned char lds[2][2][128 * FBK];
+
+  const unsigned char* Atile = A + (long long)tile_m * 128 * K;
+  const unsigned char* Btile = B + (long long)tile_n * 128 * K;
+  const int sld = K / 32;  // scale row stride
+  const unsigned char* Astile = As + (long long)tile_m * 128 * sld;
+  const unsigned char* Bstile = Bs + (long long)tile_n * 128 * sld;
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[i][j][r] = 0.f;
+
+  const int m_base = wm * 64;
+  const int n_base = wn * 64;
+
+  const int ntiles = K / FBK;
+  stage_fp8_tile(Atile, K, lds[0][0], tid);
+  stage_fp8_tile(Btile, K, lds[0][1], tid);
+  __syncthreads();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      stage_fp8_tile(Atile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][0], tid);
+      stage_fp8_tile(Btile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][1], tid);
+    }
+    const unsigned char* Al = lds[buf][0];
+    const unsigned char* Bl = lds[buf][1];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {  // FBK=128 in 2 MFMA steps of K=64
+      // per-lane K block index for the scale: (t*128 + kk*64 + lhi*32)/32
+      const int sblk = t * 4 + kk * 2 + lhi;
+      i32x8 af[2], bf[2];
+      int sa[2], sb[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        const int row = m_base + mi * 32 + l31;
+        af[mi] = read_mx_frag(Al, row, kk * 4 + lhi * 2);
+        sa[mi] = Astile[(long long)row * sld + sblk];
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int row = n_base + ni * 32 + l31;
+        bf[ni] = read_mx_frag(Bl, row, kk * 4 + lhi * 2);
+        sb[ni] = Bstile[(long long)row * sld + sblk];
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0, sa[mi], 0, sb[ni]);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // C/D 32x32 map: col = lane&31, row = (r&3)+8*(r>>2)+4*(lane>>5)
+  const long long c_col0 = (long long)tile_n * 128 + n_base + l31;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long long row = (long long)tile_m * 128 + m_base + mi * 32
+                            + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        crow[c_col0 + ni * 32] = f2bf(acc[mi][ni][r]);
+    }
+}

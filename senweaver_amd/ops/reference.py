@@ -127,3 +127,33 @@ def gemm_bt_fp8_ref(a_q, a_s, b_q, b_s):
     bf = b_q.view(torch.float8_e4m3fn).float()
     c = (af @ bf.t()) * a_s.unsqueeze(1) * b_s.unsqueeze(0)
     return c.to(torch.bfloat16)
+
+
+def quant_mxfp8_ref(x: torch.Tensor):
+    """OCP MX quantization: per-32-element block, e8m0 scale = 2^(floor(log2(amax))-8).
+
+    Returns (q_uint8[rows,K], scales_uint8[rows,K//32]) with scale byte =
+    exponent + 127 (e8m0 biased).
+    """
+    x2 = x.reshape(-1, x.shape[-1]).float()
+    rows, K = x2.shape
+    blk = x2.view(rows, K // 32, 32)
+    amax = blk.abs().amax(dim=-1)
+    e = torch.where(amax > 0, torch.floor(torch.log2(amax)) - 8,
+                    torch.full_like(amax, -127.0))
+    # non-saturating variant: OCP's floor-8 scale leaves amax/X in [256,512),
+    # clipping up to 12.5% off the block max; step one finer when it would
+    e = torch.where(amax * torch.exp2(-e) > FP8_MAX, e + 1, e)
+    e = e.clamp(-127, 127)
+    scales = (e + 127).to(torch.uint8)
+    q = (blk * torch.exp2(-e).unsqueeze(-1)).clamp(-FP8_MAX, FP8_MAX)
+    q = q.to(torch.float8_e4m3fn).view(torch.uint8).reshape(rows, K)
+    return q, scales
+
+
+def gemm_bt_mxfp8_ref(a_q, a_s, b_q, b_s):
+    def deq(q, s):
+        rows, K = q.shape
+        f = q.view(torch.float8_e4m3fn).float().view(rows, K // 32, 32)
+        return (f * torch.exp2(s.float() - 127).unsqueeze(-1)).reshape(rows, K)
+    return (deq(a_q, a_s) @ deq(b_q, b_s).t()).to(torch.bfloat16)
