@@ -22,6 +22,7 @@ extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char
 extern "C" __global__ void gemm_bt_fp8_kernel(const unsigned char*, const float*, const unsigned char*, const float*, ushort*, int, int, int);
 extern "C" __global__ void quant_mxfp8_kernel(const ushort*, unsigned char*, unsigned char*, int);
 extern "C" __global__ void gemm_bt_mxfp8_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_mxfp8_256_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m4(const ushort*, const ushort*, ushort*, int, int, int);
@@ -277,11 +278,19 @@ torch::Tensor gemm_bt_mxfp8(torch::Tensor Aq, torch::Tensor As,
               "mxfp8 gemm needs M,N%128, K%128; got ", M, "x", N, "x", K);
   TORCH_CHECK(As.size(1) == K / 32 && Bs.size(1) == K / 32);
   auto C = torch::empty({M, N}, Aq.options().dtype(torch::kBFloat16));
-  const int nwg = (M / 128) * (N / 128);
- hipLaunchKernelGGL(( gemm_bt_mxfp8_kernel), dim3(dim3(nwg)), dim3(dim3(256)), 0, cur_stream(), 
-      Aq.data_ptr<unsigned char>(), As.data_ptr<unsigned char>(),
-      Bq.data_ptr<unsigned char>(), Bs.data_ptr<unsigned char>(),
-      bf16_mut(C), M, N, K);
+  const int nwg256 = (M % 256 == 0 && N % 256 == 0) ? (M / 256) * (N / 256) : 0;
+  if (nwg256 >= 160) {
+   hipLaunchKernelGGL(( gemm_bt_mxfp8_256_kernel), dim3(dim3(nwg256)), dim3(dim3(512)), 0, cur_stream(), 
+        Aq.data_ptr<unsigned char>(), As.data_ptr<unsigned char>(),
+        Bq.data_ptr<unsigned char>(), Bs.data_ptr<unsigned char>(),
+        bf16_mut(C), M, N, K);
+  } else {
+    const int nwg = (M / 128) * (N / 128);
+   hipLaunchKernelGGL(( gemm_bt_mxfp8_kernel), dim3(dim3(nwg)), dim3(dim3(256)), 0, cur_stream(), 
+        Aq.data_ptr<unsigned char>(), As.data_ptr<unsigned char>(),
+        Bq.data_ptr<unsigned char>(), Bs.data_ptr<unsigned char>(),
+        bf16_mut(C), M, N, K);
+  }
   HIP_CHECK_KERNEL();
   return C;
 }

@@ -230,12 +230,20 @@ quant_mxfp8_kernel(const ushort* __restrict__ x, unsigned char* __restrict__ q,
 // C[M,N] = dequant(A_q, A_s) @ dequant(B_q, B_s)^T, bf16 out.
 // 128x128 tile, BK=128 fp8 bytes, 4 waves (2x2, 64x64 per wave = 2x2 frags
 // of 32x32), glds double buffer + one __syncthreads per K-tile.
+//
+// Fragment layout (probed empirically, benchmarks/probe_mx.py): per 64-k
+// instruction, lane half lhi holds k in [lhi*16, lhi*16+16) u
+// [32+lhi*16, 32+lhi*16+16) — 16 bytes from EACH of the two 32-element
+// scale blocks — while the per-lane scale register of half lhi supplies
+// the e8m0 scale for block lhi.  So the two 16-B chunks a lane reads are
+// (kk*4 + lhi) and (kk*4 + lhi + 2), and the scale byte is block
+// (t*4 + kk*2 + lhi) of that fragment's row.
 __device__ __forceinline__ i32x8 read_mx_frag(const unsigned char* lds_tile,
                                               int row, int c0) {
-  // two consecutive 16-B chunks (c0, c0+1), each XOR-swizzled
+  // two 16-B chunks (c0, c0+2) — one from each scale block — XOR-swizzled
   i32x8 out;
   const int p0 = (c0 ^ (row & 7)) * 16;
-  const int p1 = ((c0 + 1) ^ (row & 7)) * 16;
+  const int p1 = ((c0 + 2) ^ (row & 7)) * 16;
   *reinterpret_cast<int4*>(&out) = *reinterpret_cast<const int4*>(lds_tile + row * FBK + p0);
   *(reinterpret_cast<int4*>(&out) + 1) = *reinterpret_cast<const int4*>(lds_tile + row * FBK + p1);
   return out;
@@ -286,34 +294,54 @@ gemm_bt_mxfp8_kernel(const unsigned char* __restrict__ A, const unsigned char* _
   const int ntiles = K / FBK;
   stage_fp8_tile(Atile, K, lds[0][0], tid);
   stage_fp8_tile(Btile, K, lds[0][1], tid);
+
+  // scale rows this lane's fragments use (dword = the 4 e8m0 bytes of one
+  // 128-deep K-tile); prefetched one tile ahead so the byte extracts in the
+  // MFMA loop never wait on global memory
+  const unsigned char* As_row[2] = {
+      Astile + (long long)(m_base + 0 * 32 + l31) * sld,
+      Astile + (long long)(m_base + 1 * 32 + l31) * sld};
+  const unsigned char* Bs_row[2] = {
+      Bstile + (long long)(n_base + 0 * 32 + l31) * sld,
+      Bstile + (long long)(n_base + 1 * 32 + l31) * sld};
+  unsigned sa_dw[2], sb_dw[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    sa_dw[i] = *reinterpret_cast<const unsigned*>(As_row[i]);
+    sb_dw[i] = *reinterpret_cast<const unsigned*>(Bs_row[i]);
+  }
   __syncthreads();
 
   int buf = 0;
   for (int t = 0; t < ntiles; ++t) {
+    unsigned na[2], nb[2];
     if (t + 1 < ntiles) {
       stage_fp8_tile(Atile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][0], tid);
       stage_fp8_tile(Btile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][1], tid);
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        na[i] = *reinterpret_cast<const unsigned*>(As_row[i] + (t + 1) * 4);
+        nb[i] = *reinterpret_cast<const unsigned*>(Bs_row[i] + (t + 1) * 4);
+      }
     }
     const unsigned char* Al = lds[buf][0];
     const unsigned char* Bl = lds[buf][1];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {  // FBK=128 in 2 MFMA steps of K=64
-      // per-lane K block index for the scale: (t*128 + kk*64 + lhi*32)/32
-      const int sblk = t * 4 + kk * 2 + lhi;
+      // lane half lhi supplies the scale of block (kk*2 + lhi) of this tile
+      const int sh = 8 * (kk * 2 + lhi);
       i32x8 af[2], bf[2];
       int sa[2], sb[2];
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi) {
-        const int row = m_base + mi * 32 + l31;
-        af[mi] = read_mx_frag(Al, row, kk * 4 + lhi * 2);
-        sa[mi] = Astile[(long long)row * sld + sblk];
+        af[mi] = read_mx_frag(Al, m_base + mi * 32 + l31, kk * 4 + lhi);
+        sa[mi] = (sa_dw[mi] >> sh) & 0xff;
       }
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
-        const int row = n_base + ni * 32 + l31;
-        bf[ni] = read_mx_frag(Bl, row, kk * 4 + lhi * 2);
-        sb[ni] = Bstile[(long long)row * sld + sblk];
+        bf[ni] = read_mx_frag(Bl, n_base + ni * 32 + l31, kk * 4 + lhi);
+        sb[ni] = (sb_dw[ni] >> sh) & 0xff;
       }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
@@ -325,6 +353,8 @@ gemm_bt_mxfp8_kernel(const unsigned char* __restrict__ A, const unsigned char* _
     __builtin_amdgcn_s_setprio(0);
     __syncthreads();
     buf ^= 1;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) { sa_dw[i] = na[i]; sb_dw[i] = nb[i]; }
   }
 
   // C/D 32x32 map: col = lane&31, row = (r&3)+8*(r>>2)+4*(lane>>5)
@@ -334,6 +364,144 @@ gemm_bt_mxfp8_kernel(const unsigned char* __restrict__ A, const unsigned char* _
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const long long row = (long long)tile_m * 128 + m_base + mi * 32
+                            + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        crow[c_col0 + ni * 32] = f2bf(acc[mi][ni][r]);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// 256x256-tile MX fp8 GEMM: 8 waves (2 m x 4 n), wave tile 128x64 as 4x2
+// frags of 32x32.  Raises the MFMA:ds_read ratio from 8:4 to 8:6-per-
+// double-the-work (6 frag reads feed 8 scaled MFMAs vs 4 feeding 4) and
+// halves barriers per flop; 1 block/CU (acc alone is 128 VGPRs).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void stage_fp8_tile_256(
+    const unsigned char* __restrict__ src, long long ldK, unsigned char* lds_tile,
+    int tid) {
+  // 256 rows x 8 chunks(16B) = 2048 chunks / 512 threads = 4 glds each
+  const int wave_chunk = tid & ~63;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = i * 512 + tid;
+    const int row = s >> 3;
+    const int c = (s & 7) ^ (row & 7);
+    const unsigned char* g = src + (long long)row * ldK + c * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(lds_tile + (long long)(i * 512 + wave_chunk) * 16),
+        16, 0, 0);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_mxfp8_256_kernel(const unsigned char* __restrict__ A, const unsigned char* __restrict__ As,
+                         const unsigned char* __restrict__ B, const unsigned char* __restrict__ Bs,
+                         ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / 256) * (N / 256);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / 256;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;      // 0..1 -> 128-row half
+  const int wn = wid & 3;       // 0..3 -> 64-col quarter
+  const int l31 = lane & 31;
+  const int lhi = lane >> 5;
+
+  __shared__ __attribute__((aligned(16))) unsigned char lds[2][2][256 * FBK];
+
+  const unsigned char* Atile = A + (long long)tile_m * 256 * K;
+  const unsigned char* Btile = B + (long long)tile_n * 256 * K;
+  const int sld = K / 32;
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[i][j][r] = 0.f;
+
+  const int m_base = wm * 128;
+  const int n_base = wn * 64;
+
+  const unsigned char* As_row[4];
+  const unsigned char* Bs_row[2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    As_row[i] = As + ((long long)tile_m * 256 + m_base + i * 32 + l31) * sld;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+    Bs_row[i] = Bs + ((long long)tile_n * 256 + n_base + i * 32 + l31) * sld;
+  unsigned sa_dw[4], sb_dw[2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) sa_dw[i] = *reinterpret_cast<const unsigned*>(As_row[i]);
+#pragma unroll
+  for (int i = 0; i < 2; ++i) sb_dw[i] = *reinterpret_cast<const unsigned*>(Bs_row[i]);
+
+  const int ntiles = K / FBK;
+  stage_fp8_tile_256(Atile, K, lds[0][0], tid);
+  stage_fp8_tile_256(Btile, K, lds[0][1], tid);
+  __syncthreads();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    unsigned na[4], nb[2];
+    if (t + 1 < ntiles) {
+      stage_fp8_tile_256(Atile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][0], tid);
+      stage_fp8_tile_256(Btile + (long long)(t + 1) * FBK, K, lds[buf ^ 1][1], tid);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) na[i] = *reinterpret_cast<const unsigned*>(As_row[i] + (t + 1) * 4);
+#pragma unroll
+      for (int i = 0; i < 2; ++i) nb[i] = *reinterpret_cast<const unsigned*>(Bs_row[i] + (t + 1) * 4);
+    }
+    const unsigned char* Al = lds[buf][0];
+    const unsigned char* Bl = lds[buf][1];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int sh = 8 * (kk * 2 + lhi);
+      i32x8 af[4], bf[2];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = read_mx_frag(Al, m_base + mi * 32 + l31, kk * 4 + lhi);
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        bf[ni] = read_mx_frag(Bl, n_base + ni * 32 + l31, kk * 4 + lhi);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0,
+              (int)((sa_dw[mi] >> sh) & 0xff), 0, (int)((sb_dw[ni] >> sh) & 0xff));
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    buf ^= 1;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) sa_dw[i] = na[i];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) sb_dw[i] = nb[i];
+  }
+
+  const long long c_col0 = (long long)tile_n * 256 + n_base + l31;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long long row = (long long)tile_m * 256 + m_base + mi * 32
                             + (r & 3) + 8 * (r >> 2) + 4 * lhi;
       ushort* crow = C + row * N;
 #pragma unroll

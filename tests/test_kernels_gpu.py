@@ -265,6 +265,25 @@ def test_gemm_mxfp8_gpu(dev):
     assert rel < 0.05
 
 
+def test_gemm_mxfp8_256_tile_gpu(dev):
+    # big-tile dispatch kicks in at >=160 256x256 tiles; check it against a
+    # GPU-side exact dequantized product
+    torch.manual_seed(8)
+    M, N, K = 4096, 4096, 1024  # 16x16=256 tiles
+    a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    b = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+    aq, asc = ops.quant_mxfp8(a)
+    bq, bsc = ops.quant_mxfp8(b)
+    c = ops.gemm_bt_mxfp8(aq, asc, bq, bsc)
+
+    def deq(q, s):
+        f = q.view(torch.float8_e4m3fn).float().view(q.shape[0], -1, 32)
+        return (f * torch.exp2(s.float() - 127).unsqueeze(-1)).reshape(q.shape[0], -1)
+
+    exact = deq(aq, asc) @ deq(bq, bsc).t()
+    torch.testing.assert_close(c.float(), exact, atol=1.0, rtol=5e-2)
+
+
 def test_mxfp8_model_forward_gpu(dev):
     from senweaver_amd.models import tiny_debug
     from senweaver_amd.models.llama import LlamaModel
