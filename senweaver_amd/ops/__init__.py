@@ -9,6 +9,7 @@ reference implementations (tests / CPU-only engine mode).
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -107,16 +108,40 @@ def swiglu(gateup: torch.Tensor) -> torch.Tensor:
     return ref.swiglu_ref(gateup)
 
 
+_GEMM_IMPL = os.environ.get("SENWEAVER_GEMM", "auto")  # auto | hip
+
+
 def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    """C[M,N] = A[M,K] @ B[N,K]^T.  Pads M to the kernel tile on GPU."""
+    """C[M,N] = A[M,K] @ B[N,K]^T.
+
+    Dispatch (measured on MI355X, profiles/r01_gemm_dispatch.txt): decode-
+    shaped M<=16 goes to our wave-per-row GEMV kernel (4.4 TB/s, ~2x blas on
+    skinny); plain prefill GEMMs go to hipBLASLt (1.28-1.58 PF/s at bench
+    shapes vs our tiled kernel's 0.88-1.15 — the guide's rule: hand-write the
+    fused/special ops, use the vendor library for plain GEMMs it wins).
+    ``SENWEAVER_GEMM=hip`` forces the in-repo tiled kernel everywhere.
+    """
+    if not _on_gpu(a):
+        return ref.gemm_bt_ref(a, b)
+    M = a.shape[0]
+    if M <= 16 and b.shape[1] % 512 == 0 and b.shape[0] % 4 == 0:
+        return hip_ext().gemm_bt(a.contiguous(), b.contiguous())
+    if _GEMM_IMPL != "hip":
+        return a @ b.t()
+    return gemm_bt_tiled(a, b)
+
+
+def gemm_bt_tiled(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """The in-repo MFMA tiled GEMM (128^2 4-wave / 256^2 8-wave LDS tiles).
+
+    Pads M to the tile.  Kept first-class for fused-epilogue growth and A/B
+    benchmarking against hipBLASLt (benchmarks/kernel_bench.py gemm).
+    """
     if not _on_gpu(a):
         return ref.gemm_bt_ref(a, b)
     ext = hip_ext()
-    M = a.shape[0]
-    if M <= 16 and b.shape[1] % 512 == 0 and b.shape[0] % 4 == 0:
-        return ext.gemm_bt(a.contiguous(), b.contiguous())
+    M, N = a.shape[0], b.shape[0]
     # pad M to the 256-tile when N allows it and the grid fills the chip
-    N = b.shape[0]
     tile = 256 if (N % 256 == 0 and ((M + 255) // 256) * (N // 256) >= 160) else 128
     pad = (-M) % tile
     if pad:

@@ -76,7 +76,7 @@ def test_gemm_layout_sanity(dev):
          torch.arange(K, device=dev).unsqueeze(0) * 0.001).bfloat16()
     b = (torch.arange(N, device=dev).unsqueeze(1) * 0.02 -
          torch.arange(K, device=dev).unsqueeze(0) * 0.003).bfloat16()
-    c = ops.gemm_bt(a, b)
+    c = ops.gemm_bt_tiled(a, b)
     c_ref = ref.gemm_bt_ref(a, b)
     torch.testing.assert_close(c.float(), c_ref.float(), atol=5e-2, rtol=2e-2)
 
@@ -86,7 +86,7 @@ def test_gemm_layout_sanity(dev):
 def test_gemm_random(dev, M, N, K):
     a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
     b = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
-    c = ops.gemm_bt(a, b)
+    c = ops.gemm_bt_tiled(a, b)
     c_ref = ref.gemm_bt_ref(a, b)
     torch.testing.assert_close(c.float(), c_ref.float(), atol=0.5, rtol=3e-2)
 
@@ -95,7 +95,7 @@ def test_gemm_pad_m(dev):
     # M not a multiple of 128 goes through the host-side pad
     a = torch.randn(300, 512, dtype=torch.bfloat16, device=dev)
     b = torch.randn(256, 512, dtype=torch.bfloat16, device=dev)
-    c = ops.gemm_bt(a, b)
+    c = ops.gemm_bt_tiled(a, b)
     assert c.shape == (300, 256)
     torch.testing.assert_close(c.float(), ref.gemm_bt_ref(a, b).float(), atol=0.5, rtol=3e-2)
 
