@@ -152,8 +152,8 @@ class LlamaModel:
             qb, kb = ops.rope_scatter_qkv(qkv, self.cos_sin, positions,
                                           self.local_heads, self.local_kv_heads,
                                           c.head_dim, B, S)
-            vview = qkv[:, qs + kvs:].reshape(B, S, self.local_kv_heads, c.head_dim)
             if cache is not None:
+                vview = qkv[:, qs + kvs:].reshape(B, S, self.local_kv_heads, c.head_dim)
                 kflat = kb.permute(0, 2, 1, 3).reshape(T, self.local_kv_heads, c.head_dim)
                 vflat = vview.reshape(T, self.local_kv_heads, c.head_dim)
                 for b in range(B):
@@ -161,10 +161,20 @@ class LlamaModel:
                     cache.append(li, seqs[b], kflat[b * S: b * S + n],
                                  vflat[b * S: b * S + n].contiguous(),
                                  advance_len=(li == c.num_layers - 1))
-            # V^T computed directly: the attention kernel consumes [B,Hk,D,S]
-            vt = vview.permute(0, 2, 3, 1).contiguous()
+            # V^T [B,Hk,D,S] straight from the qkv slice (LDS-tiled transpose)
+            vt = ops.vt_from_qkv(qkv, self.local_heads, self.local_kv_heads,
+                                 c.head_dim, B, S)
             if S % 128 == 0:
                 ot = ops.attn_fwd_t(qb, kb, vt, self.scale)  # O^T [B,Hq_local,D,S]
+                if self.quant == "bf16":
+                    # o_proj directly off the O^T view: hipBLASLt takes the
+                    # transposed A natively — no [T, qsize] copy materialized
+                    o = torch.matmul(ot.reshape(B, self.local_q_size, S).transpose(1, 2),
+                                     L["o"].t()).reshape(T, -1)
+                    self.tp.all_reduce_(o)
+                    h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
+                    hidden = self._ffn(h, L)
+                    continue
                 attn = ot.permute(0, 3, 1, 2).reshape(T, self.local_q_size).contiguous()
             else:  # v1 kernel path for 64-granular sequences
                 ab = ops.attn_fwd(qb, kb, None, self.scale, vt=vt)

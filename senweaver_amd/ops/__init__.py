@@ -102,6 +102,19 @@ def rope_scatter_qkv(qkv: torch.Tensor, cos_sin, positions, Hq: int, Hk: int,
             kr.reshape(B, S, Hk, D).permute(0, 2, 1, 3).contiguous())
 
 
+def vt_from_qkv(qkv: torch.Tensor, Hq: int, Hk: int, D: int, B: int, S: int
+                ) -> torch.Tensor:
+    """V^T [B,Hk,D,S] straight from the fused qkv projection output.
+
+    LDS-tiled 32x32 transpose (coalesced both ways) instead of a strided
+    permute().contiguous() copy."""
+    if _on_gpu(qkv):
+        return hip_ext().vt_from_qkv(qkv.contiguous(), Hq, Hk, D, B, S)
+    T = qkv.shape[0]
+    v = qkv[:, (Hq + Hk) * D:(Hq + 2 * Hk) * D].reshape(B, S, Hk, D)
+    return v.permute(0, 2, 3, 1).contiguous()
+
+
 def swiglu(gateup: torch.Tensor) -> torch.Tensor:
     if _on_gpu(gateup):
         return hip_ext().swiglu(gateup.contiguous())

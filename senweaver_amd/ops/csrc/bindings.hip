@@ -11,6 +11,7 @@
 extern "C" __global__ void rmsnorm_fwd_kernel(const ushort*, const ushort*, ushort*, ushort*, int, float);
 extern "C" __global__ void rope_fwd_kernel(ushort*, ushort*, const float*, const int*, int, int, int);
 extern "C" __global__ void rope_scatter_kernel(const ushort*, long long, ushort*, ushort*, const float*, const int*, int, int, int, int);
+extern "C" __global__ void transpose_v_kernel(const ushort*, long long, long long, ushort*, int, int, int, int);
 extern "C" __global__ void swiglu_fwd_kernel(const ushort*, ushort*, long long, int);
 extern "C" __global__ void add_bf16_kernel(const ushort*, const ushort*, ushort*, long long);
 extern "C" __global__ void argmax_rows_kernel(const ushort*, int*, int);
@@ -117,6 +118,21 @@ std::vector<torch::Tensor> rope_scatter_qkv(torch::Tensor qkv, torch::Tensor cos
       (int)D, (int)S);
   HIP_CHECK_KERNEL();
   return {qo, ko};
+}
+
+torch::Tensor vt_from_qkv(torch::Tensor qkv, int64_t Hq, int64_t Hk, int64_t D,
+                          int64_t B, int64_t S) {
+  check_bf16(qkv, "qkv");
+  const long long ld = qkv.size(1);
+  TORCH_CHECK(qkv.size(0) == B * S && ld >= (Hq + 2 * Hk) * D);
+  auto vt = torch::empty({B, Hk, D, S}, qkv.options());
+  const long long v_off = (Hq + Hk) * D;
+  transpose_v_kernel<<<dim3((unsigned)((S + 31) / 32), (unsigned)((D + 31) / 32),
+                           (unsigned)(B * Hk)),
+                       dim3(256), 0, cur_stream()>>>(
+      bf16_ptr(qkv), ld, v_off, bf16_mut(vt), (int)B, (int)S, (int)Hk, (int)D);
+  HIP_CHECK_KERNEL();
+  return vt;
 }
 
 // ---------------- SwiGLU ----------------
@@ -382,6 +398,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual += x; y = rmsnorm(residual)");
   m.def("rope_inplace", &rope_inplace, "RoPE in place over q,k");
   m.def("rope_scatter_qkv", &rope_scatter_qkv, "fused RoPE + [B,H,S,D] scatter from qkv");
+  m.def("vt_from_qkv", &vt_from_qkv, "LDS-tiled V^T [B,Hk,D,S] from the qkv slice");
   m.def("swiglu", &swiglu, "silu(gate)*up from fused gateup");
   m.def("add_bf16", &add_bf16, "a + b (bf16)");
   m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");

@@ -183,3 +183,38 @@ add_bf16_kernel(const ushort* __restrict__ a, const ushort* __restrict__ b,
     *reinterpret_cast<bf16x8*>(y + idx * 8) = out;
   }
 }
+
+// ---------------------------------------------------------------------------
+// V^T extraction: qkv [T, ld] (v slice at v_off) -> vt [B, Hk, D, S].
+// LDS-tiled 32x32 transpose: reads are contiguous along d, writes contiguous
+// along s — replaces a strided torch permute().contiguous() that ran ~4x off
+// bandwidth.  Guards handle S not a multiple of 32.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+transpose_v_kernel(const ushort* __restrict__ qkv, long long ld, long long v_off,
+                   ushort* __restrict__ vt, int B, int S, int Hk, int D) {
+  __shared__ ushort tile[32][33];  // +1 pad kills bank conflicts
+  const int bh = blockIdx.z;       // b * Hk + hk
+  const int b = bh / Hk;
+  const int hk = bh % Hk;
+  const int s0 = blockIdx.x * 32;
+  const int d0 = blockIdx.y * 32;
+  const int tx = threadIdx.x & 31;   // along d on read, along s on write
+  const int ty0 = threadIdx.x >> 5;  // 8 rows per pass
+
+  const ushort* src = qkv + (long long)b * S * ld + v_off + (long long)hk * D;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = s0 + ty0 + i * 8;
+    if (s < S && d0 + tx < D)
+      tile[ty0 + i * 8][tx] = src[(long long)s * ld + d0 + tx];
+  }
+  __syncthreads();
+  ushort* dst = vt + (((long long)b * Hk + hk) * D) * S;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int d = d0 + ty0 + i * 8;
+    if (d < D && s0 + tx < S)
+      dst[(long long)d * S + s0 + tx] = tile[tx][ty0 + i * 8];
+  }
+}

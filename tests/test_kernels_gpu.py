@@ -354,3 +354,13 @@ def test_attn_fwd_v2_spiked(dev):
     o_ref = ref.attn_fwd_ref(q, k, v, scale, causal=True)
     torch.testing.assert_close(ot.transpose(-1, -2).float(), o_ref.float(),
                                atol=8e-2, rtol=8e-2)
+
+
+def test_vt_from_qkv_gpu(dev):
+    B, S, Hq, Hk, D = 2, 192, 4, 2, 128  # S not a multiple of 32-tile grid edge
+    ld = (Hq + 2 * Hk) * D
+    qkv = torch.randn(B * S, ld, dtype=torch.bfloat16, device=dev)
+    vt = ops.vt_from_qkv(qkv, Hq, Hk, D, B, S)
+    v = qkv[:, (Hq + Hk) * D:].reshape(B, S, Hk, D)
+    ref_vt = v.permute(0, 2, 3, 1).contiguous()
+    assert torch.equal(vt, ref_vt)
