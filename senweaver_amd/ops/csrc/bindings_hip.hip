@@ -112,7 +112,9 @@ std::vector<torch::Tensor> rope_scatter_qkv(torch::Tensor qkv, torch::Tensor cos
   TORCH_CHECK(T == B * S && cos_sin.size(1) == D && ld >= (Hq + 2 * Hk) * D);
   auto qo = torch::empty({B, Hq, S, D}, qkv.options());
   auto ko = torch::empty({B, Hk, S, D}, qkv.options());
- hipLaunchKernelGGL(( rope_scatter_kernel), dim3(dim3((unsigned)T, (unsigned)(Hq + Hk))), dim3(dim3(64)), 0, cur_stream(), 
+  const int heads_per_blk = (int)(256 / (D / 2));
+ hipLaunchKernelGGL(( rope_scatter_kernel), dim3(dim3((unsigned)T, (unsigned)((Hq + Hk + heads_per_blk - 1) / heads_per_blk))),
+                        dim3(dim3(256)), 0, cur_stream(), 
       bf16_ptr(qkv), ld, bf16_mut(qo), bf16_mut(ko),
       cos_sin.data_ptr<float>(), positions.data_ptr<int>(), (int)Hq, (int)Hk,
       (int)D, (int)S);
@@ -145,9 +147,8 @@ torch::Tensor swiglu(torch::Tensor gateup) {
   auto sizes = gateup.sizes().vec();
   sizes.back() = inter;
   auto y = torch::empty(sizes, gateup.options());
-  const long long total = rows * (inter / 8);
-  const int blocks = (int)std::min<long long>((total + 255) / 256, 2048);
- hipLaunchKernelGGL(( swiglu_fwd_kernel), dim3(dim3(blocks)), dim3(dim3(256)), 0, cur_stream(), 
+  const int nvec_blocks = (inter / 8 + 255) / 256;
+ hipLaunchKernelGGL(( swiglu_fwd_kernel), dim3(dim3(nvec_blocks, (unsigned)rows)), dim3(dim3(256)), 0, cur_stream(), 
       bf16_ptr(gateup), bf16_mut(y), rows, inter);
   HIP_CHECK_KERNEL();
   return y;

@@ -106,21 +106,26 @@ rope_fwd_kernel(ushort* __restrict__ q, ushort* __restrict__ k,
 // qkv GEMM's natural layout), applies rotate-half RoPE, and writes the
 // attention kernel's [B, H, S, D] layout directly — replacing a rope pass
 // plus two transpose copies (one full q/k read+write each) with one pass.
-// Grid: (T, Hq+Hk), 64 lanes (D=128: lane d pairs with d+64).
+// Grid: (T, ceil((Hq+Hk)/heads_per_blk)), 256 threads covering several heads.
 // ---------------------------------------------------------------------------
-extern "C" __global__ void __launch_bounds__(64)
+extern "C" __global__ void __launch_bounds__(256)
 rope_scatter_kernel(const ushort* __restrict__ qkv, long long ld,
                     ushort* __restrict__ q_out, ushort* __restrict__ k_out,
                     const float* __restrict__ cos_sin,
                     const int* __restrict__ positions, int Hq, int Hk, int D,
                     int S) {
   // reads q/k heads straight out of the fused qkv GEMM output (row stride
-  // ld = q_size + 2*kv_size) — no separate contiguous slice copies
+  // ld = q_size + 2*kv_size) — no separate contiguous slice copies.
+  // 256-thread blocks cover (256 / (D/2)) heads each: adjacent heads read
+  // adjacent qkv segments, so loads coalesce across the whole block (the
+  // one-head-per-64-thread-block version ran ~1.9 TB/s in situ).
   const int t = blockIdx.x;
-  const int h = blockIdx.y;
+  const int half = D / 2;
+  const int heads_per_blk = 256 / half;
+  const int h = blockIdx.y * heads_per_blk + threadIdx.x / half;
+  if (h >= Hq + Hk) return;
   const int b = t / S;
   const int s = t % S;
-  const int half = D / 2;
   const ushort* src;
   ushort* dst;
   if (h < Hq) {
@@ -131,14 +136,13 @@ rope_scatter_kernel(const ushort* __restrict__ qkv, long long ld,
     dst = k_out + (((long long)b * Hk + (h - Hq)) * S + s) * D;
   }
   const float* cs = cos_sin + (long long)positions[t] * D;
-  for (int d = threadIdx.x; d < half; d += blockDim.x) {
-    const float c = cs[d];
-    const float sn = cs[half + d];
-    const float x1 = bf2f(src[d]);
-    const float x2 = bf2f(src[d + half]);
-    dst[d] = f2bf(x1 * c - x2 * sn);
-    dst[d + half] = f2bf(x2 * c + x1 * sn);
-  }
+  const int d = threadIdx.x % half;
+  const float c = cs[d];
+  const float sn = cs[half + d];
+  const float x1 = bf2f(src[d]);
+  const float x2 = bf2f(src[d + half]);
+  dst[d] = f2bf(x1 * c - x2 * sn);
+  dst[d + half] = f2bf(x2 * c + x1 * sn);
 }
 
 // ---------------------------------------------------------------------------
@@ -148,24 +152,23 @@ rope_scatter_kernel(const ushort* __restrict__ qkv, long long ld,
 extern "C" __global__ void __launch_bounds__(256)
 swiglu_fwd_kernel(const ushort* __restrict__ gateup, ushort* __restrict__ y,
                   long long rows, int inter) {
+  // 2D grid (vec-chunks, rows): no div/mod per element — the grid-stride
+  // version's 64-bit divide was throttling this to ~1.4 TB/s in situ
   const int nvec = inter / 8;
-  const long long total = rows * nvec;
-  for (long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-       idx < total; idx += (long long)gridDim.x * blockDim.x) {
-    const long long r = idx / nvec;
-    const int i = (int)(idx % nvec);
-    bf16x8 g = *reinterpret_cast<const bf16x8*>(gateup + r * 2 * inter + i * 8);
-    bf16x8 u = *reinterpret_cast<const bf16x8*>(gateup + r * 2 * inter + inter + i * 8);
-    bf16x8 out;
+  const long long r = blockIdx.y;
+  const int i = blockIdx.x * 256 + threadIdx.x;
+  if (i >= nvec) return;
+  bf16x8 g = *reinterpret_cast<const bf16x8*>(gateup + r * 2 * inter + i * 8);
+  bf16x8 u = *reinterpret_cast<const bf16x8*>(gateup + r * 2 * inter + inter + i * 8);
+  bf16x8 out;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float gf = bf2f(g.v[j]);
-      float uf = bf2f(u.v[j]);
-      float silu = gf / (1.f + __expf(-gf));
-      out.v[j] = f2bf(silu * uf);
-    }
-    *reinterpret_cast<bf16x8*>(y + r * inter + i * 8) = out;
+  for (int j = 0; j < 8; ++j) {
+    float gf = bf2f(g.v[j]);
+    float uf = bf2f(u.v[j]);
+    float silu = gf / (1.f + __expf(-gf));
+    out.v[j] = f2bf(silu * uf);
   }
+  *reinterpret_cast<bf16x8*>(y + r * inter + i * 8) = out;
 }
 
 // ---------------------------------------------------------------------------
