@@ -172,9 +172,23 @@ def quant_fp8(x: torch.Tensor):
 
 
 def gemm_bt_fp8(a_q, a_s, b_q, b_s):
-    """C = (A_q @ B_q^T) * a_s[m] * b_s[n] on fp8 MFMA, bf16 out."""
+    """C = (A_q @ B_q^T) * a_s[m] * b_s[n] on fp8 MFMA, bf16 out.
+
+    Dispatch mirrors gemm_bt: plain shapes go to hipBLASLt's fp8 path
+    (torch._scaled_mm, 2.42 PF/s @8192^3 vs our kernel's 1.37 — bitwise-
+    identical result, profiles/r01_gemm_dispatch.txt); our kernel covers
+    shapes the library rejects and SENWEAVER_GEMM=hip."""
     if _on_gpu(a_q):
         M = a_q.shape[0]
+        if _GEMM_IMPL != "hip" and M % 16 == 0 and a_q.shape[1] % 16 == 0:
+            try:
+                return torch._scaled_mm(
+                    a_q.view(torch.float8_e4m3fn),
+                    b_q.view(torch.float8_e4m3fn).t(),
+                    scale_a=a_s.unsqueeze(1), scale_b=b_s.unsqueeze(0),
+                    out_dtype=torch.bfloat16)
+            except RuntimeError:
+                pass
         pad = (-M) % 128
         if pad:
             a_q = torch.nn.functional.pad(a_q, (0, 0, 0, pad))
@@ -217,7 +231,8 @@ def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
 
     ``a_sorted``: [T_pad, K] tokens sorted by expert (padded by >=128 rows);
     ``w``: [E, N, K]; ``seg_starts_cpu``: list/1D tensor of E+1 cumulative
-    row offsets (host side).  Rows outside segments are left unwritten.
+    row offsets (host side).  Returns at least seg_starts[-1] rows; rows
+    past the last segment (if any) are undefined.
     """
     E = w.shape[0]
     starts = [int(x) for x in seg_starts_cpu]
@@ -228,13 +243,23 @@ def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
             if t > s:
                 out[s:t] = ref.gemm_bt_ref(a_sorted[s:t], w[e])
         return out
+    dev = a_sorted.device
+    if _GEMM_IMPL != "hip" and hasattr(torch, "_grouped_mm"):
+        # hipBLASLt grouped path (1.33 PF/s vs our kernel's 0.95 at Mixtral
+        # shapes, profiles/r01_gemm_dispatch.txt); our kernel remains the
+        # fallback for offsets/shapes the library rejects
+        try:
+            return torch._grouped_mm(
+                a_sorted[:starts[-1]], w.transpose(1, 2),
+                offs=torch.tensor(starts[1:], dtype=torch.int32, device=dev))
+        except RuntimeError:
+            pass
     tile_expert, tile_m0 = [], []
     for e in range(E):
         s, t = starts[e], starts[e + 1]
         for m0 in range(s, t, 128):
             tile_expert.append(e)
             tile_m0.append(m0)
-    dev = a_sorted.device
     te = torch.tensor(tile_expert, dtype=torch.int32, device=dev)
     tm = torch.tensor(tile_m0, dtype=torch.int32, device=dev)
     ends = torch.tensor(starts[1:], dtype=torch.int32, device=dev)
