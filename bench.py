@@ -141,12 +141,10 @@ def main() -> int:
     my_mask = torch.stack(my_mask).to(dev_t) if my_cands else None
 
     def one_step(step_idx: int):
-        # teacher-forced scoring of this rank's candidate shard (micro-batched)
-        mb = args.micro_batch
-        lps = []
-        for i in range(0, my_tok.shape[0], mb):
-            lps.append(backend.score_token_batch(my_tok[i:i + mb], my_mask[i:i + mb]))
-        lps = torch.cat(lps).tolist() if lps else []
+        # teacher-forced scoring of this rank's candidate shard: microbatches
+        # pipelined over two HIP streams, ONE host sync per step
+        lps = backend.score_microbatches(my_tok, my_mask, args.micro_batch)
+        lps = lps.cpu().tolist()
         my_scores = []
         for j, ci in enumerate(my_cands):
             chunk = lps[j * args.rollouts:(j + 1) * args.rollouts]

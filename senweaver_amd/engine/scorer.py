@@ -80,39 +80,81 @@ class LlamaBackend:
     # ------------------------------------------------------------------
     def sequence_logprobs(self, sequences: Sequence[Tuple[List[int], List[bool]]]
                           ) -> List[float]:
-        """Returns the mean per-token logprob of masked tokens per sequence."""
-        results: List[float] = []
-        mb = self.micro_batch
-        for i in range(0, len(sequences), mb):
-            chunk = sequences[i: i + mb]
-            results.extend(self._logprobs_chunk(chunk))
-        return results
+        """Returns the mean per-token logprob of masked tokens per sequence.
 
-    def _logprobs_chunk(self, chunk) -> List[float]:
-        B = len(chunk)
-        S = _pad64(max(len(ids) for ids, _ in chunk))
-        tokens = torch.zeros(B, S, dtype=torch.long)
-        mask = torch.zeros(B, S, dtype=torch.bool)
-        for b, (ids, m) in enumerate(chunk):
-            tokens[b, : len(ids)] = torch.tensor(ids, dtype=torch.long)
-            mask[b, : len(m)] = torch.tensor(m, dtype=torch.bool)
-        return self.score_token_batch(tokens.to(self.device), mask.to(self.device)).tolist()
+        Microbatches are pipelined over two HIP streams (attention/elemwise
+        work of one co-schedules with the GEMMs of the next) and synced ONCE
+        at the end instead of per-chunk."""
+        mb = self.micro_batch
+        chunks = [sequences[i: i + mb] for i in range(0, len(sequences), mb)]
+        outs = []
+        for chunk, stream in zip(chunks, self._stream_cycle(len(chunks))):
+            B = len(chunk)
+            S = _pad64(max(len(ids) for ids, _ in chunk))
+            tokens = torch.zeros(B, S, dtype=torch.long)
+            mask = torch.zeros(B, S, dtype=torch.bool)
+            for b, (ids, m) in enumerate(chunk):
+                tokens[b, : len(ids)] = torch.tensor(ids, dtype=torch.long)
+                mask[b, : len(m)] = torch.tensor(m, dtype=torch.bool)
+            outs.append(self._on_stream(
+                stream, tokens.to(self.device), mask.to(self.device)))
+        return torch.cat(outs).cpu().tolist() if outs else []
+
+    def score_microbatches(self, tokens: torch.Tensor, mask: torch.Tensor,
+                           micro_batch: Optional[int] = None) -> torch.Tensor:
+        """Device-resident scores for [N, S] token/mask tensors, microbatched
+        over two streams; ONE host sync when the caller reads the result."""
+        mb = micro_batch or self.micro_batch
+        N = tokens.shape[0]
+        if N == 0:
+            return torch.zeros(0, dtype=torch.float32, device=tokens.device)
+        idxs = list(range(0, N, mb))
+        outs = []
+        for i, stream in zip(idxs, self._stream_cycle(len(idxs))):
+            outs.append(self._on_stream(stream, tokens[i:i + mb], mask[i:i + mb]))
+        return torch.cat(outs)
+
+    # two alternating HIP streams (None entries = current stream / CPU)
+    def _stream_cycle(self, n: int):
+        if self.device.type != "cuda" or n <= 1:
+            return [None] * n
+        if getattr(self, "_streams", None) is None:
+            self._streams = [torch.cuda.Stream(), torch.cuda.Stream()]
+        return [self._streams[i % 2] for i in range(n)]
+
+    def _on_stream(self, stream, tokens, mask):
+        if stream is None:
+            return self.score_token_batch_device(tokens, mask)
+        cur = torch.cuda.current_stream()
+        stream.wait_stream(cur)
+        with torch.cuda.stream(stream):
+            out = self.score_token_batch_device(tokens, mask)
+        cur.wait_stream(stream)
+        return out
 
     def score_token_batch(self, tokens: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
+        """As score_token_batch_device, synced to CPU."""
+        return self.score_token_batch_device(tokens, mask).cpu()
+
+    def score_token_batch_device(self, tokens: torch.Tensor, mask: torch.Tensor
+                                 ) -> torch.Tensor:
         """Mean per-token logprob of masked positions per sequence.
 
         tokens/mask: [B, S] on the engine device (S % 64 == 0).  Fully
         vectorized: only assistant-position hidden rows go through the
         lm_head GEMM (a ~4x lm_head saving at the bench's 25% mask).
-        Returns f32 [B] on CPU.
+        Returns f32 [B] on the engine device (no host sync).
         """
         B, S = tokens.shape
-        hidden = self.model.prefill(tokens)  # [B, S, H]
+        # mask processing FIRST: nonzero() host-syncs its stream, so doing it
+        # before the prefill is enqueued keeps the two-stream pipeline 2-deep
+        # (the sync only waits on work from two microbatches back)
         m = mask.clone()
         m[:, 0] = False  # position 0 has no predictor
         flat_pos = m.reshape(-1).nonzero(as_tuple=False).squeeze(1)  # b*S + p
         if flat_pos.numel() == 0:
-            return torch.zeros(B, dtype=torch.float32)
+            return torch.zeros(B, dtype=torch.float32, device=tokens.device)
+        hidden = self.model.prefill(tokens)  # [B, S, H]
         # p >= 1 within every sequence (position 0 masked off above), so
         # flat_pos - 1 stays inside the same sequence's rows
         rows = hidden.reshape(B * S, -1)[flat_pos - 1]
@@ -124,7 +166,7 @@ class LlamaBackend:
         counts = torch.zeros(B, dtype=torch.float32, device=lp.device)
         sums.index_add_(0, seq_of, lp.float())
         counts.index_add_(0, seq_of, torch.ones_like(lp, dtype=torch.float32))
-        return (sums / counts.clamp(min=1)).cpu()
+        return sums / counts.clamp(min=1)
 
     # ------------------------------------------------------------------
     # PromptOptimizerBackend API

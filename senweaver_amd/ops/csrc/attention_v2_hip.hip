@@ -117,6 +117,7 @@ attn_fwd_v2_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
     __syncthreads();
 
     // ---- S^T[kv64][q32] via two 32x32 subtiles ----
+    __builtin_amdgcn_s_setprio(1);
     f32x16 st[2];
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
@@ -146,6 +147,7 @@ attn_fwd_v2_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
         vals[sub * 16 + r] = v;
         tile_max = fmaxf(tile_max, v);
       }
+    __builtin_amdgcn_s_setprio(0);
     // combine the lane pair (other 32 kv of this q) — ONE cross-lane op
     tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
     const float m_new = fmaxf(m_run, tile_max);
@@ -153,7 +155,8 @@ attn_fwd_v2_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
     float rsum = 0.f;
 #pragma unroll
     for (int i = 0; i < 32; ++i) {
-      const float p = (vals[i] == -INFINITY) ? 0.f : __expf(vals[i] - m_new);
+      // exp(-inf - m_new) underflows to 0 — masked slots need no select
+      const float p = __expf(vals[i] - m_new);
       vals[i] = p;
       rsum += p;
     }
@@ -190,6 +193,7 @@ attn_fwd_v2_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
         w[g * 4 + 3] = (unsigned)r1[1];
       }
       // ---- PV: O^T[d][q] += V^T[d][kv16] @ P^T[kv16][q] for the 2 groups ----
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int g = 0; g < 2; ++g) {
         short8 pfrag;
@@ -206,6 +210,7 @@ attn_fwd_v2_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
           o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pfrag, o_acc[db], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();  // tile consumed; safe to restage
   }
