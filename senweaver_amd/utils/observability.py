@@ -136,3 +136,90 @@ class PerformanceMonitor:
             m: {"count": len(v), "avgMs": sum(v) / len(v), "maxMs": max(v)}
             for m, v in self._samples.items() if v
         }
+
+
+class MetricsPoller:
+    """Periodic metrics heartbeat (reference MetricsPollService: a 15-min
+    interval that captures a poll event so dashboards see liveness even in
+    idle sessions).  Thread-timer based; start()/stop(); the capture payload
+    includes the aggregate debug counts so each heartbeat is self-contained.
+    """
+
+    DEFAULT_INTERVAL_S = 15 * 60
+
+    def __init__(self, metrics: MetricsService,
+                 interval_s: float = DEFAULT_INTERVAL_S,
+                 extra_props: Optional[Callable[[], Dict[str, Any]]] = None) -> None:
+        self._metrics = metrics
+        self._interval = interval_s
+        self._extra = extra_props
+        self._timer = None
+        self._running = False
+        self.polls = 0
+
+    def start(self) -> None:
+        if self._running:
+            return
+        self._running = True
+        self._schedule()
+
+    def _schedule(self) -> None:
+        import threading
+        self._timer = threading.Timer(self._interval, self._fire)
+        self._timer.daemon = True
+        self._timer.start()
+
+    def _fire(self) -> None:
+        if not self._running:
+            return
+        self.poll_once()
+        self._schedule()
+
+    def poll_once(self) -> None:
+        props: Dict[str, Any] = {"poll": self.polls, **self._metrics.debug_info()}
+        if self._extra is not None:
+            try:
+                props.update(self._extra())
+            except Exception:
+                pass
+        self._metrics.capture("metrics_poll_heartbeat", props)
+        self.polls += 1
+
+    def stop(self) -> None:
+        self._running = False
+        if self._timer is not None:
+            self._timer.cancel()
+            self._timer = None
+
+
+class trace_range:
+    """Host-side analog of the reference's perf spans that also shows up in
+    rocprofv3 timelines: wraps a region in a ROCtx/NVTX range when CUDA(=HIP)
+    is available, and records the wall time into a PerformanceMonitor.
+
+    Usage: ``with trace_range("score_batch", monitor): ...``
+    """
+
+    def __init__(self, name: str, monitor: Optional[PerformanceMonitor] = None) -> None:
+        self.name = name
+        self._monitor = monitor
+        self._nvtx = False
+        self._t0 = 0.0
+
+    def __enter__(self) -> "trace_range":
+        try:
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.nvtx.range_push(self.name)
+                self._nvtx = True
+        except Exception:
+            pass
+        self._t0 = time.perf_counter()
+        return self
+
+    def __exit__(self, *exc) -> None:
+        if self._monitor is not None:
+            self._monitor.record(self.name, (time.perf_counter() - self._t0) * 1000)
+        if self._nvtx:
+            import torch
+            torch.cuda.nvtx.range_pop()

@@ -270,3 +270,27 @@ def test_stats_shape(fixed_clock, seq_uuid):
     assert stats["totalReports"] == 1
     assert stats["beamSearchActive"] is False
     assert stats["beamBestScore"] is None
+
+
+def test_pack_rules_budget_property():
+    # property test: for arbitrary rule lists the packed content
+    # never exceeds the 2000-char budget, is exactly the '\n'-join of the
+    # first `included` rules, and packing stops at the FIRST rule that would
+    # overflow (reference loop semantics — no skip-and-continue)
+    from hypothesis import given, strategies as st
+
+    @given(st.lists(st.text(alphabet=st.characters(blacklist_categories=("Cs",)),
+                            min_size=0, max_size=700), max_size=30))
+    def prop(rules):
+        content, included = pack_rules(rules)
+        assert len(content) <= 2000
+        assert 0 <= included <= len(rules)
+        expected = ""
+        for r in rules[:included]:
+            expected = expected + ("\n" if expected else "") + r
+        assert content == expected
+        if included < len(rules):
+            overflow = content + ("\n" if content else "") + rules[included]
+            assert len(overflow) > 2000
+
+    prop()

@@ -77,3 +77,69 @@ def test_scm_commit_message(tmp_path):
     assert "hello" in diff
     msg = svc.generate_commit_message()
     assert msg  # produced by the backend
+
+
+# ---------------- AIRegexService ----------------
+
+class _ScriptedBackend:
+    def __init__(self, reply):
+        self.reply = reply
+
+    def generate(self, prompt, max_new_tokens=64):
+        if isinstance(self.reply, Exception):
+            raise self.reply
+        return self.reply
+
+
+def test_airegex_valid_generation():
+    from senweaver_amd.features.airegex import AIRegexService
+    svc = AIRegexService(_ScriptedBackend(r"\d{3}-\d{4}"))
+    r = svc.generate("a phone number like 555-1234")
+    assert r.pattern == r"\d{3}-\d{4}" and not r.is_fallback
+
+
+def test_airegex_slash_delimited_and_fenced():
+    from senweaver_amd.features.airegex import AIRegexService
+    svc = AIRegexService(_ScriptedBackend("```\n/foo+bar/gi\n```"))
+    r = svc.generate("foo then bar")
+    assert r.pattern == "foo+bar" and not r.is_fallback
+
+
+def test_airegex_invalid_falls_back_literal():
+    from senweaver_amd.features.airegex import AIRegexService
+    import re
+    svc = AIRegexService(_ScriptedBackend("([unclosed"))
+    r = svc.generate("a (weird) thing")
+    assert r.is_fallback and re.search(r.pattern, "a (weird) thing")
+
+
+def test_airegex_offline_fallback():
+    from senweaver_amd.features.airegex import AIRegexService
+    svc = AIRegexService(None)
+    r = svc.generate("c++ code")
+    assert r.is_fallback
+    import re
+    assert re.search(r.pattern, "some c++ code here")
+
+
+# ---------------- MetricsPoller ----------------
+
+def test_metrics_poller_heartbeat():
+    from senweaver_amd.utils.observability import MetricsPoller, MetricsService
+    ms = MetricsService()
+    ms.capture("some_event", {"a": 1})
+    p = MetricsPoller(ms, interval_s=9999, extra_props=lambda: {"gpu": 1})
+    p.poll_once()
+    p.poll_once()
+    beats = [e for e in ms.events if e["event"] == "metrics_poll_heartbeat"]
+    assert len(beats) == 2
+    assert beats[0]["poll"] == 0 and beats[1]["poll"] == 1
+    assert beats[0]["gpu"] == 1 and beats[0]["totalEvents"] >= 1
+
+
+def test_trace_range_records():
+    from senweaver_amd.utils.observability import PerformanceMonitor, trace_range
+    mon = PerformanceMonitor(enabled=True)
+    with trace_range("messageTrimming", mon):
+        pass
+    assert "messageTrimming" in mon.summary()
