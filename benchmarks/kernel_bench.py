@@ -74,9 +74,11 @@ def bench_attn():
         v = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
         vt = v.transpose(-1, -2).contiguous()
         scale = 1.0 / math.sqrt(D)
-        t = timeit(lambda: ops.attn_fwd_t(q, k, vt, scale))
+        ext = ops.hip_ext()
+        t2 = timeit(lambda: ext.attn_fwd_v2(q, k, vt, scale))
+        t3 = timeit(lambda: ext.attn_fwd_v3(q, k, vt, scale))
         flops = 2 * 2 * B * H * S * S * D / 2  # causal half
-        print(f"ATTN B{B} H{H} S{S}: {flops / t / 1e12:7.1f} TF/s  ({t * 1e3:.2f} ms)")
+        print(f"ATTN B{B} H{H} S{S}: v2 {flops / t2 / 1e12:7.1f}  v3 {flops / t3 / 1e12:7.1f} TF/s")
         tsdpa = timeit(lambda: torch.nn.functional.scaled_dot_product_attention(
             q, k, v, is_causal=True, enable_gqa=True))
         print(f"  torch sdpa:        {flops / tsdpa / 1e12:7.1f} TF/s")

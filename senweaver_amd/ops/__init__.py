@@ -122,6 +122,7 @@ def swiglu(gateup: torch.Tensor) -> torch.Tensor:
 
 
 _GEMM_IMPL = os.environ.get("SENWEAVER_GEMM", "auto")  # auto | hip
+_ATTN_IMPL = os.environ.get("SENWEAVER_ATTN", "auto")  # auto | v2
 
 
 def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
@@ -296,8 +297,14 @@ def attn_fwd_t(q: torch.Tensor, k: torch.Tensor, vt: torch.Tensor,
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
-        return hip_ext().attn_fwd_v2(q.contiguous(), k.contiguous(),
-                                     vt.contiguous(), scale)
+        ext = hip_ext()
+        # v3 (8 waves sharing the staged K/V tile) when the grid still fills
+        # the chip at 2 blocks/CU; v2's smaller blocks otherwise
+        if _ATTN_IMPL != "v2" and q.shape[2] >= 256 and                 ((q.shape[2] + 255) // 256) * q.shape[0] * q.shape[1] >= 512:
+            return ext.attn_fwd_v3(q.contiguous(), k.contiguous(),
+                                   vt.contiguous(), scale)
+        return ext.attn_fwd_v2(q.contiguous(), k.contiguous(),
+                               vt.contiguous(), scale)
     v = vt.transpose(-1, -2).contiguous()
     o = ref.attn_fwd_ref(q, k, v, scale, causal=True)
     return o.transpose(-1, -2).contiguous()

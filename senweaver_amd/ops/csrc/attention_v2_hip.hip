@@ -225,3 +225,174 @@ attn_fwd_v2_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
       OTh[(long long)d * S + q_lane] = f2bf(o_acc[db][r] * inv_l);
     }
 }
+
+// ---------------------------------------------------------------------------
+// v3: same per-wave structure as v2, but EIGHT waves per block (512 threads,
+// 256 q rows) sharing one staged K/V tile — staging bytes and barrier count
+// per flop halve at unchanged per-wave VGPR (124 -> still 4 waves/SIMD via
+// 2 blocks/CU).  S % 128 == 0; q rows past S are inert (guarded load/store).
+// ---------------------------------------------------------------------------
+#define V3_QBLK 256
+
+__device__ __forceinline__ void v3_stage(const ushort* __restrict__ src,
+                                         long long ld, int chunks_per_row,
+                                         ushort* lds_tile, int chunks_total,
+                                         int tid) {
+  const int wave_chunk = tid & ~63;
+  for (int s0 = 0; s0 < chunks_total; s0 += 512) {
+    const int s = s0 + tid;
+    if (s >= chunks_total) break;
+    const int row = s / chunks_per_row;
+    const int cl = s % chunks_per_row;
+    const int c = ((cl & 7) ^ (row & 7)) | (cl & ~7);
+    const ushort* g = src + (long long)row * ld + c * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(lds_tile + ((long long)(s0 + wave_chunk)) * 8),
+        16, 0, 0);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+attn_fwd_v3_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
+                   const ushort* __restrict__ VT, ushort* __restrict__ OT,
+                   int B, int H, int Hk, int S, float scale) {
+  const int qb = gridDim.x - 1 - blockIdx.x;  // heaviest-first
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int kvh = h / (H / Hk);
+  const int D = 128;
+
+  const ushort* Qh = Q + (((long long)b * H + h) * S) * D;
+  const ushort* Kh = K + (((long long)b * Hk + kvh) * S) * D;
+  const ushort* VTh = VT + (((long long)b * Hk + kvh) * D) * S;
+  ushort* OTh = OT + (((long long)b * H + h) * D) * S;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int l31 = lane & 31;
+  const int lhi = lane >> 5;
+  const int q0 = qb * V3_QBLK + wid * 32;
+  const int q_lane = q0 + l31;
+  const bool live = q_lane < S;
+
+  __shared__ __attribute__((aligned(16))) ushort smem[V2_KVBLK * 128 + 128 * V2_KVBLK];
+  ushort* k_lds = smem;
+  ushort* vt_lds = smem + V2_KVBLK * 128;
+
+  short8 qf[8];
+  {
+    const long long qrow = (long long)(live ? q_lane : 0) * D;
+#pragma unroll
+    for (int st = 0; st < 8; ++st)
+      qf[st] = *reinterpret_cast<const short8*>(Qh + qrow + st * 16 + lhi * 8);
+  }
+
+  float m_run = -INFINITY, l_run = 0.f;
+  f32x16 o_acc[4];
+#pragma unroll
+  for (int db = 0; db < 4; ++db)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) o_acc[db][r] = 0.f;
+
+  const int kv_end = min(S, qb * V3_QBLK + V3_QBLK);
+  for (int kv0 = 0; kv0 < kv_end; kv0 += V2_KVBLK) {
+    v3_stage(Kh + (long long)kv0 * D, D, 16, k_lds, V2_KVBLK * 16, tid);
+    v3_stage(VTh + kv0, S, 8, vt_lds, 128 * 8, tid);
+    __syncthreads();
+
+    __builtin_amdgcn_s_setprio(1);
+    f32x16 st[2];
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      f32x16 acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+#pragma unroll
+      for (int stp = 0; stp < 8; ++stp) {
+        short8 kf = v2_read(k_lds, sub * 32 + l31, stp * 2 + lhi, 16);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp], acc, 0, 0, 0);
+      }
+      st[sub] = acc;
+    }
+
+    float vals[32];
+    float tile_max = -INFINITY;
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv = kv0 + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+        float v = st[sub][r] * scale;
+        if (kv > q_lane) v = -INFINITY;
+        vals[sub * 16 + r] = v;
+        tile_max = fmaxf(tile_max, v);
+      }
+    __builtin_amdgcn_s_setprio(0);
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+    const float m_new = fmaxf(m_run, tile_max);
+    const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+    float rsum = 0.f;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) {
+      const float p = __expf(vals[i] - m_new);
+      vals[i] = p;
+      rsum += p;
+    }
+    rsum += __shfl_xor(rsum, 32, 64);
+    l_run = l_run * alpha + rsum;
+    m_run = m_new;
+#pragma unroll
+    for (int db = 0; db < 4; ++db)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
+
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      unsigned w[8];
+#pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        const int base = sub * 16 + g * 8;
+        unsigned x0 = cvt_pk_bf16(vals[base + 0], vals[base + 1]);
+        unsigned y0 = cvt_pk_bf16(vals[base + 4], vals[base + 5]);
+        unsigned x1 = cvt_pk_bf16(vals[base + 2], vals[base + 3]);
+        unsigned y1 = cvt_pk_bf16(vals[base + 6], vals[base + 7]);
+        auto r0 = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
+        auto r1 = __builtin_amdgcn_permlane32_swap(x1, y1, false, false);
+        w[g * 4 + 0] = (unsigned)r0[0];
+        w[g * 4 + 1] = (unsigned)r1[0];
+        w[g * 4 + 2] = (unsigned)r0[1];
+        w[g * 4 + 3] = (unsigned)r1[1];
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        short8 pfrag;
+        unsigned* pw = reinterpret_cast<unsigned*>(&pfrag);
+        pw[0] = w[g * 4 + 0];
+        pw[1] = w[g * 4 + 1];
+        pw[2] = w[g * 4 + 2];
+        pw[3] = w[g * 4 + 3];
+        const int kvg = sub * 32 + g * 16;
+#pragma unroll
+        for (int db = 0; db < 4; ++db) {
+          short8 vf = v2_read(vt_lds, db * 32 + l31, (kvg >> 3) + lhi, 8);
+          o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pfrag, o_acc[db], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+
+  if (!live) return;
+  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+  for (int db = 0; db < 4; ++db)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+      OTh[(long long)d * S + q_lane] = f2bf(o_acc[db][r] * inv_l);
+    }
+}

@@ -31,6 +31,7 @@ extern "C" __global__ void gemv_bt_bf16_v2_m8(const ushort*, const ushort*, usho
 extern "C" __global__ void gemv_bt_bf16_v2_m16(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
 extern "C" __global__ void attn_fwd_v2_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
+extern "C" __global__ void attn_fwd_v3_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
 extern "C" __global__ void paged_decode_attn_kernel(const ushort*, const ushort*, const ushort*, ushort*,
                                                     const int*, const int*, int, int, int, float);
 
@@ -348,6 +349,23 @@ torch::Tensor attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
   return ot;
 }
 
+// v3: v2 structure with 8 waves sharing one staged K/V tile (256 q rows)
+torch::Tensor attn_fwd_v3(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
+                          double scale) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(vt, "vt");
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hk = k.size(1);
+  TORCH_CHECK(D == 128 && S % 128 == 0 && H % Hk == 0);
+  TORCH_CHECK(vt.size(1) == Hk && vt.size(2) == D && vt.size(3) == S);
+  auto ot = torch::empty({B, H, D, S}, q.options());
+  attn_fwd_v3_kernel<<<dim3((S + 255) / 256, H, B), dim3(512), 0, cur_stream()>>>(
+      bf16_ptr(q), bf16_ptr(k), bf16_ptr(vt), bf16_mut(ot), B, H, Hk, S, (float)scale);
+  HIP_CHECK_KERNEL();
+  return ot;
+}
+
 // ---------------- Paged decode attention ----------------
 torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor kcache,
                                 torch::Tensor vcache, torch::Tensor block_table,
@@ -410,6 +428,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt_mxfp8", &gemm_bt_mxfp8, "MX block-scaled fp8 MFMA GEMM (32x32x64)");
   m.def("attn_fwd", &attn_fwd, "causal flash attention fwd (D=128, GQA)");
   m.def("attn_fwd_v2", &attn_fwd_v2, "swapped-QK^T attention; O^T out [B,H,D,S]");
+  m.def("attn_fwd_v3", &attn_fwd_v3, "v2 with 8-wave shared K/V tiles; O^T out");
   m.def("paged_decode_attn", &paged_decode_attn, "paged decode attention");
   m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");
   m.def("target_logprob", &target_logprob, "fused log_softmax gather");
