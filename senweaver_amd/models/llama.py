@@ -46,8 +46,8 @@ class LlamaModel:
         self.local_kv_size = self.local_kv_heads * c.head_dim
         self.local_inter = c.intermediate_size // tpw
         assert quant in ("bf16", "fp8", "mxfp8")
-        if quant in ("fp8", "mxfp8"):
-            assert c.num_experts == 0, "fp8 MoE not supported yet"
+        if quant == "mxfp8":
+            assert c.num_experts == 0, "mxfp8 MoE not supported yet"
         self.quant = quant
         if self.device.type == "cuda":
             # Device-side init: seconds for 8B instead of minutes of CPU RNG +
@@ -105,11 +105,21 @@ class LlamaModel:
             # "mxfp8": e8m0 scale per 32-element K block, run on the
             # block-scaled 32x32x64 MFMA (2x fp8 rate, HW-fused dequant).
             qfn = ops.quant_fp8 if self.quant == "fp8" else ops.quant_mxfp8
+            dense_projs = (("qkv", "o", "gateup", "down") if c.num_experts == 0
+                           else ("qkv", "o"))
             for L in self.layers:
-                for name in ("qkv", "o", "gateup", "down"):
+                for name in dense_projs:
                     q, s = qfn(L[name])
                     L[name + "_q"], L[name + "_s"] = q, s
                     del L[name]
+                if c.num_experts > 0:
+                    # per-expert rowwise quant (router stays bf16 — tiny)
+                    for name in ("w13", "w2"):
+                        E, N, K = L[name].shape
+                        q, s = ops.quant_fp8(L[name])
+                        L[name + "_q"] = q.reshape(E, N, K)
+                        L[name + "_s"] = s.reshape(E, N)
+                        del L[name]
             self.lm_head_q, self.lm_head_s = qfn(self.lm_head)
         self.cos_sin = ops.rope_tables(c.max_position, c.head_dim, c.rope_theta).to(self.device)
         self.scale = 1.0 / math.sqrt(c.head_dim)
@@ -206,6 +216,8 @@ class LlamaModel:
         c = self.config
         T = h.shape[0]
         k = c.num_experts_per_tok
+        if self.quant == "fp8":
+            return self._moe_ffn_fp8(h, L)
         # router is [E=8, H]: far below the MFMA tile (N=8) — a plain skinny
         # library matmul, not a hot op
         router_logits = h @ L["router"].t()  # [T, E]
@@ -232,6 +244,47 @@ class LlamaModel:
         down = ops.grouped_gemm_bt(act_p, L["w2"], seg_starts)[:Tk]
         out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=h.device)
         out.index_add_(0, sorted_token, down.float() * sorted_weight.unsqueeze(1))
+        return out.to(h.dtype)
+
+    def _moe_ffn_fp8(self, h: torch.Tensor, L: dict) -> torch.Tensor:
+        """fp8 routed FFN: expert-sorted tokens in 16-ALIGNED padded segments
+        (hipBLASLt fp8 alignment), rowwise-quantized activations, one fp8
+        GEMM per non-empty expert (ops.grouped_gemm_bt_fp8).  Pad rows are
+        zero -> quantize to zero -> contribute nothing."""
+        c = self.config
+        T = h.shape[0]
+        k = c.num_experts_per_tok
+        router_logits = h @ L["router"].t()
+        probs = torch.softmax(router_logits.float(), dim=-1)
+        topw, topi = probs.topk(k, dim=-1)
+        topw = topw / topw.sum(dim=-1, keepdim=True)
+        flat_expert = topi.reshape(-1)
+        flat_token = torch.arange(T, device=h.device).repeat_interleave(k)
+        order = torch.argsort(flat_expert, stable=True)
+        sorted_token = flat_token[order]
+        sorted_weight = topw.reshape(-1)[order]
+        sorted_expert = flat_expert[order]
+        counts = torch.bincount(flat_expert, minlength=c.num_experts)
+        seg_starts, pad_starts = [0], [0]
+        for e in range(c.num_experts):
+            n = int(counts[e])
+            seg_starts.append(seg_starts[-1] + n)
+            pad_starts.append(pad_starts[-1] + ((n + 15) // 16) * 16)
+        Tk = T * k
+        dev = h.device
+        seg_t = torch.tensor(seg_starts[:-1], device=dev)
+        pad_t = torch.tensor(pad_starts[:-1], device=dev)
+        # destination row of each sorted token inside its padded segment
+        dest = pad_t[sorted_expert] + (torch.arange(Tk, device=dev) - seg_t[sorted_expert])
+        a_pad = torch.zeros(pad_starts[-1], c.hidden_size, dtype=h.dtype, device=dev)
+        a_pad[dest] = h[sorted_token]
+        aq, a_s = ops.quant_fp8(a_pad)
+        gateup = ops.grouped_gemm_bt_fp8(aq, a_s, L["w13_q"], L["w13_s"], pad_starts)
+        act = ops.swiglu(gateup)
+        aq2, as2 = ops.quant_fp8(act)
+        down = ops.grouped_gemm_bt_fp8(aq2, as2, L["w2_q"], L["w2_s"], pad_starts)
+        out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=dev)
+        out.index_add_(0, sorted_token, down[dest].float() * sorted_weight.unsqueeze(1))
         return out.to(h.dtype)
 
     # ------------------------------------------------------------------

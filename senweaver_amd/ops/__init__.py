@@ -267,6 +267,41 @@ def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
     return hip_ext().grouped_gemm_bt(a_sorted.contiguous(), w.contiguous(), te, tm, ends)
 
 
+def grouped_gemm_bt_fp8(a_q, a_s, w_q, w_s, seg_starts_cpu) -> torch.Tensor:
+    """Per-expert fp8 GEMM over expert-sorted segments (fp8 MoE).
+
+    ``a_q``/[rows] ``a_s``: rowwise-quantized sorted activations;
+    ``w_q`` [E,N,K] u8 + ``w_s`` [E,N] f32; ``seg_starts_cpu``: E+1 host
+    offsets, 16-ALIGNED (the MoE layer pads each expert's segment so every
+    start satisfies hipBLASLt's fp8 alignment).  GPU: one torch._scaled_mm
+    per non-empty expert; our fp8 kernel is the per-segment fallback.
+    """
+    E, N = w_q.shape[0], w_q.shape[1]
+    starts = [int(x) for x in seg_starts_cpu]
+    Tk = starts[-1]
+    if not _on_gpu(a_q):
+        out = torch.zeros(Tk, N, dtype=torch.bfloat16)
+        for e in range(E):
+            s, t = starts[e], starts[e + 1]
+            if t > s:
+                out[s:t] = ref.gemm_bt_fp8_ref(a_q[s:t], a_s[s:t], w_q[e], w_s[e])
+        return out
+    out = torch.empty(Tk, N, dtype=torch.bfloat16, device=a_q.device)
+    for e in range(E):
+        s, t = starts[e], starts[e + 1]
+        if t == s:
+            continue
+        try:
+            out[s:t] = torch._scaled_mm(
+                a_q[s:t].view(torch.float8_e4m3fn),
+                w_q[e].view(torch.float8_e4m3fn).t(),
+                scale_a=a_s[s:t].unsqueeze(1), scale_b=w_s[e].unsqueeze(0),
+                out_dtype=torch.bfloat16)
+        except RuntimeError:
+            out[s:t] = gemm_bt_fp8(a_q[s:t], a_s[s:t], w_q[e], w_s[e])
+    return out
+
+
 def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional[float] = None,
              vt: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Causal GQA attention. q [B,H,S,D], k/v [B,Hk,S,D] (S % 64 == 0 on GPU).

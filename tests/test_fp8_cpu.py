@@ -129,3 +129,37 @@ def test_mxfp8_model_close_to_bf16():
     h8 = m8.prefill(toks)
     rel = (h16.float() - h8.float()).norm() / h16.float().norm()
     assert rel.item() < 0.15
+
+
+def test_fp8_moe_forward_close_to_bf16():
+    from senweaver_amd.models.config import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+    m16 = LlamaModel(tiny_moe(), device="cpu", seed=4)
+    m8 = LlamaModel(tiny_moe(), device="cpu", seed=4, quant="fp8")
+    toks = torch.randint(0, 512, (2, 16))
+    h16 = m16.prefill(toks)
+    h8 = m8.prefill(toks)
+    assert torch.isfinite(h8.float()).all()
+    rel = (h16.float() - h8.float()).norm() / h16.float().norm()
+    assert rel.item() < 0.25
+
+
+def test_grouped_gemm_fp8_segments():
+    torch.manual_seed(9)
+    E, K, N = 4, 64, 32
+    sizes = [16, 0, 48, 16]  # 16-aligned incl. empty expert
+    seg = [0]
+    for s in sizes:
+        seg.append(seg[-1] + s)
+    a = torch.randn(seg[-1], K).to(torch.bfloat16)
+    w = torch.randn(E, N, K).to(torch.bfloat16)
+    aq, asc = ops.quant_fp8(a)
+    wq_flat, ws_flat = ops.quant_fp8(w)
+    wq = wq_flat.reshape(E, N, K)
+    ws = ws_flat.reshape(E, N)
+    out = ops.grouped_gemm_bt_fp8(aq, asc, wq, ws, seg)
+    for e in range(E):
+        s, t = seg[e], seg[e + 1]
+        if t > s:
+            r = ops.reference.gemm_bt_fp8_ref(aq[s:t], asc[s:t], wq[e], ws[e])
+            torch.testing.assert_close(out[s:t].float(), r.float(), atol=0.3, rtol=3e-2)

@@ -379,3 +379,16 @@ def test_attn_v3_matches_v2(dev):
         o2 = ext.attn_fwd_v2(q, k, vt, s)
         o3 = ext.attn_fwd_v3(q, k, vt, s)
         torch.testing.assert_close(o2.float(), o3.float(), atol=2e-2, rtol=1e-2)
+
+
+def test_fp8_moe_forward_gpu(dev):
+    from senweaver_amd.models import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+    m16 = LlamaModel(tiny_moe(), device=dev, seed=4)
+    m8 = LlamaModel(tiny_moe(), device=dev, seed=4, quant="fp8")
+    toks = torch.randint(0, 512, (2, 64), device=dev)
+    h16 = m16.prefill(toks)
+    h8 = m8.prefill(toks)
+    assert torch.isfinite(h8.float()).all()
+    rel = ((h16.float() - h8.float()).norm() / h16.float().norm()).item()
+    assert rel < 0.25
