@@ -197,3 +197,33 @@ def test_distributed_beam_identical_state():
     assert results[0] == results[1]
     assert results[0]["round"] == 2
     assert len(results[0]["beam"]) == 4
+
+
+def _payload_tp_prefill_fp8(rank, world):
+    import torch
+    from senweaver_amd.models import tiny_tp
+    from senweaver_amd.models.llama import LlamaModel
+    from senweaver_amd.parallel.tp import TPContext
+
+    tp = TPContext.from_default_group()
+    model = LlamaModel(tiny_tp(), device="cpu", seed=5, tp=tp, quant="fp8")
+    ref16 = LlamaModel(tiny_tp(), device="cpu", seed=5)
+    tokens = torch.randint(0, 512, (1, 64), generator=torch.Generator().manual_seed(9))
+    hidden = model.prefill(tokens).float()
+    h16 = ref16.prefill(tokens).float()
+    rel = ((hidden - h16).norm() / h16.norm()).item()
+    return [rel] + hidden.sum(-1).squeeze(0).tolist()[:8]
+
+
+def test_tp2_fp8_agrees_and_tracks_bf16():
+    """config-5 cross: TP sharding composed with fp8 projections.
+
+    Row-parallel shards (o/down) quantize over their LOCAL K-half, so fp8
+    TP=k is a different (finer) quantization grouping than TP=1 — bitwise
+    equality is not expected.  The contract: every rank produces the
+    identical result (collective determinism) and the fp8 TP output stays
+    within fp8 tolerance of the bf16 model."""
+    results = _run_dist("_payload_tp_prefill_fp8")
+    for rank, vals in results.items():
+        assert vals[0] < 0.15  # rel vs bf16
+    assert results[0] == pytest.approx(results[1], abs=1e-5)
