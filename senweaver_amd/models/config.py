@@ -84,6 +84,16 @@ def tiny_moe(vocab: int = 512) -> ModelConfig:
     )
 
 
+def tiny_moe_tp(vocab: int = 512) -> ModelConfig:
+    """CPU-testable MoE model with TP-divisible heads/experts' FFN."""
+    return ModelConfig(
+        name="tiny-moe-tp", hidden_size=256, intermediate_size=512,
+        num_layers=2, num_heads=2, num_kv_heads=2, head_dim=128,
+        vocab_size=vocab, max_position=2048, num_experts=4,
+        num_experts_per_tok=2,
+    )
+
+
 def tiny_tp(vocab: int = 512) -> ModelConfig:
     """CPU-testable TP model (heads divisible by 2)."""
     return ModelConfig(

@@ -39,7 +39,6 @@ class LlamaModel:
             assert c.num_heads % tpw == 0 and c.num_kv_heads % tpw == 0, \
                 f"TP={tpw} must divide heads ({c.num_heads}/{c.num_kv_heads})"
             assert c.intermediate_size % tpw == 0
-            assert c.num_experts == 0, "MoE+TP combo not supported yet"
         self.local_heads = c.num_heads // tpw
         self.local_kv_heads = c.num_kv_heads // tpw
         self.local_q_size = self.local_heads * c.head_dim
@@ -85,10 +84,24 @@ class LlamaModel:
             }
             del qkv_full, o_full
             if c.num_experts > 0:
-                # Mixtral-style MoE: fused gate|up per expert + down, router
+                # Mixtral-style MoE: fused gate|up per expert + down, router.
+                # TP shards WITHIN each expert (Megatron column gate|up / row
+                # down); the router is replicated, and since every rank sees
+                # the identical hidden states, routing agrees by construction.
                 layer["router"] = W(c.num_experts, c.hidden_size)
-                layer["w13"] = W(c.num_experts, 2 * c.intermediate_size, c.hidden_size)
-                layer["w2"] = W(c.num_experts, c.hidden_size, c.intermediate_size)
+                w13_full = W(c.num_experts, 2 * c.intermediate_size, c.hidden_size)
+                w2_full = W(c.num_experts, c.hidden_size, c.intermediate_size)
+                if tpw > 1:
+                    layer["w13"] = torch.stack([
+                        shard_gateup(w13_full[e], c.intermediate_size, self.tp)
+                        for e in range(c.num_experts)]).contiguous()
+                    layer["w2"] = torch.stack([
+                        shard_rows(w2_full[e], self.tp, dim=1)
+                        for e in range(c.num_experts)]).contiguous()
+                    del w13_full, w2_full
+                else:
+                    layer["w13"] = w13_full
+                    layer["w2"] = w2_full
             else:
                 gateup_full = W(2 * c.intermediate_size, c.hidden_size)
                 down_full = W(c.hidden_size, c.intermediate_size)
@@ -239,12 +252,14 @@ class LlamaModel:
         a_sorted[:Tk] = h[sorted_token]
         gateup = ops.grouped_gemm_bt(a_sorted, L["w13"], seg_starts)
         act = ops.swiglu(gateup[:Tk])
-        act_p = torch.zeros(Tk + pad, c.intermediate_size, dtype=h.dtype, device=h.device)
+        act_p = torch.zeros(Tk + pad, act.shape[1], dtype=h.dtype, device=h.device)
         act_p[:Tk] = act
         down = ops.grouped_gemm_bt(act_p, L["w2"], seg_starts)[:Tk]
         out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=h.device)
         out.index_add_(0, sorted_token, down.float() * sorted_weight.unsqueeze(1))
-        return out.to(h.dtype)
+        res = out.to(h.dtype)
+        self.tp.all_reduce_(res)  # row-parallel down partial sum (TP)
+        return res
 
     def _moe_ffn_fp8(self, h: torch.Tensor, L: dict) -> torch.Tensor:
         """fp8 routed FFN: expert-sorted tokens in 16-ALIGNED padded segments
@@ -285,7 +300,9 @@ class LlamaModel:
         down = ops.grouped_gemm_bt_fp8(aq2, as2, L["w2_q"], L["w2_s"], pad_starts)
         out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=dev)
         out.index_add_(0, sorted_token, down[dest].float() * sorted_weight.unsqueeze(1))
-        return out.to(h.dtype)
+        res = out.to(h.dtype)
+        self.tp.all_reduce_(res)  # row-parallel down partial sum (TP)
+        return res
 
     # ------------------------------------------------------------------
     # Decode: one new token per sequence.  tokens [B], positions [B].

@@ -227,3 +227,33 @@ def test_tp2_fp8_agrees_and_tracks_bf16():
     for rank, vals in results.items():
         assert vals[0] < 0.15  # rel vs bf16
     assert results[0] == pytest.approx(results[1], abs=1e-5)
+
+
+def _payload_moe_tp_prefill(rank, world):
+    import torch
+    from senweaver_amd.models import tiny_moe_tp
+    from senweaver_amd.models.llama import LlamaModel
+    from senweaver_amd.parallel.tp import TPContext
+
+    tp = TPContext.from_default_group()
+    model = LlamaModel(tiny_moe_tp(), device="cpu", seed=6, tp=tp)
+    tokens = torch.randint(0, 512, (1, 64), generator=torch.Generator().manual_seed(11))
+    hidden = model.prefill(tokens)
+    return hidden.float().sum(-1).squeeze(0).tolist()[:8]
+
+
+def test_moe_tp2_matches_tp1():
+    """MoE + TP: per-expert Megatron sharding reproduces the TP=1 model
+    (replicated router -> identical routing; column gate|up + row down)."""
+    import torch
+    from senweaver_amd.models import tiny_moe_tp
+    from senweaver_amd.models.llama import LlamaModel
+
+    ref_model = LlamaModel(tiny_moe_tp(), device="cpu", seed=6)
+    tokens = torch.randint(0, 512, (1, 64), generator=torch.Generator().manual_seed(11))
+    ref = ref_model.prefill(tokens).float().sum(-1).squeeze(0).tolist()[:8]
+
+    results = _run_dist("_payload_moe_tp_prefill")
+    for rank, vals in results.items():
+        assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
+    assert results[0] == pytest.approx(results[1], abs=1e-5)
