@@ -93,7 +93,9 @@ def main() -> int:
     ap.add_argument("--beam-width", type=int, default=4)
     ap.add_argument("--branch-factor", type=int, default=4)
     ap.add_argument("--rollouts", type=int, default=4)
-    ap.add_argument("--micro-batch", type=int, default=4  # measured best on MI355X: 16 chunks of 4 pipeline across the two HIP streams (0.627 vs 0.612 its at mb8))
+    # mb4 measured best on MI355X: 16 chunks of 4 pipeline across the two
+    # HIP streams (repeatable 0.627 vs 0.612 it/s at mb8)
+    ap.add_argument("--micro-batch", type=int, default=4)
     ap.add_argument("--quant", default="bf16", choices=["bf16", "fp8", "mxfp8"])
     ap.add_argument("--tp", action="store_true",
                     help="tensor-parallel scoring over the whole world (config 5) "
