@@ -1,9 +1,16 @@
-"""Op dispatch: hand-written gfx950 HIP kernels on GPU, fp32 torch on CPU.
+"""Op dispatch for the gfx950 compute path.
 
-On a GPU box the HIP extension is REQUIRED — ops raise if a CUDA(=HIP) tensor
-arrives and the extension is missing, so nothing silently falls back to eager
-PyTorch on the hardware the kernels target.  CPU tensors use the fp32
-reference implementations (tests / CPU-only engine mode).
+Fused and shape-special ops (attention, paged decode, RoPE/RMSNorm/SwiGLU,
+V^T extraction, MX/rowwise fp8 quant, decode GEMV) run the in-repo HIP
+kernels; PLAIN GEMMs go to whichever implementation measured faster on
+MI355X — hipBLASLt for prefill shapes, our kernels elsewhere — per
+profiles/r01_gemm_dispatch.txt (SENWEAVER_GEMM=hip forces in-repo
+everywhere).
+
+On a GPU box the HIP extension is REQUIRED — ops raise if a CUDA(=HIP)
+tensor arrives and the extension is missing, so nothing silently falls back
+to eager PyTorch on the hardware the kernels target.  CPU tensors use the
+fp32 reference implementations (tests / CPU-only engine mode).
 """
 
 from __future__ import annotations
@@ -129,7 +136,7 @@ def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """C[M,N] = A[M,K] @ B[N,K]^T.
 
     Dispatch (measured on MI355X, profiles/r01_gemm_dispatch.txt): decode-
-    shaped M<=16 goes to our wave-per-row GEMV kernel (4.4 TB/s, ~2x blas on
+    shaped M<=16 goes to our wave-per-row GEMV kernel (5.2 TB/s in situ, ~2x blas on
     skinny); plain prefill GEMMs go to hipBLASLt (1.28-1.58 PF/s at bench
     shapes vs our tiled kernel's 0.88-1.15 — the guide's rule: hand-write the
     fused/special ops, use the vendor library for plain GEMMs it wins).
