@@ -38,12 +38,18 @@ __device__ __forceinline__ void v2_stage(const ushort* __restrict__ src,
                                          long long ld, int chunks_per_row,
                                          ushort* lds_tile, int chunks_total,
                                          int tid) {
+  // XOR swizzle over the FULL chunk index (mask = chunks_per_row-1): a
+  // 16-lane ds_read_b128 phase reads 16 consecutive rows at one logical
+  // chunk, and a full-width XOR maps them to 16 distinct bank groups
+  // (the 3-bit XOR left rows r and r+8 on the same banks: 2-way conflict,
+  // PMC SQ_LDS_BANK_CONFLICT ~1.1e9 cycles at the bench shape)
+  const int cmask = chunks_per_row - 1;
   const int wave_chunk = tid & ~63;
   for (int s0 = 0; s0 < chunks_total; s0 += 256) {
     const int s = s0 + tid;
     const int row = s / chunks_per_row;
     const int cl = s % chunks_per_row;
-    const int c = ((cl & 7) ^ (row & 7)) | (cl & ~7);
+    const int c = (cl ^ row) & cmask;
     const ushort* g = src + (long long)row * ld + c * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)g,
@@ -54,7 +60,7 @@ __device__ __forceinline__ void v2_stage(const ushort* __restrict__ src,
 
 __device__ __forceinline__ short8 v2_read(const ushort* lds_tile, int row,
                                           int c, int chunks_per_row) {
-  const int phys = ((c & 7) ^ (row & 7)) | (c & ~7);
+  const int phys = (c ^ row) & (chunks_per_row - 1);
   return *reinterpret_cast<const short8*>(lds_tile + (row * chunks_per_row + phys) * 8);
 }
 
@@ -238,13 +244,14 @@ __device__ __forceinline__ void v3_stage(const ushort* __restrict__ src,
                                          long long ld, int chunks_per_row,
                                          ushort* lds_tile, int chunks_total,
                                          int tid) {
+  const int cmask = chunks_per_row - 1;
   const int wave_chunk = tid & ~63;
   for (int s0 = 0; s0 < chunks_total; s0 += 512) {
     const int s = s0 + tid;
     if (s >= chunks_total) break;
     const int row = s / chunks_per_row;
     const int cl = s % chunks_per_row;
-    const int c = ((cl & 7) ^ (row & 7)) | (cl & ~7);
+    const int c = (cl ^ row) & cmask;
     const ushort* g = src + (long long)row * ld + c * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)g,
