@@ -284,9 +284,11 @@ attn_fwd_v3_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
   const int q_lane = q0 + l31;
   const bool live = q_lane < S;
 
-  __shared__ __attribute__((aligned(16))) ushort smem[V2_KVBLK * 128 + 128 * V2_KVBLK];
-  ushort* k_lds = smem;
-  ushort* vt_lds = smem + V2_KVBLK * 128;
+  // double-buffered K/V tiles: 2 x 32 KiB — at 2 blocks/CU (VGPR-limited)
+  // this still fits 160 KiB LDS, so the prefetch of tile t+1 overlaps the
+  // MFMA work of tile t for free (v2/v3 single-buffer relied on occupancy
+  // alone; PMC showed 3.99e9 parked cycles at the bench shape)
+  __shared__ __attribute__((aligned(16))) ushort smem[2][V2_KVBLK * 128 + 128 * V2_KVBLK];
 
   short8 qf[8];
   {
@@ -304,10 +306,19 @@ attn_fwd_v3_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
     for (int r = 0; r < 16; ++r) o_acc[db][r] = 0.f;
 
   const int kv_end = min(S, qb * V3_QBLK + V3_QBLK);
+  v3_stage(Kh, D, 16, smem[0], V2_KVBLK * 16, tid);
+  v3_stage(VTh, S, 8, smem[0] + V2_KVBLK * 128, 128 * 8, tid);
+  __syncthreads();
+  int buf = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += V2_KVBLK) {
-    v3_stage(Kh + (long long)kv0 * D, D, 16, k_lds, V2_KVBLK * 16, tid);
-    v3_stage(VTh + kv0, S, 8, vt_lds, 128 * 8, tid);
-    __syncthreads();
+    if (kv0 + V2_KVBLK < kv_end) {
+      v3_stage(Kh + (long long)(kv0 + V2_KVBLK) * D, D, 16, smem[buf ^ 1],
+               V2_KVBLK * 16, tid);
+      v3_stage(VTh + kv0 + V2_KVBLK, S, 8, smem[buf ^ 1] + V2_KVBLK * 128,
+               128 * 8, tid);
+    }
+    const ushort* k_lds = smem[buf];
+    const ushort* vt_lds = smem[buf] + V2_KVBLK * 128;
 
     __builtin_amdgcn_s_setprio(1);
     f32x16 st[2];
@@ -391,6 +402,7 @@ attn_fwd_v3_kernel(const ushort* __restrict__ Q, const ushort* __restrict__ K,
       __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();
+    buf ^= 1;
   }
 
   if (!live) return;
