@@ -67,7 +67,7 @@ def bench_fp8():
 
 def bench_attn():
     dev = torch.device("cuda:0")
-    for (B, H, Hk, S) in [(4, 32, 8, 2048), (1, 32, 8, 8192)]:
+    for (B, H, Hk, S) in [(4, 32, 8, 2048), (2, 32, 8, 4096), (1, 32, 8, 8192)]:
         D = 128
         q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=dev)
         k = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
