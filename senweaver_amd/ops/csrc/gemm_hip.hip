@@ -305,3 +305,102 @@ GEMV2_INST(2)
 GEMV2_INST(4)
 GEMV2_INST(8)
 GEMV2_INST(16)
+
+// ---------------------------------------------------------------------------
+// 256x256-tile on v_mfma_f32_32x32x16_bf16: identical FLOPs, LDS traffic and
+// geometry to gemm_bt_bf16_256_kernel (8 waves, wave tile 128x64, BK=64,
+// glds dbuf, one barrier/K-tile) but HALF the MFMA instruction count (32
+// 32x32x16 per K-tile vs 64 16x16x32) — attacking the issue-stall share the
+// PMC log attributes to the 16x16 version.  Fragment maps follow the
+// probed/validated attention + MX-kernel layouts: A/B lane row = base+l31,
+// k = step*16 + lhi*8 + e; C col = lane&31 (N), row = (r&3)+8*(r>>2)+4*lhi.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16g;
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_256x32_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / 256) * (N / 256);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / 256;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;  // 0..1
+  const int wn = wid & 3;   // 0..3
+  const int l31 = lane & 31;
+  const int lhi = lane >> 5;
+
+  __shared__ __attribute__((aligned(16))) ushort lds[2][2][256 * 64];
+
+  const ushort* Atile = A + (long long)tile_m * 256 * K;
+  const ushort* Btile = B + (long long)tile_n * 256 * K;
+
+  f32x16g acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[i][j][r] = 0.f;
+
+  const int m_base = wm * 128;
+  const int n_base = wn * 64;
+
+  const int ntiles = K / BK;
+  stage_tile_glds_512(Atile, K, lds[0][0], tid);
+  stage_tile_glds_512(Btile, K, lds[0][1], tid);
+  __syncthreads();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      stage_tile_glds_512(Atile + (long long)(t + 1) * BK, K, lds[buf ^ 1][0], tid);
+      stage_tile_glds_512(Btile + (long long)(t + 1) * BK, K, lds[buf ^ 1][1], tid);
+    }
+    const ushort* Al = lds[buf][0];
+    const ushort* Bl = lds[buf][1];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int stp = 0; stp < 4; ++stp) {  // BK=64 in 4 K=16 steps
+      short8 af[4], bf[2];
+      const int c = stp * 2 + lhi;
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        af[mi] = read_frag(Al, m_base + mi * 32 + l31, c);
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        bf[ni] = read_frag(Bl, n_base + ni * 32 + l31, c);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const long long c_col0 = (long long)tile_n * 256 + n_base + l31;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long long row = (long long)tile_m * 256 + m_base + mi * 32
+                            + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        crow[c_col0 + ni * 32] = f2bf(acc[mi][ni][r]);
+    }
+}
