@@ -19,6 +19,7 @@ extern "C" __global__ void target_logprob_kernel(const ushort*, const int*, floa
 extern "C" __global__ void gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256x32_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_256sg_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
 extern "C" __global__ void gemm_bt_fp8_kernel(const unsigned char*, const float*, const unsigned char*, const float*, ushort*, int, int, int);
@@ -432,6 +433,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     HIP_CHECK_KERNEL();
     return C;
   }, "256-tile GEMM on 32x32x16 MFMA (A/B experiment)");
+  m.def("gemm_bt_256sg", [](torch::Tensor a, torch::Tensor b) {
+    check_bf16(a, "a"); check_bf16(b, "b");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0);
+    auto C = torch::empty({M, N}, a.options());
+    gemm_bt_bf16_256sg_kernel<<<dim3((M / 256) * (N / 256)), dim3(512), 0, cur_stream()>>>(
+        bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K);
+    HIP_CHECK_KERNEL();
+    return C;
+  }, "256-tile GEMM with sched_group_barrier interleave (A/B experiment)");
   m.def("grouped_gemm_bt", &grouped_gemm_bt, "segment-grouped C = A @ W[e]^T (MoE)");
   m.def("quant_fp8", &quant_fp8, "row-wise bf16 -> e4m3 + scale");
   m.def("gemm_bt_fp8", &gemm_bt_fp8, "fp8 MFMA GEMM with row/col rescale");

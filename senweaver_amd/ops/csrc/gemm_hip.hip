@@ -404,3 +404,103 @@ gemm_bt_bf16_256x32_kernel(const ushort* __restrict__ A, const ushort* __restric
         crow[c_col0 + ni * 32] = f2bf(acc[mi][ni][r]);
     }
 }
+
+// ---------------------------------------------------------------------------
+// A/B experiment: 256-tile 16x16x32 kernel with explicit scheduling groups —
+// __builtin_amdgcn_sched_group_barrier interleaves each MFMA with a DS read
+// so fragment-load latency hides under matrix ops instead of clustering.
+// SCHED_GROUP_BARRIER(mask, size, syncID): mask 0x8 = MFMA, 0x100 = DS read.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_256sg_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                          ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / 256) * (N / 256);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / 256;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;
+  const int wn = wid & 3;
+
+  __shared__ __attribute__((aligned(16))) ushort lds[2][2][256 * 64];
+
+  const ushort* Atile = A + (long long)tile_m * 256 * K;
+  const ushort* Btile = B + (long long)tile_n * 256 * K;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int m_base = wm * 128;
+  const int n_base = wn * 64;
+  const int frag_row = lane & 15;
+  const int frag_kgrp = lane >> 4;
+
+  const int ntiles = K / BK;
+  stage_tile_glds_512(Atile, K, lds[0][0], tid);
+  stage_tile_glds_512(Btile, K, lds[0][1], tid);
+  __syncthreads();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      stage_tile_glds_512(Atile + (long long)(t + 1) * BK, K, lds[buf ^ 1][0], tid);
+      stage_tile_glds_512(Btile + (long long)(t + 1) * BK, K, lds[buf ^ 1][1], tid);
+    }
+    const ushort* Al = lds[buf][0];
+    const ushort* Bl = lds[buf][1];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      short8 af[8], bf[4];
+      const int c = kk * 4 + frag_kgrp;
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+        af[mi] = read_frag(Al, m_base + mi * 16 + frag_row, c);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bf[ni] = read_frag(Bl, n_base + ni * 16 + frag_row, c);
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      // interleave: 12 DS reads paired through the 32 MFMAs of this step
+#pragma unroll
+      for (int g = 0; g < 12; ++g) {
+        __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);  // 1 DS read
+        __builtin_amdgcn_sched_group_barrier(0x008, 2, 0);  // 2 MFMAs
+      }
+      __builtin_amdgcn_sched_group_barrier(0x008, 8, 0);    // trailing MFMAs
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const long long c_row0 = (long long)tile_m * 256 + m_base + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * 256 + n_base + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}
