@@ -243,18 +243,29 @@ class LlamaModel:
         sorted_token = flat_token[order]
         sorted_weight = topw.reshape(-1)[order]
         counts = torch.bincount(flat_expert, minlength=c.num_experts)
-        seg_starts = [0]
-        for e in range(c.num_experts):
-            seg_starts.append(seg_starts[-1] + int(counts[e]))
         Tk = T * k
-        pad = 128  # last-tile overread margin for the grouped kernel
-        a_sorted = torch.zeros(Tk + pad, c.hidden_size, dtype=h.dtype, device=h.device)
-        a_sorted[:Tk] = h[sorted_token]
-        gateup = ops.grouped_gemm_bt(a_sorted, L["w13"], seg_starts)
-        act = ops.swiglu(gateup[:Tk])
-        act_p = torch.zeros(Tk + pad, act.shape[1], dtype=h.dtype, device=h.device)
-        act_p[:Tk] = act
-        down = ops.grouped_gemm_bt(act_p, L["w2"], seg_starts)[:Tk]
+        if h.is_cuda:
+            # sync-free routing: device-side cumulative ends feed the grouped
+            # GEMM directly (the per-expert int(counts[e]) reads were 8 tiny
+            # D2H syncs per MoE layer, visible as copyBuffer storms in the
+            # Mixtral profile and stalling the two-stream pipeline)
+            offs = torch.cumsum(counts, 0).to(torch.int32)
+            a_sorted = h[sorted_token]
+            gateup = ops.grouped_gemm_bt(a_sorted, L["w13"], offs)[:Tk]
+            act = ops.swiglu(gateup)
+            down = ops.grouped_gemm_bt(act, L["w2"], offs)[:Tk]
+        else:
+            seg_starts = [0]
+            for e in range(c.num_experts):
+                seg_starts.append(seg_starts[-1] + int(counts[e]))
+            pad = 128  # last-tile overread margin for the grouped kernel
+            a_sorted = torch.zeros(Tk + pad, c.hidden_size, dtype=h.dtype, device=h.device)
+            a_sorted[:Tk] = h[sorted_token]
+            gateup = ops.grouped_gemm_bt(a_sorted, L["w13"], seg_starts)
+            act = ops.swiglu(gateup[:Tk])
+            act_p = torch.zeros(Tk + pad, act.shape[1], dtype=h.dtype, device=h.device)
+            act_p[:Tk] = act
+            down = ops.grouped_gemm_bt(act_p, L["w2"], seg_starts)[:Tk]
         out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=h.device)
         out.index_add_(0, sorted_token, down.float() * sorted_weight.unsqueeze(1))
         res = out.to(h.dtype)
@@ -279,10 +290,11 @@ class LlamaModel:
         sorted_token = flat_token[order]
         sorted_weight = topw.reshape(-1)[order]
         sorted_expert = flat_expert[order]
-        counts = torch.bincount(flat_expert, minlength=c.num_experts)
+        # one D2H transfer for all expert counts (16-aligned padding needs
+        # host-side sizes; the bf16 path is fully sync-free)
+        counts = torch.bincount(flat_expert, minlength=c.num_experts).cpu().tolist()
         seg_starts, pad_starts = [0], [0]
-        for e in range(c.num_experts):
-            n = int(counts[e])
+        for n in counts:
             seg_starts.append(seg_starts[-1] + n)
             pad_starts.append(pad_starts[-1] + ((n + 15) // 16) * 16)
         Tk = T * k

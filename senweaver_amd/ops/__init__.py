@@ -237,13 +237,22 @@ def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
                     ) -> torch.Tensor:
     """Segment-grouped C = A_seg @ W[e]^T for MoE.
 
-    ``a_sorted``: [T_pad, K] tokens sorted by expert (padded by >=128 rows);
-    ``w``: [E, N, K]; ``seg_starts_cpu``: list/1D tensor of E+1 cumulative
-    row offsets (host side).  Returns at least seg_starts[-1] rows; rows
-    past the last segment (if any) are undefined.
+    ``a_sorted``: tokens sorted by expert; ``w``: [E, N, K].
+    ``seg_starts_cpu`` is either a host list/1D tensor of E+1 cumulative row
+    offsets, or a DEVICE int32 tensor of E cumulative segment ENDS — the
+    device form keeps the whole routed FFN sync-free on the hipBLASLt
+    grouped path (torch._grouped_mm takes device offsets directly; the
+    per-expert ``int(counts[e])`` host reads were 8 tiny D2H syncs per MoE
+    layer).  With host offsets, ``a_sorted`` may carry >=128 pad rows for
+    the in-repo kernel's tile overread; with device offsets it must be
+    exactly the routed rows.  Returns at least the routed rows; rows past
+    the last segment (if any) are undefined.
     """
     E = w.shape[0]
-    starts = [int(x) for x in seg_starts_cpu]
+    dev_offs = (isinstance(seg_starts_cpu, torch.Tensor)
+                and seg_starts_cpu.is_cuda)
+    if not dev_offs:
+        starts = [int(x) for x in seg_starts_cpu]
     if not _on_gpu(a_sorted):
         out = torch.zeros(a_sorted.shape[0], w.shape[1], dtype=a_sorted.dtype)
         for e in range(E):
@@ -257,11 +266,18 @@ def grouped_gemm_bt(a_sorted: torch.Tensor, w: torch.Tensor, seg_starts_cpu,
         # shapes, profiles/r01_gemm_dispatch.txt); our kernel remains the
         # fallback for offsets/shapes the library rejects
         try:
+            if dev_offs:
+                return torch._grouped_mm(a_sorted, w.transpose(1, 2),
+                                         offs=seg_starts_cpu)
             return torch._grouped_mm(
                 a_sorted[:starts[-1]], w.transpose(1, 2),
                 offs=torch.tensor(starts[1:], dtype=torch.int32, device=dev))
         except RuntimeError:
             pass
+    if dev_offs:
+        # fallback needs host offsets (tile map is host-built) + pad rows
+        starts = [0] + seg_starts_cpu.cpu().tolist()
+        a_sorted = torch.nn.functional.pad(a_sorted, (0, 0, 0, 128))
     tile_expert, tile_m0 = [], []
     for e in range(E):
         s, t = starts[e], starts[e + 1]
