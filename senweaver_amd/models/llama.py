@@ -254,6 +254,13 @@ class LlamaModel:
             gateup = ops.grouped_gemm_bt(a_sorted, L["w13"], offs)[:Tk]
             act = ops.swiglu(gateup)
             down = ops.grouped_gemm_bt(act, L["w2"], offs)[:Tk]
+            # fused weighted combine: inverse permutation makes each output
+            # row a private k-way sum (replaces zeros + f32 index_add + cast)
+            inv = torch.argsort(order).to(torch.int32).reshape(T, k)
+            res = ops.hip_ext().moe_combine(down.contiguous(), inv,
+                                            sorted_weight.float().contiguous())
+            self.tp.all_reduce_(res)
+            return res
         else:
             seg_starts = [0]
             for e in range(c.num_experts):

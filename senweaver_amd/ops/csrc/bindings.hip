@@ -21,6 +21,7 @@ extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*,
 extern "C" __global__ void gemm_bt_bf16_256x32_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256sg_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
+extern "C" __global__ void moe_combine_kernel(const ushort*, const int*, const float*, ushort*, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
 extern "C" __global__ void gemm_bt_fp8_kernel(const unsigned char*, const float*, const unsigned char*, const float*, ushort*, int, int, int);
 extern "C" __global__ void quant_mxfp8_kernel(const ushort*, unsigned char*, unsigned char*, int);
@@ -444,6 +445,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return C;
   }, "256-tile GEMM with sched_group_barrier interleave (A/B experiment)");
   m.def("grouped_gemm_bt", &grouped_gemm_bt, "segment-grouped C = A @ W[e]^T (MoE)");
+  m.def("moe_combine", [](torch::Tensor down, torch::Tensor pos, torch::Tensor weight) {
+    check_bf16(down, "down");
+    TORCH_CHECK(pos.scalar_type() == torch::kInt32 && pos.dim() == 2);
+    TORCH_CHECK(weight.scalar_type() == torch::kFloat32);
+    const long long T = pos.size(0);
+    const int k = pos.size(1), H = down.size(1);
+    TORCH_CHECK(H % 8 == 0);
+    auto out = torch::empty({T, (long)H}, down.options());
+    moe_combine_kernel<<<dim3((unsigned)T), dim3(256), 0, cur_stream()>>>(
+        bf16_ptr(down), pos.data_ptr<int>(), weight.data_ptr<float>(),
+        bf16_mut(out), H, k);
+    HIP_CHECK_KERNEL();
+    return out;
+  }, "routed-FFN weighted combine via inverse permutation");
   m.def("quant_fp8", &quant_fp8, "row-wise bf16 -> e4m3 + scale");
   m.def("gemm_bt_fp8", &gemm_bt_fp8, "fp8 MFMA GEMM with row/col rescale");
   m.def("quant_mxfp8", &quant_mxfp8, "OCP MX quant: bf16 -> e4m3 + e8m0 per-32 scales");

@@ -392,3 +392,21 @@ def test_fp8_moe_forward_gpu(dev):
     assert torch.isfinite(h8.float()).all()
     rel = ((h16.float() - h8.float()).norm() / h16.float().norm()).item()
     assert rel < 0.25
+
+
+def test_moe_combine_gpu(dev):
+    torch.manual_seed(12)
+    T, k, H = 256, 2, 512
+    Tk = T * k
+    down = torch.randn(Tk, H, dtype=torch.bfloat16, device=dev)
+    # a routing permutation: positions of each token's k entries
+    perm = torch.randperm(Tk, device=dev)
+    inv = torch.argsort(perm).to(torch.int32).reshape(T, k)
+    w = torch.rand(Tk, device=dev)
+    out = ops.hip_ext().moe_combine(down, inv, w)
+    ref_out = torch.zeros(T, H, dtype=torch.float32, device=dev)
+    tok_of_pos = perm // k  # position p holds token perm[p]//k? no — build from inv
+    for j in range(k):
+        pos = inv[:, j].long()
+        ref_out += w[pos].unsqueeze(1) * down[pos].float()
+    torch.testing.assert_close(out.float(), ref_out, atol=5e-2, rtol=2e-2)
