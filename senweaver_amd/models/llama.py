@@ -317,9 +317,16 @@ class LlamaModel:
         act = ops.swiglu(gateup)
         aq2, as2 = ops.quant_fp8(act)
         down = ops.grouped_gemm_bt_fp8(aq2, as2, L["w2_q"], L["w2_s"], pad_starts)
-        out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=dev)
-        out.index_add_(0, sorted_token, down[dest].float() * sorted_weight.unsqueeze(1))
-        res = out.to(h.dtype)
+        if h.is_cuda:
+            # fused weighted combine over the PADDED layout (see _moe_ffn)
+            inv_pad = dest[torch.argsort(order)].to(torch.int32).reshape(T, k)
+            w_pad = torch.zeros(pad_starts[-1], dtype=torch.float32, device=dev)
+            w_pad[dest] = sorted_weight.float()
+            res = ops.hip_ext().moe_combine(down.contiguous(), inv_pad, w_pad)
+        else:
+            out = torch.zeros(T, c.hidden_size, dtype=torch.float32, device=dev)
+            out.index_add_(0, sorted_token, down[dest].float() * sorted_weight.unsqueeze(1))
+            res = out.to(h.dtype)
         self.tp.all_reduce_(res)  # row-parallel down partial sum (TP)
         return res
 
