@@ -9,7 +9,7 @@ Capability-compatible with the reference's static registry
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Union
 
 

@@ -14,8 +14,8 @@ from __future__ import annotations
 import threading
 import time
 import uuid
-from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional
+from dataclasses import dataclass
+from typing import Any, Dict, List, Optional
 
 MAX_PARALLEL_SUBAGENTS = 8
 MAX_SUBAGENT_DEPTH = 4

@@ -11,7 +11,7 @@ from __future__ import annotations
 import math
 import re
 import time
-from typing import Dict, Optional
+from typing import Dict
 
 CHAT_RETRIES = 5
 BASE_RETRY_DELAY_MS = 3000

@@ -10,7 +10,7 @@ replay: token id, position, KV slot, block table, context length.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import torch
 

@@ -8,7 +8,7 @@ and the decode kernel reads a position's per-head row as one contiguous
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 

@@ -16,7 +16,6 @@ negative-reward ones score higher.
 
 from __future__ import annotations
 
-import math
 from typing import List, Optional, Sequence, Tuple
 
 import torch

@@ -19,7 +19,6 @@ gloo, bit-matching is not required there).
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 import torch.distributed as dist

@@ -14,7 +14,7 @@ import subprocess
 import threading
 import time
 import uuid
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
 DAEMON_BIN = os.path.join(_HERE, "_bin", "senweaver_daemon")

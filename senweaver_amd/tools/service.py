@@ -20,7 +20,7 @@ import threading
 import time
 import uuid
 from dataclasses import dataclass
-from typing import Any, Callable, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 from .registry import APPROVAL_TYPE_OF_TOOL, BUILTIN_TOOLS, is_builtin_tool
 

@@ -19,7 +19,6 @@ import uuid
 from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional, Tuple
 
-from ..tools.registry import BUILTIN_TOOLS
 
 OnText = Callable[..., None]  # kwargs: full_text, full_reasoning, tool_call
 OnFinal = Callable[..., None]

@@ -22,9 +22,9 @@ from __future__ import annotations
 import threading
 import uuid
 from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
-from .grammar import RawToolCall, ReasoningExtractor, XMLToolExtractor
+from .grammar import ReasoningExtractor, XMLToolExtractor
 
 
 @dataclass

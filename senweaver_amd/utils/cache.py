@@ -4,7 +4,7 @@ from __future__ import annotations
 
 import time
 from collections import OrderedDict
-from typing import Any, Generic, Hashable, Optional, TypeVar
+from typing import Generic, Hashable, Optional, TypeVar
 
 K = TypeVar("K", bound=Hashable)
 V = TypeVar("V")
