@@ -247,3 +247,38 @@ def test_full_rl_loop_feedback_to_injected_rules(tmp_path):
     sysmsg = conv.generate_system_message("agent")
     assert "# APO Optimized Rules" in sysmsg
     assert rules[0].strip().lstrip("- ")[:30] in sysmsg
+
+
+def test_multi_agent_traced_turn(tmp_path):
+    """SURVEY §3.4 / BASELINE config 4: a main-agent turn that spawns a
+    subagent — the spawn runs through the tool path with depth/limit checks
+    and lands in the SAME trace as a tool_call span."""
+    from senweaver_amd.agents.subagents import SubagentRunner
+    from senweaver_amd.tools import ToolsService
+    from senweaver_amd.transport import LLMMessageService
+
+    main_responses = [
+        "Delegating. <spawn_subagent><label>research</label>"
+        "<task_prompt>summarize hello.txt</task_prompt></spawn_subagent>",
+        "Subagent reported back; done.",
+    ]
+    sub_responses = ["the file greets the world", "summary: a greeting"]
+    (tmp_path / "hello.txt").write_text("hello world\n")
+
+    main_backend = ScriptedBackend(main_responses)
+    sub_backend = ScriptedBackend(sub_responses)
+    runner = SubagentRunner(LLMMessageService(sub_backend))
+    tools = ToolsService(str(tmp_path), subagent_runner=runner)
+    tc = TraceCollector(storage=MemoryStorage())
+    svc = ChatThreadService(LLMMessageService(main_backend), tools, tc,
+                            settings=GlobalSettings(auto_approve={}),
+                            sleep=lambda s: None)
+    thread = svc.open_thread()
+    svc.add_user_message_and_stream_response(thread.id, "delegate the summary")
+    tool_msg = next(m for m in thread.messages if m.role == "tool")
+    assert tool_msg.tool_success is True
+
+    trace = tc.get_all_traces()[0]
+    tool_spans = [s for s in trace.spans if s.type == "tool_call"]
+    assert any(s.data.get("toolName") == "spawn_subagent" for s in tool_spans)
+    assert trace.summary.total_tool_calls == 1
