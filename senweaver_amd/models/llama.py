@@ -45,8 +45,6 @@ class LlamaModel:
         self.local_kv_size = self.local_kv_heads * c.head_dim
         self.local_inter = c.intermediate_size // tpw
         assert quant in ("bf16", "fp8", "mxfp8")
-        if quant == "mxfp8":
-            assert c.num_experts == 0, "mxfp8 MoE not supported yet"
         self.quant = quant
         if self.device.type == "cuda":
             # Device-side init: seconds for 8B instead of minutes of CPU RNG +
@@ -126,12 +124,13 @@ class LlamaModel:
                     L[name + "_q"], L[name + "_s"] = q, s
                     del L[name]
                 if c.num_experts > 0:
-                    # per-expert rowwise quant (router stays bf16 — tiny)
+                    # per-expert quant (router stays bf16 — tiny)
                     for name in ("w13", "w2"):
                         E, N, K = L[name].shape
-                        q, s = ops.quant_fp8(L[name])
+                        q, s = qfn(L[name])
                         L[name + "_q"] = q.reshape(E, N, K)
-                        L[name + "_s"] = s.reshape(E, N)
+                        L[name + "_s"] = (s.reshape(E, N) if self.quant == "fp8"
+                                          else s.reshape(E, N, K // 32))
                         del L[name]
             self.lm_head_q, self.lm_head_s = qfn(self.lm_head)
         self.cos_sin = ops.rope_tables(c.max_position, c.head_dim, c.rope_theta).to(self.device)
@@ -229,7 +228,7 @@ class LlamaModel:
         c = self.config
         T = h.shape[0]
         k = c.num_experts_per_tok
-        if self.quant == "fp8":
+        if self.quant in ("fp8", "mxfp8"):
             return self._moe_ffn_fp8(h, L)
         # router is [E=8, H]: far below the MFMA tile (N=8) — a plain skinny
         # library matmul, not a hot op
@@ -312,11 +311,14 @@ class LlamaModel:
         dest = pad_t[sorted_expert] + (torch.arange(Tk, device=dev) - seg_t[sorted_expert])
         a_pad = torch.zeros(pad_starts[-1], c.hidden_size, dtype=h.dtype, device=dev)
         a_pad[dest] = h[sorted_token]
-        aq, a_s = ops.quant_fp8(a_pad)
-        gateup = ops.grouped_gemm_bt_fp8(aq, a_s, L["w13_q"], L["w13_s"], pad_starts)
+        qfn = ops.quant_fp8 if self.quant == "fp8" else ops.quant_mxfp8
+        gfn = (ops.grouped_gemm_bt_fp8 if self.quant == "fp8"
+               else ops.grouped_gemm_bt_mxfp8)
+        aq, a_s = qfn(a_pad)
+        gateup = gfn(aq, a_s, L["w13_q"], L["w13_s"], pad_starts)
         act = ops.swiglu(gateup)
-        aq2, as2 = ops.quant_fp8(act)
-        down = ops.grouped_gemm_bt_fp8(aq2, as2, L["w2_q"], L["w2_s"], pad_starts)
+        aq2, as2 = qfn(act)
+        down = gfn(aq2, as2, L["w2_q"], L["w2_s"], pad_starts)
         if h.is_cuda:
             # fused weighted combine over the PADDED layout (see _moe_ffn)
             inv_pad = dest[torch.argsort(order)].to(torch.int32).reshape(T, k)

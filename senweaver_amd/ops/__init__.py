@@ -325,6 +325,27 @@ def grouped_gemm_bt_fp8(a_q, a_s, w_q, w_s, seg_starts_cpu) -> torch.Tensor:
     return out
 
 
+def grouped_gemm_bt_mxfp8(a_q, a_s, w_q, w_s, seg_starts_cpu) -> torch.Tensor:
+    """Per-expert MX block-scaled fp8 GEMM over expert-sorted segments.
+
+    ``w_q`` [E,N,K] u8 + ``w_s`` [E,N,K//32] e8m0 bytes; activations
+    rowwise-MX-quantized.  One gemm_bt_mxfp8 per non-empty expert (the
+    32x32x64 scaled MFMA applies the scales in hardware).
+    """
+    E, N = w_q.shape[0], w_q.shape[1]
+    starts = [int(x) for x in seg_starts_cpu]
+    Tk = starts[-1]
+    if not _on_gpu(a_q):
+        out = torch.zeros(Tk, N, dtype=torch.bfloat16)
+    else:
+        out = torch.empty(Tk, N, dtype=torch.bfloat16, device=a_q.device)
+    for e in range(E):
+        s, t = starts[e], starts[e + 1]
+        if t > s:
+            out[s:t] = gemm_bt_mxfp8(a_q[s:t], a_s[s:t], w_q[e], w_s[e])
+    return out
+
+
 def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional[float] = None,
              vt: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Causal GQA attention. q [B,H,S,D], k/v [B,Hk,S,D] (S % 64 == 0 on GPU).

@@ -163,3 +163,16 @@ def test_grouped_gemm_fp8_segments():
         if t > s:
             r = ops.reference.gemm_bt_fp8_ref(aq[s:t], asc[s:t], wq[e], ws[e])
             torch.testing.assert_close(out[s:t].float(), r.float(), atol=0.3, rtol=3e-2)
+
+
+def test_mxfp8_moe_forward_close_to_bf16():
+    from senweaver_amd.models.config import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+    m16 = LlamaModel(tiny_moe(), device="cpu", seed=4)
+    m8 = LlamaModel(tiny_moe(), device="cpu", seed=4, quant="mxfp8")
+    toks = torch.randint(0, 512, (2, 16))
+    h16 = m16.prefill(toks)
+    h8 = m8.prefill(toks)
+    assert torch.isfinite(h8.float()).all()
+    rel = (h16.float() - h8.float()).norm() / h16.float().norm()
+    assert rel.item() < 0.2

@@ -410,3 +410,16 @@ def test_moe_combine_gpu(dev):
         pos = inv[:, j].long()
         ref_out += w[pos].unsqueeze(1) * down[pos].float()
     torch.testing.assert_close(out.float(), ref_out, atol=5e-2, rtol=2e-2)
+
+
+def test_mxfp8_moe_forward_gpu(dev):
+    from senweaver_amd.models import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+    m16 = LlamaModel(tiny_moe(), device=dev, seed=4)
+    m8 = LlamaModel(tiny_moe(), device=dev, seed=4, quant="mxfp8")
+    toks = torch.randint(0, 512, (2, 64), device=dev)
+    h16 = m16.prefill(toks)
+    h8 = m8.prefill(toks)
+    assert torch.isfinite(h8.float()).all()
+    rel = ((h16.float() - h8.float()).norm() / h16.float().norm()).item()
+    assert rel < 0.2
