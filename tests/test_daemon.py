@@ -32,19 +32,19 @@ def daemon():
     else:
         proc.kill()
         pytest.fail("daemon socket never appeared")
-    yield sock
+    yield sock, proc.pid
     proc.terminate()
     proc.wait(timeout=10)
 
 
 def test_daemon_ping(daemon):
-    c = DaemonClient(daemon)
+    c = DaemonClient(daemon[0])
     assert c.ping(timeout=120)  # first ping waits for engine import/ready
     c.close()
 
 
 def test_daemon_stream_and_final(daemon):
-    c = DaemonClient(daemon)
+    c = DaemonClient(daemon[0])
     assert c.ping(timeout=120)
     texts = []
     done = threading.Event()
@@ -64,7 +64,7 @@ def test_daemon_stream_and_final(daemon):
 
 
 def test_daemon_abort(daemon):
-    c = DaemonClient(daemon)
+    c = DaemonClient(daemon[0])
     assert c.ping(timeout=120)
     rid = c.send_llm_message([{"role": "user", "content": "long task"}],
                              max_new_tokens=64)
@@ -120,3 +120,36 @@ def test_performance_monitor_slo():
     with pm.timer("systemMessageGeneration"):
         pass
     assert pm.summary()["systemMessageGeneration"]["count"] == 1
+
+
+def test_daemon_worker_crash_restart(daemon):
+    """Kill the engine worker mid-session: the daemon detects the exit,
+    respawns it (daemon.cpp waitpid/restart path), and subsequent requests
+    are served by the fresh worker."""
+    import signal
+
+    sock, daemon_pid = daemon
+    c = DaemonClient(sock)
+    assert c.ping(timeout=120)
+    # the worker is THE child of OUR daemon process (exact pid, no patterns)
+    out = subprocess.run(["pgrep", "-P", str(daemon_pid)], capture_output=True,
+                         text=True)
+    pids = [int(x) for x in out.stdout.split()]
+    assert pids, "engine worker child process not found"
+    os.kill(pids[0], signal.SIGKILL)
+
+    done = threading.Event()
+    texts = []
+    deadline = time.time() + 180
+    while time.time() < deadline:
+        done.clear()
+        texts.clear()
+        c.send_llm_message([{"role": "user", "content": "after restart"}],
+                           on_final=lambda m: (texts.append(m["fullText"]), done.set()),
+                           on_error=lambda m: done.set(),
+                           max_new_tokens=4)
+        if done.wait(timeout=60) and texts:
+            break
+        time.sleep(1.0)
+    assert texts, "no successful response after worker restart"
+    c.close()
