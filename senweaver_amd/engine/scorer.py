@@ -22,6 +22,7 @@ import torch
 
 from .. import ops
 from ..apo.optimizer import rollout_weight
+from ..utils.observability import trace_range
 from ..apo.schema import RolloutResult
 from ..models.config import ModelConfig, get_config
 from ..models.llama import LlamaModel
@@ -145,27 +146,31 @@ class LlamaBackend:
         Returns f32 [B] on the engine device (no host sync).
         """
         B, S = tokens.shape
-        # mask processing FIRST: nonzero() host-syncs its stream, so doing it
-        # before the prefill is enqueued keeps the two-stream pipeline 2-deep
-        # (the sync only waits on work from two microbatches back)
-        m = mask.clone()
-        m[:, 0] = False  # position 0 has no predictor
-        flat_pos = m.reshape(-1).nonzero(as_tuple=False).squeeze(1)  # b*S + p
-        if flat_pos.numel() == 0:
-            return torch.zeros(B, dtype=torch.float32, device=tokens.device)
-        hidden = self.model.prefill(tokens)  # [B, S, H]
-        # p >= 1 within every sequence (position 0 masked off above), so
-        # flat_pos - 1 stays inside the same sequence's rows
-        rows = hidden.reshape(B * S, -1)[flat_pos - 1]
-        logits = self.model.logits(rows)
-        targets = tokens.reshape(-1)[flat_pos].to(torch.int32)
-        lp = ops.target_logprob(logits, targets)
-        seq_of = torch.div(flat_pos, S, rounding_mode="floor")
-        sums = torch.zeros(B, dtype=torch.float32, device=lp.device)
-        counts = torch.zeros(B, dtype=torch.float32, device=lp.device)
-        sums.index_add_(0, seq_of, lp.float())
-        counts.index_add_(0, seq_of, torch.ones_like(lp, dtype=torch.float32))
-        return sums / counts.clamp(min=1)
+        # trace_range = rocTX/NVTX span: shows up in rocprofv3 timelines
+        # alongside the kernels (SURVEY §5.1's perf-span analog)
+        with trace_range(f"score_b{B}_s{S}"):
+            # mask processing FIRST: nonzero() host-syncs its stream, so
+            # doing it before the prefill is enqueued keeps the two-stream
+            # pipeline 2-deep (the sync only waits on work from two
+            # microbatches back)
+            m = mask.clone()
+            m[:, 0] = False  # position 0 has no predictor
+            flat_pos = m.reshape(-1).nonzero(as_tuple=False).squeeze(1)
+            if flat_pos.numel() == 0:
+                return torch.zeros(B, dtype=torch.float32, device=tokens.device)
+            hidden = self.model.prefill(tokens)  # [B, S, H]
+            # p >= 1 within every sequence (position 0 masked off above), so
+            # flat_pos - 1 stays inside the same sequence's rows
+            rows = hidden.reshape(B * S, -1)[flat_pos - 1]
+            logits = self.model.logits(rows)
+            targets = tokens.reshape(-1)[flat_pos].to(torch.int32)
+            lp = ops.target_logprob(logits, targets)
+            seq_of = torch.div(flat_pos, S, rounding_mode="floor")
+            sums = torch.zeros(B, dtype=torch.float32, device=lp.device)
+            counts = torch.zeros(B, dtype=torch.float32, device=lp.device)
+            sums.index_add_(0, seq_of, lp.float())
+            counts.index_add_(0, seq_of, torch.ones_like(lp, dtype=torch.float32))
+            return sums / counts.clamp(min=1)
 
     # ------------------------------------------------------------------
     # PromptOptimizerBackend API
