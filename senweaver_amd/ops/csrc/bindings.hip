@@ -116,7 +116,8 @@ std::vector<torch::Tensor> rope_scatter_qkv(torch::Tensor qkv, torch::Tensor cos
   TORCH_CHECK(T == B * S && cos_sin.size(1) == D && ld >= (Hq + 2 * Hk) * D);
   auto qo = torch::empty({B, Hq, S, D}, qkv.options());
   auto ko = torch::empty({B, Hk, S, D}, qkv.options());
-  const int heads_per_blk = (int)(256 / (D / 2));
+  TORCH_CHECK((D / 2) % 8 == 0, "rope_scatter needs D/2 % 8 == 0");
+  const int heads_per_blk = (int)(256 / (D / 16));
   rope_scatter_kernel<<<dim3((unsigned)T, (unsigned)((Hq + Hk + heads_per_blk - 1) / heads_per_blk)),
                         dim3(256), 0, cur_stream()>>>(
       bf16_ptr(qkv), ld, bf16_mut(qo), bf16_mut(ko),
