@@ -134,7 +134,7 @@ torch::Tensor vt_from_qkv(torch::Tensor qkv, int64_t Hq, int64_t Hk, int64_t D,
   TORCH_CHECK(qkv.size(0) == B * S && ld >= (Hq + 2 * Hk) * D);
   auto vt = torch::empty({B, Hk, D, S}, qkv.options());
   const long long v_off = (Hq + Hk) * D;
- hipLaunchKernelGGL(( transpose_v_kernel), dim3((unsigned)((S + 31) / 32), (unsigned)((D + 31) / 32),
+ hipLaunchKernelGGL(( transpose_v_kernel), dim3((unsigned)((S + 31) / 32), (unsigned)((D + 63) / 64),
                            (unsigned)(B * Hk)),
                        dim3(dim3(256)), 0, cur_stream(), 
       bf16_ptr(qkv), ld, v_off, bf16_mut(vt), (int)B, (int)S, (int)Hk, (int)D);

@@ -206,28 +206,49 @@ add_bf16_kernel(const ushort* __restrict__ a, const ushort* __restrict__ b,
 extern "C" __global__ void __launch_bounds__(256)
 transpose_v_kernel(const ushort* __restrict__ qkv, long long ld, long long v_off,
                    ushort* __restrict__ vt, int B, int S, int Hk, int D) {
-  __shared__ ushort tile[32][33];  // +1 pad kills bank conflicts
+  // 16-B vector loads AND stores (the elementwise version's 2-B scalar
+  // stores ran ~0.9 TB/s).  32 s-rows x 64 d-cols per block keeps all 256
+  // threads busy in both phases; row pad of 8 (72-ushort stride) preserves
+  // vector alignment and spreads rows over bank groups.
+  __shared__ ushort tile[32][72];
   const int bh = blockIdx.z;       // b * Hk + hk
   const int b = bh / Hk;
   const int hk = bh % Hk;
   const int s0 = blockIdx.x * 32;
-  const int d0 = blockIdx.y * 32;
-  const int tx = threadIdx.x & 31;   // along d on read, along s on write
-  const int ty0 = threadIdx.x >> 5;  // 8 rows per pass
-
+  const int d0 = blockIdx.y * 64;
   const ushort* src = qkv + (long long)b * S * ld + v_off + (long long)hk * D;
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int s = s0 + ty0 + i * 8;
-    if (s < S && d0 + tx < D)
-      tile[ty0 + i * 8][tx] = src[(long long)s * ld + d0 + tx];
+  {
+    const int row = threadIdx.x >> 3;       // 0..31 along s
+    const int sub = threadIdx.x & 7;        // 8 chunks of 8 along d
+    const int s = s0 + row;
+    const int d = d0 + sub * 8;
+    if (s < S && d < D) {
+      if (d + 7 < D) {
+        *reinterpret_cast<bf16x8*>(&tile[row][sub * 8]) =
+            *reinterpret_cast<const bf16x8*>(src + (long long)s * ld + d);
+      } else {
+        for (int j = 0; j < 8 && d + j < D; ++j)
+          tile[row][sub * 8 + j] = src[(long long)s * ld + d + j];
+      }
+    }
   }
   __syncthreads();
   ushort* dst = vt + (((long long)b * Hk + hk) * D) * S;
+  {
+    const int row = threadIdx.x >> 2;       // 0..63 along d
+    const int sub = threadIdx.x & 3;        // 4 chunks of 8 along s
+    const int d = d0 + row;
+    const int s = s0 + sub * 8;
+    if (d < D && s < S) {
+      bf16x8 o;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int d = d0 + ty0 + i * 8;
-    if (d < D && s0 + tx < S)
-      dst[(long long)d * S + s0 + tx] = tile[tx][ty0 + i * 8];
+      for (int j = 0; j < 8; ++j) o.v[j] = tile[sub * 8 + j][row];
+      if (s + 7 < S) {
+        *reinterpret_cast<bf16x8*>(dst + (long long)d * S + s) = o;
+      } else {
+        for (int j = 0; j < 8 && s + j < S; ++j)
+          dst[(long long)d * S + s + j] = o.v[j];
+      }
+    }
   }
 }
