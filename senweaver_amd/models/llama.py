@@ -21,7 +21,9 @@ import torch
 from .. import ops
 from .config import ModelConfig
 
-if False:  # type-checking only; avoids the engine<->models import cycle
+from typing import TYPE_CHECKING
+
+if TYPE_CHECKING:  # avoids the engine<->models import cycle
     from ..engine.kvcache import PagedKVCache
 
 
