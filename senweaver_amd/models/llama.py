@@ -358,14 +358,13 @@ class LlamaModel:
             else:
                 h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
             qkv = self._linear(h, L, "qkv")
-            qs, kvs = self.local_q_size, self.local_kv_size
-            q = qkv[:, :qs].reshape(B, self.local_heads, c.head_dim).contiguous()
-            k = qkv[:, qs: qs + kvs].reshape(B, self.local_kv_heads, c.head_dim).contiguous()
-            v = qkv[:, qs + kvs:].reshape(B, self.local_kv_heads, c.head_dim).contiguous()
-            ops.rope_inplace(q, k, self.cos_sin, pos32)
             P = kcaches[li].shape[0]
-            kcaches[li].view(P * 16, self.local_kv_heads, c.head_dim).index_copy_(0, slot, k)
-            vcaches[li].view(P * 16, self.local_kv_heads, c.head_dim).index_copy_(0, slot, v)
+            # fused slice+rope+cache-append (one kernel instead of six)
+            q = ops.rope_kv_append(
+                qkv, self.cos_sin, pos32, slot,
+                kcaches[li].view(P * 16, self.local_kv_heads, c.head_dim),
+                vcaches[li].view(P * 16, self.local_kv_heads, c.head_dim),
+                self.local_heads, self.local_kv_heads, c.head_dim)
             attn = ops.paged_decode_attn(q, kcaches[li], vcaches[li], bt, ctx, self.scale)
             attn = attn.reshape(B, self.local_q_size)
             o = self._linear(attn, L, "o")

@@ -109,6 +109,30 @@ def rope_scatter_qkv(qkv: torch.Tensor, cos_sin, positions, Hq: int, Hk: int,
             kr.reshape(B, S, Hk, D).permute(0, 2, 1, 3).contiguous())
 
 
+def rope_kv_append(qkv: torch.Tensor, cos_sin, positions, slot,
+                   kcache_flat, vcache_flat, Hq: int, Hk: int, D: int
+                   ) -> torch.Tensor:
+    """Decode fused head dispatch: qkv row -> rope(q,k); q returned
+    [B,Hq,D] contiguous; k/v written straight into the paged cache rows at
+    ``slot`` (caches viewed [P*16, Hk, D]).  Replaces six ~5 us decode
+    kernels per layer (3 slice copies + rope + 2 index_copy)."""
+    if _on_gpu(qkv):
+        return hip_ext().rope_kv_append(qkv.contiguous(), cos_sin,
+                                        positions.to(torch.int32),
+                                        slot.to(torch.int32),
+                                        kcache_flat, vcache_flat, Hq, Hk, D)
+    B = qkv.shape[0]
+    qs, kvs = Hq * D, Hk * D
+    q = qkv[:, :qs].reshape(B, Hq, D).clone()
+    k = qkv[:, qs:qs + kvs].reshape(B, Hk, D).clone()
+    v = qkv[:, qs + kvs:qs + 2 * kvs].reshape(B, Hk, D)
+    qr = ref.rope_ref(q, cos_sin, positions)
+    kr = ref.rope_ref(k, cos_sin, positions)
+    kcache_flat.index_copy_(0, slot.long(), kr)
+    vcache_flat.index_copy_(0, slot.long(), v.contiguous())
+    return qr
+
+
 def vt_from_qkv(qkv: torch.Tensor, Hq: int, Hk: int, D: int, B: int, S: int
                 ) -> torch.Tensor:
     """V^T [B,Hk,D,S] straight from the fused qkv projection output.

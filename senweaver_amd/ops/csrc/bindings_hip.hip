@@ -12,6 +12,7 @@ extern "C" __global__ void rmsnorm_fwd_kernel(const ushort*, const ushort*, usho
 extern "C" __global__ void rope_fwd_kernel(ushort*, ushort*, const float*, const int*, int, int, int);
 extern "C" __global__ void rope_scatter_kernel(const ushort*, long long, ushort*, ushort*, const float*, const int*, int, int, int, int);
 extern "C" __global__ void transpose_v_kernel(const ushort*, long long, long long, ushort*, int, int, int, int);
+extern "C" __global__ void rope_kv_append_kernel(const ushort*, long long, ushort*, ushort*, ushort*, const float*, const int*, const int*, int, int, int);
 extern "C" __global__ void swiglu_fwd_kernel(const ushort*, ushort*, long long, int);
 extern "C" __global__ void add_bf16_kernel(const ushort*, const ushort*, ushort*, long long);
 extern "C" __global__ void argmax_rows_kernel(const ushort*, int*, int);
@@ -140,6 +141,30 @@ torch::Tensor vt_from_qkv(torch::Tensor qkv, int64_t Hq, int64_t Hk, int64_t D,
       bf16_ptr(qkv), ld, v_off, bf16_mut(vt), (int)B, (int)S, (int)Hk, (int)D);
   HIP_CHECK_KERNEL();
   return vt;
+}
+
+torch::Tensor rope_kv_append(torch::Tensor qkv, torch::Tensor cos_sin,
+                             torch::Tensor positions, torch::Tensor slot,
+                             torch::Tensor kcache, torch::Tensor vcache,
+                             int64_t Hq, int64_t Hk, int64_t D) {
+  check_bf16(qkv, "qkv");
+  const int B = qkv.size(0);
+  const long long ld = qkv.size(1);
+  TORCH_CHECK(ld >= (Hq + 2 * Hk) * D && (D / 2) % 8 == 0);
+  TORCH_CHECK(positions.scalar_type() == torch::kInt32 &&
+              slot.scalar_type() == torch::kInt32);
+  auto q = torch::empty({(long)B, Hq, D}, qkv.options());
+  const int heads_per_blk = (int)(256 / (D / 16));
+  const long long H_total = Hq + 2 * Hk;
+ hipLaunchKernelGGL(( rope_kv_append_kernel), dim3(dim3((unsigned)B, (unsigned)((H_total + heads_per_blk - 1) / heads_per_blk))),
+                          dim3(dim3(256)), 0, cur_stream(), 
+      bf16_ptr(qkv), ld, bf16_mut(q),
+      reinterpret_cast<ushort*>(kcache.data_ptr()),
+      reinterpret_cast<ushort*>(vcache.data_ptr()),
+      cos_sin.data_ptr<float>(), positions.data_ptr<int>(),
+      slot.data_ptr<int>(), (int)Hq, (int)Hk, (int)D);
+  HIP_CHECK_KERNEL();
+  return q;
 }
 
 // ---------------- SwiGLU ----------------
@@ -422,6 +447,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_inplace", &rope_inplace, "RoPE in place over q,k");
   m.def("rope_scatter_qkv", &rope_scatter_qkv, "fused RoPE + [B,H,S,D] scatter from qkv");
   m.def("vt_from_qkv", &vt_from_qkv, "LDS-tiled V^T [B,Hk,D,S] from the qkv slice");
+  m.def("rope_kv_append", &rope_kv_append, "decode: slice+RoPE+KV-append in one kernel");
   m.def("swiglu", &swiglu, "silu(gate)*up from fused gateup");
   m.def("add_bf16", &add_bf16, "a + b (bf16)");
   m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");

@@ -253,3 +253,69 @@ transpose_v_kernel(const ushort* __restrict__ qkv, long long ld, long long v_off
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Decode-step fused head dispatch: reads the fused qkv projection row
+// directly (strided slices), applies rotate-half RoPE to q/k, and writes
+//   q  -> q_out [B, Hq, D] (contiguous, ready for paged attention)
+//   k  -> kcache[slot[b]] rows, v -> vcache[slot[b]] rows
+// One kernel replacing SIX per decode layer (3 slice-contiguous copies +
+// rope_inplace + 2 cache index_copy), each of which sat at the ~5 us
+// small-kernel floor (profiles/r01_decode_breakdown.txt).
+// Caches viewed as [P*16, Hk, D]; 16-B/lane vectorized.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+rope_kv_append_kernel(const ushort* __restrict__ qkv, long long ld,
+                      ushort* __restrict__ q_out, ushort* __restrict__ kcache,
+                      ushort* __restrict__ vcache,
+                      const float* __restrict__ cos_sin,
+                      const int* __restrict__ positions,
+                      const int* __restrict__ slot,
+                      int Hq, int Hk, int D) {
+  const int b = blockIdx.x;
+  const int half = D / 2;
+  const int lanes_per_head = half / 8;
+  const int heads_per_blk = 256 / lanes_per_head;
+  const int h = blockIdx.y * heads_per_blk + threadIdx.x / lanes_per_head;
+  const int H_total = Hq + 2 * Hk;
+  if (h >= H_total) return;
+  const int d0 = (threadIdx.x % lanes_per_head) * 8;
+
+  const ushort* src;
+  ushort* dst;
+  bool do_rope = true;
+  if (h < Hq) {
+    src = qkv + (long long)b * ld + (long long)h * D;
+    dst = q_out + ((long long)b * Hq + h) * D;
+  } else if (h < Hq + Hk) {
+    src = qkv + (long long)b * ld + (long long)(Hq + (h - Hq)) * D;
+    dst = kcache + ((long long)slot[b] * Hk + (h - Hq)) * D;
+  } else {
+    src = qkv + (long long)b * ld + (long long)(Hq + Hk + (h - Hq - Hk)) * D;
+    dst = vcache + ((long long)slot[b] * Hk + (h - Hq - Hk)) * D;
+    do_rope = false;
+  }
+  if (!do_rope) {
+    // v: plain 16-B copies of both halves' chunks
+    *reinterpret_cast<bf16x8*>(dst + d0) =
+        *reinterpret_cast<const bf16x8*>(src + d0);
+    *reinterpret_cast<bf16x8*>(dst + d0 + half) =
+        *reinterpret_cast<const bf16x8*>(src + d0 + half);
+    return;
+  }
+  const float* cs = cos_sin + (long long)positions[b] * D;
+  bf16x8 x1 = *reinterpret_cast<const bf16x8*>(src + d0);
+  bf16x8 x2 = *reinterpret_cast<const bf16x8*>(src + d0 + half);
+  bf16x8 o1, o2;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const float c = cs[d0 + j];
+    const float sn = cs[half + d0 + j];
+    const float a = bf2f(x1.v[j]);
+    const float bb = bf2f(x2.v[j]);
+    o1.v[j] = f2bf(a * c - bb * sn);
+    o2.v[j] = f2bf(bb * c + a * sn);
+  }
+  *reinterpret_cast<bf16x8*>(dst + d0) = o1;
+  *reinterpret_cast<bf16x8*>(dst + d0 + half) = o2;
+}
