@@ -351,13 +351,24 @@ class LlamaModel:
         B = tokens.shape[0]
         hidden = self.embed[tokens]
         residual = None
+        # fused add+norm+GEMV path: bf16 dense decode with GEMV-eligible
+        # shapes (one kernel replaces fused_add_rmsnorm + gemv per norm)
+        fuse_ng = (self.quant == "bf16" and B <= 16
+                   and c.hidden_size % 512 == 0
+                   and (self.local_q_size + 2 * self.local_kv_size) % 4 == 0
+                   and 2 * self.local_inter % 4 == 0)
         for li, L in enumerate(self.layers):
-            if residual is None:
-                residual = hidden.clone()
-                h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
+            if fuse_ng:
+                qkv, residual = ops.gemv_norm_bt(hidden, residual,
+                                                 L["input_norm"], L["qkv"],
+                                                 c.rms_eps)
             else:
-                h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
-            qkv = self._linear(h, L, "qkv")
+                if residual is None:
+                    residual = hidden.clone()
+                    h = ops.rmsnorm(hidden, L["input_norm"], c.rms_eps)
+                else:
+                    h = ops.fused_add_rmsnorm(hidden, residual, L["input_norm"], c.rms_eps)
+                qkv = self._linear(h, L, "qkv")
             P = kcaches[li].shape[0]
             # fused slice+rope+cache-append (one kernel instead of six)
             q = ops.rope_kv_append(
@@ -369,8 +380,16 @@ class LlamaModel:
             attn = attn.reshape(B, self.local_q_size)
             o = self._linear(attn, L, "o")
             self.tp.all_reduce_(o)
-            h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
-            hidden = self._ffn(h, L)
+            if fuse_ng and c.num_experts == 0:
+                gateup, residual = ops.gemv_norm_bt(o, residual,
+                                                    L["post_norm"], L["gateup"],
+                                                    c.rms_eps)
+                act = ops.swiglu(gateup)
+                hidden = self._linear(act, L, "down")
+                self.tp.all_reduce_(hidden)
+            else:
+                h = ops.fused_add_rmsnorm(o, residual, L["post_norm"], c.rms_eps)
+                hidden = self._ffn(h, L)
         return ops.fused_add_rmsnorm(hidden, residual, self.final_norm, c.rms_eps)
 
     def decode_step(self, tokens: torch.Tensor, positions: torch.Tensor,

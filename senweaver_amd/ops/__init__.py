@@ -133,6 +133,23 @@ def rope_kv_append(qkv: torch.Tensor, cos_sin, positions, slot,
     return qr
 
 
+def gemv_norm_bt(x, res_in, normw, b, eps: float = 1e-5):
+    """Decode fused residual-add + RMSNorm + GEMV.
+
+    Returns (C [M,N], res_new [M,K] = x (+ res_in)).  RMSNorm is a linear
+    row scale, so the raw dot and sum-of-squares accumulate in one k-pass
+    and the scale applies after the wave reduction — one kernel instead of
+    fused_add_rmsnorm + gemv.  M <= 16, K %% 512 == 0, N %% 4 == 0."""
+    if _on_gpu(x):
+        c, res_new = hip_ext().gemv_norm_bt(x.contiguous(),
+                                            res_in.contiguous() if res_in is not None else None,
+                                            normw.contiguous(), b.contiguous(), eps)
+        return c, res_new
+    t = x + res_in if res_in is not None else x.clone()
+    y = ref.rmsnorm_ref(t, normw, eps)
+    return ref.gemm_bt_ref(y, b), t
+
+
 def vt_from_qkv(qkv: torch.Tensor, Hq: int, Hk: int, D: int, B: int, S: int
                 ) -> torch.Tensor:
     """V^T [B,Hk,D,S] straight from the fused qkv projection output.

@@ -33,6 +33,11 @@ extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, usho
 extern "C" __global__ void gemv_bt_bf16_v2_m4(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m8(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m16(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_norm_bt_bf16_m1(const ushort*, const ushort*, const ushort*, const ushort*, ushort*, ushort*, int, int, int, float);
+extern "C" __global__ void gemv_norm_bt_bf16_m2(const ushort*, const ushort*, const ushort*, const ushort*, ushort*, ushort*, int, int, int, float);
+extern "C" __global__ void gemv_norm_bt_bf16_m4(const ushort*, const ushort*, const ushort*, const ushort*, ushort*, ushort*, int, int, int, float);
+extern "C" __global__ void gemv_norm_bt_bf16_m8(const ushort*, const ushort*, const ushort*, const ushort*, ushort*, ushort*, int, int, int, float);
+extern "C" __global__ void gemv_norm_bt_bf16_m16(const ushort*, const ushort*, const ushort*, const ushort*, ushort*, ushort*, int, int, int, float);
 extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
 extern "C" __global__ void attn_fwd_v2_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
 extern "C" __global__ void attn_fwd_v3_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
@@ -342,6 +347,51 @@ torch::Tensor gemm_bt_mxfp8(torch::Tensor Aq, torch::Tensor As,
   return C;
 }
 
+// decode fused add+RMSNorm+GEMV: returns (C [M,N], res_out=x(+res) [M,K])
+std::vector<torch::Tensor> gemv_norm_bt(torch::Tensor x, c10::optional<torch::Tensor> res_in,
+                                        torch::Tensor normw, torch::Tensor B, double eps) {
+  check_bf16(x, "x");
+  check_bf16(normw, "normw");
+  check_bf16(B, "B");
+  const int M = x.size(0), K = x.size(1), N = B.size(0);
+  TORCH_CHECK(M <= 16 && N % 4 == 0 && K % 512 == 0 && B.size(1) == K);
+  auto C = torch::empty({M, N}, x.options());
+  auto res_out = torch::empty({M, K}, x.options());
+  int MM = M <= 1 ? 1 : M <= 2 ? 2 : M <= 4 ? 4 : M <= 8 ? 8 : 16;
+  torch::Tensor xp = x, rp;
+  const ushort* rptr = nullptr;
+  if (res_in.has_value()) {
+    check_bf16(*res_in, "res_in");
+    rp = *res_in;
+  }
+  if (M < MM) {
+    xp = torch::zeros({MM, (long)K}, x.options());
+    xp.narrow(0, 0, M).copy_(x);
+    if (res_in.has_value()) {
+      auto r2 = torch::zeros({MM, (long)K}, x.options());
+      r2.narrow(0, 0, M).copy_(rp);
+      rp = r2;
+    }
+  }
+  if (res_in.has_value()) rptr = bf16_ptr(rp);
+  const dim3 grid((N + 3) / 4);
+  auto launch = [&](auto kern) {
+    kern<<<grid, dim3(256), 0, cur_stream()>>>(bf16_ptr(xp), rptr,
+                                               bf16_ptr(normw), bf16_ptr(B),
+                                               bf16_mut(C), bf16_mut(res_out),
+                                               M, N, K, (float)eps);
+  };
+  switch (MM) {
+    case 1: launch(gemv_norm_bt_bf16_m1); break;
+    case 2: launch(gemv_norm_bt_bf16_m2); break;
+    case 4: launch(gemv_norm_bt_bf16_m4); break;
+    case 8: launch(gemv_norm_bt_bf16_m8); break;
+    default: launch(gemv_norm_bt_bf16_m16); break;
+  }
+  HIP_CHECK_KERNEL();
+  return {C, res_out};
+}
+
 // ---------------- Flash attention prefill ----------------
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
                        double scale) {
@@ -448,6 +498,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_scatter_qkv", &rope_scatter_qkv, "fused RoPE + [B,H,S,D] scatter from qkv");
   m.def("vt_from_qkv", &vt_from_qkv, "LDS-tiled V^T [B,Hk,D,S] from the qkv slice");
   m.def("rope_kv_append", &rope_kv_append, "decode: slice+RoPE+KV-append in one kernel");
+  m.def("gemv_norm_bt", &gemv_norm_bt, "decode: fused add+RMSNorm+GEMV");
   m.def("swiglu", &swiglu, "silu(gate)*up from fused gateup");
   m.def("add_bf16", &add_bf16, "a + b (bf16)");
   m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");

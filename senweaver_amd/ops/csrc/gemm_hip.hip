@@ -504,3 +504,80 @@ gemm_bt_bf16_256sg_kernel(const ushort* __restrict__ A, const ushort* __restrict
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Decode fused residual-add + RMSNorm + GEMV:
+//   t[m] = x[m] (+ res_in[m]);  C[m][n] = (sum_k t*normw*B[n]) * rsqrt(ss/K+eps)
+// RMSNorm is a per-row LINEAR scale, so the raw dot and the sum of squares
+// accumulate in the SAME k-pass and the scale applies after the wave
+// reduction — one kernel replaces (fused_add_rmsnorm + gemv), two of the
+// ~5 us launches per decode layer.  Block 0 additionally writes
+// res_out[m] = t[m] (a separate buffer: no cross-block write/read race).
+// ---------------------------------------------------------------------------
+#define GEMV_NORM_INST(MM)                                                    \
+  extern "C" __global__ void __launch_bounds__(256)                           \
+  gemv_norm_bt_bf16_m##MM(const ushort* x, const ushort* res_in,              \
+                          const ushort* normw, const ushort* B, ushort* C,    \
+                          ushort* res_out, int M, int N, int K, float eps) {  \
+    const int wid = threadIdx.x >> 6;                                         \
+    const int lane = threadIdx.x & 63;                                        \
+    const int n = blockIdx.x * 4 + wid;                                       \
+    if (n >= N) return;                                                       \
+    const ushort* brow = B + (long long)n * K;                                \
+    float dot[MM], ss[MM];                                                    \
+    _Pragma("unroll") for (int m = 0; m < MM; ++m) { dot[m] = 0.f; ss[m] = 0.f; } \
+    for (int k = lane * 8; k < K; k += 64 * 8) {                              \
+      bf16x8 bv = *reinterpret_cast<const bf16x8*>(brow + k);                 \
+      bf16x8 wv = *reinterpret_cast<const bf16x8*>(normw + k);                \
+      float bw[8];                                                            \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j)                           \
+          bw[j] = bf2f(bv.v[j]) * bf2f(wv.v[j]);                              \
+      _Pragma("unroll") for (int m = 0; m < MM; ++m) {                        \
+        bf16x8 av = *reinterpret_cast<const bf16x8*>(x + (long long)m * K + k); \
+        float t[8];                                                           \
+        if (res_in != nullptr) {                                              \
+          bf16x8 rv = *reinterpret_cast<const bf16x8*>(res_in + (long long)m * K + k); \
+          _Pragma("unroll") for (int j = 0; j < 8; ++j)                       \
+              t[j] = bf2f(av.v[j]) + bf2f(rv.v[j]);                           \
+        } else {                                                              \
+          _Pragma("unroll") for (int j = 0; j < 8; ++j) t[j] = bf2f(av.v[j]); \
+        }                                                                     \
+        float d = 0.f, s2 = 0.f;                                              \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j) {                       \
+          d += t[j] * bw[j];                                                  \
+          s2 += t[j] * t[j];                                                  \
+        }                                                                     \
+        dot[m] += d;                                                          \
+        ss[m] += s2;                                                          \
+      }                                                                       \
+    }                                                                         \
+    _Pragma("unroll") for (int m = 0; m < MM; ++m) {                          \
+      float d = wave_reduce_sum(dot[m]);                                      \
+      float s2 = wave_reduce_sum(ss[m]);                                      \
+      if (lane == 0 && m < M)                                                 \
+        C[(long long)m * N + n] = f2bf(d * rsqrtf(s2 / (float)K + eps));      \
+    }                                                                         \
+    if (blockIdx.x == 0) {                                                    \
+      /* waves stride rows; lanes stride k: res_out = t (vectorized) */       \
+      for (int m = wid; m < M; m += 4) {                                      \
+        for (int k = lane * 8; k < K; k += 64 * 8) {                          \
+          bf16x8 av = *reinterpret_cast<const bf16x8*>(x + (long long)m * K + k); \
+          bf16x8 o;                                                           \
+          if (res_in != nullptr) {                                            \
+            bf16x8 rv = *reinterpret_cast<const bf16x8*>(res_in + (long long)m * K + k); \
+            _Pragma("unroll") for (int j = 0; j < 8; ++j)                     \
+                o.v[j] = f2bf(bf2f(av.v[j]) + bf2f(rv.v[j]));                 \
+          } else {                                                            \
+            o = av;                                                           \
+          }                                                                   \
+          *reinterpret_cast<bf16x8*>(res_out + (long long)m * K + k) = o;     \
+        }                                                                     \
+      }                                                                       \
+    }                                                                         \
+  }
+
+GEMV_NORM_INST(1)
+GEMV_NORM_INST(2)
+GEMV_NORM_INST(4)
+GEMV_NORM_INST(8)
+GEMV_NORM_INST(16)

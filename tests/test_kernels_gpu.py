@@ -423,3 +423,19 @@ def test_mxfp8_moe_forward_gpu(dev):
     assert torch.isfinite(h8.float()).all()
     rel = ((h16.float() - h8.float()).norm() / h16.float().norm()).item()
     assert rel < 0.2
+
+
+def test_gemv_norm_gpu(dev):
+    torch.manual_seed(13)
+    for M, with_res in [(1, True), (4, True), (3, False), (16, True)]:
+        K, N = 4096, 1024
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+        res = torch.randn(M, K, dtype=torch.bfloat16, device=dev) if with_res else None
+        w = torch.randn(K, dtype=torch.bfloat16, device=dev)
+        b = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+        c, res_new = ops.gemv_norm_bt(x, res, w, b, 1e-5)
+        t = (x + res) if with_res else x
+        y = ref.rmsnorm_ref(t.cpu(), w.cpu(), 1e-5)
+        c_ref = ref.gemm_bt_ref(y, b.cpu())
+        torch.testing.assert_close(c.float().cpu(), c_ref.float(), atol=0.5, rtol=3e-2)
+        torch.testing.assert_close(res_new.float().cpu(), t.float().cpu(), atol=2e-2, rtol=1e-2)
