@@ -353,7 +353,9 @@ class LlamaModel:
         residual = None
         # fused add+norm+GEMV path: bf16 dense decode with GEMV-eligible
         # shapes (one kernel replaces fused_add_rmsnorm + gemv per norm)
+        import os as _os
         fuse_ng = (self.quant == "bf16" and B <= 16
+                   and _os.environ.get("SENWEAVER_DECODE_NORMFUSE", "1") == "1"
                    and c.hidden_size % 512 == 0
                    and (self.local_q_size + 2 * self.local_kv_size) % 4 == 0
                    and 2 * self.local_inter % 4 == 0)
