@@ -353,9 +353,13 @@ class LlamaModel:
         residual = None
         # fused add+norm+GEMV path: bf16 dense decode with GEMV-eligible
         # shapes (one kernel replaces fused_add_rmsnorm + gemv per norm)
+        # add+norm+GEMV fusion measured SLOWER (206 vs 218 tok/s same box):
+        # every GEMV wave re-reads x/res/normw rows from L2 (3 rows instead
+        # of 1), outweighing the two saved ~5us launches per layer.  Kept
+        # behind the env switch as a logged A/B (profiles/r01_pmc_summary).
         import os as _os
         fuse_ng = (self.quant == "bf16" and B <= 16
-                   and _os.environ.get("SENWEAVER_DECODE_NORMFUSE", "1") == "1"
+                   and _os.environ.get("SENWEAVER_DECODE_NORMFUSE", "0") == "1"
                    and c.hidden_size % 512 == 0
                    and (self.local_q_size + 2 * self.local_kv_size) % 4 == 0
                    and 2 * self.local_inter % 4 == 0)
