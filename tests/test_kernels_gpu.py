@@ -439,3 +439,27 @@ def test_gemv_norm_gpu(dev):
         c_ref = ref.gemm_bt_ref(y, b.cpu())
         torch.testing.assert_close(c.float().cpu(), c_ref.float(), atol=0.5, rtol=3e-2)
         torch.testing.assert_close(res_new.float().cpu(), t.float().cpu(), atol=2e-2, rtol=1e-2)
+
+
+def test_rope_kv_append_gpu(dev):
+    torch.manual_seed(14)
+    B, Hq, Hk, D, P = 4, 4, 2, 128, 8
+    ld = (Hq + 2 * Hk) * D
+    qkv = torch.randn(B, ld, dtype=torch.bfloat16, device=dev)
+    cs = ops.rope_tables(256, D, 500000.0).to(dev)
+    pos = torch.tensor([3, 17, 42, 99], dtype=torch.int32, device=dev)
+    slot = torch.tensor([5, 21, 60, 100], dtype=torch.int32, device=dev)
+    kc = torch.zeros(P * 16, Hk, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.zeros(P * 16, Hk, D, dtype=torch.bfloat16, device=dev)
+    q = ops.rope_kv_append(qkv, cs, pos, slot, kc, vc, Hq, Hk, D)
+    # reference: manual slice + rope + index_copy
+    qs, kvs = Hq * D, Hk * D
+    q_ref = ref.rope_ref(qkv[:, :qs].reshape(B, Hq, D).cpu(), cs.cpu(), pos.cpu())
+    k_ref = ref.rope_ref(qkv[:, qs:qs + kvs].reshape(B, Hk, D).cpu(), cs.cpu(), pos.cpu())
+    v_ref = qkv[:, qs + kvs:].reshape(B, Hk, D).cpu()
+    torch.testing.assert_close(q.cpu().float(), q_ref.float(), atol=2e-2, rtol=1e-2)
+    for i in range(B):
+        torch.testing.assert_close(kc[slot[i].long()].cpu().float(), k_ref[i].float(),
+                                   atol=2e-2, rtol=1e-2)
+        torch.testing.assert_close(vc[slot[i].long()].cpu().float(), v_ref[i].float(),
+                                   atol=0, rtol=0)
