@@ -143,44 +143,78 @@ class ReasoningExtractor:
         self._found2 = False
 
     def feed(self, cumulative_text: str) -> Tuple[str, str, bool]:
-        """Feed the cumulative raw text; returns (text, reasoning, emit)."""
+        """Feed the cumulative raw text; returns (text, reasoning, emit).
+
+        Single consumed-index state machine: ``_latest_idx`` is the position
+        in the cumulative string that has been classified so far, and each
+        call loops through as many state transitions (text -> reasoning ->
+        text) as the string allows — so a one-shot feed of the complete text
+        splits fully, and arbitrary chunk boundaries give identical results
+        (chunking-invariance property test in tests/test_grammar.py).  A
+        trailing partial open/close tag is held back until more text (or
+        finalize) resolves it.
+        """
         t1, t2 = self.tags
         s = cumulative_text
-        if not self._found1:
-            if ends_with_any_prefix_of(s, t1):
-                return self.full_text, self.full_reasoning, False  # wait for more
-            idx = s.find(t1)
-            if idx != -1:
-                self._found1 = True
-                self.full_text += s[:idx]
-                self._latest_idx = idx + len(t1)
-                return self.full_text, self.full_reasoning, True
-            self.full_text = s
-            self._latest_idx = len(s)
-            return self.full_text, self.full_reasoning, True
-        if not self._found2:
-            p = ends_with_any_prefix_of(s, t2)
-            if p and p != t2:
-                return self.full_text, self.full_reasoning, False
-            idx = s.find(t2, self._latest_idx)
-            if idx != -1:
-                self._found2 = True
-                self.full_reasoning += s[self._latest_idx: idx]
-                self._latest_idx = idx + len(t2)
-                return self.full_text, self.full_reasoning, True
+        emitted = False
+        while True:
+            if not self._found1:
+                idx = s.find(t1, self._latest_idx)
+                if idx != -1:
+                    if idx > self._latest_idx:
+                        self.full_text += s[self._latest_idx: idx]
+                        emitted = True
+                    self._latest_idx = idx + len(t1)
+                    self._found1 = True
+                    continue
+                keep = len(s)
+                p = ends_with_any_prefix_of(s, t1)
+                if p:
+                    keep = len(s) - len(p)
+                if keep > self._latest_idx:
+                    self.full_text += s[self._latest_idx: keep]
+                    self._latest_idx = keep
+                    emitted = True
+                break
+            if not self._found2:
+                idx = s.find(t2, self._latest_idx)
+                if idx != -1:
+                    if idx > self._latest_idx:
+                        self.full_reasoning += s[self._latest_idx: idx]
+                        emitted = True
+                    self._latest_idx = idx + len(t2)
+                    self._found2 = True
+                    continue
+                keep = len(s)
+                p = ends_with_any_prefix_of(s, t2)
+                if p:
+                    keep = len(s) - len(p)
+                if keep > self._latest_idx:
+                    self.full_reasoning += s[self._latest_idx: keep]
+                    self._latest_idx = keep
+                    emitted = True
+                break
             if len(s) > self._latest_idx:
-                self.full_reasoning += s[self._latest_idx:]
+                self.full_text += s[self._latest_idx:]
                 self._latest_idx = len(s)
-            return self.full_text, self.full_reasoning, True
-        if len(s) > self._latest_idx:
-            self.full_text += s[self._latest_idx:]
-            self._latest_idx = len(s)
-        return self.full_text, self.full_reasoning, True
+                emitted = True
+            break
+        return self.full_text, self.full_reasoning, emitted
 
     def finalize(self, cumulative_text: str) -> Tuple[str, str]:
         """Final (text, reasoning) after the stream ends (extractGrammar :117-127):
-        the accumulated split already excludes the think tags themselves."""
+        the accumulated split already excludes the think tags themselves.
+        A trailing partial tag prefix is flushed as literal content of the
+        active region (the stream is over — it can never complete)."""
         self.feed(cumulative_text)
+        s = cumulative_text
+        if self._latest_idx < len(s):
+            tail = s[self._latest_idx:]
+            if self._found1 and not self._found2:
+                self.full_reasoning += tail
+            else:
+                self.full_text += tail
+            self._latest_idx = len(s)
         return self.full_text, self.full_reasoning
 
 

@@ -101,3 +101,59 @@ def test_xml_disabled_for_normal_mode():
 def test_trim_before_after_newlines():
     assert trim_before_and_after_newlines("  \ncontent\n  ") == "content"
     assert trim_before_and_after_newlines("inline") == "inline"
+
+
+def test_reasoning_chunking_invariance_property():
+    """Property (hypothesis): feeding the SAME cumulative text in arbitrary
+    chunk splits yields the same final (text, reasoning) as one-shot."""
+    from hypothesis import given, strategies as st
+    from senweaver_amd.transport.grammar import ReasoningExtractor
+
+    body = st.text(alphabet="ab<>/think ", min_size=0, max_size=60)
+
+    @given(body, st.lists(st.integers(min_value=1, max_value=10), max_size=12))
+    def prop(raw, steps):
+        full = f"pre {raw}<think>deep {raw}</think> post {raw}"
+        one = ReasoningExtractor()
+        t1, r1, _ = one.feed(full)
+        t1f, r1f = one.full_text, one.full_reasoning
+        inc = ReasoningExtractor()
+        i = 0
+        for s in steps:
+            i = min(len(full), i + s)
+            inc.feed(full[:i])
+        inc.feed(full)
+        assert inc.full_text == t1f
+        assert inc.full_reasoning == r1f
+
+    prop()
+
+
+def test_xml_tool_chunking_invariance_property():
+    """Property: a tool call embedded in the stream is detected with the
+    same name/params regardless of chunk boundaries."""
+    from hypothesis import given, strategies as st
+    from senweaver_amd.transport.grammar import XMLToolExtractor
+
+    @given(st.text(alphabet="abc XY\n", min_size=0, max_size=40),
+           st.lists(st.integers(min_value=1, max_value=7), max_size=20))
+    def prop(prefix, steps):
+        full = (prefix + "<read_file><uri>some/file.txt</uri>"
+                "</read_file>")
+        one = XMLToolExtractor("agent")
+        one.feed(full)
+        ref_call = one.latest_tool_call
+        assert ref_call is not None and ref_call.name == "read_file"
+
+        inc = XMLToolExtractor("agent")
+        i = 0
+        for s in steps:
+            i = min(len(full), i + s)
+            inc.feed(full[:i])
+        inc.feed(full)
+        call = inc.latest_tool_call
+        assert call is not None
+        assert call.name == ref_call.name
+        assert call.raw_params == ref_call.raw_params
+
+    prop()
