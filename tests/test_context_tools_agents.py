@@ -304,3 +304,31 @@ def test_edit_agent_tool(tmp_path):
     (tmp_path / "x.txt").write_text("ORIGINAL-CONTENT")
     svc.call_tool("edit_agent", {"uri": "x.txt", "mode": "edit", "description": "improve"})
     assert (tmp_path / "x.txt").read_text() != "ORIGINAL-CONTENT"
+
+
+def test_prepare_messages_never_overflows_property():
+    """Property (SURVEY §5.7 behavioral contract): for arbitrary histories the
+    fitted messages always fit the input-char budget, and the last user
+    message is always present."""
+    from hypothesis import given, settings, strategies as st
+    from senweaver_amd.context.fitting import (
+        CHARS_PER_TOKEN, Msg, prepare_messages, reserved_output_tokens,
+    )
+
+    msg = st.tuples(st.sampled_from(["user", "assistant", "tool"]),
+                    st.text(alphabet="xyz \n", min_size=1, max_size=4000))
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.lists(msg, min_size=1, max_size=40), st.sampled_from([16384, 32768, 200000]))
+    def prop(raw, cw):
+        msgs = [Msg(r, c) for r, c in raw]
+        if not any(m.role == "user" for m in msgs):
+            msgs.append(Msg("user", "final question"))
+        sysmsg, fitted = prepare_messages(msgs, "You are helpful.", cw)
+        reserved = reserved_output_tokens(cw, None)
+        budget = (cw - reserved) * CHARS_PER_TOKEN
+        total = len(sysmsg) + sum(len(m.content) for m in fitted if m.role != "system")
+        assert total <= budget * 1.01  # fitting guarantee (85% margin inside)
+        assert any(m.role == "user" for m in fitted), "last user message kept"
+
+    prop()

