@@ -22,25 +22,63 @@ SYSTEM_CACHE_TTL_MS = 30_000
 
 
 def base_system_message(chat_mode: str, workspace_overview: str = "",
-                        system_info: str = "") -> str:
-    """Compact analog of chat_systemMessage (prompts.ts:806-1237): mode
-    header + system info + files overview.  Tool XML definitions are
-    appended by the transport's tool registry when tools are enabled."""
+                        system_info: str = "",
+                        workspace_folders: Optional[List[str]] = None,
+                        active_file: Optional[str] = None,
+                        open_files: Optional[List[str]] = None,
+                        terminal_ids: Optional[List[str]] = None,
+                        os_name: str = "",
+                        now: Optional[str] = None,
+                        mcp_tool_names: Optional[List[str]] = None,
+                        custom_api_description: str = "",
+                        supports_vision: bool = True) -> str:
+    """Analog of chat_systemMessage (prompts.ts:806-1237) with the same
+    information slots in the same order: mode header, <system_info> (OS,
+    current date/time, workspace folders, active file, open files, agent-mode
+    persistent terminal IDs), <files_overview>, then MCP/custom-API/vision
+    notes.  Tool XML definitions are appended by the transport's tool
+    registry when tools are enabled; the prose itself is original (the
+    reference's ~1800-line wording is product copy)."""
     headers = {
         "normal": "You are SenWeaver-AMD, an expert coding assistant. Answer questions "
                   "about the user's codebase; you may read files but not modify them.",
         "agent": "You are SenWeaver-AMD in agent mode: complete the user's task end to "
                  "end using the available tools. Verify your work before finishing.",
         "designer": "You are SenWeaver-AMD in designer mode: build and refine UI "
-                    "components and their backing APIs.",
+                    "components and their backing APIs. Plan the complete set of "
+                    "related pages before generating, then produce them one by one.",
         "gather": "You are SenWeaver-AMD in gather mode: collect and summarize relevant "
                   "context from the codebase. Read-only.",
     }
     parts = [headers.get(chat_mode, headers["normal"])]
     if system_info:
         parts.append(f"<system_info>\n{system_info}\n</system_info>")
+    else:
+        info_lines = []
+        if os_name:
+            info_lines.append(f"- Operating System: {os_name}")
+        if now:
+            info_lines.append(f"- Current date/time: {now} (reference point for "
+                              "relative dates like today/tomorrow/this week)")
+        info_lines.append("- Workspace folders:\n"
+                          + ("\n".join(workspace_folders) if workspace_folders
+                             else "NO FOLDERS OPEN"))
+        info_lines.append(f"- Active file:\n{active_file or 'NO ACTIVE FILE'}")
+        info_lines.append("- Open files:\n"
+                          + ("\n".join(open_files) if open_files else "NO OPENED FILES"))
+        if chat_mode == "agent" and terminal_ids:
+            info_lines.append("- Persistent terminal IDs available for commands: "
+                              + ", ".join(terminal_ids))
+        parts.append("<system_info>\n" + "\n".join(info_lines) + "\n</system_info>")
     if workspace_overview:
         parts.append(f"<files_overview>\n{workspace_overview}\n</files_overview>")
+    if mcp_tool_names:
+        parts.append("MCP tools available in this session: " + ", ".join(mcp_tool_names))
+    if custom_api_description:
+        parts.append(f"Custom API: {custom_api_description}")
+    if not supports_vision:
+        parts.append("This model has no vision capability: use the image-description "
+                     "tool when the user provides images.")
     return "\n\n".join(parts)
 
 

@@ -332,3 +332,20 @@ def test_prepare_messages_never_overflows_property():
         assert any(m.role == "user" for m in fitted), "last user message kept"
 
     prop()
+
+
+def test_base_system_message_full_slots():
+    from senweaver_amd.context.pipeline import base_system_message
+    msg = base_system_message(
+        "agent", workspace_folders=["/w/project"], active_file="/w/project/a.py",
+        open_files=["/w/project/a.py", "/w/project/b.py"],
+        terminal_ids=["t1", "t2"], os_name="Linux", now="2026-09-13 12:00",
+        mcp_tool_names=["db_query"], supports_vision=False)
+    assert "<system_info>" in msg and "</system_info>" in msg
+    assert "/w/project" in msg and "a.py" in msg
+    assert "t1, t2" in msg
+    assert "db_query" in msg
+    assert "no vision" in msg.lower() or "vision" in msg
+    # normal mode: no terminal section
+    msg2 = base_system_message("normal", terminal_ids=["t1"])
+    assert "t1" not in msg2
