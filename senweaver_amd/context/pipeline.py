@@ -116,13 +116,17 @@ class ConvertToLLMMessages:
         self._cached_msg = None
 
     def generate_system_message(self, chat_mode: str, workspace_overview: str = "",
-                                system_info: str = "") -> str:
-        key = f"{chat_mode}:{hash(workspace_overview)}:{hash(system_info)}"
+                                system_info: str = "", **env_slots) -> str:
+        """``env_slots`` pass through to base_system_message (workspace
+        folders, active/open files, terminal IDs, MCP tools, ...)."""
+        key = (f"{chat_mode}:{hash(workspace_overview)}:{hash(system_info)}:"
+               f"{hash(tuple(sorted((k, str(v)) for k, v in env_slots.items())))}")
         now = self._clock()
         if (self._cached_msg is not None and self._cached_key == key
                 and now - self._cached_at < SYSTEM_CACHE_TTL_MS):
             return self._cached_msg
-        msg = base_system_message(chat_mode, workspace_overview, system_info)
+        msg = base_system_message(chat_mode, workspace_overview, system_info,
+                                  **env_slots)
         ma = multi_agent_section(chat_mode)
         if ma:
             msg = msg + "\n\n" + ma
