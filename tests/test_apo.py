@@ -294,3 +294,25 @@ def test_pack_rules_budget_property():
             assert len(overflow) > 2000
 
     prop()
+
+
+def test_gradient_gate_on_auto_analyze(fixed_clock, seq_uuid):
+    """try_auto_analyze triggers a textual gradient only when goodRate < 0.7
+    AND feedbacks >= 15 (reference apoService.ts:468 gate)."""
+    from senweaver_amd.apo.optimizer import LocalGradientEngine, StubBackend
+
+    # bad case: 18 bad vs 4 good -> goodRate < 0.7, feedbacks 22 >= 15
+    engine = LocalGradientEngine(StubBackend())
+    tc, apo, _ = make_services(fixed_clock, seq_uuid, optimizer=engine)
+    seed_bad_traces(tc, n_bad=18, n_good=4)
+    report = apo.try_auto_analyze()
+    assert report is not None and report.good_rate < 0.7
+    assert apo.get_textual_gradients(), "gradient should have been requested"
+
+    # good case: high goodRate -> no gradient despite many feedbacks
+    engine2 = LocalGradientEngine(StubBackend())
+    tc2, apo2, _ = make_services(fixed_clock, seq_uuid, optimizer=engine2)
+    seed_bad_traces(tc2, n_bad=2, n_good=20)
+    report2 = apo2.try_auto_analyze()
+    assert report2 is not None and report2.good_rate >= 0.7
+    assert not apo2.get_textual_gradients()
