@@ -100,3 +100,57 @@ def test_js_stringify_quirks():
     assert js_stringify({"a": 1.0, "b": None, "c": "好"}) == '{"a":1,"b":null,"c":"好"}'
     assert js_stringify(float("nan")) == "null"
     assert js_stringify([0.5, 2.0]) == "[0.5,2]"
+
+
+def test_file_storage_restart_resume(tmp_path):
+    """The reference's durability contract (README:143 'restart loses
+    nothing'): beam state + segments + traces written through FileStorage
+    reload in fresh service instances from the same path."""
+    from senweaver_amd.storage import FileStorage
+
+    path = str(tmp_path / "state.json")
+
+    def services(storage):
+        t = {"now": 1700000000000}
+
+        def clock():
+            t["now"] += 10
+            return t["now"]
+        n = {"i": 0}
+
+        def uuid_fn():
+            n["i"] += 1
+            return f"00000000-0000-4000-8000-{n['i']:012d}"
+        tc = TraceCollector(storage=storage, clock=clock, uuid_fn=uuid_fn)
+        return tc, APOService(tc, storage=storage, clock=clock, uuid_fn=uuid_fn)
+
+    s1 = FileStorage(path)
+    tc1, apo1 = services(s1)
+    tid = tc1.start_trace("th", {"chatMode": "agent"})
+    tc1.record_user_message("th", 0, "persist me")
+    tc1.end_trace(tid)
+    tc1.flush()
+    apo1.ensure_beam_state()
+    apo1.apply_beam_update({
+        "beam": [VersionedPromptTemplate("v7", "- durable rule", 0.9, 1700000000000)],
+        "bestPrompt": VersionedPromptTemplate("v7", "- durable rule", 0.9, 1700000000000),
+        "bestScore": 0.9,
+        "round": 2,
+    })
+    apo1.flush()
+    s1.flush()
+
+    # fresh process: everything reloads
+    s2 = FileStorage(path)
+    tc2, apo2 = services(s2)
+    assert len(tc2.get_all_traces()) == 1
+    state = apo2.get_beam_state()
+    assert state is not None and state.current_round == 2
+    assert state.history_best_prompt.content == "- durable rule"
+    assert apo2.get_optimized_rules()  # optimized segment survived
+
+    # corrupted file: silent tolerant load (reference storage-load path)
+    with open(path, "w") as f:
+        f.write("{not json")
+    s3 = FileStorage(path)
+    assert s3.get("senweaver.apo.beamState") is None
