@@ -257,3 +257,27 @@ def test_moe_tp2_matches_tp1():
     for rank, vals in results.items():
         assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
     assert results[0] == pytest.approx(results[1], abs=1e-5)
+
+
+def test_bench_torchrun_world2_contract():
+    """Launch bench.py through torch.distributed.run exactly as the driver
+    does (world 2, gloo on CPU, tiny model): rank 0 must print one JSON line
+    with the contract fields and dp2 parallelism."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29641", "bench.py", "--gpus", "2", "--steps", "1",
+         "--warmup", "0", "--model", "tiny-debug", "--seq-len", "256"],
+        cwd=repo, capture_output=True, text=True, timeout=420)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [ln for ln in out.stdout.splitlines() if ln.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["metric"].startswith("APO iterations/sec")
+    assert d["n_gpus"] == 2 and d["scaling"] == "strong"
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0 and d["higher_is_better"] is True
