@@ -5,7 +5,9 @@ One step = one APO beam iteration at the reference's workload shape
 (apoService.ts defaults: beamWidth=4 x branchFactor=4 = 16 candidate prompts,
 gradientBatchSize=4 rollouts): every candidate is scored against every
 rollout by teacher-forced log-prob through the Llama-3-8B bf16 backbone
-(hand-written gfx950 HIP kernels), candidates sharded over ranks
+(hand-written gfx950 HIP kernels for attention/norm/rope/decode and
+hipBLASLt for plain projection GEMMs — measured dispatch,
+profiles/r01_gemm_dispatch.txt), candidates sharded over ranks
 (candidate-parallel DP over RCCL/xGMI), score vector one-shot all-reduced,
 Top-K selected and beam state updated on every rank.
 
