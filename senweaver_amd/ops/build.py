@@ -40,10 +40,21 @@ def build(verbose: bool = False):
 
 
 def load_prebuilt():
-    """Load the already-built .so without invoking the compiler (GPU box path)."""
+    """Load the already-built .so without invoking the compiler (GPU box path).
+
+    Refuses a STALE .so (any csrc source or header newer than the binary):
+    returning None makes the caller fall through to build(), whose ninja
+    dependency check recompiles.  Without this check an edited kernel that
+    failed to compile would silently keep running the old binary.
+    """
     so = os.path.join(BUILD_DIR, EXT_NAME + ".so")
     if not os.path.exists(so):
         return None
+    so_mtime = os.path.getmtime(so)
+    for f in os.listdir(CSRC):
+        if f.endswith((".hip", ".h", ".cuh")):
+            if os.path.getmtime(os.path.join(CSRC, f)) > so_mtime:
+                return None  # stale -> rebuild path
     import importlib.util
 
     spec = importlib.util.spec_from_file_location(EXT_NAME, so)
