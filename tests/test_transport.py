@@ -1,0 +1,90 @@
+"""LLMMessageService: streaming contract, instant abort, synchronous mode."""
+
+import threading
+import time
+
+from senweaver_amd.transport import LLMMessageService
+from senweaver_amd.transport.service import LLMChatMessage
+
+
+class SlowBackend:
+    """Streams one char at a time; optionally blocks until released."""
+
+    def __init__(self, text, step_delay=0.0, gate=None):
+        self.text = text
+        self.step_delay = step_delay
+        self.gate = gate  # threading.Event to wait on mid-stream
+
+    def stream_generate(self, prompt, max_new_tokens, should_stop, on_chunk):
+        acc = ""
+        for i, ch in enumerate(self.text):
+            if should_stop():
+                return acc
+            if self.gate is not None and i == 3:
+                self.gate.wait(timeout=10)
+            acc += ch
+            on_chunk(acc)
+            if self.step_delay:
+                time.sleep(self.step_delay)
+        return acc
+
+
+def msgs(content="hi"):
+    return [LLMChatMessage(role="user", content=content)]
+
+
+def test_streaming_cumulative_contract_and_final():
+    svc = LLMMessageService(SlowBackend("hello <think>why</think> world"))
+    texts, finals = [], []
+    done = threading.Event()
+    svc.send_llm_message(
+        msgs(), on_text=lambda **m: texts.append((m["full_text"], m["full_reasoning"])),
+        on_final_message=lambda **m: (finals.append(m), done.set()),
+        on_error=lambda **m: done.set())
+    assert done.wait(5)
+    assert finals and finals[0]["full_text"] == "hello  world"
+    assert finals[0]["full_reasoning"] == "why"
+    # cumulative: each full_text extends the previous
+    for (a, _), (b, _) in zip(texts, texts[1:]):
+        assert b.startswith(a)
+
+
+def test_instant_abort_no_round_trip():
+    """abort() resolves client-side immediately: the abort
+    callback fires even while the backend is still blocked mid-stream."""
+    gate = threading.Event()
+    svc = LLMMessageService(SlowBackend("abcdefgh", gate=gate))
+    aborted = threading.Event()
+    got_final = threading.Event()
+    rid = svc.send_llm_message(
+        msgs(), on_text=lambda **m: None,
+        on_final_message=lambda **m: got_final.set(),
+        on_error=lambda **m: None,
+        on_abort=lambda: aborted.set())
+    time.sleep(0.05)     # let the stream reach the gate
+    t0 = time.time()
+    svc.abort(rid)       # returns instantly — no waiting on the backend
+    assert time.time() - t0 < 0.1
+    gate.set()           # backend resumes, sees should_stop, exits
+    assert aborted.wait(5)
+    assert not got_final.is_set()
+
+
+def test_synchronous_mode_and_empty_error():
+    svc = LLMMessageService(SlowBackend(""))
+    errors = []
+    svc.send_llm_message(msgs(), on_text=lambda **m: None,
+                         on_final_message=lambda **m: None,
+                         on_error=lambda **m: errors.append(m["message"]),
+                         synchronous=True)
+    assert errors and "empty" in errors[0].lower()
+
+
+def test_wait_joins_thread():
+    svc = LLMMessageService(SlowBackend("xyz", step_delay=0.01))
+    done = threading.Event()
+    rid = svc.send_llm_message(msgs(), on_text=lambda **m: None,
+                               on_final_message=lambda **m: done.set(),
+                               on_error=lambda **m: done.set())
+    svc.wait(rid, timeout=5)
+    assert done.is_set()
