@@ -145,3 +145,28 @@ def test_candidate_parallel_scorer_single_rank(backend):
     scores = scorer(["- a", "- b", "- c"], [r])
     direct = backend.score_batch(["- a", "- b", "- c"], [r])
     assert scores == pytest.approx(direct, abs=1e-6)
+
+
+def test_kvcache_page_reuse_and_exhaustion():
+    """Pages freed by free_seq are reused; exhaustion raises loudly (the
+    'out of pages' path that once leaked pages from aborted prefills)."""
+    from senweaver_amd.models import tiny_debug
+    cfg = tiny_debug()
+    cache = PagedKVCache(cfg, num_pages=4, device="cpu")
+    import torch as t
+
+    def fill(seq, n):
+        k = t.zeros(n, cfg.num_kv_heads, cfg.head_dim, dtype=t.bfloat16)
+        for layer in range(cfg.num_layers):
+            cache.append(layer, seq, k, k.clone(),
+                         advance_len=(layer == cfg.num_layers - 1))
+
+    s1 = cache.new_seq()
+    fill(s1, 16 * 4)  # consumes all 4 pages
+    with pytest.raises(RuntimeError):
+        s2 = cache.new_seq()
+        fill(s2, 16)
+    cache.free_seq(s1)
+    s3 = cache.new_seq()
+    fill(s3, 16 * 2)  # reuses freed pages
+    assert cache.ctx_lens_tensor([s3]).tolist() == [32]
