@@ -67,6 +67,11 @@ def build_synthetic_workload(cfg, vocab: int, n_candidates: int, n_rollouts: int
     return sequences
 
 
+# Golden finalReward for the fixed 100-span synthetic trace below; the 9-dim
+# reward semantics (traceCollectorService.ts:668-788) must not drift.
+FINAL_REWARD_GOLDEN = 0.49500000000000005
+
+
 def final_reward_check() -> float:
     """Config-1 sanity value: 9-dim finalReward over a 100-span synthetic trace."""
     from senweaver_amd.trace import TraceCollector
@@ -82,7 +87,12 @@ def final_reward_check() -> float:
         tc.record_llm_call("bench", 1, input_tokens=2000, output_tokens=500, duration=900)
     tc.end_trace(tid)
     tc.record_user_feedback("bench", 1, "good")
-    return tc.get_all_traces()[0].summary.final_reward
+    reward = tc.get_all_traces()[0].summary.final_reward
+    if abs(reward - FINAL_REWARD_GOLDEN) > 1e-12:
+        raise SystemExit(
+            f"reward regression: finalReward(100 spans) = {reward!r}, "
+            f"golden = {FINAL_REWARD_GOLDEN!r}")
+    return reward
 
 
 def main() -> int:
@@ -188,7 +198,8 @@ def main() -> int:
     if rank == 0:
         ips = args.steps / elapsed
         result = {
-            "metric": "APO iterations/sec (beam=4, Llama-3-8B scorer)",
+            "metric": (f"APO iterations/sec (beam={args.beam_width}, "
+                       f"{cfg.name} scorer, {args.quant})"),
             "value": ips,
             "unit": "iterations/s",
             "n_gpus": n_gpus,
