@@ -73,7 +73,13 @@ std::string json_field(const std::string& line, const std::string& key) {
 struct Client {
   int fd;
   std::string inbuf;
+  std::string outbuf;  // pending events not yet accepted by the socket
 };
+
+// A consumer that stops reading gets its events buffered up to this cap and
+// is then dropped (its requests aborted) — the daemon never buffers
+// unboundedly for a stalled client.
+constexpr size_t kMaxClientOutbuf = 8u << 20;  // 8 MiB
 
 class Daemon {
  public:
@@ -145,8 +151,37 @@ class Daemon {
     size_t off = 0;
     while (off < s.size()) {
       ssize_t n = write(fd, s.data() + off, s.size() - off);
-      if (n <= 0) return false;
+      if (n <= 0) {
+        if (n < 0 && errno == EINTR) continue;
+        return false;
+      }
       off += (size_t)n;
+    }
+    return true;
+  }
+
+  // Queue an event line for a client; non-blocking flush.  Returns false if
+  // the client is gone or stalled past the buffer cap (caller drops it).
+  bool send_to_client(int fd, const std::string& line) {
+    auto it = clients_.find(fd);
+    if (it == clients_.end()) return false;
+    Client& c = it->second;
+    c.outbuf += line;
+    c.outbuf += '\n';
+    if (!flush_client(c)) return false;
+    return c.outbuf.size() <= kMaxClientOutbuf;
+  }
+
+  static bool flush_client(Client& c) {
+    while (!c.outbuf.empty()) {
+      ssize_t n = write(c.fd, c.outbuf.data(), c.outbuf.size());
+      if (n > 0) {
+        c.outbuf.erase(0, (size_t)n);
+        continue;
+      }
+      if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return true;
+      if (n < 0 && errno == EINTR) continue;
+      return false;  // broken pipe / closed
     }
     return true;
   }
@@ -156,7 +191,7 @@ class Daemon {
     const std::string rid = json_field(line, "requestId");
     if (method == "sendLLMMessage") {
       if (rid.empty()) {
-        write_all(fd, "{\"event\":\"onError\",\"message\":\"missing requestId\"}\n");
+        send_to_client(fd, "{\"event\":\"onError\",\"message\":\"missing requestId\"}");
         return;
       }
       request_client_[rid] = fd;
@@ -168,33 +203,37 @@ class Daemon {
       aborted_.insert(rid);
       request_client_.erase(rid);
       write_all(worker_in_, line + "\n");
-      write_all(fd, "{\"event\":\"onAbort\",\"requestId\":\"" + rid + "\"}\n");
+      send_to_client(fd, "{\"event\":\"onAbort\",\"requestId\":\"" + rid + "\"}");
     } else if (method == "list" || method == "ping") {
       write_all(worker_in_, line + "\n");
       request_client_[method] = fd;  // single in-flight list/ping per method
     } else if (method == "shutdown") {
-      write_all(fd, "{\"event\":\"shuttingDown\"}\n");
+      send_to_client(fd, "{\"event\":\"shuttingDown\"}");
       shutdown_all();
       exit(0);
     } else {
-      write_all(fd, "{\"event\":\"onError\",\"message\":\"unknown method\"}\n");
+      send_to_client(fd, "{\"event\":\"onError\",\"message\":\"unknown method\"}");
     }
   }
 
   void handle_worker_line(const std::string& line) {
     const std::string rid = json_field(line, "requestId");
     const std::string ev = json_field(line, "event");
+    const bool terminal =
+        ev == "onFinalMessage" || ev == "onError" || ev == "onAbort";
+    // aborted ids: suppress events but reap the tombstone once the worker
+    // acknowledges the end of the request, so aborted_ stays bounded
+    if (!rid.empty() && aborted_.count(rid)) {
+      if (terminal) aborted_.erase(rid);
+      return;
+    }
     std::string key = rid;
     if (key.empty()) key = ev == "listResult" ? "list" : ev == "pong" ? "ping" : "";
     auto it = request_client_.find(key);
     if (it == request_client_.end()) return;           // client gone
-    if (aborted_.count(rid)) return;                   // aborted: drop events
-    if (!write_all(it->second, line + "\n")) {
-      request_client_.erase(it);
-      return;
-    }
-    if (ev == "onFinalMessage" || ev == "onError" || ev == "listResult" || ev == "pong")
-      request_client_.erase(key);
+    int cfd = it->second;
+    if (terminal || ev == "listResult" || ev == "pong") request_client_.erase(it);
+    if (!send_to_client(cfd, line)) drop_client(cfd);
   }
 
   void drop_client(int fd) {
@@ -202,7 +241,14 @@ class Daemon {
     clients_.erase(fd);
     for (auto it = request_client_.begin(); it != request_client_.end();) {
       if (it->second == fd) {
-        aborted_.insert(it->first);
+        const std::string& rid = it->first;
+        if (rid != "list" && rid != "ping") {
+          // stop the engine decoding for a dead client; tombstone the id so
+          // late events are suppressed until the worker acks
+          aborted_.insert(rid);
+          write_all(worker_in_,
+                    "{\"method\":\"abort\",\"requestId\":\"" + rid + "\"}\n");
+        }
         it = request_client_.erase(it);
       } else {
         ++it;
@@ -227,24 +273,43 @@ class Daemon {
       std::vector<pollfd> fds;
       fds.push_back({listen_fd_, POLLIN, 0});
       fds.push_back({worker_out_, POLLIN, 0});
-      for (auto& [fd, c] : clients_) fds.push_back({fd, POLLIN, 0});
+      for (auto& [fd, c] : clients_)
+        fds.push_back({fd, (short)(POLLIN | (c.outbuf.empty() ? 0 : POLLOUT)), 0});
       if (poll(fds.data(), fds.size(), 1000) < 0) {
         if (errno == EINTR) continue;
         break;
       }
-      // worker died?
+      // worker died?  Error out every in-flight request (the reference
+      // channel errors the request back to the caller — a respawned worker
+      // knows nothing about old requestIds, so waiting clients would hang),
+      // then respawn.
       int status;
       if (worker_pid_ > 0 && waitpid(worker_pid_, &status, WNOHANG) == worker_pid_) {
         fprintf(stderr, "[daemon] engine worker exited (%d); restarting\n", status);
         close(worker_in_);
         close(worker_out_);
         worker_pid_ = -1;
+        worker_buf_.clear();
+        std::map<std::string, int> inflight;
+        inflight.swap(request_client_);
+        aborted_.clear();  // old worker's ids can never arrive again
+        for (auto& [rid, cfd] : inflight) {
+          const char* ev = (rid == "list") ? "listResult" : (rid == "ping") ? "pong" : "onError";
+          std::string msg = std::string("{\"event\":\"") + ev +
+                            "\",\"requestId\":\"" + rid +
+                            "\",\"message\":\"engine worker crashed; request lost\"}";
+          if (!send_to_client(cfd, msg)) drop_client(cfd);
+        }
         if (!spawn_worker()) break;
         continue;
       }
       if (fds[0].revents & POLLIN) {
         int cfd = accept(listen_fd_, nullptr, nullptr);
-        if (cfd >= 0) clients_[cfd] = Client{cfd, ""};
+        if (cfd >= 0) {
+          int fl = fcntl(cfd, F_GETFL, 0);
+          fcntl(cfd, F_SETFL, fl | O_NONBLOCK);
+          clients_[cfd] = Client{cfd, "", ""};
+        }
       }
       if (fds[1].revents & (POLLIN | POLLHUP)) {
         char buf[65536];
@@ -259,11 +324,19 @@ class Daemon {
         }
       }
       for (size_t i = 2; i < fds.size(); ++i) {
-        if (!(fds[i].revents & (POLLIN | POLLHUP | POLLERR))) continue;
         int fd = fds[i].fd;
+        if (fds[i].revents & POLLOUT) {
+          auto it = clients_.find(fd);
+          if (it != clients_.end() && !flush_client(it->second)) {
+            drop_client(fd);
+            continue;
+          }
+        }
+        if (!(fds[i].revents & (POLLIN | POLLHUP | POLLERR))) continue;
         char buf[65536];
         ssize_t n = read(fd, buf, sizeof(buf));
         if (n <= 0) {
+          if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) continue;
           drop_client(fd);
           continue;
         }

@@ -153,3 +153,30 @@ def test_daemon_worker_crash_restart(daemon):
         time.sleep(1.0)
     assert texts, "no successful response after worker restart"
     c.close()
+
+
+def test_daemon_worker_crash_errors_inflight(daemon):
+    """An in-flight request whose worker dies must receive onError (not hang):
+    daemon.cpp errors out every outstanding requestId before respawning, the
+    same contract as the reference channel erroring the request back."""
+    import signal
+
+    sock, daemon_pid = daemon
+    c = DaemonClient(sock)
+    assert c.ping(timeout=120)
+    out = subprocess.run(["pgrep", "-P", str(daemon_pid)], capture_output=True,
+                         text=True)
+    pids = [int(x) for x in out.stdout.split()]
+    assert pids, "engine worker child process not found"
+
+    done = threading.Event()
+    errors = []
+    rid = c.send_llm_message([{"role": "user", "content": "doomed request"}],
+                             on_final=lambda m: done.set(),
+                             on_error=lambda m: (errors.append(m), done.set()),
+                             max_new_tokens=512)
+    time.sleep(0.3)  # let the request reach the worker
+    os.kill(pids[0], signal.SIGKILL)
+    assert done.wait(timeout=60), "in-flight request hung after worker crash"
+    assert errors and "crash" in errors[0].get("message", "")
+    c.close()
