@@ -1,5 +1,5 @@
 from .kvcache import PagedKVCache, PAGE_SIZE
-from .tokenizer import HashTokenizer
+from .tokenizer import BPETokenizer, HashTokenizer
 from .scorer import LlamaBackend
 
-__all__ = ["PagedKVCache", "PAGE_SIZE", "HashTokenizer", "LlamaBackend"]
+__all__ = ["PagedKVCache", "PAGE_SIZE", "BPETokenizer", "HashTokenizer", "LlamaBackend"]

@@ -45,7 +45,7 @@ class LlamaBackend:
         self.device = torch.device(device)
         self.config = config
         self.model = LlamaModel(config, device=self.device, seed=seed, tp=tp, quant=quant)
-        self.tokenizer = tok.HashTokenizer(config.vocab_size)
+        self.tokenizer = tok.for_vocab(config.vocab_size)
         self.max_seq = min(max_seq, config.max_position)
         self.micro_batch = micro_batch
 
