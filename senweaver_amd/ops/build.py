@@ -18,7 +18,8 @@ EXT_NAME = "senweaver_amd_hip"
 SOURCES = [
     os.path.join(CSRC, f)
     for f in ("bindings.hip", "elemwise.hip", "sampling.hip", "gemm.hip",
-              "attention.hip", "attention_v2.hip", "decode_attention.hip", "moe.hip", "fp8.hip")
+              "gemm_pipe.hip", "attention.hip", "attention_v2.hip",
+              "decode_attention.hip", "moe.hip", "fp8.hip")
 ]
 
 

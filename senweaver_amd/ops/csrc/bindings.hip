@@ -21,6 +21,7 @@ extern "C" __global__ void gemm_bt_bf16_kernel(const ushort*, const ushort*, ush
 extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256x32_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256sg_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_8ph_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void moe_combine_kernel(const ushort*, const int*, const float*, ushort*, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
@@ -234,12 +235,16 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
     return C;
   }
   TORCH_CHECK(K % 64 == 0, "gemm_bt needs K % 64 == 0; got K=", K);
-  // 256-tile kernel runs 1 block/CU (128 KiB LDS): it needs >=~160 blocks to
-  // fill 256 CUs; below that the 2-block/CU 128-tile kernel wins.
+  // 256-tile kernels run 1 block/CU (128 KiB LDS): they need >=~160 blocks
+  // to fill 256 CUs; below that the 2-block/CU 128-tile kernel wins.
   if (M % 256 == 0 && N % 256 == 0 && (M / 256) * (N / 256) >= 160) {
     const int nwg = (M / 256) * (N / 256);
-    gemm_bt_bf16_256_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
-        bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+    if (K % 128 == 0)  // deep-pipelined 8-phase schedule (counted vmcnt)
+      gemm_bt_bf16_8ph_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
+          bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+    else
+      gemm_bt_bf16_256_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
+          bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
   } else {
     TORCH_CHECK(M % 128 == 0 && N % 128 == 0,
                 "gemm_bt needs M,N % 128 == 0 (pad host-side); got ",
@@ -502,6 +507,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu", &swiglu, "silu(gate)*up from fused gateup");
   m.def("add_bf16", &add_bf16, "a + b (bf16)");
   m.def("gemm_bt", &gemm_bt, "C = A @ B^T (bf16 MFMA)");
+  m.def("gemm_bt_8ph", [](torch::Tensor a, torch::Tensor b) {
+    check_bf16(a, "a"); check_bf16(b, "b");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 128 == 0);
+    auto C = torch::empty({M, N}, a.options());
+    gemm_bt_bf16_8ph_kernel<<<dim3((M / 256) * (N / 256)), dim3(512), 0, cur_stream()>>>(
+        bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K);
+    HIP_CHECK_KERNEL();
+    return C;
+  }, "256-tile 8-phase deep-pipelined GEMM (counted vmcnt)");
   m.def("gemm_bt_256x32", [](torch::Tensor a, torch::Tensor b) {
     check_bf16(a, "a"); check_bf16(b, "b");
     const int M = a.size(0), K = a.size(1), N = b.size(0);

@@ -1,0 +1,244 @@
+// 8-phase software-pipelined bf16 MFMA GEMM for gfx950 (MI355X).
+//
+//   C[M,N] = A[M,K] @ B[N,K]^T   (both operands K-contiguous, bf16, f32 acc)
+//
+// This is the deep-pipelined 256x256 structure the CDNA4 guide documents as
+// the plain-HIP top tier (~1.3-1.5 PF/s on random data), replacing the
+// one-barrier-per-K-tile glds double-buffer whose ~1.0-1.2 PF ceiling is the
+// single vmcnt(0)+barrier drain per K-step (measured round 1:
+// profiles/r01_pmc_summary.txt — 57% MFMA-busy, stall = the barrier drain).
+//
+// Structure per 512-thread block (8 waves as 2M x 4N, wave tile 128x64):
+//   - K advances in BK=64 tiles; each tile's operands are staged as FOUR
+//     16 KiB half-tiles (k-split: A/B x k-halves of 32), stream-ordered
+//     [B-kh0, A-kh0, B-kh1, A-kh1], double-buffered (2 x 4 slots = 128 KiB).
+//   - 4 phases per K-tile, each phase: {4-or-8 ds_read_b128 fragment loads,
+//     issue ONE half-tile prefetch (2 global_load_lds_dwordx4/thread),
+//     s_barrier, 16 MFMA under s_setprio(1), s_barrier}.  Phases pair one
+//     wave's loads with other waves' MFMAs on the same SIMD (role split).
+//   - vmcnt is counted, ONCE per K-tile (s_waitcnt vmcnt(6) at the tile
+//     boundary = 3 half-tiles left in flight), never 0 in the main loop;
+//     raw s_barrier (not __syncthreads) so in-flight DMA crosses barriers.
+//     Prefetch runs 3-7 half-tiles ahead; the boundary wait guarantees the
+//     whole NEXT tile has landed while tile t+2's halves stream.
+//   - Overwrite legality is phase-exact: slot j of the current buffer has
+//     its last ds_read at phase j's top, and tile t+2's half j is issued at
+//     phase j+1, after the closing barrier of phase j drained every wave's
+//     reads (stream order == per-phase consumption order, by construction).
+//   - LDS swizzle: a k-half slot is [256 rows][4 chunks of 16 B]; chunk c of
+//     row r lands at slot H[c] ^ ((r>>2)&3), H = {0,3,1,2}.  This makes all
+//     four of ds_read_b128's 16-lane groups conflict-free for the fragment
+//     read pattern (row = base + lane&15, c = lane>>4) — verified per group
+//     against the (addr/4)%64 banking of the microarch guide.  The inverse
+//     permutation is applied to the global SOURCE address (glds destinations
+//     are lane-linear), staying within each row's 64 B k-half segment so
+//     cacheline coalescing is preserved.
+//   - XCD-aware bijective blockIdx remap (neighbor tiles share L2 panels).
+//
+// Constraints: M % 256 == 0 (host pads), N % 256 == 0, K % 128 == 0
+// (even K-tile count keeps the double-buffer flip compile-time).
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define PBM 256
+#define PBN 256
+#define PBK 64
+// one half-tile slot: 256 rows x 32 k = 256 x 4 chunks x 16 B = 16 KiB
+#define SLOT_USHORT (256 * 32)
+// chunk-permutation tables packed as nibbles: H = {0,3,1,2}, H^-1 = {0,2,3,1}
+#define H_PACK 0x2130u
+#define HINV_PACK 0x1320u
+
+// Issue one half-tile (16 KiB) as 2 global_load_lds_dwordx4 per thread.
+// row/c precomputed per thread outside the loop; k0 = t*64 + kh*32.
+__device__ __forceinline__ void pipe_stage_half(
+    const ushort* __restrict__ op, long long ldK, int k0,
+    ushort* lds_slot, const int row[2], const int cofs[2], int wave_chunk) {
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const ushort* g = op + (long long)row[i] * ldK + k0 + cofs[i];
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(lds_slot +
+            (long long)(i * 512 + wave_chunk) * 8),
+        16, 0, 0);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                        ushort* __restrict__ C, int M, int N, int K) {
+  // ---- XCD-aware bijective remap (T1) ----
+  const int nwg = (M / PBM) * (N / PBN);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / PBN;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;            // 0..1 -> 128-row half of A
+  const int wn = wid & 3;             // 0..3 -> 64-row quarter of B
+  const int l15 = lane & 15;
+  const int kgrp = lane >> 4;         // chunk within a k-half (0..3)
+
+  // slots in stream order per buf: 0=B-kh0, 1=A-kh0, 2=B-kh1, 3=A-kh1
+  __shared__ __attribute__((aligned(16))) ushort lds[2][4][SLOT_USHORT];
+
+  const ushort* Atile = A + (long long)tile_m * PBM * K;
+  const ushort* Btile = B + (long long)tile_n * PBN * K;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // fragment-read byte offset within a slot (constant per lane):
+  //   addr(row, c) = row*64 + (H[c] ^ ((row>>2)&3))*16, row = base + l15,
+  //   and base>>2 == 0 (mod 4) for every frag base, so r4 = (l15>>2)&3.
+  const int swz = ((H_PACK >> (kgrp * 4)) & 0xF) ^ ((l15 >> 2) & 3);
+  const int frag_off = l15 * 64 + swz * 16;          // bytes
+  const int a_off = wm * 128 * 64 + frag_off;        // + mi*16*64
+  const int b_off = wn * 64 * 64 + frag_off;         // + ni*16*64
+
+  // staging source coords per thread (i = 0,1): row s>>2, chunk from H^-1
+  int st_row[2], st_cofs[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int s = i * 512 + tid;
+    const int r = s >> 2;
+    const int c = (HINV_PACK >> ((((s & 3) ^ ((r >> 2) & 3)) * 4))) & 0xF;
+    st_row[i] = r;
+    st_cofs[i] = c * 8;  // elements
+  }
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / PBK;  // even (K % 128 == 0)
+
+  // ---- prologue: halves 0..3 (tile 0), vmcnt(4); halves 4..6, vmcnt(6) ----
+  // half h: tile h>>2, slot h&3; slot&1 ? A : B; k-half (slot>>1)&1
+#define ISSUE_HALF(TGT, SLOT, BUF)                                            \
+  do {                                                                        \
+    if ((TGT) < ntiles) {                                                     \
+      const int k0_ = (TGT) * PBK + (((SLOT) >> 1) & 1) * 32;                 \
+      if ((SLOT) & 1)                                                         \
+        pipe_stage_half(Atile, K, k0_, &lds[(BUF)][(SLOT)][0], st_row,        \
+                        st_cofs, wave_chunk);                                 \
+      else                                                                    \
+        pipe_stage_half(Btile, K, k0_, &lds[(BUF)][(SLOT)][0], st_row,        \
+                        st_cofs, wave_chunk);                                 \
+    }                                                                         \
+  } while (0)
+
+  ISSUE_HALF(0, 0, 0);
+  ISSUE_HALF(0, 1, 0);
+  ISSUE_HALF(0, 2, 0);
+  ISSUE_HALF(0, 3, 0);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  ISSUE_HALF(1, 0, 1);
+  ISSUE_HALF(1, 1, 1);
+  ISSUE_HALF(1, 2, 1);
+  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  // ---- main loop: 8 phases per 2 K-tiles, buf flips are compile-time ----
+  // Phase p of tile t: ds_read/MFMA quadrant p; issue half (4t + 7 + p).
+  //   p0: A-kh0 mi0..3 + B-kh0 (8 reads) | issue A-kh1(t+1) -> buf^1 slot 3
+  //   p1: A-kh0 mi4..7      (4 reads)    | issue B-kh0(t+2) -> buf   slot 0
+  //   p2: A-kh1 mi0..3 + B-kh1 (8 reads) | issue A-kh0(t+2) -> buf   slot 1
+  //   p3: A-kh1 mi4..7      (4 reads)    | issue B-kh1(t+2) -> buf   slot 2
+#define LOAD_A4(DST, BUF, KH_SLOT, MI0)                                       \
+  _Pragma("unroll") for (int j = 0; j < 4; ++j) {                             \
+    DST[j] = *reinterpret_cast<const short8*>(                                \
+        reinterpret_cast<const char*>(&lds[(BUF)][(KH_SLOT)][0]) + a_off +    \
+        ((MI0) + j) * 1024);                                                  \
+  }
+#define LOAD_B4(DST, BUF, KH_SLOT)                                            \
+  _Pragma("unroll") for (int j = 0; j < 4; ++j) {                             \
+    DST[j] = *reinterpret_cast<const short8*>(                                \
+        reinterpret_cast<const char*>(&lds[(BUF)][(KH_SLOT)][0]) + b_off +    \
+        j * 1024);                                                            \
+  }
+#define MFMA16(MI0)                                                           \
+  __builtin_amdgcn_s_setprio(1);                                              \
+  _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                            \
+  _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                           \
+      acc[(MI0) + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(          \
+          af[mi], bf[ni], acc[(MI0) + mi][ni], 0, 0, 0);                      \
+  __builtin_amdgcn_s_setprio(0);
+
+#define TILE4(T, BUF)                                                         \
+  do {                                                                        \
+    short8 af[4], bf[4];                                                      \
+    /* phase 0 */                                                             \
+    LOAD_A4(af, BUF, 1, 0);                                                   \
+    LOAD_B4(bf, BUF, 0);                                                      \
+    ISSUE_HALF((T) + 1, 3, (BUF) ^ 1);                                        \
+    __builtin_amdgcn_s_barrier();                                             \
+    MFMA16(0);                                                                \
+    __builtin_amdgcn_s_barrier();                                             \
+    /* phase 1 */                                                             \
+    LOAD_A4(af, BUF, 1, 4);                                                   \
+    ISSUE_HALF((T) + 2, 0, BUF);                                              \
+    __builtin_amdgcn_s_barrier();                                             \
+    MFMA16(4);                                                                \
+    __builtin_amdgcn_s_barrier();                                             \
+    /* phase 2 */                                                             \
+    LOAD_A4(af, BUF, 3, 0);                                                   \
+    LOAD_B4(bf, BUF, 2);                                                      \
+    ISSUE_HALF((T) + 2, 1, BUF);                                              \
+    __builtin_amdgcn_s_barrier();                                             \
+    MFMA16(0);                                                                \
+    __builtin_amdgcn_s_barrier();                                             \
+    /* phase 3 */                                                             \
+    LOAD_A4(af, BUF, 3, 4);                                                   \
+    ISSUE_HALF((T) + 2, 2, BUF);                                              \
+    __builtin_amdgcn_s_barrier();                                             \
+    MFMA16(4);                                                                \
+    __builtin_amdgcn_s_barrier();                                             \
+  } while (0)
+
+  for (int t = 0; t < ntiles; t += 2) {
+    // tile boundary waits: whole next tile landed, <=3 half-tiles in flight
+    if (t == ntiles - 2)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    TILE4(t, 0);
+    if (t + 2 >= ntiles)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    TILE4(t + 1, 1);
+  }
+#undef TILE4
+#undef MFMA16
+#undef LOAD_A4
+#undef LOAD_B4
+#undef ISSUE_HALF
+
+  // ---- epilogue: C-fragment map col = lane&15, row = (lane>>4)*4 + e ----
+  const long long c_row0 = (long long)tile_m * PBM + wm * 128 + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * PBN + wn * 64 + l15;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}

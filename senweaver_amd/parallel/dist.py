@@ -33,7 +33,10 @@ def init_from_env(device: Optional[torch.device] = None) -> tuple:
     world = int(os.environ["WORLD_SIZE"])
     backend = "nccl" if torch.cuda.is_available() else "gloo"
     if torch.cuda.is_available():
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        # modulo: ranks may oversubscribe one device (RCCL permits multiple
+        # ranks per GPU — how the world>1 path is burned in on a 1-GPU box)
+        local = int(os.environ.get("LOCAL_RANK", rank))
+        torch.cuda.set_device(local % torch.cuda.device_count())
     dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=600))
     return rank, world
 
