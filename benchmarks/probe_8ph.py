@@ -1,0 +1,77 @@
+"""Within-probe interleaved A/B of the 8-phase GEMM structural variants.
+
+Variants (gemm_pipe.hip): 0 = 2 barriers/phase (template as written),
+1 = 1 barrier/phase, 2 = V0 without setprio, 3 = V1 + static young-half
+prio.  Baselines: old one-barrier-per-K-tile 256-tile kernel + hipBLASLt.
+Interleaved rounds in one process (methodology rule 24); random data
+(rule 25).  Pass "pmc" to run ONLY variant 0 a few times for a rocprofv3
+--pmc pass.
+"""
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, ".")
+from senweaver_amd import ops  # noqa: E402
+
+dev = "cuda:0"
+ext = ops.hip_ext()
+
+pmc_mode = len(sys.argv) > 1 and sys.argv[1] == "pmc"
+
+shapes = [(4096, 4096, 4096)] if pmc_mode else [(4096, 4096, 4096),
+                                                (8192, 8192, 8192),
+                                                (2048, 28672, 4096)]
+
+for (M, N, K) in shapes:
+    a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    b = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+    fl = 2 * M * N * K / 1e12
+
+    if pmc_mode:
+        # a handful of dispatches of V0 + best variant + blas for the counter pass
+        for _ in range(3):
+            ext.gemm_bt_8ph_v(a, b, 0)
+        for _ in range(3):
+            ext.gemm_bt_8ph_v(a, b, 1)
+        for _ in range(3):
+            a @ b.t()
+        torch.cuda.synchronize()
+        print("pmc dispatches done")
+        break
+
+    arms = {
+        "v0-2bar": lambda: ext.gemm_bt_8ph_v(a, b, 0),
+        "v1-1bar": lambda: ext.gemm_bt_8ph_v(a, b, 1),
+        "v2-nopr": lambda: ext.gemm_bt_8ph_v(a, b, 2),
+        "v3-stat": lambda: ext.gemm_bt_8ph_v(a, b, 3),
+        "blas   ": lambda: a @ b.t(),
+    }
+    # numerics check each variant once vs blas
+    cb = (a @ b.t()).float()
+    for name, fn in arms.items():
+        c = fn().float()
+        err = (c - cb).abs().max().item()
+        assert err <= max(1.5, 0.01 * cb.abs().max().item()), (name, err)
+    # warmup
+    for fn in arms.values():
+        for _ in range(3):
+            fn()
+    torch.cuda.synchronize()
+    # interleaved rounds
+    times = {k: [] for k in arms}
+    for rnd in range(8):
+        for name, fn in arms.items():
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(5):
+                fn()
+            torch.cuda.synchronize()
+            times[name].append((time.perf_counter() - t0) / 5)
+    line = f"{M}x{N}x{K}:"
+    for name, ts in times.items():
+        ts.sort()
+        med = ts[len(ts) // 2]
+        line += f"  {name} {fl / med:7.1f}/{fl / ts[0]:7.1f}"
+    print(line + "   (TF/s med/max)")

@@ -22,6 +22,9 @@ extern "C" __global__ void gemm_bt_bf16_256_kernel(const ushort*, const ushort*,
 extern "C" __global__ void gemm_bt_bf16_256x32_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_256sg_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_8ph_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_8ph_v1_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_8ph_v2_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_8ph_v3_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void moe_combine_kernel(const ushort*, const int*, const float*, ushort*, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
@@ -517,6 +520,25 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     HIP_CHECK_KERNEL();
     return C;
   }, "256-tile 8-phase deep-pipelined GEMM (counted vmcnt)");
+  m.def("gemm_bt_8ph_v", [](torch::Tensor a, torch::Tensor b, long var) {
+    check_bf16(a, "a"); check_bf16(b, "b");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 128 == 0);
+    auto C = torch::empty({M, N}, a.options());
+    const dim3 g((M / 256) * (N / 256)), blk(512);
+    switch (var) {
+      case 1: gemm_bt_bf16_8ph_v1_kernel<<<g, blk, 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+      case 2: gemm_bt_bf16_8ph_v2_kernel<<<g, blk, 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+      case 3: gemm_bt_bf16_8ph_v3_kernel<<<g, blk, 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+      default: gemm_bt_bf16_8ph_kernel<<<g, blk, 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+    }
+    HIP_CHECK_KERNEL();
+    return C;
+  }, "8-phase GEMM structural variants (A/B probe)");
   m.def("gemm_bt_256x32", [](torch::Tensor a, torch::Tensor b) {
     check_bf16(a, "a"); check_bf16(b, "b");
     const int M = a.size(0), K = a.size(1), N = b.size(0);

@@ -68,9 +68,18 @@ __device__ __forceinline__ void pipe_stage_half(
   }
 }
 
-extern "C" __global__ void __launch_bounds__(512, 1)
-gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
-                        ushort* __restrict__ C, int M, int N, int K) {
+// Structural variants (within-probe A/B; VAR=0 is the shipped kernel):
+//   0: two raw barriers per phase ({reads,glds} | BAR | MFMA | BAR)
+//   1: ONE raw barrier per phase ({reads, glds, MFMA} | BAR) — reads and
+//      MFMAs co-scheduled inside the segment; overwrite-legality window is
+//      still <1 phase because a wave reaches the barrier only after its
+//      MFMAs (which drained its reads)
+//   2: VAR 0 without s_setprio (isolates T5)
+//   3: VAR 1 with static young-half priority (waves 4-7 setprio(1) once)
+template <int VAR>
+__device__ __forceinline__ void
+gemm8ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
+             ushort* __restrict__ C, int M, int N, int K) {
   // ---- XCD-aware bijective remap (T1) ----
   const int nwg = (M / PBM) * (N / PBN);
   int wgid = blockIdx.x;
@@ -150,6 +159,8 @@ gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__
   ISSUE_HALF(1, 2, 1);
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   __builtin_amdgcn_s_barrier();
+  if (VAR == 3 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static young-half priority (T5)
 
   // ---- main loop: 8 phases per 2 K-tiles, buf flips are compile-time ----
   // Phase p of tile t: ds_read/MFMA quadrant p; issue half (4t + 7 + p).
@@ -170,12 +181,14 @@ gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__
         j * 1024);                                                            \
   }
 #define MFMA16(MI0)                                                           \
-  __builtin_amdgcn_s_setprio(1);                                              \
+  if (VAR == 0 || VAR == 2) __builtin_amdgcn_s_barrier();                     \
+  if (VAR != 2 && VAR != 3) __builtin_amdgcn_s_setprio(1);                    \
   _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                            \
   _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                           \
       acc[(MI0) + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(          \
           af[mi], bf[ni], acc[(MI0) + mi][ni], 0, 0, 0);                      \
-  __builtin_amdgcn_s_setprio(0);
+  if (VAR != 2 && VAR != 3) __builtin_amdgcn_s_setprio(0);                    \
+  __builtin_amdgcn_s_barrier();
 
 #define TILE4(T, BUF)                                                         \
   do {                                                                        \
@@ -184,28 +197,20 @@ gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__
     LOAD_A4(af, BUF, 1, 0);                                                   \
     LOAD_B4(bf, BUF, 0);                                                      \
     ISSUE_HALF((T) + 1, 3, (BUF) ^ 1);                                        \
-    __builtin_amdgcn_s_barrier();                                             \
     MFMA16(0);                                                                \
-    __builtin_amdgcn_s_barrier();                                             \
     /* phase 1 */                                                             \
     LOAD_A4(af, BUF, 1, 4);                                                   \
     ISSUE_HALF((T) + 2, 0, BUF);                                              \
-    __builtin_amdgcn_s_barrier();                                             \
     MFMA16(4);                                                                \
-    __builtin_amdgcn_s_barrier();                                             \
     /* phase 2 */                                                             \
     LOAD_A4(af, BUF, 3, 0);                                                   \
     LOAD_B4(bf, BUF, 2);                                                      \
     ISSUE_HALF((T) + 2, 1, BUF);                                              \
-    __builtin_amdgcn_s_barrier();                                             \
     MFMA16(0);                                                                \
-    __builtin_amdgcn_s_barrier();                                             \
     /* phase 3 */                                                             \
     LOAD_A4(af, BUF, 3, 4);                                                   \
     ISSUE_HALF((T) + 2, 2, BUF);                                              \
-    __builtin_amdgcn_s_barrier();                                             \
     MFMA16(4);                                                                \
-    __builtin_amdgcn_s_barrier();                                             \
   } while (0)
 
   for (int t = 0; t < ntiles; t += 2) {
@@ -241,4 +246,28 @@ gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__
         crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
     }
   }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                        ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_body<0>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v1_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_body<1>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v2_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_body<2>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v3_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_body<3>(A, B, C, M, N, K);
 }
