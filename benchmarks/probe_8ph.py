@@ -7,12 +7,13 @@ Interleaved rounds in one process (methodology rule 24); random data
 (rule 25).  Pass "pmc" to run ONLY variant 0 a few times for a rocprofv3
 --pmc pass.
 """
+import os
 import sys
 import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from senweaver_amd import ops  # noqa: E402
 
 dev = "cuda:0"
@@ -43,9 +44,9 @@ for (M, N, K) in shapes:
 
     arms = {
         "v0-2bar": lambda: ext.gemm_bt_8ph_v(a, b, 0),
-        "v1-1bar": lambda: ext.gemm_bt_8ph_v(a, b, 1),
-        "v2-nopr": lambda: ext.gemm_bt_8ph_v(a, b, 2),
         "v3-stat": lambda: ext.gemm_bt_8ph_v(a, b, 3),
+        "v4-full1": lambda: ext.gemm_bt_8ph_v(a, b, 4),
+        "v5-full2": lambda: ext.gemm_bt_8ph_v(a, b, 5),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas

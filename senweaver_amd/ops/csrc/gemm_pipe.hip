@@ -248,10 +248,202 @@ gemm8ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Second-generation staging: FULL-tile stage units with 128-byte rows.
+//
+// The k-half slots above stage 64-B row segments (half a cacheline per
+// request).  This body stages each operand TILE (256 rows x 64 k = 32 KiB)
+// as one unit of 4 glds/thread with full 128-B-contiguous rows, halving the
+// request count at equal bytes.  Phase-local consumption still holds:
+//   reads:  P0 {A rows, kk0 chunks | B kk0}, P1 {A kk0 mi4..7},
+//           P2 {kk1...}, P3 {...} — B unit last read P2-top (kk1 B frags),
+//           A unit last read P3-top.
+//   issue:  A(t+1) at P0 (into buf^1, old data dead since t-1 P3);
+//           B(t+2) at P3 (into buf, B(t) last read P2-top).
+//   wait:   vmcnt(4) at each tile boundary leaves only B(t+2)'s 4 loads in
+//           flight -> A(t+1), B(t+1) landed.
+// LDS swizzle for 128-B rows: chunk c of row r lands at (c + 2*(r>>1)) & 7
+// — conflict-free for all four ds_read_b128 lane groups (verified against
+// the (addr/4)%64 banking as before); kk=1 read offsets are kk=0's ^ 64 B.
+// BARS: 1 = one barrier/phase + static young-half prio, 2 = two barriers +
+// per-cluster setprio.
+// ---------------------------------------------------------------------------
+template <int BARS>
+__device__ __forceinline__ void
+gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                  ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / PBM) * (N / PBN);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / PBN;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;
+  const int wn = wid & 3;
+  const int l15 = lane & 15;
+  const int kgrp = lane >> 4;
+
+  // slots: 0 = B tile, 1 = A tile (32 KiB each = 256 rows x 8 chunks)
+  __shared__ __attribute__((aligned(16))) ushort lds[2][2][256 * 64];
+
+  const ushort* Atile = A + (long long)tile_m * PBM * K;
+  const ushort* Btile = B + (long long)tile_n * PBN * K;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // fragment read offset: row*128 + ((c + 2*(row>>1))&7)*16, c = kk*4+kgrp;
+  // row = base + l15 with base % 16 == 0, so the swizzle is per-lane const
+  // and the kk=1 offset is the kk=0 offset XOR 64.
+  const int swz0 = (kgrp + 2 * ((l15 >> 1) & 3)) & 7;
+  const int frag0 = l15 * 128 + swz0 * 16;               // bytes, kk = 0
+  const int a_off = wm * 128 * 128 + frag0;              // + mi*16*128
+  const int b_off = wn * 64 * 128 + frag0;               // + ni*16*128
+
+  // staging source coords (i = 0..3): linear slot s = i*512 + tid covers
+  // row = s>>3, dest chunk sc = s&7; source chunk = (sc - 2*(row>>1)) & 7
+  int st_row[4], st_cofs[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = i * 512 + tid;
+    const int r = s >> 3;
+    const int c = ((s & 7) - 2 * ((r >> 1) & 3)) & 7;
+    st_row[i] = r;
+    st_cofs[i] = c * 8;  // elements
+  }
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / PBK;
+
+#define ISSUE_TILE(TGT, SLOT, BUF)                                           \
+  do {                                                                       \
+    if ((TGT) < ntiles) {                                                    \
+      const int k0_ = (TGT) * PBK;                                           \
+      const ushort* op_ = (SLOT) ? Atile : Btile;                            \
+      ushort* dst_ = &lds[(BUF)][(SLOT)][0];                                 \
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+        const ushort* g = op_ + (long long)st_row[i] * K + k0_ + st_cofs[i]; \
+        __builtin_amdgcn_global_load_lds(                                    \
+            (const __attribute__((address_space(1))) unsigned int*)g,        \
+            (__attribute__((address_space(3))) unsigned int*)(dst_ +         \
+                (long long)(i * 512 + wave_chunk) * 8),                      \
+            16, 0, 0);                                                       \
+      }                                                                      \
+    }                                                                        \
+  } while (0)
+
+  // prologue: B(0), A(0), B(1); wait all but B(1)'s 4 loads
+  ISSUE_TILE(0, 0, 0);
+  ISSUE_TILE(0, 1, 0);
+  ISSUE_TILE(1, 0, 1);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  if (BARS == 1 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static young-half priority
+
+#define LOAD_A4F(DST, BUF, MI0, KX)                                          \
+  _Pragma("unroll") for (int j = 0; j < 4; ++j) {                            \
+    DST[j] = *reinterpret_cast<const short8*>(                               \
+        reinterpret_cast<const char*>(&lds[(BUF)][1][0]) +                   \
+        ((a_off + ((MI0) + j) * 2048) ^ ((KX) * 64)));                       \
+  }
+#define LOAD_B4F(DST, BUF, KX)                                               \
+  _Pragma("unroll") for (int j = 0; j < 4; ++j) {                            \
+    DST[j] = *reinterpret_cast<const short8*>(                               \
+        reinterpret_cast<const char*>(&lds[(BUF)][0][0]) +                   \
+        ((b_off + j * 2048) ^ ((KX) * 64)));                                 \
+  }
+#define MFMA16F(MI0)                                                         \
+  if (BARS == 2) __builtin_amdgcn_s_barrier();                               \
+  if (BARS == 2) __builtin_amdgcn_s_setprio(1);                              \
+  _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                          \
+  _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                          \
+      acc[(MI0) + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(         \
+          af[mi], bf[ni], acc[(MI0) + mi][ni], 0, 0, 0);                     \
+  if (BARS == 2) __builtin_amdgcn_s_setprio(0);                              \
+  __builtin_amdgcn_s_barrier();
+
+#define TILE4F(T, BUF)                                                       \
+  do {                                                                       \
+    short8 af[4], bf[4];                                                     \
+    /* phase 0: A kk0 mi0..3 + B kk0; issue A(T+1) into buf^1 */             \
+    LOAD_A4F(af, BUF, 0, 0);                                                 \
+    LOAD_B4F(bf, BUF, 0);                                                    \
+    ISSUE_TILE((T) + 1, 1, (BUF) ^ 1);                                       \
+    MFMA16F(0);                                                              \
+    /* phase 1: A kk0 mi4..7 */                                              \
+    LOAD_A4F(af, BUF, 4, 0);                                                 \
+    MFMA16F(4);                                                              \
+    /* phase 2: A kk1 mi0..3 + B kk1 */                                      \
+    LOAD_A4F(af, BUF, 0, 1);                                                 \
+    LOAD_B4F(bf, BUF, 1);                                                    \
+    MFMA16F(0);                                                              \
+    /* phase 3: A kk1 mi4..7; issue B(T+2) into buf */                       \
+    LOAD_A4F(af, BUF, 4, 1);                                                 \
+    ISSUE_TILE((T) + 2, 0, BUF);                                             \
+    MFMA16F(4);                                                              \
+  } while (0)
+
+  for (int t = 0; t < ntiles; t += 2) {
+    if (t >= ntiles - 2)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    TILE4F(t, 0);
+    if (t + 2 >= ntiles)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    TILE4F(t + 1, 1);
+  }
+#undef TILE4F
+#undef MFMA16F
+#undef LOAD_A4F
+#undef LOAD_B4F
+#undef ISSUE_TILE
+
+  const long long c_row0 = (long long)tile_m * PBM + wm * 128 + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * PBN + wn * 64 + l15;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}
+
 extern "C" __global__ void __launch_bounds__(512, 1)
 gemm_bt_bf16_8ph_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
                         ushort* __restrict__ C, int M, int N, int K) {
   gemm8ph_body<0>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v4_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_full_body<1>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v5_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_full_body<2>(A, B, C, M, N, K);
 }
 
 extern "C" __global__ void __launch_bounds__(512, 1)
