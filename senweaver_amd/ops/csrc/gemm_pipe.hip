@@ -207,23 +207,30 @@ gemm8ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
     LOAD_B4(bf, BUF, 2);                                                      \
     ISSUE_HALF((T) + 2, 1, BUF);                                              \
     MFMA16(0);                                                                \
-    /* phase 3 */                                                             \
+    /* phase 3 + tile-boundary wait.  The vmcnt must sit BEFORE this tile's \
+       closing barrier: vmcnt is per-wave, and the next tile's ds_reads are \
+       only safe once EVERY wave has drained its own staging DMA — wait,    \
+       then rendezvous.  (A wait after the barrier lets a fast wave read    \
+       slots whose glds were issued by a still-computing slow wave: seen as \
+       nondeterministic corruption at 4k before the fix.) */                  \
     LOAD_A4(af, BUF, 3, 4);                                                   \
     ISSUE_HALF((T) + 2, 2, BUF);                                              \
-    MFMA16(4);                                                                \
+    if (VAR == 0 || VAR == 2) __builtin_amdgcn_s_barrier();                   \
+    if (VAR != 2 && VAR != 3) __builtin_amdgcn_s_setprio(1);                  \
+    _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                          \
+    _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                          \
+        acc[4 + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(            \
+            af[mi], bf[ni], acc[4 + mi][ni], 0, 0, 0);                        \
+    if (VAR != 2 && VAR != 3) __builtin_amdgcn_s_setprio(0);                  \
+    if ((T) >= ntiles - 2)                                                    \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
+    else                                                                      \
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");                        \
+    __builtin_amdgcn_s_barrier();                                             \
   } while (0)
 
   for (int t = 0; t < ntiles; t += 2) {
-    // tile boundary waits: whole next tile landed, <=3 half-tiles in flight
-    if (t == ntiles - 2)
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     TILE4(t, 0);
-    if (t + 2 >= ntiles)
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     TILE4(t + 1, 1);
   }
 #undef TILE4
@@ -389,22 +396,28 @@ gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
     LOAD_A4F(af, BUF, 0, 1);                                                 \
     LOAD_B4F(bf, BUF, 1);                                                    \
     MFMA16F(0);                                                              \
-    /* phase 3: A kk1 mi4..7; issue B(T+2) into buf */                       \
+    /* phase 3: A kk1 mi4..7; issue B(T+2); tile-boundary wait BEFORE the  \
+       closing barrier (vmcnt is per-wave: wait, then rendezvous — a wait  \
+       after the barrier lets a fast wave read slots whose DMA a slow wave \
+       has not yet drained; nondeterministic corruption at 4k before fix) */ \
     LOAD_A4F(af, BUF, 4, 1);                                                 \
     ISSUE_TILE((T) + 2, 0, BUF);                                             \
-    MFMA16F(4);                                                              \
+    if (BARS == 2) __builtin_amdgcn_s_barrier();                             \
+    if (BARS == 2) __builtin_amdgcn_s_setprio(1);                            \
+    _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                         \
+    _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                         \
+        acc[4 + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
+            af[mi], bf[ni], acc[4 + mi][ni], 0, 0, 0);                       \
+    if (BARS == 2) __builtin_amdgcn_s_setprio(0);                            \
+    if ((T) >= ntiles - 2)                                                   \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                       \
+    else                                                                     \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                       \
+    __builtin_amdgcn_s_barrier();                                            \
   } while (0)
 
   for (int t = 0; t < ntiles; t += 2) {
-    if (t >= ntiles - 2)
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     TILE4F(t, 0);
-    if (t + 2 >= ntiles)
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     TILE4F(t + 1, 1);
   }
 #undef TILE4F
