@@ -46,7 +46,8 @@ for (M, N, K) in shapes:
         "v0-2bar": lambda: ext.gemm_bt_8ph_v(a, b, 0),
         "v3-stat": lambda: ext.gemm_bt_8ph_v(a, b, 3),
         "v4-full1": lambda: ext.gemm_bt_8ph_v(a, b, 4),
-        "v5-full2": lambda: ext.gemm_bt_8ph_v(a, b, 5),
+        "v6-sprd": lambda: ext.gemm_bt_8ph_v(a, b, 6),
+        "v7-clpr": lambda: ext.gemm_bt_8ph_v(a, b, 7),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas

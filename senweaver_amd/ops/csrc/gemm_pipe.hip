@@ -272,10 +272,15 @@ gemm8ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
 // LDS swizzle for 128-B rows: chunk c of row r lands at (c + 2*(r>>1)) & 7
 // — conflict-free for all four ds_read_b128 lane groups (verified against
 // the (addr/4)%64 banking as before); kk=1 read offsets are kk=0's ^ 64 B.
-// BARS: 1 = one barrier/phase + static young-half prio, 2 = two barriers +
-// per-cluster setprio.
+// MODE: 1 = one barrier/phase + static young-half prio, bunched issues
+//          (A(t+1) x4 at P0, B(t+2) x4 at P3), vmcnt(4);
+//       2 = two barriers/phase + per-cluster setprio, bunched, vmcnt(4);
+//       3 = like 1 but glds SPREAD one-or-two per phase (B(t+1)q2+A(t+1)q1
+//           at P0, A q2 at P1, B(t+2)q1 at P3, its q2 at the next P0) with
+//           vmcnt(2) boundaries;
+//       4 = one barrier/phase + per-cluster setprio, bunched, vmcnt(4).
 // ---------------------------------------------------------------------------
-template <int BARS>
+template <int MODE>
 __device__ __forceinline__ void
 gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
                   ushort* __restrict__ C, int M, int N, int K) {
@@ -333,13 +338,13 @@ gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
 
   const int ntiles = K / PBK;
 
-#define ISSUE_TILE(TGT, SLOT, BUF)                                           \
+#define ISSUE_PART(TGT, SLOT, BUF, I0, NI)                                   \
   do {                                                                       \
     if ((TGT) < ntiles) {                                                    \
       const int k0_ = (TGT) * PBK;                                           \
       const ushort* op_ = (SLOT) ? Atile : Btile;                            \
       ushort* dst_ = &lds[(BUF)][(SLOT)][0];                                 \
-      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+      _Pragma("unroll") for (int i = (I0); i < (I0) + (NI); ++i) {           \
         const ushort* g = op_ + (long long)st_row[i] * K + k0_ + st_cofs[i]; \
         __builtin_amdgcn_global_load_lds(                                    \
             (const __attribute__((address_space(1))) unsigned int*)g,        \
@@ -349,14 +354,21 @@ gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
       }                                                                      \
     }                                                                        \
   } while (0)
+#define ISSUE_TILE(TGT, SLOT, BUF) ISSUE_PART(TGT, SLOT, BUF, 0, 4)
 
-  // prologue: B(0), A(0), B(1); wait all but B(1)'s 4 loads
+  // prologue: B(0), A(0), then prime the next-tile stream
   ISSUE_TILE(0, 0, 0);
   ISSUE_TILE(0, 1, 0);
-  ISSUE_TILE(1, 0, 1);
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  if (MODE == 3) {
+    ISSUE_PART(1, 0, 1, 0, 2);  // B(1) rows 0..127
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  } else {
+    ISSUE_TILE(1, 0, 1);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  }
   __builtin_amdgcn_s_barrier();
-  if (BARS == 1 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+  if ((MODE == 1 || MODE == 3) &&
+      __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);  // static young-half priority
 
 #define LOAD_A4F(DST, BUF, MI0, KX)                                          \
@@ -372,25 +384,31 @@ gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
         ((b_off + j * 2048) ^ ((KX) * 64)));                                 \
   }
 #define MFMA16F(MI0)                                                         \
-  if (BARS == 2) __builtin_amdgcn_s_barrier();                               \
-  if (BARS == 2) __builtin_amdgcn_s_setprio(1);                              \
+  if (MODE == 2) __builtin_amdgcn_s_barrier();                               \
+  if (MODE == 2 || MODE == 4) __builtin_amdgcn_s_setprio(1);                 \
   _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                          \
   _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                          \
       acc[(MI0) + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(         \
           af[mi], bf[ni], acc[(MI0) + mi][ni], 0, 0, 0);                     \
-  if (BARS == 2) __builtin_amdgcn_s_setprio(0);                              \
+  if (MODE == 2 || MODE == 4) __builtin_amdgcn_s_setprio(0);                 \
   __builtin_amdgcn_s_barrier();
 
 #define TILE4F(T, BUF)                                                       \
   do {                                                                       \
     short8 af[4], bf[4];                                                     \
-    /* phase 0: A kk0 mi0..3 + B kk0; issue A(T+1) into buf^1 */             \
+    /* phase 0: A kk0 mi0..3 + B kk0; issue next-tile stages */              \
     LOAD_A4F(af, BUF, 0, 0);                                                 \
     LOAD_B4F(bf, BUF, 0);                                                    \
-    ISSUE_TILE((T) + 1, 1, (BUF) ^ 1);                                       \
+    if (MODE == 3) {                                                         \
+      ISSUE_PART((T) + 1, 0, (BUF) ^ 1, 2, 2); /* B(t+1) rows 128..255 */    \
+      ISSUE_PART((T) + 1, 1, (BUF) ^ 1, 0, 2); /* A(t+1) rows 0..127 */      \
+    } else {                                                                 \
+      ISSUE_TILE((T) + 1, 1, (BUF) ^ 1);                                     \
+    }                                                                        \
     MFMA16F(0);                                                              \
     /* phase 1: A kk0 mi4..7 */                                              \
     LOAD_A4F(af, BUF, 4, 0);                                                 \
+    if (MODE == 3) ISSUE_PART((T) + 1, 1, (BUF) ^ 1, 2, 2);                  \
     MFMA16F(4);                                                              \
     /* phase 2: A kk1 mi0..3 + B kk1 */                                      \
     LOAD_A4F(af, BUF, 0, 1);                                                 \
@@ -401,16 +419,21 @@ gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
        after the barrier lets a fast wave read slots whose DMA a slow wave \
        has not yet drained; nondeterministic corruption at 4k before fix) */ \
     LOAD_A4F(af, BUF, 4, 1);                                                 \
-    ISSUE_TILE((T) + 2, 0, BUF);                                             \
-    if (BARS == 2) __builtin_amdgcn_s_barrier();                             \
-    if (BARS == 2) __builtin_amdgcn_s_setprio(1);                            \
+    if (MODE == 3)                                                           \
+      ISSUE_PART((T) + 2, 0, BUF, 0, 2);  /* B(t+2) rows 0..127 */           \
+    else                                                                     \
+      ISSUE_TILE((T) + 2, 0, BUF);                                           \
+    if (MODE == 2) __builtin_amdgcn_s_barrier();                             \
+    if (MODE == 2 || MODE == 4) __builtin_amdgcn_s_setprio(1);               \
     _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                         \
     _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                         \
         acc[4 + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
             af[mi], bf[ni], acc[4 + mi][ni], 0, 0, 0);                       \
-    if (BARS == 2) __builtin_amdgcn_s_setprio(0);                            \
+    if (MODE == 2 || MODE == 4) __builtin_amdgcn_s_setprio(0);               \
     if ((T) >= ntiles - 2)                                                   \
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                       \
+    else if (MODE == 3)                                                      \
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");                       \
     else                                                                     \
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                       \
     __builtin_amdgcn_s_barrier();                                            \
@@ -425,6 +448,7 @@ gemm8ph_full_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
 #undef LOAD_A4F
 #undef LOAD_B4F
 #undef ISSUE_TILE
+#undef ISSUE_PART
 
   const long long c_row0 = (long long)tile_m * PBM + wm * 128 + (lane >> 4) * 4;
   const long long c_col0 = (long long)tile_n * PBN + wn * 64 + l15;
@@ -457,6 +481,18 @@ extern "C" __global__ void __launch_bounds__(512, 1)
 gemm_bt_bf16_8ph_v5_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
                            ushort* __restrict__ C, int M, int N, int K) {
   gemm8ph_full_body<2>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v6_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_full_body<3>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v7_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm8ph_full_body<4>(A, B, C, M, N, K);
 }
 
 extern "C" __global__ void __launch_bounds__(512, 1)
