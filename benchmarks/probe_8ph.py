@@ -35,7 +35,7 @@ for (M, N, K) in shapes:
         for _ in range(3):
             ext.gemm_bt_8ph_v(a, b, 4)
         for _ in range(3):
-            ext.gemm_bt_8ph_v(a, b, 6)
+            ext.gemm_bt_8ph_v(a, b, 8)
         for _ in range(3):
             a @ b.t()
         torch.cuda.synchronize()
@@ -43,11 +43,10 @@ for (M, N, K) in shapes:
         break
 
     arms = {
-        "v0-2bar": lambda: ext.gemm_bt_8ph_v(a, b, 0),
-        "v3-stat": lambda: ext.gemm_bt_8ph_v(a, b, 3),
         "v4-full1": lambda: ext.gemm_bt_8ph_v(a, b, 4),
         "v6-sprd": lambda: ext.gemm_bt_8ph_v(a, b, 6),
-        "v7-clpr": lambda: ext.gemm_bt_8ph_v(a, b, 7),
+        "v8-2ph ": lambda: ext.gemm_bt_8ph_v(a, b, 8),
+        "v9-1ph ": lambda: ext.gemm_bt_8ph_v(a, b, 9),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas

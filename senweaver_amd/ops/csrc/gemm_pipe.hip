@@ -483,10 +483,177 @@ gemm_bt_bf16_8ph_v5_kernel(const ushort* __restrict__ A, const ushort* __restric
   gemm8ph_full_body<2>(A, B, C, M, N, K);
 }
 
+// ---------------------------------------------------------------------------
+// Low-barrier generation: 2 phases per K-tile (or 1), full-tile staging,
+// A double-buffered + B ring-of-3 across the WHOLE 160 KiB LDS.
+//
+// PMC on the 4-phase bodies shows 32% SQ_WAIT_ANY (parked at 4-8 barrier
+// generations/tile) vs hipBLASLt's 8.6%.  This body cuts to 2 barriers/tile:
+//   P0: {A(t) kk0 8x b128 + B(t) kk0 4x | issue A(t+1)->slot (t+1)&1 |
+//        32 MFMA kk0 | BAR}
+//   P1: {kk1 reads | issue B(t+2)->slot 2+(t+2)%3 | 32 MFMA kk1 |
+//        vmcnt(4) | BAR}
+// The B ring gives B(t+2) a 4-phase flight window; A(t+1) flies 1 full
+// 32-MFMA phase (~1.1k cycles > HBM latency).  vmcnt(4) at the tile end
+// leaves exactly B(t+2) in flight; everything the next tile reads is landed
+// before the rendezvous (wait-then-barrier, as in the 4-phase fix).
+// PHASES=1 merges both phases: one barrier/K-tile, counted vmcnt — the
+// old structure minus its vmcnt(0) drain.
+// ---------------------------------------------------------------------------
+template <int PHASES>
+__device__ __forceinline__ void
+gemm2ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
+             ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / PBM) * (N / PBN);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / PBN;
+  const int tile_m = wgid / tiles_n;
+  const int tile_n = wgid % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;
+  const int wn = wid & 3;
+  const int l15 = lane & 15;
+  const int kgrp = lane >> 4;
+
+  // 5 x 32 KiB slots = the full 160 KiB: A in 0,1 (dbuf); B ring in 2,3,4
+  __shared__ __attribute__((aligned(16))) ushort lds[5][256 * 64];
+
+  const ushort* Atile = A + (long long)tile_m * PBM * K;
+  const ushort* Btile = B + (long long)tile_n * PBN * K;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int swz0 = (kgrp + 2 * ((l15 >> 1) & 3)) & 7;
+  const int frag0 = l15 * 128 + swz0 * 16;
+  const int a_off = wm * 128 * 128 + frag0;
+  const int b_off = wn * 64 * 128 + frag0;
+
+  int st_row[4], st_cofs[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = i * 512 + tid;
+    const int r = s >> 3;
+    const int c = ((s & 7) - 2 * ((r >> 1) & 3)) & 7;
+    st_row[i] = r;
+    st_cofs[i] = c * 8;
+  }
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / PBK;
+
+#define ISSUE2(TGT, OP, SLOT)                                                \
+  do {                                                                       \
+    if ((TGT) < ntiles) {                                                    \
+      const int k0_ = (TGT) * PBK;                                           \
+      ushort* dst_ = &lds[(SLOT)][0];                                        \
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+        const ushort* g = (OP) + (long long)st_row[i] * K + k0_ + st_cofs[i];\
+        __builtin_amdgcn_global_load_lds(                                    \
+            (const __attribute__((address_space(1))) unsigned int*)g,        \
+            (__attribute__((address_space(3))) unsigned int*)(dst_ +         \
+                (long long)(i * 512 + wave_chunk) * 8),                      \
+            16, 0, 0);                                                       \
+      }                                                                      \
+    }                                                                        \
+  } while (0)
+
+  // prologue: B(0)->2, A(0)->0, B(1)->3; wait all but B(1)
+  ISSUE2(0, Btile, 2);
+  ISSUE2(0, Atile, 0);
+  ISSUE2(1, Btile, 3);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static young-half priority
+
+#define LOAD_A8(DST, ASLOT, KX)                                              \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                            \
+    DST[j] = *reinterpret_cast<const short8*>(                               \
+        reinterpret_cast<const char*>(&lds[0][0]) + (ASLOT) * 32768 +        \
+        ((a_off + j * 2048) ^ ((KX) * 64)));                                 \
+  }
+#define LOAD_B4R(DST, BSLOT, KX)                                             \
+  _Pragma("unroll") for (int j = 0; j < 4; ++j) {                            \
+    DST[j] = *reinterpret_cast<const short8*>(                               \
+        reinterpret_cast<const char*>(&lds[0][0]) + (BSLOT) * 32768 +        \
+        ((b_off + j * 2048) ^ ((KX) * 64)));                                 \
+  }
+#define MFMA32(KX)                                                           \
+  _Pragma("unroll") for (int mi = 0; mi < 8; ++mi)                           \
+  _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                           \
+      acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                 \
+          af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int aslot = t & 1;
+    const int bslot = 2 + t % 3;
+    const int bslot2 = 2 + (t + 2) % 3;
+    short8 af[8], bf[4];
+    // phase 0: kk0
+    LOAD_A8(af, aslot, 0);
+    LOAD_B4R(bf, bslot, 0);
+    ISSUE2(t + 1, Atile, aslot ^ 1);
+    MFMA32(0);
+    if (PHASES == 2) __builtin_amdgcn_s_barrier();
+    // phase 1: kk1
+    LOAD_A8(af, aslot, 1);
+    LOAD_B4R(bf, bslot, 1);
+    ISSUE2(t + 2, Btile, bslot2);
+    MFMA32(1);
+    if (t >= ntiles - 2)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+#undef MFMA32
+#undef LOAD_A8
+#undef LOAD_B4R
+#undef ISSUE2
+
+  const long long c_row0 = (long long)tile_m * PBM + wm * 128 + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * PBN + wn * 64 + l15;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}
+
 extern "C" __global__ void __launch_bounds__(512, 1)
 gemm_bt_bf16_8ph_v6_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
                            ushort* __restrict__ C, int M, int N, int K) {
   gemm8ph_full_body<3>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v8_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm2ph_body<2>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v9_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                           ushort* __restrict__ C, int M, int N, int K) {
+  gemm2ph_body<1>(A, B, C, M, N, K);
 }
 
 extern "C" __global__ void __launch_bounds__(512, 1)
