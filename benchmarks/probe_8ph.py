@@ -35,7 +35,7 @@ for (M, N, K) in shapes:
         for _ in range(3):
             ext.gemm_bt_8ph_v(a, b, 4)
         for _ in range(3):
-            ext.gemm_bt_8ph_v(a, b, 8)
+            ext.gemm_bt_8ph_v(a, b, 9)
         for _ in range(3):
             a @ b.t()
         torch.cuda.synchronize()
@@ -44,9 +44,9 @@ for (M, N, K) in shapes:
 
     arms = {
         "v4-full1": lambda: ext.gemm_bt_8ph_v(a, b, 4),
-        "v6-sprd": lambda: ext.gemm_bt_8ph_v(a, b, 6),
         "v8-2ph ": lambda: ext.gemm_bt_8ph_v(a, b, 8),
         "v9-1ph ": lambda: ext.gemm_bt_8ph_v(a, b, 9),
+        "v11-grp": lambda: ext.gemm_bt_8ph_v(a, b, 11),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas

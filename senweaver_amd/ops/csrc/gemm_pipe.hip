@@ -500,7 +500,7 @@ gemm_bt_bf16_8ph_v5_kernel(const ushort* __restrict__ A, const ushort* __restric
 // PHASES=1 merges both phases: one barrier/K-tile, counted vmcnt — the
 // old structure minus its vmcnt(0) drain.
 // ---------------------------------------------------------------------------
-template <int PHASES>
+template <int PHASES, int GROUPED = 0>
 __device__ __forceinline__ void
 gemm2ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
              ushort* __restrict__ C, int M, int N, int K) {
@@ -512,8 +512,23 @@ gemm2ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
     wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
   }
   const int tiles_n = N / PBN;
-  const int tile_m = wgid / tiles_n;
-  const int tile_n = wgid % tiles_n;
+  int tile_m, tile_n;
+  if (GROUPED) {
+    // grouped-m flat order: each XCD's contiguous wgid chunk becomes a
+    // GM-row x n 2D block, so its L2 holds GM A-panels + chunk/GM B-panels
+    // instead of 1-2 A-panels + a whole row of B-panels.
+    const int GM = 8;
+    const int tiles_m = M / PBM;
+    const int group = wgid / (GM * tiles_n);
+    const int rem = wgid % (GM * tiles_n);
+    const int g0 = group * GM;
+    const int gh = (tiles_m - g0 < GM) ? (tiles_m - g0) : GM;
+    tile_m = g0 + rem % gh;
+    tile_n = rem / gh;
+  } else {
+    tile_m = wgid / tiles_n;
+    tile_n = wgid % tiles_n;
+  }
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -654,6 +669,12 @@ extern "C" __global__ void __launch_bounds__(512, 1)
 gemm_bt_bf16_8ph_v9_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
                            ushort* __restrict__ C, int M, int N, int K) {
   gemm2ph_body<1>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v11_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                            ushort* __restrict__ C, int M, int N, int K) {
+  gemm2ph_body<1, 1>(A, B, C, M, N, K);
 }
 
 extern "C" __global__ void __launch_bounds__(512, 1)
