@@ -43,10 +43,11 @@ for (M, N, K) in shapes:
         break
 
     arms = {
-        "v4-full1": lambda: ext.gemm_bt_8ph_v(a, b, 4),
-        "v8-2ph ": lambda: ext.gemm_bt_8ph_v(a, b, 8),
+
         "v9-1ph ": lambda: ext.gemm_bt_8ph_v(a, b, 9),
         "v11-grp": lambda: ext.gemm_bt_8ph_v(a, b, 11),
+        "v12-npr": lambda: ext.gemm_bt_8ph_v(a, b, 12),
+        "v13-16w": lambda: ext.gemm_bt_8ph_v(a, b, 13),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas

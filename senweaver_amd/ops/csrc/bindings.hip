@@ -32,6 +32,8 @@ extern "C" __global__ void gemm_bt_bf16_8ph_v7_kernel(const ushort*, const ushor
 extern "C" __global__ void gemm_bt_bf16_8ph_v8_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_8ph_v9_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_8ph_v11_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_8ph_v12_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_8ph_v13_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void moe_combine_kernel(const ushort*, const int*, const float*, ushort*, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
@@ -553,6 +555,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       case 9: gemm_bt_bf16_8ph_v9_kernel<<<g, blk, 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
       case 11: gemm_bt_bf16_8ph_v11_kernel<<<g, blk, 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+      case 12: gemm_bt_bf16_8ph_v12_kernel<<<g, blk, 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+      case 13: gemm_bt_bf16_8ph_v13_kernel<<<dim3((M / 256) * (N / 256)), dim3(1024), 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
       default: gemm_bt_bf16_8ph_kernel<<<g, blk, 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;

@@ -500,7 +500,7 @@ gemm_bt_bf16_8ph_v5_kernel(const ushort* __restrict__ A, const ushort* __restric
 // PHASES=1 merges both phases: one barrier/K-tile, counted vmcnt — the
 // old structure minus its vmcnt(0) drain.
 // ---------------------------------------------------------------------------
-template <int PHASES, int GROUPED = 0>
+template <int PHASES, int GROUPED = 0, int PRIO = 1>
 __device__ __forceinline__ void
 gemm2ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
              ushort* __restrict__ C, int M, int N, int K) {
@@ -590,7 +590,7 @@ gemm2ph_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
   ISSUE2(1, Btile, 3);
   asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+  if (PRIO && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);  // static young-half priority
 
 #define LOAD_A8(DST, ASLOT, KX)                                              \
@@ -675,6 +675,153 @@ extern "C" __global__ void __launch_bounds__(512, 1)
 gemm_bt_bf16_8ph_v11_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
                             ushort* __restrict__ C, int M, int N, int K) {
   gemm2ph_body<1, 1>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_8ph_v12_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                            ushort* __restrict__ C, int M, int N, int K) {
+  gemm2ph_body<1, 1, 0>(A, B, C, M, N, K);
+}
+
+// 16-wave (1024-thread) single-barrier body: 4 waves/SIMD, wave tile 64x64
+// (4x4 fragments, 64 acc VGPRs) — rendezvous skew hides behind 4-way wave
+// interleave on each SIMD.  Same 5-slot LDS (A dbuf + B ring-3), same
+// swizzle; staging is 2 glds/thread per 32 KiB unit.
+__device__ __forceinline__ void
+gemm2ph_wide_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                  ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / PBM) * (N / PBN);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / PBN;
+  const int GM = 8;
+  const int tiles_m = M / PBM;
+  const int group = wgid / (GM * tiles_n);
+  const int rem = wgid % (GM * tiles_n);
+  const int g0 = group * GM;
+  const int gh = (tiles_m - g0 < GM) ? (tiles_m - g0) : GM;
+  const int tile_m = g0 + rem % gh;
+  const int tile_n = rem / gh;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // 0..15
+  const int wm = wid >> 2;            // 0..3 -> 64-row band of A
+  const int wn = wid & 3;             // 0..3 -> 64-row band of B
+  const int l15 = lane & 15;
+  const int kgrp = lane >> 4;
+
+  __shared__ __attribute__((aligned(16))) ushort lds[5][256 * 64];
+
+  const ushort* Atile = A + (long long)tile_m * PBM * K;
+  const ushort* Btile = B + (long long)tile_n * PBN * K;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int swz0 = (kgrp + 2 * ((l15 >> 1) & 3)) & 7;
+  const int frag0 = l15 * 128 + swz0 * 16;
+  const int a_off = wm * 64 * 128 + frag0;
+  const int b_off = wn * 64 * 128 + frag0;
+
+  int st_row[2], st_cofs[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int s = i * 1024 + tid;
+    const int r = s >> 3;
+    const int c = ((s & 7) - 2 * ((r >> 1) & 3)) & 7;
+    st_row[i] = r;
+    st_cofs[i] = c * 8;
+  }
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / PBK;
+
+#define ISSUE2W(TGT, OP, SLOT)                                               \
+  do {                                                                       \
+    if ((TGT) < ntiles) {                                                    \
+      const int k0_ = (TGT) * PBK;                                           \
+      ushort* dst_ = &lds[(SLOT)][0];                                        \
+      _Pragma("unroll") for (int i = 0; i < 2; ++i) {                        \
+        const ushort* g = (OP) + (long long)st_row[i] * K + k0_ + st_cofs[i];\
+        __builtin_amdgcn_global_load_lds(                                    \
+            (const __attribute__((address_space(1))) unsigned int*)g,        \
+            (__attribute__((address_space(3))) unsigned int*)(dst_ +         \
+                (long long)(i * 1024 + wave_chunk) * 8),                     \
+            16, 0, 0);                                                       \
+      }                                                                      \
+    }                                                                        \
+  } while (0)
+
+  ISSUE2W(0, Btile, 2);
+  ISSUE2W(0, Atile, 0);
+  ISSUE2W(1, Btile, 3);
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 512)
+    __builtin_amdgcn_s_setprio(1);  // static young-half priority
+
+#define LOAD_W4(DST, SLOTBASE, OFF, KX)                                      \
+  _Pragma("unroll") for (int j = 0; j < 4; ++j) {                            \
+    DST[j] = *reinterpret_cast<const short8*>(                               \
+        reinterpret_cast<const char*>(&lds[0][0]) + (SLOTBASE) * 32768 +     \
+        (((OFF) + j * 2048) ^ ((KX) * 64)));                                 \
+  }
+#define MFMA16W(KX)                                                         \
+  _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                          \
+  _Pragma("unroll") for (int ni = 0; ni < 4; ++ni)                          \
+      acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                \
+          af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int aslot = t & 1;
+    const int bslot = 2 + t % 3;
+    const int bslot2 = 2 + (t + 2) % 3;
+    short8 af[4], bf[4];
+    LOAD_W4(af, aslot, a_off, 0);
+    LOAD_W4(bf, bslot, b_off, 0);
+    ISSUE2W(t + 1, Atile, aslot ^ 1);
+    MFMA16W(0);
+    LOAD_W4(af, aslot, a_off, 1);
+    LOAD_W4(bf, bslot, b_off, 1);
+    ISSUE2W(t + 2, Btile, bslot2);
+    MFMA16W(1);
+    if (t >= ntiles - 2)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+#undef MFMA16W
+#undef LOAD_W4
+#undef ISSUE2W
+
+  const long long c_row0 = (long long)tile_m * PBM + wm * 64 + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * PBN + wn * 64 + l15;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(1024, 4)
+gemm_bt_bf16_8ph_v13_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                            ushort* __restrict__ C, int M, int N, int K) {
+  gemm2ph_wide_body(A, B, C, M, N, K);
 }
 
 extern "C" __global__ void __launch_bounds__(512, 1)
