@@ -33,9 +33,7 @@ for (M, N, K) in shapes:
     if pmc_mode:
         # a handful of dispatches of the leading variants + blas for counters
         for _ in range(3):
-            ext.gemm_bt_8ph_v(a, b, 4)
-        for _ in range(3):
-            ext.gemm_bt_8ph_v(a, b, 9)
+            ext.gemm_bt_8ph_v(a, b, 13)
         for _ in range(3):
             a @ b.t()
         torch.cuda.synchronize()
@@ -48,6 +46,7 @@ for (M, N, K) in shapes:
         "v11-grp": lambda: ext.gemm_bt_8ph_v(a, b, 11),
         "v12-npr": lambda: ext.gemm_bt_8ph_v(a, b, 12),
         "v13-16w": lambda: ext.gemm_bt_8ph_v(a, b, 13),
+        "v14-16n": lambda: ext.gemm_bt_8ph_v(a, b, 14),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas

@@ -687,6 +687,7 @@ gemm_bt_bf16_8ph_v12_kernel(const ushort* __restrict__ A, const ushort* __restri
 // (4x4 fragments, 64 acc VGPRs) — rendezvous skew hides behind 4-way wave
 // interleave on each SIMD.  Same 5-slot LDS (A dbuf + B ring-3), same
 // swizzle; staging is 2 glds/thread per 32 KiB unit.
+template <int PRIO>
 __device__ __forceinline__ void
 gemm2ph_wide_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
                   ushort* __restrict__ C, int M, int N, int K) {
@@ -765,7 +766,7 @@ gemm2ph_wide_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
   ISSUE2W(1, Btile, 3);
   asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 512)
+  if (PRIO && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 512)
     __builtin_amdgcn_s_setprio(1);  // static young-half priority
 
 #define LOAD_W4(DST, SLOTBASE, OFF, KX)                                      \
@@ -821,7 +822,13 @@ gemm2ph_wide_body(const ushort* __restrict__ A, const ushort* __restrict__ B,
 extern "C" __global__ void __launch_bounds__(1024, 4)
 gemm_bt_bf16_8ph_v13_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
                             ushort* __restrict__ C, int M, int N, int K) {
-  gemm2ph_wide_body(A, B, C, M, N, K);
+  gemm2ph_wide_body<1>(A, B, C, M, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(1024, 4)
+gemm_bt_bf16_8ph_v14_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                            ushort* __restrict__ C, int M, int N, int K) {
+  gemm2ph_wide_body<0>(A, B, C, M, N, K);
 }
 
 extern "C" __global__ void __launch_bounds__(512, 1)
