@@ -252,8 +252,12 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
   // to fill 256 CUs; below that the 2-block/CU 128-tile kernel wins.
   if (M % 256 == 0 && N % 256 == 0 && (M / 256) * (N / 256) >= 160) {
     const int nwg = (M / 256) * (N / 256);
-    if (K % 128 == 0)  // deep-pipelined 8-phase schedule (counted vmcnt)
-      gemm_bt_bf16_8ph_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
+    if (K % 128 == 0)
+      // 16-wave single-barrier-per-K-tile pipeline (A dbuf + B ring-3 over
+      // the full 160 KiB LDS, counted vmcnt) — the A/B-measured best
+      // (benchmarks/probe_8ph.py: 1.21/1.31/1.17 PF at 4k/8k/gateup
+      // vs 1.05/1.10/1.06 for the round-1 structure)
+      gemm_bt_bf16_8ph_v14_kernel<<<dim3(nwg), dim3(1024), 0, cur_stream()>>>(
           bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
     else
       gemm_bt_bf16_256_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(

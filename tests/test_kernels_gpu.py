@@ -118,6 +118,24 @@ def test_gemm_8ph_random(dev, M, N, K):
         torch.testing.assert_close(c.float(), c_ref.float(), atol=0.5, rtol=3e-2)
 
 
+@pytest.mark.parametrize("M,N,K", [(256, 256, 128), (512, 512, 512),
+                                   (2048, 6144, 4096), (4096, 4096, 4096),
+                                   (2048, 4096, 14336)])
+def test_gemm_pipe16w_random(dev, M, N, K):
+    """The shipped 16-wave pipeline kernel (v13/v14 = the gemm_bt dispatch
+    for big shapes): numerics + determinism race screen."""
+    ext = ops.hip_ext()
+    for seed in (0, 1):
+        g = torch.Generator(device="cpu").manual_seed(seed)
+        a = torch.randn(M, K, generator=g).bfloat16().to(dev)
+        b = torch.randn(N, K, generator=g).bfloat16().to(dev)
+        c = ext.gemm_bt_8ph_v(a, b, 14)
+        c2 = ext.gemm_bt_8ph_v(a, b, 14)
+        assert torch.equal(c, c2), "nondeterministic output (pipeline race)"
+        c_ref = ref.gemm_bt_ref(a, b)
+        torch.testing.assert_close(c.float(), c_ref.float(), atol=0.5, rtol=3e-2)
+
+
 def test_gemm_pad_m(dev):
     # M not a multiple of 128 goes through the host-side pad
     a = torch.randn(300, 512, dtype=torch.bfloat16, device=dev)
