@@ -76,3 +76,18 @@ for (M, N, K) in shapes:
         med = ts[len(ts) // 2]
         line += f"  {name} {fl / med:7.1f}/{fl / ts[0]:7.1f}"
     print(line + "   (TF/s med/max)")
+
+# MX fp8 pipeline A/B (appended): pipe kernel vs old 256 vs _scaled_mm
+if not pmc_mode:
+    for (M, N, K) in [(4096, 4096, 4096), (8192, 8192, 8192), (2048, 28672, 4096)]:
+        a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+        b = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+        amq, ams = ops.quant_mxfp8(a)
+        bmq, bms = ops.quant_mxfp8(b)
+        fl = 2 * M * N * K / 1e12
+        c_new = ops.gemm_bt_mxfp8(amq, ams, bmq, bms).float()
+        c_ref = (a.float() @ b.float().t())
+        err = (c_new - c_ref).abs().max().item()
+        rel = ((c_new - c_ref).norm() / c_ref.norm()).item()
+        t_new = timeit(lambda: ops.gemm_bt_mxfp8(amq, ams, bmq, bms))
+        print(f"MXpipe {M}x{N}x{K}: {fl / t_new:7.1f} TF/s  rel={rel:.4f} maxerr={err:.2f}")

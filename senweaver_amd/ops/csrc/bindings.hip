@@ -42,6 +42,7 @@ extern "C" __global__ void gemm_bt_fp8_kernel(const unsigned char*, const float*
 extern "C" __global__ void quant_mxfp8_kernel(const ushort*, unsigned char*, unsigned char*, int);
 extern "C" __global__ void gemm_bt_mxfp8_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_mxfp8_256_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_mxfp8_pipe_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m4(const ushort*, const ushort*, ushort*, int, int, int);
@@ -354,7 +355,9 @@ torch::Tensor gemm_bt_mxfp8(torch::Tensor Aq, torch::Tensor As,
   auto C = torch::empty({M, N}, Aq.options().dtype(torch::kBFloat16));
   const int nwg256 = (M % 256 == 0 && N % 256 == 0) ? (M / 256) * (N / 256) : 0;
   if (nwg256 >= 160) {
-    gemm_bt_mxfp8_256_kernel<<<dim3(nwg256), dim3(512), 0, cur_stream()>>>(
+    // single-barrier-per-K-tile pipeline (A dbuf + B ring-3, counted vmcnt,
+    // asm-prefetched e8m0 scales) at the 2x MX MFMA rate
+    gemm_bt_mxfp8_pipe_kernel<<<dim3(nwg256), dim3(512), 0, cur_stream()>>>(
         Aq.data_ptr<unsigned char>(), As.data_ptr<unsigned char>(),
         Bq.data_ptr<unsigned char>(), Bs.data_ptr<unsigned char>(),
         bf16_mut(C), M, N, K);

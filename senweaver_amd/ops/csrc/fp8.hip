@@ -509,3 +509,246 @@ gemm_bt_mxfp8_256_kernel(const unsigned char* __restrict__ A, const unsigned cha
         crow[c_col0 + ni * 32] = f2bf(acc[mi][ni][r]);
     }
 }
+
+// ---------------------------------------------------------------------------
+// MX fp8 16-wave single-barrier pipeline — the v14 bf16 structure
+// (gemm_pipe.hip) at the 2x block-scaled MFMA rate.
+//
+// 1024 threads / 16 waves (4 per SIMD), wave tile 64x64 as 2x2 fragments of
+// v_mfma_scale_f32_32x32x64_f8f6f4; FBK=128 K-tile = one 32 KiB stage unit
+// per operand (256 rows x 128 fp8 = full 128-B rows); A double-buffered +
+// B ring-of-3 across the whole 160 KiB LDS; ONE barrier per K-tile with the
+// counted vmcnt placed BEFORE it (wait-then-rendezvous).
+//
+// LDS swizzle: chunk c of row r at slot (c + (r>>1)) & 7 — conflict-free
+// for the 32-row fragment read pattern (row = base + lane&31, chunks
+// kk*4+lhi and +2): within each ds_read_b128 16-lane group the same-parity
+// rows hit 8 distinct 16-B slots (derived against the (addr/4)%64 banking).
+//
+// e8m0 scales ([row][K/32] bytes; 1 dword covers one K-tile's 4 blocks) are
+// prefetched one tile ahead with INLINE-ASM global_load_dword: hipcc's wait
+// bookkeeping cannot see them, so they never trigger the
+// vmcnt(0)-at-ordinary-load-use drain that would serialize the glds
+// pipeline; the tile-boundary s_waitcnt vmcnt(2) (which names the scale
+// registers as operands) is what guarantees they have landed.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned char* __restrict__ As,
+                          const unsigned char* __restrict__ B, const unsigned char* __restrict__ Bs,
+                          ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / 256) * (N / 256);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / 256;
+  const int GM = 8;
+  const int tiles_m = M / 256;
+  const int grp = wgid / (GM * tiles_n);
+  const int rem = wgid % (GM * tiles_n);
+  const int g0 = grp * GM;
+  const int gh = (tiles_m - g0 < GM) ? (tiles_m - g0) : GM;
+  const int tile_m = g0 + rem % gh;
+  const int tile_n = rem / gh;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;       // 0..7
+  const int wm = wid >> 2;        // 0..1 -> 128-row half of A
+  const int wn = wid & 3;         // 0..3 -> 64-row band of B
+  const int l31 = lane & 31;
+  const int lhi = lane >> 5;
+
+  // 5 x 32 KiB: A in slots 0,1 (dbuf); B ring in 2,3,4
+  __shared__ __attribute__((aligned(16))) unsigned char lds[5][256 * FBK];
+
+  const unsigned char* Atile = A + (long long)tile_m * 256 * K;
+  const unsigned char* Btile = B + (long long)tile_n * 256 * K;
+  const int sld = K / 32;
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[i][j][r] = 0.f;
+
+  // fragment-read swizzle constant: (c + (row>>1)) & 7 with row = base+l31,
+  // base % 32 == 0 -> per-lane constant part is (l31>>1) & 7
+  const int swp = (l31 >> 1) & 7;
+  const int a_row0 = wm * 128;  // + mi*32 + l31
+  const int b_row0 = wn * 64;
+
+  // staging source coords (i = 0..3): slot s = i*512 + tid, row = s>>3,
+  // dest chunk sc = s&7; source chunk = (sc - (row>>1)) & 7.  Kept as ONE
+  // 32-bit byte offset per i (row*K + chunk*16): 4 VGPRs total instead of
+  // four 64-bit pointer pairs, which spilled into the K-loop.
+  unsigned st_off[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = i * 512 + tid;
+    const int r = s >> 3;
+    st_off[i] = (unsigned)r * (unsigned)K + ((((s & 7) - (r >> 1)) & 7) * 16);
+  }
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / FBK;
+
+#define MXISSUE(TGT, OP, SLOT)                                               \
+  do {                                                                       \
+    if ((TGT) < ntiles) {                                                    \
+      const int k0_ = (TGT) * FBK;                                           \
+      unsigned char* dst_ = &lds[(SLOT)][0];                                 \
+      const unsigned char* opk_ = (OP) + k0_;                                \
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+        const unsigned char* g = opk_ + st_off[i];                           \
+        __builtin_amdgcn_global_load_lds(                                    \
+            (const __attribute__((address_space(1))) unsigned int*)g,        \
+            (__attribute__((address_space(3))) unsigned int*)(dst_ +         \
+                (long long)(i * 512 + wave_chunk) * 16),                     \
+            16, 0, 0);                                                       \
+      }                                                                      \
+    }                                                                        \
+  } while (0)
+
+  // Scale access via SRSRC buffer loads (T8/T20): the descriptor lives in
+  // SGPRs (built from readfirstlane'd kernarg pointers, provably uniform),
+  // each lane carries only a 32-bit voffset -> 4 VGPRs instead of 4
+  // pointer pairs.  sld bytes per row; one dword per K-tile (FBK=128).
+  const unsigned as_bytes = (unsigned)M * (unsigned)sld;
+  const unsigned bs_bytes = (unsigned)N * (unsigned)sld;
+  const auto rsrc_a = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)As, 0, as_bytes, 0x00020000);
+  const auto rsrc_b = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)Bs, 0, bs_bytes, 0x00020000);
+  unsigned voa[4], vob[2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    voa[i] = (unsigned)((tile_m * 256 + a_row0 + i * 32 + l31) * sld);
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+    vob[i] = (unsigned)((tile_n * 256 + b_row0 + i * 32 + l31) * sld);
+  // tile-0 scales with plain buffer loads (before any glds is outstanding)
+  unsigned sa_cur[4], sb_cur[2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    sa_cur[i] = __builtin_amdgcn_raw_buffer_load_b32(rsrc_a, voa[i], 0, 0);
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+    sb_cur[i] = __builtin_amdgcn_raw_buffer_load_b32(rsrc_b, vob[i], 0, 0);
+
+  MXISSUE(0, Btile, 2);
+  MXISSUE(0, Atile, 0);
+  MXISSUE(1, Btile, 3);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+#define MXFRAG(DST, SLOT, ROW, C0)                                           \
+  do {                                                                       \
+    const unsigned char* base_ = &lds[(SLOT)][0] + (long long)(ROW) * FBK;   \
+    const int p0_ = ((((C0) & 7) + swp) & 7) * 16;                           \
+    const int p1_ = (((((C0) + 2) & 7) + swp) & 7) * 16;                     \
+    *reinterpret_cast<int4*>(&DST) =                                         \
+        *reinterpret_cast<const int4*>(base_ + p0_);                         \
+    *(reinterpret_cast<int4*>(&DST) + 1) =                                   \
+        *reinterpret_cast<const int4*>(base_ + p1_);                         \
+  } while (0)
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int aslot = t & 1;
+    const int bslot = 2 + t % 3;
+    const int bslot2 = 2 + (t + 2) % 3;
+    // prefetch next tile's scale dwords via asm loads (invisible to hipcc's
+    // waitcnt bookkeeping -> no pipeline-draining use-waits); drained by the
+    // boundary vmcnt(2) below, which names them as operands.
+    unsigned sa_n[4], sb_n[2];
+    if (t + 1 < ntiles) {
+      const unsigned so = (unsigned)((t + 1) * 4);
+      asm volatile("buffer_load_dword %0, %6, %12, %14 offen\n\t"
+                   "buffer_load_dword %1, %7, %12, %14 offen\n\t"
+                   "buffer_load_dword %2, %8, %12, %14 offen\n\t"
+                   "buffer_load_dword %3, %9, %12, %14 offen\n\t"
+                   "buffer_load_dword %4, %10, %13, %14 offen\n\t"
+                   "buffer_load_dword %5, %11, %13, %14 offen"
+                   : "=&v"(sa_n[0]), "=&v"(sa_n[1]), "=&v"(sa_n[2]),
+                     "=&v"(sa_n[3]), "=&v"(sb_n[0]), "=&v"(sb_n[1])
+                   : "v"(voa[0]), "v"(voa[1]), "v"(voa[2]), "v"(voa[3]),
+                     "v"(vob[0]), "v"(vob[1]),
+                     "s"(rsrc_a), "s"(rsrc_b), "s"(so)
+                   : "memory");
+    } else {
+      sa_n[0] = sa_n[1] = sa_n[2] = sa_n[3] = sb_n[0] = sb_n[1] = 0;
+    }
+    i32x8 af, bf[2];
+    // kk = 0 (K 0..63 of the tile)
+    MXFRAG(bf[0], bslot, b_row0 + 0 * 32 + l31, 0 * 4 + lhi);
+    MXFRAG(bf[1], bslot, b_row0 + 1 * 32 + l31, 0 * 4 + lhi);
+    MXISSUE(t + 1, Atile, aslot ^ 1);
+    {
+      const int sh = 8 * (0 * 2 + lhi);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        MXFRAG(af, aslot, a_row0 + mi * 32 + l31, 0 * 4 + lhi);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              af, bf[ni], acc[mi][ni], 0, 0, 0,
+              (int)((sa_cur[mi] >> sh) & 0xff), 0,
+              (int)((sb_cur[ni] >> sh) & 0xff));
+      }
+    }
+    // kk = 1 (K 64..127)
+    MXFRAG(bf[0], bslot, b_row0 + 0 * 32 + l31, 1 * 4 + lhi);
+    MXFRAG(bf[1], bslot, b_row0 + 1 * 32 + l31, 1 * 4 + lhi);
+    MXISSUE(t + 2, Btile, bslot2);
+    {
+      const int sh = 8 * (1 * 2 + lhi);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        MXFRAG(af, aslot, a_row0 + mi * 32 + l31, 1 * 4 + lhi);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              af, bf[ni], acc[mi][ni], 0, 0, 0,
+              (int)((sa_cur[mi] >> sh) & 0xff), 0,
+              (int)((sb_cur[ni] >> sh) & 0xff));
+      }
+    }
+    // boundary: wait (drains A(t+1), B(t+1) glds AND the scale loads; only
+    // B(t+2)'s 2 glds stay in flight), then rendezvous
+    if (t >= ntiles - 2)
+      asm volatile("s_waitcnt vmcnt(0)"
+                   : "+v"(sa_n[0]), "+v"(sa_n[1]), "+v"(sa_n[2]),
+                     "+v"(sa_n[3]), "+v"(sb_n[0]), "+v"(sb_n[1])
+                   :: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)"
+                   : "+v"(sa_n[0]), "+v"(sa_n[1]), "+v"(sa_n[2]),
+                     "+v"(sa_n[3]), "+v"(sb_n[0]), "+v"(sb_n[1])
+                   :: "memory");
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int i = 0; i < 4; ++i) sa_cur[i] = sa_n[i];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) sb_cur[i] = sb_n[i];
+  }
+#undef MXFRAG
+#undef MXISSUE
+
+  // 32x32 C map: col = l31 (+ni*32), row = (r&3) + 8*(r>>2) + 4*lhi
+  const long long c_col0 = (long long)tile_n * 256 + b_row0 + l31;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long long row = (long long)tile_m * 256 + a_row0 + mi * 32
+                            + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        crow[c_col0 + ni * 32] = f2bf(acc[mi][ni][r]);
+    }
+}
