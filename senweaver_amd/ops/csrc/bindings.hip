@@ -355,9 +355,12 @@ torch::Tensor gemm_bt_mxfp8(torch::Tensor Aq, torch::Tensor As,
   auto C = torch::empty({M, N}, Aq.options().dtype(torch::kBFloat16));
   const int nwg256 = (M % 256 == 0 && N % 256 == 0) ? (M / 256) * (N / 256) : 0;
   if (nwg256 >= 160) {
-    // single-barrier-per-K-tile pipeline (A dbuf + B ring-3, counted vmcnt,
-    // asm-prefetched e8m0 scales) at the 2x MX MFMA rate
-    gemm_bt_mxfp8_pipe_kernel<<<dim3(nwg256), dim3(512), 0, cur_stream()>>>(
+    // NOTE: a single-barrier pipeline port (gemm_bt_mxfp8_pipe_kernel, kept
+    // for reference) measured SLOWER (918-998 TF/s vs 1187-1325): the
+    // 32x32x64 f32x16 accumulators fragment the register budget at both
+    // 512-thr (72 B/lane spill) and 1024-thr (176 B/lane) geometries.
+    // Measured-and-documented rejection; see profiles/r02_gemm_pipeline.txt.
+    gemm_bt_mxfp8_256_kernel<<<dim3(nwg256), dim3(512), 0, cur_stream()>>>(
         Aq.data_ptr<unsigned char>(), As.data_ptr<unsigned char>(),
         Bq.data_ptr<unsigned char>(), Bs.data_ptr<unsigned char>(),
         bf16_mut(C), M, N, K);

@@ -425,9 +425,9 @@ gemm_bt_mxfp8_256_kernel(const unsigned char* __restrict__ A, const unsigned cha
   const unsigned char* Btile = B + (long long)tile_n * 256 * K;
   const int sld = K / 32;
 
-  f32x16 acc[4][2];
+  f32x16 acc[2][2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j)
 #pragma unroll
@@ -532,7 +532,7 @@ gemm_bt_mxfp8_256_kernel(const unsigned char* __restrict__ A, const unsigned cha
 // pipeline; the tile-boundary s_waitcnt vmcnt(2) (which names the scale
 // registers as operands) is what guarantees they have landed.
 // ---------------------------------------------------------------------------
-extern "C" __global__ void __launch_bounds__(512, 1)
+extern "C" __global__ void __launch_bounds__(1024, 4)
 gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned char* __restrict__ As,
                           const unsigned char* __restrict__ B, const unsigned char* __restrict__ Bs,
                           ushort* __restrict__ C, int M, int N, int K) {
@@ -555,8 +555,8 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;       // 0..7
-  const int wm = wid >> 2;        // 0..1 -> 128-row half of A
+  const int wid = tid >> 6;       // 0..15
+  const int wm = wid >> 2;        // 0..3 -> 64-row band of A
   const int wn = wid & 3;         // 0..3 -> 64-row band of B
   const int l31 = lane & 31;
   const int lhi = lane >> 5;
@@ -568,9 +568,9 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
   const unsigned char* Btile = B + (long long)tile_n * 256 * K;
   const int sld = K / 32;
 
-  f32x16 acc[4][2];
+  f32x16 acc[2][2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j)
 #pragma unroll
@@ -579,17 +579,17 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
   // fragment-read swizzle constant: (c + (row>>1)) & 7 with row = base+l31,
   // base % 32 == 0 -> per-lane constant part is (l31>>1) & 7
   const int swp = (l31 >> 1) & 7;
-  const int a_row0 = wm * 128;  // + mi*32 + l31
+  const int a_row0 = wm * 64;   // + mi*32 + l31
   const int b_row0 = wn * 64;
 
   // staging source coords (i = 0..3): slot s = i*512 + tid, row = s>>3,
   // dest chunk sc = s&7; source chunk = (sc - (row>>1)) & 7.  Kept as ONE
   // 32-bit byte offset per i (row*K + chunk*16): 4 VGPRs total instead of
   // four 64-bit pointer pairs, which spilled into the K-loop.
-  unsigned st_off[4];
+  unsigned st_off[2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int s = i * 512 + tid;
+  for (int i = 0; i < 2; ++i) {
+    const int s = i * 1024 + tid;
     const int r = s >> 3;
     st_off[i] = (unsigned)r * (unsigned)K + ((((s & 7) - (r >> 1)) & 7) * 16);
   }
@@ -603,12 +603,12 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
       const int k0_ = (TGT) * FBK;                                           \
       unsigned char* dst_ = &lds[(SLOT)][0];                                 \
       const unsigned char* opk_ = (OP) + k0_;                                \
-      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                        \
+      _Pragma("unroll") for (int i = 0; i < 2; ++i) {                        \
         const unsigned char* g = opk_ + st_off[i];                           \
         __builtin_amdgcn_global_load_lds(                                    \
             (const __attribute__((address_space(1))) unsigned int*)g,        \
             (__attribute__((address_space(3))) unsigned int*)(dst_ +         \
-                (long long)(i * 512 + wave_chunk) * 16),                     \
+                (long long)(i * 1024 + wave_chunk) * 16),                    \
             16, 0, 0);                                                       \
       }                                                                      \
     }                                                                        \
@@ -624,26 +624,24 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
       (void*)As, 0, as_bytes, 0x00020000);
   const auto rsrc_b = __builtin_amdgcn_make_buffer_rsrc(
       (void*)Bs, 0, bs_bytes, 0x00020000);
-  unsigned voa[4], vob[2];
+  unsigned voa[2], vob[2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i) {
     voa[i] = (unsigned)((tile_m * 256 + a_row0 + i * 32 + l31) * sld);
-#pragma unroll
-  for (int i = 0; i < 2; ++i)
     vob[i] = (unsigned)((tile_n * 256 + b_row0 + i * 32 + l31) * sld);
+  }
   // tile-0 scales with plain buffer loads (before any glds is outstanding)
-  unsigned sa_cur[4], sb_cur[2];
+  unsigned sa_cur[2], sb_cur[2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i) {
     sa_cur[i] = __builtin_amdgcn_raw_buffer_load_b32(rsrc_a, voa[i], 0, 0);
-#pragma unroll
-  for (int i = 0; i < 2; ++i)
     sb_cur[i] = __builtin_amdgcn_raw_buffer_load_b32(rsrc_b, vob[i], 0, 0);
+  }
 
   MXISSUE(0, Btile, 2);
   MXISSUE(0, Atile, 0);
   MXISSUE(1, Btile, 3);
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
 #define MXFRAG(DST, SLOT, ROW, C0)                                           \
@@ -664,23 +662,19 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
     // prefetch next tile's scale dwords via asm loads (invisible to hipcc's
     // waitcnt bookkeeping -> no pipeline-draining use-waits); drained by the
     // boundary vmcnt(2) below, which names them as operands.
-    unsigned sa_n[4], sb_n[2];
+    unsigned sa_n[2], sb_n[2];
     if (t + 1 < ntiles) {
       const unsigned so = (unsigned)((t + 1) * 4);
-      asm volatile("buffer_load_dword %0, %6, %12, %14 offen\n\t"
-                   "buffer_load_dword %1, %7, %12, %14 offen\n\t"
-                   "buffer_load_dword %2, %8, %12, %14 offen\n\t"
-                   "buffer_load_dword %3, %9, %12, %14 offen\n\t"
-                   "buffer_load_dword %4, %10, %13, %14 offen\n\t"
-                   "buffer_load_dword %5, %11, %13, %14 offen"
-                   : "=&v"(sa_n[0]), "=&v"(sa_n[1]), "=&v"(sa_n[2]),
-                     "=&v"(sa_n[3]), "=&v"(sb_n[0]), "=&v"(sb_n[1])
-                   : "v"(voa[0]), "v"(voa[1]), "v"(voa[2]), "v"(voa[3]),
-                     "v"(vob[0]), "v"(vob[1]),
+      asm volatile("buffer_load_dword %0, %4, %8, %10 offen\n\t"
+                   "buffer_load_dword %1, %5, %8, %10 offen\n\t"
+                   "buffer_load_dword %2, %6, %9, %10 offen\n\t"
+                   "buffer_load_dword %3, %7, %9, %10 offen"
+                   : "=&v"(sa_n[0]), "=&v"(sa_n[1]), "=&v"(sb_n[0]), "=&v"(sb_n[1])
+                   : "v"(voa[0]), "v"(voa[1]), "v"(vob[0]), "v"(vob[1]),
                      "s"(rsrc_a), "s"(rsrc_b), "s"(so)
                    : "memory");
     } else {
-      sa_n[0] = sa_n[1] = sa_n[2] = sa_n[3] = sb_n[0] = sb_n[1] = 0;
+      sa_n[0] = sa_n[1] = sb_n[0] = sb_n[1] = 0;
     }
     i32x8 af, bf[2];
     // kk = 0 (K 0..63 of the tile)
@@ -690,7 +684,7 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
     {
       const int sh = 8 * (0 * 2 + lhi);
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
+      for (int mi = 0; mi < 2; ++mi) {
         MXFRAG(af, aslot, a_row0 + mi * 32 + l31, 0 * 4 + lhi);
 #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
@@ -707,7 +701,7 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
     {
       const int sh = 8 * (1 * 2 + lhi);
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
+      for (int mi = 0; mi < 2; ++mi) {
         MXFRAG(af, aslot, a_row0 + mi * 32 + l31, 1 * 4 + lhi);
 #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
@@ -721,19 +715,15 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
     // B(t+2)'s 2 glds stay in flight), then rendezvous
     if (t >= ntiles - 2)
       asm volatile("s_waitcnt vmcnt(0)"
-                   : "+v"(sa_n[0]), "+v"(sa_n[1]), "+v"(sa_n[2]),
-                     "+v"(sa_n[3]), "+v"(sb_n[0]), "+v"(sb_n[1])
+                   : "+v"(sa_n[0]), "+v"(sa_n[1]), "+v"(sb_n[0]), "+v"(sb_n[1])
                    :: "memory");
     else
-      asm volatile("s_waitcnt vmcnt(4)"
-                   : "+v"(sa_n[0]), "+v"(sa_n[1]), "+v"(sa_n[2]),
-                     "+v"(sa_n[3]), "+v"(sb_n[0]), "+v"(sb_n[1])
+      asm volatile("s_waitcnt vmcnt(2)"
+                   : "+v"(sa_n[0]), "+v"(sa_n[1]), "+v"(sb_n[0]), "+v"(sb_n[1])
                    :: "memory");
     __builtin_amdgcn_s_barrier();
 #pragma unroll
-    for (int i = 0; i < 4; ++i) sa_cur[i] = sa_n[i];
-#pragma unroll
-    for (int i = 0; i < 2; ++i) sb_cur[i] = sb_n[i];
+    for (int i = 0; i < 2; ++i) { sa_cur[i] = sa_n[i]; sb_cur[i] = sb_n[i]; }
   }
 #undef MXFRAG
 #undef MXISSUE
@@ -741,7 +731,7 @@ gemm_bt_mxfp8_pipe_kernel(const unsigned char* __restrict__ A, const unsigned ch
   // 32x32 C map: col = l31 (+ni*32), row = (r&3) + 8*(r>>2) + 4*lhi
   const long long c_col0 = (long long)tile_n * 256 + b_row0 + l31;
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
+  for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const long long row = (long long)tile_m * 256 + a_row0 + mi * 32
