@@ -419,8 +419,11 @@ def attn_fwd_t(q: torch.Tensor, k: torch.Tensor, vt: torch.Tensor,
     if _on_gpu(q):
         ext = hip_ext()
         # v3 (8 waves sharing the staged K/V tile) when the grid still fills
-        # the chip at 2 blocks/CU; v2's smaller blocks otherwise
-        if _ATTN_IMPL != "v2" and q.shape[2] >= 256 and                 ((q.shape[2] + 255) // 256) * q.shape[0] * q.shape[1] >= 512:
+        # the chip at 2 blocks/CU; v2's smaller blocks otherwise.  S capped
+        # at 4096: v2's finer q-blocks win the causal load balance at 8192
+        # (404 vs 499 TF/s, profiles/r01_pmc_summary.txt) — the v3 pick at
+        # long S was a dispatch bug vs that measurement.
+        if _ATTN_IMPL != "v2" and 256 <= q.shape[2] <= 4096 and                 ((q.shape[2] + 255) // 256) * q.shape[0] * q.shape[1] >= 512:
             return ext.attn_fwd_v3(q.contiguous(), k.contiguous(),
                                    vt.contiguous(), scale)
         return ext.attn_fwd_v2(q.contiguous(), k.contiguous(),
