@@ -1,7 +1,7 @@
 """Decode (textual-gradient generation) benchmark: tokens/s for greedy decode."""
 import sys, time
 import torch
-sys.path.insert(0, ".")
+import os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from senweaver_amd.engine.scorer import LlamaBackend
 
 def main():
@@ -12,14 +12,18 @@ def main():
     torch.cuda.synchronize()
     print(f"init+warmup {time.perf_counter()-t0:.1f}s")
     n = 64
+    count = [0]
     t0 = time.perf_counter()
-    out = backend.generate("- improve the rules for tool use and verification " * 10,
-                           max_new_tokens=n)
+    out = backend.stream_generate(
+        "- improve the rules for tool use and verification " * 10,
+        max_new_tokens=n, should_stop=lambda: False,
+        on_chunk=lambda _t: count.__setitem__(0, count[0] + 1))
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
-    ntok = len(out.split()) if out else n
+    ntok = max(count[0], 1)
     print(f"decode: {ntok} tokens in {dt:.2f}s = {ntok/dt:.1f} tok/s "
-          f"({1000*dt/max(ntok,1):.1f} ms/tok)")
+          f"({1000*dt/ntok:.1f} ms/tok)")
+    print("sample:", out[:120].replace("\n", " "))
 
 if __name__ == "__main__":
     main()

@@ -25,7 +25,10 @@ class DecodeGraph:
         dev = model.device
         self.tokens = torch.zeros(batch, dtype=torch.long, device=dev)
         self.pos32 = torch.zeros(batch, dtype=torch.int32, device=dev)
-        self.slot = torch.zeros(batch, dtype=torch.long, device=dev)
+        # int32: rope_kv_append consumes slot as int32 — a long here costs a
+        # cast kernel PER LAYER per replayed step (the 4.7 us elementwise in
+        # profiles/r01_decode_breakdown2.txt, ~3% of decode)
+        self.slot = torch.zeros(batch, dtype=torch.int32, device=dev)
         self.bt = torch.zeros(batch, max_pages, dtype=torch.int32, device=dev)
         self.ctx = torch.zeros(batch, dtype=torch.int32, device=dev)
         self.graph: Optional[torch.cuda.CUDAGraph] = None

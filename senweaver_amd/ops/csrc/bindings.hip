@@ -56,8 +56,9 @@ extern "C" __global__ void gemv_norm_bt_bf16_m16(const ushort*, const ushort*, c
 extern "C" __global__ void attn_fwd_bf16_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
 extern "C" __global__ void attn_fwd_v2_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
 extern "C" __global__ void attn_fwd_v3_kernel(const ushort*, const ushort*, const ushort*, ushort*, int, int, int, int, float);
-extern "C" __global__ void paged_decode_attn_kernel(const ushort*, const ushort*, const ushort*, ushort*,
-                                                    const int*, const int*, int, int, int, float);
+extern "C" __global__ void paged_decode_attn_partial_kernel(const ushort*, const ushort*, const ushort*, float*,
+                                                            const int*, const int*, int, int, int, float);
+extern "C" __global__ void paged_decode_attn_merge_kernel(const float*, ushort*, int);
 
 namespace {
 
@@ -488,10 +489,17 @@ torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor kcache,
   TORCH_CHECK(kcache.size(1) == 16, "PAGE_SIZE=16");
   const int max_pages = block_table.size(1);
   auto o = torch::empty_like(q);
-  paged_decode_attn_kernel<<<dim3(H, B), dim3(256), 0, cur_stream()>>>(
-      bf16_ptr(q), bf16_ptr(kcache), bf16_ptr(vcache), bf16_mut(o),
+  // flash-decoding split-8: partial kernel fills the chip (H*B*8 blocks),
+  // merge kernel combines slices.  Both graph-capturable.
+  auto partials = torch::empty({B, H, 8, 2 + 128},
+                               q.options().dtype(torch::kFloat32));
+  paged_decode_attn_partial_kernel<<<dim3(H, B, 8), dim3(256), 0, cur_stream()>>>(
+      bf16_ptr(q), bf16_ptr(kcache), bf16_ptr(vcache),
+      partials.data_ptr<float>(),
       block_table.data_ptr<int>(), ctx_lens.data_ptr<int>(), H, Hk, max_pages,
       (float)scale);
+  paged_decode_attn_merge_kernel<<<dim3(H, B), dim3(64), 0, cur_stream()>>>(
+      partials.data_ptr<float>(), bf16_mut(o), H);
   HIP_CHECK_KERNEL();
   return o;
 }
