@@ -282,7 +282,9 @@ gemm_bt_bf16_256_kernel(const ushort* __restrict__ A, const ushort* __restrict__
     float acc[MM];                                                            \
     _Pragma("unroll") for (int m = 0; m < MM; ++m) acc[m] = 0.f;              \
     for (int k = lane * 8; k < K; k += 64 * 8) {                              \
-      bf16x8 bv = *reinterpret_cast<const bf16x8*>(brow + k);                 \
+      short8 bv_ = __builtin_nontemporal_load(                                \
+          reinterpret_cast<const short8*>(brow + k));                         \
+      bf16x8 bv = *reinterpret_cast<const bf16x8*>(&bv_);                     \
       float bfv[8];                                                           \
       _Pragma("unroll") for (int j = 0; j < 8; ++j) bfv[j] = bf2f(bv.v[j]);   \
       _Pragma("unroll") for (int m = 0; m < MM; ++m) {                        \
@@ -526,7 +528,9 @@ gemm_bt_bf16_256sg_kernel(const ushort* __restrict__ A, const ushort* __restrict
     float dot[MM], ss[MM];                                                    \
     _Pragma("unroll") for (int m = 0; m < MM; ++m) { dot[m] = 0.f; ss[m] = 0.f; } \
     for (int k = lane * 8; k < K; k += 64 * 8) {                              \
-      bf16x8 bv = *reinterpret_cast<const bf16x8*>(brow + k);                 \
+      short8 bv_ = __builtin_nontemporal_load(                                \
+          reinterpret_cast<const short8*>(brow + k));                         \
+      bf16x8 bv = *reinterpret_cast<const bf16x8*>(&bv_);                     \
       bf16x8 wv = *reinterpret_cast<const bf16x8*>(normw + k);                \
       float bw[8];                                                            \
       _Pragma("unroll") for (int j = 0; j < 8; ++j)                           \
