@@ -281,10 +281,16 @@ gemm_bt_bf16_256_kernel(const ushort* __restrict__ A, const ushort* __restrict__
     const ushort* brow = B + (long long)n * K;                                \
     float acc[MM];                                                            \
     _Pragma("unroll") for (int m = 0; m < MM; ++m) acc[m] = 0.f;              \
+    /* 2-deep software pipeline: the runtime-K loop otherwise keeps ONE     \
+       weight load in flight per lane (latency-bound on the small rows) */   \
+    short8 cur_ = __builtin_nontemporal_load(                                 \
+        reinterpret_cast<const short8*>(brow + lane * 8));                    \
     for (int k = lane * 8; k < K; k += 64 * 8) {                              \
-      short8 bv_ = __builtin_nontemporal_load(                                \
-          reinterpret_cast<const short8*>(brow + k));                         \
-      bf16x8 bv = *reinterpret_cast<const bf16x8*>(&bv_);                     \
+      short8 nxt_;                                                            \
+      if (k + 64 * 8 < K)                                                     \
+        nxt_ = __builtin_nontemporal_load(                                    \
+            reinterpret_cast<const short8*>(brow + k + 64 * 8));              \
+      bf16x8 bv = *reinterpret_cast<const bf16x8*>(&cur_);                    \
       float bfv[8];                                                           \
       _Pragma("unroll") for (int j = 0; j < 8; ++j) bfv[j] = bf2f(bv.v[j]);   \
       _Pragma("unroll") for (int m = 0; m < MM; ++m) {                        \
@@ -294,6 +300,7 @@ gemm_bt_bf16_256_kernel(const ushort* __restrict__ A, const ushort* __restrict__
             s += bf2f(av.v[j]) * bfv[j];                                      \
         acc[m] += s;                                                          \
       }                                                                       \
+      cur_ = nxt_;                                                            \
     }                                                                         \
     _Pragma("unroll") for (int m = 0; m < MM; ++m) {                          \
       float v = wave_reduce_sum(acc[m]);                                      \
