@@ -383,6 +383,88 @@ class ToolsService:
         return ToolResult("edit_agent", {"uri": uri, "mode": mode},
                           f"{mode} applied to {uri} ({len(new_content)} chars)")
 
+    # --- document family (offline backends: tools/documents.py) ---
+    def _tool_read_document(self, p):
+        from . import documents as docs
+        path = self._resolve(p["uri"]) if not os.path.isabs(p.get("uri", "")) \
+            else p["uri"]
+        try:
+            text = docs.read_document(path)
+        except docs.DocumentError as e:
+            raise ToolError(str(e))
+        start = int(p.get("start_index") or 0)
+        maxlen = int(p.get("max_length") or MAX_FILE_CHARS)
+        chunk = text[start: start + maxlen]
+        return ToolResult("read_document",
+                          {"uri": p["uri"], "total_chars": len(text),
+                           "start_index": start}, chunk)
+
+    def _tool_edit_document(self, p):
+        from . import documents as docs
+        path = self._resolve(p["uri"])
+        reps = p.get("replacements")
+        if isinstance(reps, str):
+            import json as _json
+            try:
+                reps = _json.loads(reps)
+            except ValueError:
+                raise ToolError("replacements must be a JSON array")
+        try:
+            docs.edit_document(path, p.get("content"), reps,
+                               backup=str(p.get("backup", "")).lower() in ("true", "1"))
+        except docs.DocumentError as e:
+            raise ToolError(str(e))
+        return ToolResult("edit_document", {"uri": p["uri"]},
+                          f"edited {p['uri']}")
+
+    def _tool_create_document(self, p):
+        from . import documents as docs
+        path = self._resolve(p.get("file_path") or "document.docx")
+        try:
+            out = docs.create_document(p.get("type") or "word", path,
+                                       p.get("document_data") or "")
+        except docs.DocumentError as e:
+            raise ToolError(str(e))
+        return ToolResult("create_document", {"path": out}, f"created {out}")
+
+    def _tool_document_convert(self, p):
+        from . import documents as docs
+        src = self._resolve(p["input_file"])
+        dst = self._resolve(p.get("output_path") or
+                            os.path.splitext(src)[0] + ".txt")
+        try:
+            out = docs.convert_document(src, dst, p.get("format"))
+        except docs.DocumentError as e:
+            raise ToolError(str(e))
+        return ToolResult("document_convert", {"path": out}, f"converted to {out}")
+
+    def _tool_document_extract(self, p):
+        from . import documents as docs
+        src = self._resolve(p["input_file"])
+        out_dir = self._resolve(p.get("output_dir") or ".")
+        try:
+            written = docs.extract_document(src, out_dir,
+                                            p.get("extract_type") or "text")
+        except docs.DocumentError as e:
+            raise ToolError(str(e))
+        return ToolResult("document_extract", {"files": written},
+                          "\n".join(written))
+
+    def _tool_document_merge(self, p):
+        from . import documents as docs
+        files = p.get("input_files")
+        if isinstance(files, str):
+            files = [f.strip() for f in files.split(",") if f.strip()]
+        if not files:
+            raise ToolError("document_merge needs input_files")
+        srcs = [self._resolve(f) for f in files]
+        dst = self._resolve(p.get("output_path") or "merged.txt")
+        try:
+            out = docs.merge_documents(srcs, dst)
+        except docs.DocumentError as e:
+            raise ToolError(str(e))
+        return ToolResult("document_merge", {"path": out}, f"merged to {out}")
+
     def _tool_skill(self, p):
         from .skills import SkillService
         svc = SkillService(self.root)

@@ -1,0 +1,118 @@
+"""Offline document/pdf tool backends.
+
+The reference serves these from the documentReader sidecar (port 3008,
+toolsService.ts:2744+); here they are stdlib OOXML/PDF implementations
+(tools/documents.py) behind the same tool names/params.
+"""
+
+import os
+import zipfile
+import zlib
+
+import pytest
+
+from senweaver_amd.tools import documents as docs
+from senweaver_amd.tools.service import ToolError, ToolsService
+
+
+@pytest.fixture()
+def ts(tmp_path):
+    return ToolsService(str(tmp_path))
+
+
+def test_docx_roundtrip(ts, tmp_path):
+    content = "# Report\n\nFirst paragraph with **bold** text.\nplain line"
+    ts.call_tool("create_document", {"type": "word", "file_path": "r.docx",
+                                     "document_data": content})
+    # a real zip with the OOXML parts
+    with zipfile.ZipFile(tmp_path / "r.docx") as z:
+        assert "word/document.xml" in z.namelist()
+        assert "[Content_Types].xml" in z.namelist()
+    out = ts.call_tool("read_document", {"uri": "r.docx"})
+    assert "# Report" in out.text          # heading style reconstructed
+    assert "bold" in out.text and "plain line" in out.text
+
+
+def test_docx_replacements_and_backup(ts, tmp_path):
+    ts.call_tool("create_document", {"type": "word", "file_path": "e.docx",
+                                     "document_data": "alpha beta gamma"})
+    ts.call_tool("edit_document", {
+        "uri": "e.docx", "backup": "true",
+        "replacements": '[{"find": "beta", "replace": "delta"}]'})
+    assert (tmp_path / "e.docx.bak").exists()
+    out = ts.call_tool("read_document", {"uri": "e.docx"})
+    assert "delta" in out.text and "beta" not in out.text
+
+
+def test_replacement_target_missing_errors(ts):
+    ts.call_tool("create_document", {"type": "word", "file_path": "x.docx",
+                                     "document_data": "abc"})
+    with pytest.raises(ToolError, match="not found"):
+        ts.call_tool("edit_document", {
+            "uri": "x.docx",
+            "replacements": '[{"find": "zzz", "replace": "q"}]'})
+
+
+def test_xlsx_roundtrip(ts):
+    ts.call_tool("create_document", {"type": "excel", "file_path": "t.xlsx",
+                                     "document_data": "name,qty\nwidget,3"})
+    out = ts.call_tool("read_document", {"uri": "t.xlsx"})
+    assert "name,qty" in out.text and "widget,3" in out.text
+
+
+def test_convert_and_merge(ts, tmp_path):
+    ts.call_tool("create_document", {"type": "word", "file_path": "a.docx",
+                                     "document_data": "doc A"})
+    (tmp_path / "b.md").write_text("doc B")
+    ts.call_tool("document_convert", {"input_file": "a.docx",
+                                      "output_path": "a.html", "format": "html"})
+    assert "<p>doc A</p>" in (tmp_path / "a.html").read_text()
+    ts.call_tool("document_merge", {"input_files": "a.docx,b.md",
+                                    "output_path": "m.txt"})
+    m = (tmp_path / "m.txt").read_text()
+    assert "doc A" in m and "doc B" in m
+
+
+def test_read_document_paging(ts, tmp_path):
+    (tmp_path / "long.txt").write_text("x" * 100)
+    out = ts.call_tool("read_document", {"uri": "long.txt", "start_index": 10,
+                                         "max_length": 5})
+    assert out.text == "xxxxx"
+    assert out.result["total_chars"] == 100
+
+
+def _tiny_pdf(path, text):
+    """Hand-built single-page PDF with a Flate-compressed text stream."""
+    stream = f"BT /F1 12 Tf 72 700 Td ({text}) Tj ET".encode()
+    comp = zlib.compress(stream)
+    objs = []
+    objs.append(b"1 0 obj\n<< /Type /Catalog /Pages 2 0 R >>\nendobj\n")
+    objs.append(b"2 0 obj\n<< /Type /Pages /Kids [3 0 R] /Count 1 >>\nendobj\n")
+    objs.append(b"3 0 obj\n<< /Type /Page /Parent 2 0 R /Contents 4 0 R >>\nendobj\n")
+    objs.append(b"4 0 obj\n<< /Length " + str(len(comp)).encode()
+                + b" /Filter /FlateDecode >>\nstream\n" + comp
+                + b"\nendstream\nendobj\n")
+    body = b"%PDF-1.4\n" + b"".join(objs) + b"trailer\n<< /Root 1 0 R >>\n%%EOF\n"
+    path.write_bytes(body)
+
+
+def test_pdf_text_extraction(ts, tmp_path):
+    _tiny_pdf(tmp_path / "doc.pdf", "Hello PDF world")
+    out = ts.call_tool("read_document", {"uri": "doc.pdf"})
+    assert "Hello PDF world" in out.text
+    assert docs.pdf_page_count(str(tmp_path / "doc.pdf")) == 1
+
+
+def test_document_extract_text(ts, tmp_path):
+    ts.call_tool("create_document", {"type": "word", "file_path": "d.docx",
+                                     "document_data": "extract me"})
+    out = ts.call_tool("document_extract", {"input_file": "d.docx",
+                                            "output_dir": "ex"})
+    assert (tmp_path / "ex" / "d.txt").read_text().strip() == "extract me"
+
+
+def test_network_tools_still_offline(ts):
+    with pytest.raises(ToolError, match="offline"):
+        ts.call_tool("web_search", {"query": "x"})
+    with pytest.raises(ToolError, match="offline"):
+        ts.call_tool("pdf_operation", {"operation": "merge"})
