@@ -22,6 +22,8 @@ each user message and at stream end (:1734, 2710).
 
 from __future__ import annotations
 
+import threading
+
 import time
 import uuid
 from dataclasses import dataclass, field
@@ -85,6 +87,9 @@ class ChatThreadService:
         self._rate_limiter = TPMRateLimiter()
 
     # ---- thread state ----
+        from ..context.compress import EnhancedContextManager
+        self._ctx_mgr = EnhancedContextManager(
+            context_limit=self.settings.context_window)
 
     def open_thread(self) -> ChatThread:
         t = ChatThread(id=str(uuid.uuid4()))
@@ -146,8 +151,12 @@ class ChatThreadService:
                 if wait_ms > 0:
                     self._sleep(wait_ms / 1000)
 
+                # in-loop per-tool compaction (chatThreadService.ts:1456-1461):
+                # above 55% occupancy the oldest/largest tool outputs are
+                # pruned before the next send
+                history = self._ctx_mgr.maybe_prune(self._history(thread))
                 sys_msg, fitted = self._converter.prepare_llm_chat_messages(
-                    self._history(thread), self.settings.chat_mode,
+                    history, self.settings.chat_mode,
                     self.settings.context_window)
                 messages = [LLMChatMessage("system", sys_msg)] + [
                     LLMChatMessage(m.role, m.content) for m in fitted]
@@ -162,11 +171,27 @@ class ChatThreadService:
                                                       model="local", provider="senweaver_amd")
                 if tool_call is None:
                     return  # conversation turn complete
+                # tool-call || system-message warmup overlap
+                # (chatThreadService.ts:1659-1684): while the tool runs, the
+                # next turn's system message is generated on a side thread so
+                # the post-tool prepare hits the cache
+                warm = threading.Thread(
+                    target=self._warm_system_message, daemon=True)
+                warm.start()
                 should_continue = self._run_tool_call(thread, tool_call)
+                warm.join(timeout=10)
                 if not should_continue:
                     return
         finally:
             thread.streaming = False
+
+
+    def _warm_system_message(self) -> None:
+        """Populate the converter's system-message cache (runs beside a tool)."""
+        try:
+            self._converter.generate_system_message(self.settings.chat_mode)
+        except Exception:
+            pass
 
     def _send_with_retries(self, thread: ChatThread, messages: List[LLMChatMessage],
                            max_new_tokens: int):

@@ -80,11 +80,15 @@ class PredictionCache:
 
 
 class EditPredictionService:
-    def __init__(self, backend, max_history: int = 8) -> None:
+    def __init__(self, backend, max_history: int = 8,
+                 context_gatherer=None) -> None:
         self._backend = backend
         self._history: List[EditEvent] = []
         self._max_history = max_history
         self.cache = PredictionCache()
+        # the editor-side half (contextGatheringService.ts): cursor-proximity
+        # snippets feed the prediction prompt
+        self._gatherer = context_gatherer
 
     def record_edit(self, uri: str, before: str, after: str) -> None:
         self._history.append(EditEvent(uri, before, after))
@@ -98,7 +102,12 @@ class EditPredictionService:
         history = "\n".join(
             f"- in {e.uri}: {e.before[:80]!r} -> {e.after[:80]!r}"
             for e in self._history[-4:])
-        prompt = (f"Recent edits:\n{history}\n\n"
+        gathered = ""
+        if self._gatherer is not None:
+            snips = self._gatherer.get_cached_snippets()[:4]
+            if snips:
+                gathered = "Workspace context:\n" + "\n---\n".join(snips) + "\n\n"
+        prompt = (f"{gathered}Recent edits:\n{history}\n\n"
                   f"Code near cursor in {uri}:\n{cursor_context[-500:]}\n\n"
                   "Predict the next edit the user will make (answer with the "
                   "edited code only):")
