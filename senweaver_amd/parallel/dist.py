@@ -31,7 +31,11 @@ def init_from_env(device: Optional[torch.device] = None) -> tuple:
         return 0, 1
     rank = int(os.environ["RANK"])
     world = int(os.environ["WORLD_SIZE"])
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    # SENWEAVER_DIST_BACKEND=gloo lets world>1 share ONE GPU (RCCL refuses
+    # co-located ranks — profiles/r02_rccl_world2.txt); default is RCCL.
+    backend = os.environ.get(
+        "SENWEAVER_DIST_BACKEND",
+        "nccl" if torch.cuda.is_available() else "gloo")
     if torch.cuda.is_available():
         # modulo: ranks may oversubscribe one device (RCCL permits multiple
         # ranks per GPU — how the world>1 path is burned in on a 1-GPU box)
