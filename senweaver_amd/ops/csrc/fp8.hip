@@ -425,9 +425,9 @@ gemm_bt_mxfp8_256_kernel(const unsigned char* __restrict__ A, const unsigned cha
   const unsigned char* Btile = B + (long long)tile_n * 256 * K;
   const int sld = K / 32;
 
-  f32x16 acc[2][2];
+  f32x16 acc[4][2];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j)
 #pragma unroll
