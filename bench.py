@@ -118,7 +118,12 @@ def main() -> int:
     rank, world = P.init_from_env()
     n_gpus = max(args.gpus, world)
     use_cuda = torch.cuda.is_available()
-    device = args.device or (f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}" if use_cuda else "cpu")
+    # modulo: ranks may oversubscribe one device (gloo-backend TP/DP burn-in
+    # on a 1-GPU box; RCCL refuses co-located ranks — profiles/r02_rccl_world2.txt)
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    if use_cuda:
+        local_rank %= torch.cuda.device_count()
+    device = args.device or (f"cuda:{local_rank}" if use_cuda else "cpu")
 
     from senweaver_amd.engine.scorer import LlamaBackend
     from senweaver_amd.models.config import get_config
