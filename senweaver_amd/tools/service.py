@@ -383,6 +383,51 @@ class ToolsService:
         return ToolResult("edit_agent", {"uri": uri, "mode": mode},
                           f"{mode} applied to {uri} ({len(new_content)} chars)")
 
+    def _tool_api_request(self, p):
+        """Custom/user API invocation (customApiService.ts + api_request tool).
+
+        Offline environment: only loopback targets are reachable — a real
+        local service (the engine daemon, an MCP HTTP server, a test
+        server) can be called; anything else raises the structured offline
+        error the other network tools use."""
+        import json as _json
+        import urllib.request
+        import urllib.error
+        from urllib.parse import urlparse
+        url = p.get("url") or ""
+        host = urlparse(url).hostname or ""
+        if host not in ("127.0.0.1", "localhost", "::1"):
+            raise ToolError(
+                "api_request can only reach loopback targets in this "
+                "environment (reference proxies it via a localhost sidecar; "
+                "offline here)")
+        method = (p.get("method") or "GET").upper()
+        body = p.get("body")
+        data = None
+        if body not in (None, ""):
+            data = body.encode() if isinstance(body, str) else _json.dumps(body).encode()
+        req = urllib.request.Request(url, data=data, method=method)
+        headers = p.get("headers")
+        if isinstance(headers, str) and headers.strip():
+            try:
+                headers = _json.loads(headers)
+            except ValueError:
+                raise ToolError("headers must be a JSON object")
+        for k, v in (headers or {}).items():
+            req.add_header(k, str(v))
+        timeout = min(int(p.get("timeout") or 10000), 60000) / 1000
+        try:
+            with urllib.request.urlopen(req, timeout=timeout) as resp:
+                text = resp.read(1_000_000).decode("utf-8", "replace")
+                status = resp.status
+        except urllib.error.HTTPError as e:
+            text = e.read(100_000).decode("utf-8", "replace")
+            status = e.code
+        except (urllib.error.URLError, OSError) as e:
+            raise ToolError(f"api_request failed: {e}")
+        return ToolResult("api_request", {"status": status},
+                          f"HTTP {status}\n{text[:MAX_FILE_CHARS]}")
+
     # --- document family (offline backends: tools/documents.py) ---
     def _tool_read_document(self, p):
         from . import documents as docs

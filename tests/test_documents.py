@@ -116,3 +116,62 @@ def test_network_tools_still_offline(ts):
         ts.call_tool("web_search", {"query": "x"})
     with pytest.raises(ToolError, match="offline"):
         ts.call_tool("pdf_operation", {"operation": "merge"})
+
+
+def test_api_request_loopback(ts):
+    """api_request executes against a real loopback HTTP server; non-local
+    targets stay structured offline errors (CustomApiService pairing)."""
+    import http.server
+    import threading
+
+    class H(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            n = int(self.headers.get("Content-Length", 0))
+            body = self.rfile.read(n)
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(b"echo:" + body)
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), H)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        out = ts.call_tool("api_request", {
+            "url": f"http://127.0.0.1:{srv.server_port}/x",
+            "method": "POST", "body": "ping"})
+        assert "echo:ping" in out.text and out.result["status"] == 200
+        with pytest.raises(ToolError, match="loopback"):
+            ts.call_tool("api_request", {"url": "https://example.com"})
+    finally:
+        srv.shutdown()
+
+
+def test_custom_api_service_registry(tmp_path):
+    from senweaver_amd.features.customapi import (
+        CustomApiField, CustomApiService, CUSTOM_API_STORAGE_KEY,
+    )
+    from senweaver_amd.storage import FileStorage
+
+    store = FileStorage(str(tmp_path / "state.json"))
+    svc = CustomApiService(storage=store, clock=lambda: 1234)
+    api = svc.add_api("Weather", "http://127.0.0.1:9999/w", "POST",
+                      "look up weather",
+                      fields=[CustomApiField("city", "string", True, "城市名")],
+                      response_description="JSON with temp")
+    assert svc.get_api(api.id).name == "Weather"
+    desc = svc.get_api_list_description()
+    assert "## Weather" in desc and "必填" in desc and "api_request" in desc
+    # restart-resume through the same storage key (FileStorage flushes on
+    # demand, like the reference's 30s/dispose flush cycle)
+    store.flush()
+    svc2 = CustomApiService(storage=FileStorage(str(tmp_path / "state.json")))
+    assert [a.name for a in svc2.get_enabled_apis()] == ["Weather"]
+    svc2.update_api(api.id, enabled=False)
+    assert svc2.get_enabled_apis() == []
+    assert svc2.get_api_list_description() == ""
+    svc2.delete_api(api.id)
+    assert svc2.state == {"apis": []}
+    assert CUSTOM_API_STORAGE_KEY == "senweaver.customApis"
