@@ -257,15 +257,20 @@ class ToolsService:
         return ToolResult("search_in_file", {"lines": nums}, ", ".join(map(str, nums)) or "(no matches)")
 
     def _tool_read_lint_errors(self, p):
-        # local lint: python syntax check for .py; otherwise none available
+        # marker-service backed (features/markers.py = _markerCheckService.ts
+        # analog): lint the file through the registered providers, publish
+        # markers, and report Error-severity diagnostics
+        from ..features.markers import MarkerService, python_lint
         path = self._resolve(p["uri"])
-        errors: List[str] = []
+        if not hasattr(self, "marker_service"):
+            self.marker_service = MarkerService()
+        markers = []
         if path.endswith(".py"):
-            import ast
-            try:
-                ast.parse(open(path, encoding="utf-8", errors="replace").read())
-            except SyntaxError as e:
-                errors.append(f"{e.lineno}: {e.msg}")
+            with open(path, encoding="utf-8", errors="replace") as f:
+                markers = python_lint(path, f.read())
+        self.marker_service.changed(path, markers)
+        errors = [f"{m.startLineNumber}: {m.message}"
+                  for m in self.marker_service.read(resource=path)]
         text = "\n".join(errors) if errors else "No lint errors found."
         return ToolResult("read_lint_errors", {"errors": errors}, text)
 
