@@ -854,3 +854,145 @@ gemm_bt_bf16_8ph_v3_kernel(const ushort* __restrict__ A, const ushort* __restric
                            ushort* __restrict__ C, int M, int N, int K) {
   gemm8ph_body<3>(A, B, C, M, N, K);
 }
+
+// ---------------------------------------------------------------------------
+// v16: TWO INDEPENDENT BLOCKS PER CU.  All prior bodies use >=128 KiB LDS
+// (1 block/CU), so every rendezvous parks the whole CU (PMC: WAIT 26-36%).
+// This body shrinks to BK=32 / 16 KiB stage units (A dbuf + B ring-3 =
+// 80 KiB) and 512 threads at <=256 VGPR -> 2 blocks/CU of 8 waves each:
+// the two blocks' barriers are independent, so one block's arrival skew
+// hides behind the other block's MFMA stream.  One phase per K-tile:
+// {A kk reads (8 b128) + B (4) | issue A(t+1), B(t+2) | 32 MFMA |
+//  vmcnt(2) | BAR}.  Same 128-B-row staging + (c + 2*(row>>1))&7 swizzle
+// (chunk c in 0..3 here: rows are 64 B... no — BK=32 bf16 = 64-B rows).
+// NOTE rows are 64 B (4 chunks): the request-granularity cost that hurt
+// v0-v3 at 8k returns; the bet is that 2-block overlap outweighs it.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(512, 2)
+gemm_bt_bf16_8ph_v16_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                            ushort* __restrict__ C, int M, int N, int K) {
+  const int nwg = (M / PBM) * (N / PBN);
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int tiles_n = N / PBN;
+  const int GM = 8;
+  const int tiles_m = M / PBM;
+  const int grp = wgid / (GM * tiles_n);
+  const int rem = wgid % (GM * tiles_n);
+  const int g0 = grp * GM;
+  const int gh = (tiles_m - g0 < GM) ? (tiles_m - g0) : GM;
+  const int tile_m = g0 + rem % gh;
+  const int tile_n = rem / gh;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;            // 0..1 -> 128-row half of A
+  const int wn = wid & 3;             // 0..3 -> 64-row band of B
+  const int l15 = lane & 15;
+  const int kgrp = lane >> 4;         // chunk 0..3 (64-B rows)
+
+  // 5 x 16 KiB slots = 80 KiB: A in 0,1; B ring in 2,3,4
+  __shared__ __attribute__((aligned(16))) ushort lds[5][256 * 32];
+
+  const ushort* Atile = A + (long long)tile_m * PBM * K;
+  const ushort* Btile = B + (long long)tile_n * PBN * K;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // 64-B rows, 4 chunks: swizzle (c + 2*(row>>1)) & 3 spreads the two
+  // same-parity rows of each ds_read_b128 lane-group class (<=2-way; the
+  // 4-chunk space cannot reach the full 8-way spread of the 128-B layout)
+  const int swz0 = (kgrp + 2 * ((l15 >> 1) & 1)) & 3;
+  const int frag0 = l15 * 64 + swz0 * 16;
+  const int a_off = wm * 128 * 64 + frag0;   // + mi*16*64
+  const int b_off = wn * 64 * 64 + frag0;    // + ni*16*64
+
+  unsigned st_off[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int s = i * 512 + tid;
+    const int r = s >> 2;
+    st_off[i] = (unsigned)r * (unsigned)K
+                + (((((s & 3) - 2 * ((r >> 1) & 1)) & 3) * 8));
+  }
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / 32;
+
+#define ISSUE16(TGT, OP, SLOT)                                               \
+  do {                                                                       \
+    if ((TGT) < ntiles) {                                                    \
+      const int k0_ = (TGT) * 32;                                            \
+      const ushort* opk_ = (OP) + k0_;                                       \
+      ushort* dst_ = &lds[(SLOT)][0];                                        \
+      _Pragma("unroll") for (int i = 0; i < 2; ++i) {                        \
+        const ushort* g = opk_ + st_off[i];                                  \
+        __builtin_amdgcn_global_load_lds(                                    \
+            (const __attribute__((address_space(1))) unsigned int*)g,        \
+            (__attribute__((address_space(3))) unsigned int*)(dst_ +         \
+                (long long)(i * 512 + wave_chunk) * 8),                      \
+            16, 0, 0);                                                       \
+      }                                                                      \
+    }                                                                        \
+  } while (0)
+
+  ISSUE16(0, Btile, 2);
+  ISSUE16(0, Atile, 0);
+  ISSUE16(1, Btile, 3);
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+#define LOAD16(DST, SLOT, OFF, MI0, NFRAG)                                   \
+  _Pragma("unroll") for (int j = 0; j < (NFRAG); ++j) {                      \
+    DST[j] = *reinterpret_cast<const short8*>(                               \
+        reinterpret_cast<const char*>(&lds[(SLOT)][0]) +                     \
+        ((OFF) + ((MI0) + j) * 1024));                                       \
+  }
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int aslot = t & 1;
+    const int bslot = 2 + t % 3;
+    const int bslot2 = 2 + (t + 2) % 3;
+    short8 af[8], bf[4];
+    LOAD16(af, aslot, a_off, 0, 8);
+    LOAD16(bf, bslot, b_off, 0, 4);
+    ISSUE16(t + 1, Atile, aslot ^ 1);
+    ISSUE16(t + 2, Btile, bslot2);
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    if (t >= ntiles - 2)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+#undef LOAD16
+#undef ISSUE16
+
+  const long long c_row0 = (long long)tile_m * PBM + wm * 128 + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * PBN + wn * 64 + l15;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}
