@@ -48,6 +48,7 @@ for (M, N, K) in shapes:
         "v13-16w": lambda: ext.gemm_bt_8ph_v(a, b, 13),
         "v14-16n": lambda: ext.gemm_bt_8ph_v(a, b, 14),
         "v16-bk32": lambda: ext.gemm_bt_8ph_v(a, b, 16),
+        "v17-2blk": lambda: ext.gemm_bt_8ph_v(a, b, 17),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas

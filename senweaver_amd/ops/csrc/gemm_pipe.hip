@@ -996,3 +996,158 @@ gemm_bt_bf16_8ph_v16_kernel(const ushort* __restrict__ A, const ushort* __restri
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// v17: the 2-blocks-per-CU premise in its proper geometry.  256x128 tile,
+// 8 waves as 4M x 2N (wave tile 64x64 -> 64 acc VGPRs), BK=32 stage units
+// (A 16 KiB dbuf + B 8 KiB ring-3 = 56 KiB LDS), __launch_bounds__(512, 4)
+// capping at 128 VGPRs -> TWO independent blocks per CU: each block's
+// per-tile vmcnt+barrier parks only ITSELF while the co-resident block's
+// MFMA stream keeps the SIMDs fed (the single-block bodies park the whole
+// CU: PMC WAIT 26-36%).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(512, 4)
+gemm_bt_bf16_8ph_v17_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                            ushort* __restrict__ C, int M, int N, int K) {
+  const int tiles_n = N / 128;
+  const int tiles_m = M / 256;
+  const int nwg = tiles_m * tiles_n;
+  int wgid = blockIdx.x;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int GM = 8;
+  const int grp = wgid / (GM * tiles_n);
+  const int rem = wgid % (GM * tiles_n);
+  const int g0 = grp * GM;
+  const int gh = (tiles_m - g0 < GM) ? (tiles_m - g0) : GM;
+  const int tile_m = g0 + rem % gh;
+  const int tile_n = rem / gh;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1;            // 0..3 -> 64-row band of A
+  const int wn = wid & 1;             // 0..1 -> 64-row band of B
+  const int l15 = lane & 15;
+  const int kgrp = lane >> 4;         // chunk 0..3 (64-B rows)
+
+  // A slots 0,1: 256x32 (16 KiB); B ring slots at byte offsets after them:
+  // one shared array (multiple __shared__ objects de-pipeline glds)
+  __shared__ __attribute__((aligned(16))) ushort lds[(2 * 256 + 3 * 128) * 32];
+#define V17_A(b) (&lds[(b) * 256 * 32])
+#define V17_B(r) (&lds[2 * 256 * 32 + (r) * 128 * 32])
+
+  const ushort* Atile = A + (long long)tile_m * 256 * K;
+  const ushort* Btile = B + (long long)tile_n * 128 * K;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int swz0 = (kgrp + 2 * ((l15 >> 1) & 1)) & 3;
+  const int frag0 = l15 * 64 + swz0 * 16;
+  const int a_off = wm * 64 * 64 + frag0;   // + mi*16*64
+  const int b_off = wn * 64 * 64 + frag0;   // + ni*16*64
+
+  unsigned st_a[2];
+  unsigned st_b;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int s = i * 512 + tid;
+    const int r = s >> 2;
+    st_a[i] = (unsigned)r * (unsigned)K
+              + (((((s & 3) - 2 * ((r >> 1) & 1)) & 3) * 8));
+  }
+  {
+    const int r = tid >> 2;
+    st_b = (unsigned)r * (unsigned)K
+           + (((((tid & 3) - 2 * ((r >> 1) & 1)) & 3) * 8));
+  }
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / 32;
+
+#define ISSUE17A(TGT, BUF)                                                   \
+  do {                                                                       \
+    if ((TGT) < ntiles) {                                                    \
+      const ushort* opk_ = Atile + (TGT) * 32;                               \
+      ushort* dst_ = V17_A(BUF);                                             \
+      _Pragma("unroll") for (int i = 0; i < 2; ++i) {                        \
+        __builtin_amdgcn_global_load_lds(                                    \
+            (const __attribute__((address_space(1))) unsigned int*)(opk_ + st_a[i]), \
+            (__attribute__((address_space(3))) unsigned int*)(dst_ +         \
+                (long long)(i * 512 + wave_chunk) * 8),                      \
+            16, 0, 0);                                                       \
+      }                                                                      \
+    }                                                                        \
+  } while (0)
+#define ISSUE17B(TGT, SLOT)                                                  \
+  do {                                                                       \
+    if ((TGT) < ntiles) {                                                    \
+      const ushort* opk_ = Btile + (TGT) * 32;                               \
+      ushort* dst_ = V17_B(SLOT);                                            \
+      __builtin_amdgcn_global_load_lds(                                      \
+          (const __attribute__((address_space(1))) unsigned int*)(opk_ + st_b), \
+          (__attribute__((address_space(3))) unsigned int*)(dst_ +           \
+              (long long)wave_chunk * 8),                                    \
+          16, 0, 0);                                                         \
+    }                                                                        \
+  } while (0)
+
+  ISSUE17B(0, 0);
+  ISSUE17A(0, 0);
+  ISSUE17B(1, 1);
+  asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int aslot = t & 1;
+    const int bslot = t % 3;
+    const int bslot2 = (t + 2) % 3;
+    short8 af[4], bf[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      af[j] = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const char*>(V17_A(aslot)) + a_off + j * 1024);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bf[j] = *reinterpret_cast<const short8*>(
+          reinterpret_cast<const char*>(V17_B(bslot)) + b_off + j * 1024);
+    ISSUE17A(t + 1, aslot ^ 1);
+    ISSUE17B(t + 2, bslot2);
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    if (t >= ntiles - 2)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+#undef ISSUE17A
+#undef ISSUE17B
+#undef V17_A
+#undef V17_B
+
+  const long long c_row0 = (long long)tile_m * 256 + wm * 64 + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * 128 + wn * 64 + l15;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long long row = c_row0 + mi * 16 + e;
+      ushort* crow = C + row * N;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        crow[c_col0 + ni * 16] = f2bf(acc[mi][ni][e]);
+    }
+  }
+}
