@@ -18,7 +18,7 @@ EXT_NAME = "senweaver_amd_hip"
 SOURCES = [
     os.path.join(CSRC, f)
     for f in ("bindings.hip", "elemwise.hip", "sampling.hip", "gemm.hip",
-              "gemm_pipe.hip", "attention.hip", "attention_v2.hip",
+              "gemm_pipe.hip", "gemm_asm.hip", "attention.hip", "attention_v2.hip",
               "decode_attention.hip", "moe.hip", "fp8.hip")
 ]
 
