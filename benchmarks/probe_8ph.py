@@ -42,13 +42,9 @@ for (M, N, K) in shapes:
 
     arms = {
 
-        "v9-1ph ": lambda: ext.gemm_bt_8ph_v(a, b, 9),
-        "v11-grp": lambda: ext.gemm_bt_8ph_v(a, b, 11),
-        "v12-npr": lambda: ext.gemm_bt_8ph_v(a, b, 12),
         "v13-16w": lambda: ext.gemm_bt_8ph_v(a, b, 13),
         "v14-16n": lambda: ext.gemm_bt_8ph_v(a, b, 14),
-        "v16-bk32": lambda: ext.gemm_bt_8ph_v(a, b, 16),
-        "v17-2blk": lambda: ext.gemm_bt_8ph_v(a, b, 17),
+        "v18-asm": lambda: ext.gemm_bt_8ph_v(a, b, 18),
         "blas   ": lambda: a @ b.t(),
     }
     # numerics check each variant once vs blas
