@@ -73,6 +73,60 @@ def gen_khalf(kx: int, tail: str) -> str:
             f'        : {clob});\n')
 
 
+def gen_tile_merged(tail: str) -> str:
+    """Both k-halves in ONE statement: all 24 ds_read_b128 issued up-front,
+    MFMAs retire behind a 4-step counted lgkmcnt ladder (16/12/4/0).
+
+    Operands: %0..%11 kh0 frags (af0..7, bf0..3), %12..%23 kh1 frags,
+    %24 aaddr0, %25 baddr0, %26 aaddr1, %27 baddr1.
+    """
+    L = []
+    for mi in range(4):
+        L.append(f"ds_read_b128 %{mi}, %24 offset:{mi * 2048}")
+    for ni in range(4):
+        L.append(f"ds_read_b128 %{8 + ni}, %25 offset:{ni * 2048}")
+    for mi in range(4, 8):
+        L.append(f"ds_read_b128 %{mi}, %24 offset:{mi * 2048}")
+    for mi in range(4):
+        L.append(f"ds_read_b128 %{12 + mi}, %26 offset:{mi * 2048}")
+    for ni in range(4):
+        L.append(f"ds_read_b128 %{20 + ni}, %27 offset:{ni * 2048}")
+    for mi in range(4, 8):
+        L.append(f"ds_read_b128 %{12 + mi}, %26 offset:{mi * 2048}")
+    L.append("s_waitcnt lgkmcnt(16)")
+    for mi in range(4):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+    L.append("s_waitcnt lgkmcnt(12)")
+    for mi in range(4, 8):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+    L.append("s_waitcnt lgkmcnt(4)")
+    for mi in range(4):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{12 + mi}, %{20 + ni}, {acc_range(mi, ni)}")
+    L.append("s_waitcnt lgkmcnt(0)")
+    for mi in range(4, 8):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{12 + mi}, %{20 + ni}, {acc_range(mi, ni)}")
+    if tail:
+        L.extend(tail.split(";"))
+    body = "\\n\\t".join(L)
+    outs = ", ".join(f'"=&v"(af[{i}])' for i in range(8))
+    outs += ", " + ", ".join(f'"=&v"(bf[{i}])' for i in range(4))
+    outs += ", " + ", ".join(f'"=&v"(af2[{i}])' for i in range(8))
+    outs += ", " + ", ".join(f'"=&v"(bf2[{i}])' for i in range(4))
+    clob = ", ".join(f'"a{i}"' for i in range(128)) + ', "memory"'
+    return (f'    asm volatile(\n        "{body}"\n'
+            f'        : {outs}\n'
+            f'        : "v"(aaddr0), "v"(baddr0), "v"(aaddr1), "v"(baddr1)\n'
+            f'        : {clob});\n')
+
+
 def main() -> None:
     init_writes = "\\n\\t".join(
         [f"v_accvgpr_write_b32 a{i}, 0" for i in range(128)] + ["s_nop 2"])
@@ -102,6 +156,8 @@ def main() -> None:
     # steady-state tile end: boundary vmcnt(4) BEFORE the barrier (v9 contract)
     kh1_steady = gen_khalf(1, tail="s_waitcnt vmcnt(4);s_barrier")
     kh1_drain = gen_khalf(1, tail="s_waitcnt vmcnt(0);s_barrier")
+    mg_steady = gen_tile_merged(tail="s_waitcnt vmcnt(4);s_barrier")
+    mg_drain = gen_tile_merged(tail="s_waitcnt vmcnt(0);s_barrier")
 
     src = f"""// GENERATED by scripts/gen_gemm_asm.py — do not edit by hand.
 //
@@ -223,7 +279,111 @@ gemm_bt_bf16_asm_kernel(const ushort* __restrict__ A, const ushort* __restrict__
   const long long c_row0 = (long long)tile_m * 256 + wm * 128 + (lane >> 4) * 4;
   const long long c_col0 = (long long)tile_n * 256 + wn * 64 + l15;
 {epilogue}}}
+
+// v19: both k-halves merged into one statement per tile — 24 reads issued
+// up-front, MFMAs retire behind a 16/12/4/0 counted lgkmcnt ladder; the
+// next tile's staging is issued BEFORE the compute statement so the DMA
+// has the whole tile in flight.
+extern "C" __global__ void __launch_bounds__(512, 1)
+gemm_bt_bf16_asm2_kernel(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                         ushort* __restrict__ C, int M, int N, int K) {{
+  const int nwg = (M / 256) * (N / 256);
+  int wgid = blockIdx.x;
+  {{
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = wgid % 8, pos = wgid / 8;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }}
+  const int tiles_n = N / 256;
+  const int GM = 8;
+  const int tiles_m = M / 256;
+  const int grp = wgid / (GM * tiles_n);
+  const int rem = wgid % (GM * tiles_n);
+  const int g0 = grp * GM;
+  const int gh = (tiles_m - g0 < GM) ? (tiles_m - g0) : GM;
+  const int tile_m = g0 + rem % gh;
+  const int tile_n = rem / gh;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2;
+  const int wn = wid & 3;
+  const int l15 = lane & 15;
+  const int kgrp = lane >> 4;
+
+  __shared__ __attribute__((aligned(16))) ushort lds[5][256 * 64];
+
+  const ushort* Atile = A + (long long)tile_m * 256 * K;
+  const ushort* Btile = B + (long long)tile_n * 256 * K;
+
+  const int swz0 = (kgrp + 2 * ((l15 >> 1) & 3)) & 7;
+  const int frag0 = l15 * 128 + swz0 * 16;
+  const int a_off = wm * 128 * 128 + frag0;
+  const int b_off = wn * 64 * 128 + frag0;
+
+  unsigned st_off[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {{
+    const int s = i * 512 + tid;
+    const int r = s >> 3;
+    st_off[i] = ((unsigned)r * (unsigned)K
+                 + ((((s & 7) - 2 * ((r >> 1) & 3)) & 7) * 8)) * 2u;  // bytes
+  }}
+  const int wave_chunk = tid & ~63;
+
+  const int ntiles = K / 64;
+
+#define AISSUE2(TGT, OP, SLOT)                                               \\
+  do {{                                                                      \\
+    if ((TGT) < ntiles) {{                                                   \\
+      const ushort* opk_ = (OP) + (TGT) * 64;                                \\
+      ushort* dst_ = &lds[(SLOT)][0];                                        \\
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {{                       \\
+        const ushort* g = (const ushort*)((const char*)opk_ + st_off[i]);    \\
+        __builtin_amdgcn_global_load_lds(                                    \\
+            (const __attribute__((address_space(1))) unsigned int*)g,        \\
+            (__attribute__((address_space(3))) unsigned int*)(dst_ +         \\
+                (long long)(i * 512 + wave_chunk) * 8),                      \\
+            16, 0, 0);                                                       \\
+      }}                                                                     \\
+    }}                                                                      \\
+  }} while (0)
+
+  AISSUE2(0, Btile, 2);
+  AISSUE2(0, Atile, 0);
+  AISSUE2(1, Btile, 3);
+  asm volatile("{init_writes}" ::: {acc_clob});
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < ntiles; ++t) {{
+    const int aslot = t & 1;
+    const int bslot = 2 + t % 3;
+    const int bslot2 = 2 + (t + 2) % 3;
+    auto* ab3 = (__attribute__((address_space(3))) char*)&lds[aslot][0];
+    auto* bb3 = (__attribute__((address_space(3))) char*)&lds[bslot][0];
+    unsigned aaddr0 = (unsigned)(unsigned long long)ab3 + (unsigned)a_off;
+    unsigned baddr0 = (unsigned)(unsigned long long)bb3 + (unsigned)b_off;
+    unsigned aaddr1 = aaddr0 ^ 64u;
+    unsigned baddr1 = baddr0 ^ 64u;
+    short8 af[8], bf[4], af2[8], bf2[4];
+    // stage FIRST: the DMA gets the whole tile's compute to hide under
+    AISSUE2(t + 1, Atile, aslot ^ 1);
+    AISSUE2(t + 2, Btile, bslot2);
+    if (t < ntiles - 2) {{
+{mg_steady}
+    }} else {{
+{mg_drain}
+    }}
+  }}
+#undef AISSUE2
+
+  const long long c_row0 = (long long)tile_m * 256 + wm * 128 + (lane >> 4) * 4;
+  const long long c_col0 = (long long)tile_n * 256 + wn * 64 + l15;
+{{epilogue2}}}}
 """
+    src = src.replace("{epilogue2}", epilogue)  # same epilogue body
     with open(os.path.abspath(OUT), "w") as f:
         f.write(src)
     print("wrote", OUT, len(src), "chars")
