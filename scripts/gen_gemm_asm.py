@@ -87,17 +87,20 @@ def gen_tile_merged(tail: str) -> str:
         L.append(f"ds_read_b128 %{8 + ni}, %25 offset:{ni * 2048}")
     for mi in range(4, 8):
         L.append(f"ds_read_b128 %{mi}, %24 offset:{mi * 2048}")
+    # first MFMA block starts once kh0's first 8 reads land (lgkmcnt is a
+    # 4-bit field: max 15 — kh1's reads are issued AFTER this block so no
+    # wait ever exceeds the field width, and they overlap these MFMAs)
+    L.append("s_waitcnt lgkmcnt(4)")
+    for mi in range(4):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
     for mi in range(4):
         L.append(f"ds_read_b128 %{12 + mi}, %26 offset:{mi * 2048}")
     for ni in range(4):
         L.append(f"ds_read_b128 %{20 + ni}, %27 offset:{ni * 2048}")
     for mi in range(4, 8):
         L.append(f"ds_read_b128 %{12 + mi}, %26 offset:{mi * 2048}")
-    L.append("s_waitcnt lgkmcnt(16)")
-    for mi in range(4):
-        for ni in range(4):
-            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
-                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
     L.append("s_waitcnt lgkmcnt(12)")
     for mi in range(4, 8):
         for ni in range(4):
