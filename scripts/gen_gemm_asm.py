@@ -387,6 +387,34 @@ gemm_bt_bf16_asm2_kernel(const ushort* __restrict__ A, const ushort* __restrict_
 {{epilogue2}}}}
 """
     src = src.replace("{epilogue2}", epilogue)  # same epilogue body
+
+    # v20/v21: v18-template variants.  v20 stages BEFORE kh0; v21 adds the
+    # static young-half s_setprio(1) (T5 static form).
+    base = src[src.index("extern \"C\" __global__ void __launch_bounds__(512, 1)\ngemm_bt_bf16_asm_kernel"):
+               src.index("// v19:")]
+    v20 = base.replace("gemm_bt_bf16_asm_kernel", "gemm_bt_bf16_asm3_kernel")
+    v20 = v20.replace("""    short8 af[8], bf[4];
+    // k-half 0: 12 reads + 32 MFMAs, counted lgkm ladder""",
+"""    short8 af[8], bf[4];
+    AISSUE(t + 1, Atile, aslot ^ 1);
+    AISSUE(t + 2, Btile, bslot2);
+    // k-half 0: 12 reads + 32 MFMAs, counted lgkm ladder""")
+    v20 = v20.replace("""    // prefetch next units while k-half 1 computes
+    AISSUE(t + 1, Atile, aslot ^ 1);
+    AISSUE(t + 2, Btile, bslot2);
+""", "")
+    v21 = base.replace("gemm_bt_bf16_asm_kernel", "gemm_bt_bf16_asm4_kernel")
+    v21 = v21.replace("""  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < ntiles; ++t) {""",
+"""  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static young-half priority (T5)
+
+  for (int t = 0; t < ntiles; ++t) {""")
+    src = src + v20 + v21
     with open(os.path.abspath(OUT), "w") as f:
         f.write(src)
     print("wrote", OUT, len(src), "chars")
