@@ -130,6 +130,68 @@ def gen_tile_merged(tail: str) -> str:
             f'        : {clob});\n')
 
 
+
+
+def gen_v22_stmt1() -> str:
+    """kh0 reads + kk0 MFMAs with kh1's reads issued under MFMA cover.
+    Outputs: %0..%11 kh0 frags, %12..%23 kh1 frags; inputs %24..%27 addrs."""
+    L = []
+    for mi in range(4):
+        L.append(f"ds_read_b128 %{mi}, %24 offset:{mi * 2048}")
+    for ni in range(4):
+        L.append(f"ds_read_b128 %{8 + ni}, %25 offset:{ni * 2048}")
+    for mi in range(4, 8):
+        L.append(f"ds_read_b128 %{mi}, %24 offset:{mi * 2048}")
+    L.append("s_waitcnt lgkmcnt(4)")
+    for mi in range(4):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+    for mi in range(4):
+        L.append(f"ds_read_b128 %{12 + mi}, %26 offset:{mi * 2048}")
+    for ni in range(4):
+        L.append(f"ds_read_b128 %{20 + ni}, %27 offset:{ni * 2048}")
+    for mi in range(4, 8):
+        L.append(f"ds_read_b128 %{12 + mi}, %26 offset:{mi * 2048}")
+    L.append("s_waitcnt lgkmcnt(12)")
+    for mi in range(4, 8):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+    body = "\\n\\t".join(L)
+    outs = ", ".join(f'"=&v"(af[{i}])' for i in range(8))
+    outs += ", " + ", ".join(f'"=&v"(bf[{i}])' for i in range(4))
+    outs += ", " + ", ".join(f'"=&v"(af2[{i}])' for i in range(8))
+    outs += ", " + ", ".join(f'"=&v"(bf2[{i}])' for i in range(4))
+    clob = ", ".join(f'"a{i}"' for i in range(128)) + ', "memory"'
+    return (f'    asm volatile(\n        "{body}"\n'
+            f'        : {outs}\n'
+            f'        : "v"(aaddr0), "v"(baddr0), "v"(aaddr1), "v"(baddr1)\n'
+            f'        : {clob});\n')
+
+
+def gen_v22_stmt2(tail: str) -> str:
+    """kk1 MFMAs consuming stmt1's kh1 fragments (inputs %0..%11)."""
+    L = ["s_waitcnt lgkmcnt(4)"]
+    for mi in range(4):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+    L.append("s_waitcnt lgkmcnt(0)")
+    for mi in range(4, 8):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+    L.extend(tail.split(";"))
+    body = "\\n\\t".join(L)
+    ins = ", ".join(f'"v"(af2[{i}])' for i in range(8))
+    ins += ", " + ", ".join(f'"v"(bf2[{i}])' for i in range(4))
+    clob = ", ".join(f'"a{i}"' for i in range(128)) + ', "memory"'
+    return (f'    asm volatile(\n        "{body}"\n'
+            f'        :\n        : {ins}\n'
+            f'        : {clob});\n')
+
+
 def main() -> None:
     init_writes = "\\n\\t".join(
         [f"v_accvgpr_write_b32 a{i}, 0" for i in range(128)] + ["s_nop 2"])
@@ -403,6 +465,20 @@ gemm_bt_bf16_asm2_kernel(const ushort* __restrict__ A, const ushort* __restrict_
     AISSUE(t + 1, Atile, aslot ^ 1);
     AISSUE(t + 2, Btile, bslot2);
 """, "")
+    st1 = gen_v22_stmt1()
+    st2_steady = gen_v22_stmt2("s_waitcnt vmcnt(4);s_barrier")
+    st2_drain = gen_v22_stmt2("s_waitcnt vmcnt(0);s_barrier")
+    v22 = base.replace("gemm_bt_bf16_asm_kernel", "gemm_bt_bf16_asm5_kernel")
+    ix0 = v22.index("    short8 af[8], bf[4];")
+    ix1 = v22.index("  }\n#undef AISSUE")
+    v22 = (v22[:ix0]
+           + "    short8 af[8], bf[4], af2[8], bf2[4];\n"
+           + st1
+           + "    AISSUE(t + 1, Atile, aslot ^ 1);\n"
+           + "    AISSUE(t + 2, Btile, bslot2);\n"
+           + "    if (t < ntiles - 2) {\n" + st2_steady
+           + "    } else {\n" + st2_drain + "    }\n"
+           + v22[ix1:])
     v21 = base.replace("gemm_bt_bf16_asm_kernel", "gemm_bt_bf16_asm4_kernel")
     v21 = v21.replace("""  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   __builtin_amdgcn_s_barrier();
@@ -414,7 +490,7 @@ gemm_bt_bf16_asm2_kernel(const ushort* __restrict__ A, const ushort* __restrict_
     __builtin_amdgcn_s_setprio(1);  // static young-half priority (T5)
 
   for (int t = 0; t < ntiles; ++t) {""")
-    src = src + v20 + v21
+    src = src + v20 + v21 + v22
     with open(os.path.abspath(OUT), "w") as f:
         f.write(src)
     print("wrote", OUT, len(src), "chars")
