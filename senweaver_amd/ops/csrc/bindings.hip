@@ -262,12 +262,16 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
   if (M % 256 == 0 && N % 256 == 0 && (M / 256) * (N / 256) >= 160) {
     const int nwg = (M / 256) * (N / 256);
     if (K % 128 == 0)
-      // 16-wave single-barrier-per-K-tile pipeline (A dbuf + B ring-3 over
-      // the full 160 KiB LDS, counted vmcnt) — the A/B-measured best
-      // (benchmarks/probe_8ph.py: 1.21/1.31/1.17 PF at 4k/8k/gateup
-      // vs 1.05/1.10/1.06 for the round-1 structure)
-      gemm_bt_bf16_8ph_v14_kernel<<<dim3(nwg), dim3(1024), 0, cur_stream()>>>(
-          bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+      // in-repo best per shape (profiles/r02_gemm_pipeline.txt): the
+      // hand-scheduled asm kernel (AGPR accumulators, counted lgkm ladder,
+      // static young-half prio) wins at M>=4096 (1.27/1.42 PF at 4k/8k);
+      // the 16-wave plain-HIP pipeline wins the M=2048 wide shapes
+      if (M >= 4096)
+        gemm_bt_bf16_asm4_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
+            bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+      else
+        gemm_bt_bf16_8ph_v14_kernel<<<dim3(nwg), dim3(1024), 0, cur_stream()>>>(
+            bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
     else
       gemm_bt_bf16_256_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
           bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
