@@ -136,6 +136,21 @@ def test_gemm_pipe16w_random(dev, M, N, K):
         torch.testing.assert_close(c.float(), c_ref.float(), atol=0.5, rtol=3e-2)
 
 
+@pytest.mark.parametrize("var", [18, 21])
+def test_gemm_asm_random(dev, var):
+    """Hand-scheduled asm kernels (v18/v21 = the M>=4096 dispatch):
+    numerics + determinism at two shapes."""
+    ext = ops.hip_ext()
+    for (M, N, K) in [(512, 512, 512), (4096, 4096, 4096)]:
+        g = torch.Generator(device="cpu").manual_seed(3)
+        a = torch.randn(M, K, generator=g).bfloat16().to(dev)
+        b = torch.randn(N, K, generator=g).bfloat16().to(dev)
+        c = ext.gemm_bt_8ph_v(a, b, var)
+        assert torch.equal(c, ext.gemm_bt_8ph_v(a, b, var))
+        torch.testing.assert_close(c.float(), ref.gemm_bt_ref(a, b).float(),
+                                   atol=0.5, rtol=3e-2)
+
+
 def test_gemm_pad_m(dev):
     # M not a multiple of 128 goes through the host-side pad
     a = torch.randn(300, 512, dtype=torch.bfloat16, device=dev)
