@@ -143,3 +143,29 @@ def test_trace_range_records():
     with trace_range("messageTrimming", mon):
         pass
     assert "messageTrimming" in mon.summary()
+
+
+def test_model_capabilities_registry():
+    """modelCapabilities.ts rebuild: record lookup, substring fallback, and
+    the reserved-output-space default semantics."""
+    from senweaver_amd.models.capabilities import (
+        DEFAULT_RESERVED_OUTPUT_TOKEN_SPACE, ModelCapabilities,
+        get_context_window, get_model_capabilities,
+        get_reserved_output_token_space,
+    )
+
+    caps = get_model_capabilities("llama-3-8b")
+    assert caps.contextWindow == 8192 and caps.supportsFIM
+    assert caps.specialToolFormat is None  # XML tool calls in agent mode
+    assert caps.supportsSystemMessage == "system-role"
+    # substring family fallback, like the reference's recognizable-model map
+    assert get_model_capabilities("my-llama-finetune").contextWindow == 8192
+    assert get_model_capabilities("mixtral-custom").contextWindow == 32768
+    # unknown -> default record; None reserved space -> 4096 default
+    unk = get_model_capabilities("who-knows")
+    assert unk.contextWindow == 4096 and unk.reservedOutputTokenSpace is None
+    assert (get_reserved_output_token_space("who-knows")
+            == DEFAULT_RESERVED_OUTPUT_TOKEN_SPACE)
+    assert get_reserved_output_token_space("llama-3-8b") == 2048
+    assert get_context_window("tiny-debug") == 2048
+    assert isinstance(caps, ModelCapabilities)
