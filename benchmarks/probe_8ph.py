@@ -45,7 +45,7 @@ for (M, N, K) in shapes:
         "v13-16w": lambda: ext.gemm_bt_8ph_v(a, b, 13),
         "v14-16n": lambda: ext.gemm_bt_8ph_v(a, b, 14),
         "v18-asm": lambda: ext.gemm_bt_8ph_v(a, b, 18),
-        "v22-spl": lambda: ext.gemm_bt_8ph_v(a, b, 22),
+        "v23-rng": lambda: ext.gemm_bt_8ph_v(a, b, 23),
         "v21-pri": lambda: ext.gemm_bt_8ph_v(a, b, 21),
         "blas   ": lambda: a @ b.t(),
     }

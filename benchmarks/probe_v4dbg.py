@@ -10,7 +10,7 @@ from senweaver_amd import ops  # noqa: E402
 dev = "cuda:0"
 ext = ops.hip_ext()
 
-for var in (22,):
+for var in (23,):
     for (M, N, K) in [(512, 512, 512), (4096, 4096, 4096)]:
         g = torch.Generator().manual_seed(7)
         a = torch.randn(M, K, generator=g).bfloat16().to(dev)

@@ -490,7 +490,36 @@ gemm_bt_bf16_asm2_kernel(const ushort* __restrict__ A, const ushort* __restrict_
     __builtin_amdgcn_s_setprio(1);  // static young-half priority (T5)
 
   for (int t = 0; t < ntiles; ++t) {""")
-    src = src + v20 + v21 + v22
+    # v23 (asm6): A ring-of-3 + B double-buffer (3*32 + 2*32 = 160 KiB):
+    # A(t+2) staged with TWO tiles of flight, B(t+1) staged at tile START
+    # with a full tile of flight — the boundary vmcnt(4) then only ever
+    # waits for loads that have had >= a full tile to land.
+    v23 = base.replace("gemm_bt_bf16_asm_kernel", "gemm_bt_bf16_asm6_kernel")
+    v23 = v23.replace("""  AISSUE(0, Btile, 2);
+  AISSUE(0, Atile, 0);
+  AISSUE(1, Btile, 3);""",
+"""  AISSUE(0, Atile, 0);   // A ring slots 0,1,2; B dbuf slots 3,4
+  AISSUE(0, Btile, 3);
+  AISSUE(1, Atile, 1);""")
+    v23 = v23.replace("""    const int aslot = t & 1;
+    const int bslot = 2 + t % 3;
+    const int bslot2 = 2 + (t + 2) % 3;""",
+"""    const int aslot = t % 3;
+    const int aslot2 = (t + 2) % 3;
+    const int bslot = 3 + (t & 1);""")
+    v23 = v23.replace("""    short8 af[8], bf[4];
+    // k-half 0: 12 reads + 32 MFMAs, counted lgkm ladder""",
+"""    short8 af[8], bf[4];
+    AISSUE(t + 1, Btile, 3 + ((t + 1) & 1));  // the other B buffer
+    // k-half 0: 12 reads + 32 MFMAs, counted lgkm ladder""")
+    v23 = v23.replace("""    // prefetch next units while k-half 1 computes
+    AISSUE(t + 1, Atile, aslot ^ 1);
+    AISSUE(t + 2, Btile, bslot2);
+""",
+"""    // A(t+2) into its ring slot: two tiles of flight
+    AISSUE(t + 2, Atile, aslot2);
+""")
+    src = src + v20 + v21 + v22 + v23
     with open(os.path.abspath(OUT), "w") as f:
         f.write(src)
     print("wrote", OUT, len(src), "chars")

@@ -42,6 +42,7 @@ extern "C" __global__ void gemm_bt_bf16_asm2_kernel(const ushort*, const ushort*
 extern "C" __global__ void gemm_bt_bf16_asm3_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_asm4_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_asm5_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_asm6_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void moe_combine_kernel(const ushort*, const int*, const float*, ushort*, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
@@ -605,6 +606,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       case 21: gemm_bt_bf16_asm4_kernel<<<g, dim3(512), 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
       case 22: gemm_bt_bf16_asm5_kernel<<<g, dim3(512), 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+      case 23: gemm_bt_bf16_asm6_kernel<<<g, dim3(512), 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
       default: gemm_bt_bf16_8ph_kernel<<<g, blk, 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
