@@ -175,3 +175,17 @@ def test_custom_api_service_registry(tmp_path):
     svc2.delete_api(api.id)
     assert svc2.state == {"apis": []}
     assert CUSTOM_API_STORAGE_KEY == "senweaver.customApis"
+
+
+def test_custom_api_description_reaches_system_message():
+    """End-to-end prompt surface: CustomApiService -> the converter's
+    custom_api_description slot (convertToLLMMessageService pairing)."""
+    from senweaver_amd.context.pipeline import ConvertToLLMMessages
+    from senweaver_amd.features.customapi import CustomApiService
+
+    svc = CustomApiService()
+    svc.add_api("Ticketing", "http://127.0.0.1:8088/t", "POST", "file a ticket")
+    conv = ConvertToLLMMessages()
+    msg = conv.generate_system_message(
+        "agent", custom_api_description=svc.get_api_list_description())
+    assert "Ticketing" in msg and "api_request" in msg
