@@ -102,6 +102,10 @@ class DaemonClient:
             time.sleep(0.05)
         return False
 
+    def shutdown(self) -> None:
+        """Ask the daemon to shut down (worker + listener)."""
+        self._send({"method": "shutdown"})
+
     def close(self) -> None:
         self._closed = True
         try:

@@ -180,3 +180,31 @@ def test_daemon_worker_crash_errors_inflight(daemon):
     assert done.wait(timeout=60), "in-flight request hung after worker crash"
     assert errors and "crash" in errors[0].get("message", "")
     c.close()
+
+
+def test_cli_singleton_start_status_stop(tmp_path, monkeypatch):
+    """Launcher CLI (server/cli.py = the Rust code-cli's engine subset):
+    singleton lock, start/status/stop, stale-lock reclaim."""
+    from senweaver_amd.server import cli
+
+    monkeypatch.setattr(cli, "STATE_DIR", str(tmp_path))
+    monkeypatch.setattr(cli, "SOCKET_PATH", str(tmp_path / "d.sock"))
+    monkeypatch.setattr(cli, "LOCK_PATH", str(tmp_path / "d.lock"))
+    monkeypatch.setattr(cli, "LOG_PATH", str(tmp_path / "d.log"))
+    monkeypatch.setenv("SENWEAVER_MODEL", "tiny-debug")
+    monkeypatch.setenv("SENWEAVER_MAX_SEQ", "128")
+
+    assert cli.read_lock() is None
+    assert cli.start() == 0
+    pid = cli.read_lock()
+    assert pid is not None
+    # second start is a no-op against the live singleton
+    assert cli.start() == 0
+    assert cli.read_lock() == pid
+    assert cli.status() == 0
+    assert cli.stop() == 0
+    assert cli.read_lock() is None
+    # stale-lock reclaim: dead pid in the lock file
+    (tmp_path / "d.lock").write_text("999999")
+    assert cli.read_lock() is None
+    assert not (tmp_path / "d.lock").exists()
