@@ -29,12 +29,20 @@ if TYPE_CHECKING:  # avoids the engine<->models import cycle
 
 class LlamaModel:
     def __init__(self, config: ModelConfig, device="cpu", dtype=torch.bfloat16, seed: int = 0,
-                 tp=None, quant: str = "bf16"):
+                 tp=None, quant: str = "bf16", ep=None):
         from ..parallel.tp import TPContext, shard_gateup, shard_qkv, shard_rows
+        from ..parallel.ep import EPContext
         self.config = config
         self.device = torch.device(device)
         self.dtype = dtype
         self.tp = tp if tp is not None else TPContext()
+        # expert parallelism (parallel/ep.py): contiguous expert shards per
+        # rank, all-to-all token routing.  Mutually exclusive with TP here
+        # (BASELINE needs neither combined; the collectives would nest).
+        self.ep = ep if ep is not None else EPContext()
+        if self.ep.world > 1:
+            assert self.tp.world == 1, "EP and TP are not combined"
+            assert config.num_experts % self.ep.world == 0
         c = config
         tpw = self.tp.world
         if tpw > 1:
@@ -98,6 +106,11 @@ class LlamaModel:
                     layer["w2"] = torch.stack([
                         shard_rows(w2_full[e], self.tp, dim=1)
                         for e in range(c.num_experts)]).contiguous()
+                    del w13_full, w2_full
+                elif self.ep.world > 1:
+                    elo, ehi = self.ep.local_experts(c.num_experts)
+                    layer["w13"] = w13_full[elo:ehi].contiguous()
+                    layer["w2"] = w2_full[elo:ehi].contiguous()
                     del w13_full, w2_full
                 else:
                     layer["w13"] = w13_full
@@ -245,6 +258,32 @@ class LlamaModel:
         sorted_weight = topw.reshape(-1)[order]
         counts = torch.bincount(flat_expert, minlength=c.num_experts)
         Tk = T * k
+        if self.ep.world > 1:
+            # EP: route expert-sorted tokens to their expert's owner rank,
+            # grouped-GEMM the LOCAL experts, route back, combine as usual
+            a_sorted = h[sorted_token]
+            x_local, local_counts, meta = self.ep.dispatch(
+                a_sorted, counts, c.num_experts)
+            Tl = x_local.shape[0]
+            pad = 128
+            a_p = torch.zeros(Tl + pad, c.hidden_size, dtype=h.dtype,
+                              device=h.device)
+            a_p[:Tl] = x_local
+            seg = [0]
+            for e in range(local_counts.shape[0]):
+                seg.append(seg[-1] + int(local_counts[e]))
+            gateup = ops.grouped_gemm_bt(a_p, L["w13"], seg)
+            act = ops.swiglu(gateup[:Tl])
+            act_p = torch.zeros(Tl + pad, act.shape[1], dtype=h.dtype,
+                                device=h.device)
+            act_p[:Tl] = act
+            down_l = ops.grouped_gemm_bt(act_p, L["w2"], seg)[:Tl]
+            down = self.ep.combine(down_l, meta)
+            res = torch.zeros(T, c.hidden_size, dtype=torch.float32,
+                              device=h.device)
+            res.index_add_(0, sorted_token,
+                           down.float() * sorted_weight.unsqueeze(-1).float())
+            return res.to(h.dtype)
         if h.is_cuda:
             # sync-free routing: device-side cumulative ends feed the grouped
             # GEMM directly (the per-expert int(counts[e]) reads were 8 tiny

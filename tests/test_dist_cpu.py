@@ -281,3 +281,35 @@ def test_bench_torchrun_world2_contract():
     assert d["n_gpus"] == 2 and d["scaling"] == "strong"
     assert d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0 and d["higher_is_better"] is True
+
+
+def _payload_ep_moe(rank, world):
+    import torch
+    from senweaver_amd.models import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+    from senweaver_amd.parallel.ep import EPContext
+
+    ep = EPContext.from_default_group()
+    model = LlamaModel(tiny_moe(), device="cpu", seed=11, ep=ep)
+    tokens = torch.randint(0, 512, (1, 48),
+                           generator=torch.Generator().manual_seed(4))
+    hidden = model.prefill(tokens)
+    return hidden.float().sum(-1).squeeze(0).tolist()[:8]
+
+
+def test_ep2_matches_replicated():
+    """EP=2 (expert-sharded, all-to-all routed) reproduces the replicated
+    single-rank MoE forward (same seed => same full weights)."""
+    import torch
+    from senweaver_amd.models import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+
+    ref_model = LlamaModel(tiny_moe(), device="cpu", seed=11)
+    tokens = torch.randint(0, 512, (1, 48),
+                           generator=torch.Generator().manual_seed(4))
+    ref = ref_model.prefill(tokens).float().sum(-1).squeeze(0).tolist()[:8]
+
+    results = _run_dist("_payload_ep_moe")
+    for rank, vals in results.items():
+        assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
+    assert results[0] == pytest.approx(results[1], abs=1e-5)

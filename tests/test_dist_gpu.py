@@ -138,3 +138,36 @@ def test_tp2_one_gpu_matches_tp1():
     for rank, vals in results.items():
         assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
     assert results[0] == pytest.approx(results[1], abs=1e-4)
+
+
+def _payload_ep2_gpu(rank, world):
+    from senweaver_amd.models import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+    from senweaver_amd.parallel.ep import EPContext
+
+    ep = EPContext.from_default_group()
+    model = LlamaModel(tiny_moe(), device="cuda:0", seed=11, ep=ep)
+    tokens = torch.randint(0, 512, (1, 48),
+                           generator=torch.Generator().manual_seed(4)).cuda()
+    hidden = model.prefill(tokens)
+    torch.cuda.synchronize()
+    return hidden.float().sum(-1).squeeze(0).cpu().tolist()[:8]
+
+
+def test_ep2_one_gpu_matches_replicated():
+    """Expert-parallel MoE with both expert shards computing on the one
+    device (collectives over gloo; RCCL all_to_all_single is the same code
+    path at world>1 on a real multi-GPU node)."""
+    from senweaver_amd.models import tiny_moe
+    from senweaver_amd.models.llama import LlamaModel
+
+    ref_model = LlamaModel(tiny_moe(), device="cuda:0", seed=11)
+    tokens = torch.randint(0, 512, (1, 48),
+                           generator=torch.Generator().manual_seed(4)).cuda()
+    ref = ref_model.prefill(tokens).float().sum(-1).squeeze(0).cpu().tolist()[:8]
+    del ref_model
+    torch.cuda.empty_cache()
+
+    results = _run_dist("_payload_ep2_gpu")
+    for rank, vals in results.items():
+        assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
