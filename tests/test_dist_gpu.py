@@ -147,7 +147,7 @@ def _payload_ep2_gpu(rank, world):
 
     ep = EPContext.from_default_group()
     model = LlamaModel(tiny_moe(), device="cuda:0", seed=11, ep=ep)
-    tokens = torch.randint(0, 512, (1, 48),
+    tokens = torch.randint(0, 512, (1, 128),
                            generator=torch.Generator().manual_seed(4)).cuda()
     hidden = model.prefill(tokens)
     torch.cuda.synchronize()
@@ -162,7 +162,7 @@ def test_ep2_one_gpu_matches_replicated():
     from senweaver_amd.models.llama import LlamaModel
 
     ref_model = LlamaModel(tiny_moe(), device="cuda:0", seed=11)
-    tokens = torch.randint(0, 512, (1, 48),
+    tokens = torch.randint(0, 512, (1, 128),
                            generator=torch.Generator().manual_seed(4)).cuda()
     ref = ref_model.prefill(tokens).float().sum(-1).squeeze(0).cpu().tolist()[:8]
     del ref_model
