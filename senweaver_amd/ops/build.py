@@ -19,6 +19,7 @@ SOURCES = [
     os.path.join(CSRC, f)
     for f in ("bindings.hip", "elemwise.hip", "sampling.hip", "gemm.hip",
               "gemm_pipe.hip", "gemm_asm.hip", "attention.hip", "attention_v2.hip",
+              "attention_v4.hip",
               "decode_attention.hip", "moe.hip", "fp8.hip")
 ]
 

@@ -486,6 +486,45 @@ torch::Tensor attn_fwd_v3(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
   return ot;
 }
 
+// v4 probe family: v3 + {defer-max, async-STAGE register staging,
+// glds-after-QK} lever matrix (attention_v4.hip header for the map)
+#define DECL_V4(VAR)                                                         \
+  extern "C" __global__ void attn_fwd_v4_##VAR##_kernel(                     \
+      const ushort*, const ushort*, const ushort*, ushort*, int, int, int,   \
+      int, float);
+DECL_V4(0) DECL_V4(1) DECL_V4(2) DECL_V4(3)
+DECL_V4(4) DECL_V4(5) DECL_V4(6) DECL_V4(7)
+DECL_V4(8) DECL_V4(9) DECL_V4(10) DECL_V4(11) DECL_V4(12) DECL_V4(13) DECL_V4(14) DECL_V4(15)
+
+torch::Tensor attn_fwd_v4(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
+                          double scale, long var) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(vt, "vt");
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hk = k.size(1);
+  TORCH_CHECK(D == 128 && S % 128 == 0 && H % Hk == 0);
+  TORCH_CHECK(vt.size(1) == Hk && vt.size(2) == D && vt.size(3) == S);
+  auto ot = torch::empty({B, H, D, S}, q.options());
+  const dim3 grid((S + 255) / 256, H, B), blk(512);
+  auto s = cur_stream();
+#define LAUNCH_V4(VAR)                                                       \
+  case VAR:                                                                  \
+    attn_fwd_v4_##VAR##_kernel<<<grid, blk, 0, s>>>(                         \
+        bf16_ptr(q), bf16_ptr(k), bf16_ptr(vt), bf16_mut(ot), B, H, Hk, S,  \
+        (float)scale);                                                       \
+    break;
+  switch (var) {
+    LAUNCH_V4(0) LAUNCH_V4(1) LAUNCH_V4(2) LAUNCH_V4(3)
+    LAUNCH_V4(4) LAUNCH_V4(5) LAUNCH_V4(6) LAUNCH_V4(7)
+    LAUNCH_V4(8) LAUNCH_V4(9) LAUNCH_V4(10) LAUNCH_V4(11) LAUNCH_V4(12) LAUNCH_V4(13) LAUNCH_V4(14) LAUNCH_V4(15)
+    default: TORCH_CHECK(false, "attn_fwd_v4: unknown variant ", var);
+  }
+#undef LAUNCH_V4
+  HIP_CHECK_KERNEL();
+  return ot;
+}
+
 // ---------------- Paged decode attention ----------------
 torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor kcache,
                                 torch::Tensor vcache, torch::Tensor block_table,
@@ -657,6 +696,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, "causal flash attention fwd (D=128, GQA)");
   m.def("attn_fwd_v2", &attn_fwd_v2, "swapped-QK^T attention; O^T out [B,H,D,S]");
   m.def("attn_fwd_v3", &attn_fwd_v3, "v2 with 8-wave shared K/V tiles; O^T out");
+  m.def("attn_fwd_v4", &attn_fwd_v4, "v4 lever-matrix probe (variant arg)");
   m.def("paged_decode_attn", &paged_decode_attn, "paged decode attention");
   m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");
   m.def("target_logprob", &target_logprob, "fused log_softmax gather");
