@@ -106,8 +106,9 @@ def bench_attn():
         ext = ops.hip_ext()
         t2 = timeit(lambda: ext.attn_fwd_v2(q, k, vt, scale))
         t3 = timeit(lambda: ext.attn_fwd_v3(q, k, vt, scale))
+        t5 = timeit(lambda: ext.attn_fwd_v5(q, k, vt, scale))
         flops = 2 * 2 * B * H * S * S * D / 2  # causal half
-        print(f"ATTN B{B} H{H} S{S}: v2 {flops / t2 / 1e12:7.1f}  v3 {flops / t3 / 1e12:7.1f} TF/s")
+        print(f"ATTN B{B} H{H} S{S}: v2 {flops / t2 / 1e12:7.1f}  v3 {flops / t3 / 1e12:7.1f}  v5 {flops / t5 / 1e12:7.1f} TF/s")
         tsdpa = timeit(lambda: torch.nn.functional.scaled_dot_product_attention(
             q, k, v, is_causal=True, enable_gqa=True))
         print(f"  torch sdpa:        {flops / tsdpa / 1e12:7.1f} TF/s")

@@ -401,7 +401,7 @@ def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional[
         if vt is None:
             vt = v.transpose(-1, -2).contiguous()
         if q.shape[2] % 128 == 0:
-            ot = hip_ext().attn_fwd_v2(q.contiguous(), k.contiguous(), vt, scale)
+            ot = attn_fwd_t(q, k, vt, scale)
             return ot.transpose(-1, -2).contiguous()
         return hip_ext().attn_fwd(q.contiguous(), k.contiguous(), vt, scale)
     if v is None:
@@ -418,13 +418,14 @@ def attn_fwd_t(q: torch.Tensor, k: torch.Tensor, vt: torch.Tensor,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
         ext = hip_ext()
-        # v3 (8 waves sharing the staged K/V tile) when the grid still fills
-        # the chip at 2 blocks/CU; v2's smaller blocks otherwise.  S capped
-        # at 4096: v2's finer q-blocks win the causal load balance at 8192
-        # (404 vs 499 TF/s, profiles/r01_pmc_summary.txt) — the v3 pick at
-        # long S was a dispatch bug vs that measurement.
-        if _ATTN_IMPL != "v2" and 256 <= q.shape[2] <= 4096 and                 ((q.shape[2] + 255) // 256) * q.shape[0] * q.shape[1] >= 512:
-            return ext.attn_fwd_v3(q.contiguous(), k.contiguous(),
+        # v5 (v3 structure + register-staged async prefetch + defer-max +
+        # softmax VALU diet + sm-split) when the grid still fills the chip;
+        # v2's smaller blocks for tiny grids.  v5 beats v2 at EVERY measured
+        # S including 8192 (580 vs ~499 TF — profiles/r02_attn_ladder.txt),
+        # so the old v2-at-8192 special case is gone.
+        if _ATTN_IMPL != "v2" and q.shape[2] >= 256 and \
+                ((q.shape[2] + 255) // 256) * q.shape[0] * q.shape[1] >= 512:
+            return ext.attn_fwd_v5(q.contiguous(), k.contiguous(),
                                    vt.contiguous(), scale)
         return ext.attn_fwd_v2(q.contiguous(), k.contiguous(),
                                vt.contiguous(), scale)

@@ -525,6 +525,26 @@ torch::Tensor attn_fwd_v4(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
   return ot;
 }
 
+// v5 = promoted winner of the v4 lever matrix (variant 15: register-staged
+// async prefetch, double LDS buffer, defer-max, softmax VALU diet, sm-split;
+// 501 TF @ B4 H32 S2048 / 580 TF @ B1 H32 S8192 causal vs v3's 434/477 —
+// profiles/r02_attn_ladder.txt)
+torch::Tensor attn_fwd_v5(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
+                          double scale) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(vt, "vt");
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hk = k.size(1);
+  TORCH_CHECK(D == 128 && S % 128 == 0 && H % Hk == 0);
+  TORCH_CHECK(vt.size(1) == Hk && vt.size(2) == D && vt.size(3) == S);
+  auto ot = torch::empty({B, H, D, S}, q.options());
+  attn_fwd_v4_15_kernel<<<dim3((S + 255) / 256, H, B), dim3(512), 0, cur_stream()>>>(
+      bf16_ptr(q), bf16_ptr(k), bf16_ptr(vt), bf16_mut(ot), B, H, Hk, S, (float)scale);
+  HIP_CHECK_KERNEL();
+  return ot;
+}
+
 // ---------------- Paged decode attention ----------------
 torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor kcache,
                                 torch::Tensor vcache, torch::Tensor block_table,
@@ -697,6 +717,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd_v2", &attn_fwd_v2, "swapped-QK^T attention; O^T out [B,H,D,S]");
   m.def("attn_fwd_v3", &attn_fwd_v3, "v2 with 8-wave shared K/V tiles; O^T out");
   m.def("attn_fwd_v4", &attn_fwd_v4, "v4 lever-matrix probe (variant arg)");
+  m.def("attn_fwd_v5", &attn_fwd_v5, "production attention (v4 winner); O^T out");
   m.def("paged_decode_attn", &paged_decode_attn, "paged decode attention");
   m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");
   m.def("target_logprob", &target_logprob, "fused log_softmax gather");

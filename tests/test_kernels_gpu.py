@@ -523,3 +523,41 @@ def test_rope_kv_append_gpu(dev):
                                    atol=2e-2, rtol=1e-2)
         torch.testing.assert_close(vc[slot[i].long()].cpu().float(), v_ref[i].float(),
                                    atol=0, rtol=0)
+
+@pytest.mark.parametrize("B,H,Hk,S", [(2, 8, 2, 256), (1, 32, 8, 1024),
+                                      (1, 8, 8, 2048)])
+def test_attn_fwd_v5(dev, B, H, Hk, S):
+    """Production prefill kernel (v4 lever-matrix winner): register-staged
+    async prefetch + defer-max + softmax VALU diet + sm-split."""
+    D = 128
+    q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+    vt = v.transpose(-1, -2).contiguous()
+    scale = 1.0 / math.sqrt(D)
+    ot = ops.hip_ext().attn_fwd_v5(q, k, vt, scale)
+    o_ref = ref.attn_fwd_ref(q, k, v, scale, causal=True)
+    torch.testing.assert_close(ot.transpose(-1, -2).float(), o_ref.float(),
+                               atol=8e-2, rtol=8e-2)
+
+
+def test_attn_fwd_v5_defer_max_paths(dev):
+    """defer-max correctness: a spike (forces the rescale branch) and a
+    slow upward drift (keeps tiles inside the defer threshold, P bounded
+    by e^8) must both match the fp32 reference."""
+    B, H, Hk, S, D = 1, 4, 4, 1024, 128
+    scale = 1.0 / math.sqrt(D)
+    for mode in ("spike", "drift"):
+        q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=dev)
+        k = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+        if mode == "spike":
+            k[:, :, 100] *= 8
+        else:
+            drift = torch.linspace(1.0, 1.5, S, device=dev)
+            k = (k.float() * drift.view(1, 1, S, 1)).to(torch.bfloat16)
+        v = torch.randn(B, Hk, S, D, dtype=torch.bfloat16, device=dev)
+        vt = v.transpose(-1, -2).contiguous()
+        ot = ops.hip_ext().attn_fwd_v5(q, k.contiguous(), vt, scale)
+        o_ref = ref.attn_fwd_ref(q, k, v, scale, causal=True)
+        torch.testing.assert_close(ot.transpose(-1, -2).float(), o_ref.float(),
+                                   atol=1e-1, rtol=1e-1)
