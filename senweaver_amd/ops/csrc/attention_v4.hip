@@ -29,10 +29,10 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 static __device__ __forceinline__ void v4_stage_glds(
     const ushort* __restrict__ src, long long ld, int chunks_per_row,
-    ushort* lds_tile, int chunks_total, int tid) {
+    ushort* lds_tile, int chunks_total, int tid, int nt = 512) {
   const int cmask = chunks_per_row - 1;
   const int wave_chunk = tid & ~63;
-  for (int s0 = 0; s0 < chunks_total; s0 += 512) {
+  for (int s0 = 0; s0 < chunks_total; s0 += nt) {
     const int s = s0 + tid;
     if (s >= chunks_total) break;
     const int row = s / chunks_per_row;
@@ -84,7 +84,8 @@ static __device__ __forceinline__ float v4_exp2(float x) {
   return r;
 }
 
-template <int STAGE, int DEFER, int DIET = 0, int SPLIT = 0>
+template <int STAGE, int DEFER, int DIET = 0, int SPLIT = 0,
+          int NT = 512, int TEPI = 0>
 static __device__ __forceinline__ void attn4_body(
     const ushort* __restrict__ Q, const ushort* __restrict__ K,
     const ushort* __restrict__ VT, ushort* __restrict__ OT, int B, int H,
@@ -105,7 +106,8 @@ static __device__ __forceinline__ void attn4_body(
   const int lane = tid & 63;
   const int l31 = lane & 31;
   const int lhi = lane >> 5;
-  const int q0 = qb * V4_QBLK + wid * 32;
+  constexpr int QBLK = NT / 2;  // 32 q rows per wave
+  const int q0 = qb * QBLK + wid * 32;
   const int q_lane = q0 + l31;
   const bool live = q_lane < S;
 
@@ -128,18 +130,18 @@ static __device__ __forceinline__ void attn4_body(
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[db][r] = 0.f;
 
-  const int kv_end = min(S, qb * V4_QBLK + V4_QBLK);
-  v4_stage_glds(Kh, D, 16, smem[0], V4_KVBLK * 16, tid);
-  v4_stage_glds(VTh, S, 8, smem[0] + V4_KVBLK * 128, 128 * 8, tid);
+  const int kv_end = min(S, qb * QBLK + QBLK);
+  v4_stage_glds(Kh, D, 16, smem[0], V4_KVBLK * 16, tid, NT);
+  v4_stage_glds(VTh, S, 8, smem[0] + V4_KVBLK * 128, 128 * 8, tid, NT);
   __syncthreads();
   int buf = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += V4_KVBLK) {
     const bool more = kv0 + V4_KVBLK < kv_end;
     if (STAGE == 0 && more) {  // v3: prefetch issued ahead of the QK reads
       v4_stage_glds(Kh + (long long)(kv0 + V4_KVBLK) * D, D, 16,
-                    smem[buf ^ 1], V4_KVBLK * 16, tid);
+                    smem[buf ^ 1], V4_KVBLK * 16, tid, NT);
       v4_stage_glds(VTh + kv0 + V4_KVBLK, S, 8,
-                    smem[buf ^ 1] + V4_KVBLK * 128, 128 * 8, tid);
+                    smem[buf ^ 1] + V4_KVBLK * 128, 128 * 8, tid, NT);
     }
     const ushort* k_lds = smem[buf];
     const ushort* vt_lds = smem[buf] + V4_KVBLK * 128;
@@ -162,19 +164,21 @@ static __device__ __forceinline__ void attn4_body(
 
     // stage issue point for the after-QK modes (K rows of tile t are
     // consumed; VT is still live until PV)
-    short8 stg[4];
+    constexpr int KC = 1024 / NT;  // per-thread chunks of K (and of VT)
+    short8 stg[2 * KC];
     if ((STAGE == 1 || STAGE == 2) && more) {
       const ushort* Kn = Kh + (long long)(kv0 + V4_KVBLK) * D;
       const ushort* VTn = VTh + kv0 + V4_KVBLK;
-      stg[0] = v4_gload(Kn, D, 16, tid);
-      stg[1] = v4_gload(Kn, D, 16, tid + 512);
-      stg[2] = v4_gload(VTn, S, 8, tid);
-      stg[3] = v4_gload(VTn, S, 8, tid + 512);
+#pragma unroll
+      for (int c = 0; c < KC; ++c) {
+        stg[c] = v4_gload(Kn, D, 16, tid + c * NT);
+        stg[KC + c] = v4_gload(VTn, S, 8, tid + c * NT);
+      }
     } else if (STAGE == 3 && more) {
       v4_stage_glds(Kh + (long long)(kv0 + V4_KVBLK) * D, D, 16,
-                    smem[buf ^ 1], V4_KVBLK * 16, tid);
+                    smem[buf ^ 1], V4_KVBLK * 16, tid, NT);
       v4_stage_glds(VTh + kv0 + V4_KVBLK, S, 8,
-                    smem[buf ^ 1] + V4_KVBLK * 128, 128 * 8, tid);
+                    smem[buf ^ 1] + V4_KVBLK * 128, 128 * 8, tid, NT);
     }
 
     float vals[32];
@@ -242,10 +246,11 @@ static __device__ __forceinline__ void attn4_body(
     if (STAGE == 2 && more) {
       ushort* kd = smem[buf ^ 1];
       ushort* vd = smem[buf ^ 1] + V4_KVBLK * 128;
-      *reinterpret_cast<short8*>(kd + (long long)tid * 8) = stg[0];
-      *reinterpret_cast<short8*>(kd + (long long)(tid + 512) * 8) = stg[1];
-      *reinterpret_cast<short8*>(vd + (long long)tid * 8) = stg[2];
-      *reinterpret_cast<short8*>(vd + (long long)(tid + 512) * 8) = stg[3];
+#pragma unroll
+      for (int c = 0; c < KC; ++c) {
+        *reinterpret_cast<short8*>(kd + (long long)(tid + c * NT) * 8) = stg[c];
+        *reinterpret_cast<short8*>(vd + (long long)(tid + c * NT) * 8) = stg[KC + c];
+      }
     }
 
     if (SPLIT) {
@@ -328,10 +333,11 @@ static __device__ __forceinline__ void attn4_body(
       if (more) {
         ushort* kd = smem[0];
         ushort* vd = smem[0] + V4_KVBLK * 128;
-        *reinterpret_cast<short8*>(kd + (long long)tid * 8) = stg[0];
-        *reinterpret_cast<short8*>(kd + (long long)(tid + 512) * 8) = stg[1];
-        *reinterpret_cast<short8*>(vd + (long long)tid * 8) = stg[2];
-        *reinterpret_cast<short8*>(vd + (long long)(tid + 512) * 8) = stg[3];
+#pragma unroll
+        for (int c = 0; c < KC; ++c) {
+          *reinterpret_cast<short8*>(kd + (long long)(tid + c * NT) * 8) = stg[c];
+          *reinterpret_cast<short8*>(vd + (long long)(tid + c * NT) * 8) = stg[KC + c];
+        }
       }
       __syncthreads();  // writes visible
     } else {
@@ -340,8 +346,36 @@ static __device__ __forceinline__ void attn4_body(
     }
   }
 
-  if (!live) return;
   const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  if (TEPI) {
+    // T21: per-lane column stores are store-ISSUE-bound (64 b16 stores).
+    // Park O in LDS [d][q-in-block] and store whole rows as b128 —
+    // QBLK*16/NT (=8) store issues per thread instead of 64.
+    __syncthreads();  // last tile's K/V reads done; smem[0] is free
+    ushort* o_lds = smem[0];  // [128][QBLK] bf16 = QBLK*256 bytes <= 64 KiB
+    const int qib = wid * 32 + l31;
+#pragma unroll
+    for (int db = 0; db < 4; ++db)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+        o_lds[d * QBLK + qib] = f2bf(o_acc[db][r] * inv_l);
+      }
+    __syncthreads();
+    const int qbase = qb * QBLK;
+    constexpr int NCH = 16 * QBLK / NT;  // b128 chunks per thread
+#pragma unroll
+    for (int c = 0; c < NCH; ++c) {
+      const int ch = tid + c * NT;           // chunk over [128][QBLK/8]
+      const int d = ch / (QBLK / 8);
+      const int qc = (ch % (QBLK / 8)) * 8;  // q offset within block
+      if (qbase + qc < S)
+        *reinterpret_cast<short8*>(OTh + (long long)d * S + qbase + qc) =
+            *reinterpret_cast<const short8*>(o_lds + d * QBLK + qc);
+    }
+    return;
+  }
+  if (!live) return;
 #pragma unroll
   for (int db = 0; db < 4; ++db)
 #pragma unroll
@@ -886,4 +920,34 @@ attn_fwd_v4_15_kernel(const ushort* __restrict__ Q,
                       float scale) {
   // reg-stage double-buffer + defer + diet + sm-split
   attn4_body<2, 1, 1, 1>(Q, K, VT, OT, B, H, Hk, S, scale);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+attn_fwd_v4_16_kernel(const ushort* __restrict__ Q,
+                      const ushort* __restrict__ K,
+                      const ushort* __restrict__ VT,
+                      ushort* __restrict__ OT, int B, int H, int Hk, int S,
+                      float scale) {
+  // 4-wave blocks (QBLK 128): two co-resident blocks per CU
+  attn4_body<2, 1, 1, 1, 256>(Q, K, VT, OT, B, H, Hk, S, scale);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+attn_fwd_v4_17_kernel(const ushort* __restrict__ Q,
+                      const ushort* __restrict__ K,
+                      const ushort* __restrict__ VT,
+                      ushort* __restrict__ OT, int B, int H, int Hk, int S,
+                      float scale) {
+  // v15 + T21 LDS-transposed epilogue
+  attn4_body<2, 1, 1, 1, 512, 1>(Q, K, VT, OT, B, H, Hk, S, scale);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+attn_fwd_v4_18_kernel(const ushort* __restrict__ Q,
+                      const ushort* __restrict__ K,
+                      const ushort* __restrict__ VT,
+                      ushort* __restrict__ OT, int B, int H, int Hk, int S,
+                      float scale) {
+  // 4-wave blocks + T21 epilogue
+  attn4_body<2, 1, 1, 1, 256, 1>(Q, K, VT, OT, B, H, Hk, S, scale);
 }
