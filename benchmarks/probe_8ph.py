@@ -13,6 +13,17 @@ import time
 
 import torch
 
+
+def timeit(fn, warm=3, it=10):
+    for _ in range(warm):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(it):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / it
+
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from senweaver_amd import ops  # noqa: E402
 
@@ -42,19 +53,21 @@ for (M, N, K) in shapes:
 
     arms = {
 
-        "v13-16w": lambda: ext.gemm_bt_8ph_v(a, b, 13),
         "v14-16n": lambda: ext.gemm_bt_8ph_v(a, b, 14),
         "v18-asm": lambda: ext.gemm_bt_8ph_v(a, b, 18),
-        "v23-rng": lambda: ext.gemm_bt_8ph_v(a, b, 23),
         "v21-pri": lambda: ext.gemm_bt_8ph_v(a, b, 21),
+        "v24-gls": lambda: ext.gemm_bt_8ph_v(a, b, 24),
         "blas   ": lambda: a @ b.t(),
     }
-    # numerics check each variant once vs blas
+    # numerics check each variant once vs blas (a wrong arm is dropped,
+    # not fatal - this is a probe)
     cb = (a @ b.t()).float()
-    for name, fn in arms.items():
-        c = fn().float()
+    for name in list(arms):
+        c = arms[name]().float()
         err = (c - cb).abs().max().item()
-        assert err <= max(1.5, 0.01 * cb.abs().max().item()), (name, err)
+        if err > max(1.5, 0.01 * cb.abs().max().item()):
+            print(f"{name}: WRONG (maxabs {err:.3f}) - dropped")
+            del arms[name]
     # warmup
     for fn in arms.values():
         for _ in range(3):

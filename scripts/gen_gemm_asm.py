@@ -73,6 +73,57 @@ def gen_khalf(kx: int, tail: str) -> str:
             f'        : {clob});\n')
 
 
+def gen_khalf_glds(kx: int, tail: str) -> str:
+    """gen_khalf + 4 raw `global_load_lds_dwordx4` issues for the NEXT
+    tile's operand, interleaved one per mi-row of the FIRST MFMA block
+    (after that row's 4 MFMAs) — inside the hand-scheduled stream, AFTER
+    the ds_read burst (the stage-first anti-pattern is glds ahead of the
+    reads; v19/v20/v23 priced it at 5-15%).  m0 carries the LDS byte
+    target, bumped 8192 per issue (4 chunks x 512 threads x 16 B).
+
+    Extra operands: %14 = 64-bit sgpr base of the staged operand's K-row
+    block, %15..%18 = per-lane byte offsets (st_off), %19 = m0 base
+    (wave-uniform, readfirstlane'd by the caller).
+    """
+    L = []
+    for mi in range(4):
+        L.append(f"ds_read_b128 %{mi}, %12 offset:{mi * 2048}")
+    for ni in range(4):
+        L.append(f"ds_read_b128 %{8 + ni}, %13 offset:{ni * 2048}")
+    for mi in range(4, 8):
+        L.append(f"ds_read_b128 %{mi}, %12 offset:{mi * 2048}")
+    L.append("s_waitcnt lgkmcnt(4)")
+    for mi in range(4):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+        if mi == 0:
+            L.append("s_mov_b32 m0, %19")
+        else:
+            L.append("s_add_u32 m0, m0, 8192")
+        L.append(f"global_load_lds_dwordx4 %{15 + mi}, %14")
+    L.append("s_waitcnt lgkmcnt(0)")
+    for mi in range(4, 8):
+        for ni in range(4):
+            L.append(f"v_mfma_f32_16x16x32_bf16 {acc_range(mi, ni)}, "
+                     f"%{mi}, %{8 + ni}, {acc_range(mi, ni)}")
+    if tail:
+        L.extend(tail.split(";"))
+    body = "\\n\\t".join(L)
+    outs = ", ".join(f'"=&v"(af[{i}])' for i in range(8))
+    outs += ", " + ", ".join(f'"=&v"(bf[{i}])' for i in range(4))
+    op = "An" if kx == 0 else "Bn"
+    m0 = "m0a" if kx == 0 else "m0b"
+    ins = (f'"v"(aaddr{kx}), "v"(baddr{kx}), "s"({op}), '
+           f'"v"(st_off[0]), "v"(st_off[1]), "v"(st_off[2]), "v"(st_off[3]), '
+           f'"s"({m0})')
+    clob = ", ".join(f'"a{i}"' for i in range(128)) + ', "memory"'
+    return (f'    asm volatile(\n        "{body}"\n'
+            f'        : {outs}\n'
+            f'        : {ins}\n'
+            f'        : {clob});\n')
+
+
 def gen_tile_merged(tail: str) -> str:
     """Both k-halves in ONE statement: all 24 ds_read_b128 issued up-front,
     MFMAs retire behind a 4-step counted lgkmcnt ladder (16/12/4/0).
@@ -519,7 +570,42 @@ gemm_bt_bf16_asm2_kernel(const ushort* __restrict__ A, const ushort* __restrict_
 """    // A(t+2) into its ring slot: two tiles of flight
     AISSUE(t + 2, Atile, aslot2);
 """)
-    src = src + v20 + v21 + v22 + v23
+    # v24 (asm7): v21 + the next tile's staging issued as RAW
+    # global_load_lds inside the asm statements (A(t+1) under kh0's first
+    # MFMA block, B(t+2) under kh1's) — A gains ~a full k-half of extra
+    # flight over the between-statements burst without ever preceding the
+    # ds_read burst.
+    kh0g = gen_khalf_glds(0, tail="")
+    kh1g = gen_khalf_glds(1, tail="s_waitcnt vmcnt(4);s_barrier")
+    v24 = base.replace("gemm_bt_bf16_asm_kernel", "gemm_bt_bf16_asm7_kernel")
+    v24 = v24.replace("""  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < ntiles; ++t) {""",
+"""  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static young-half priority (T5)
+
+  for (int t = 0; t < ntiles; ++t) {""")
+    ia = v24.index("    short8 af[8], bf[4];")
+    ib = v24.index("  }\n#undef AISSUE")
+    v24 = (v24[:ia]
+           + """    const unsigned wcb = (unsigned)((tid & ~63) * 16);
+    const ushort* An = Atile + (long long)(t + 1) * 64;
+    const ushort* Bn = Btile + (long long)(t + 2) * 64;
+    const unsigned m0a = __builtin_amdgcn_readfirstlane(
+        (unsigned)(unsigned long long)(__attribute__((address_space(3))) char*)&lds[aslot ^ 1][0] + wcb);
+    const unsigned m0b = __builtin_amdgcn_readfirstlane(
+        (unsigned)(unsigned long long)(__attribute__((address_space(3))) char*)&lds[bslot2][0] + wcb);
+    short8 af[8], bf[4];
+    if (t < ntiles - 2) {
+""" + kh0g + kh1g + """    } else if (t == ntiles - 2) {
+""" + kh0g + kh1_drain + """    } else {
+""" + kh0 + kh1_drain + """    }
+"""
+           + v24[ib:])
+    src = src + v20 + v21 + v22 + v23 + v24
     with open(os.path.abspath(OUT), "w") as f:
         f.write(src)
     print("wrote", OUT, len(src), "chars")

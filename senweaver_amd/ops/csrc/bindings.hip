@@ -43,6 +43,7 @@ extern "C" __global__ void gemm_bt_bf16_asm3_kernel(const ushort*, const ushort*
 extern "C" __global__ void gemm_bt_bf16_asm4_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_asm5_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_bf16_asm6_kernel(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemm_bt_bf16_asm7_kernel(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void grouped_gemm_bt_bf16_kernel(const ushort*, const ushort*, ushort*, const int*, const int*, const int*, int, int, int);
 extern "C" __global__ void moe_combine_kernel(const ushort*, const int*, const float*, ushort*, int, int);
 extern "C" __global__ void quant_fp8_rowwise_kernel(const ushort*, unsigned char*, float*, int);
@@ -263,11 +264,15 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
   if (M % 256 == 0 && N % 256 == 0 && (M / 256) * (N / 256) >= 160) {
     const int nwg = (M / 256) * (N / 256);
     if (K % 128 == 0)
-      // in-repo best per shape (profiles/r02_gemm_pipeline.txt): the
-      // hand-scheduled asm kernel (AGPR accumulators, counted lgkm ladder,
-      // static young-half prio) wins at M>=4096 (1.27/1.42 PF at 4k/8k);
-      // the 16-wave plain-HIP pipeline wins the M=2048 wide shapes
-      if (M >= 4096)
+      // in-repo best per shape (profiles/r02_gemm_pipeline.txt +
+      // r02_gemm_asm_glds.txt): asm7 (raw glds staged INSIDE the MFMA
+      // stream) wins big-M and wide-N shapes (1.41/1.23 PF at 8k/gateup);
+      // asm4 (between-statement staging) keeps 4096-class; the 16-wave
+      // plain-HIP pipeline keeps the remaining small-M shapes
+      if (M >= 8192 || (long long)N * 2 >= (long long)M * 7)
+        gemm_bt_bf16_asm7_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
+            bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
+      else if (M >= 4096)
         gemm_bt_bf16_asm4_kernel<<<dim3(nwg), dim3(512), 0, cur_stream()>>>(
             bf16_ptr(A), bf16_ptr(B), bf16_mut(C), M, N, K);
       else
@@ -669,6 +674,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       case 22: gemm_bt_bf16_asm5_kernel<<<g, dim3(512), 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
       case 23: gemm_bt_bf16_asm6_kernel<<<g, dim3(512), 0, cur_stream()>>>(
+                  bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
+      case 24: gemm_bt_bf16_asm7_kernel<<<g, dim3(512), 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
       default: gemm_bt_bf16_8ph_kernel<<<g, blk, 0, cur_stream()>>>(
                   bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K); break;
