@@ -53,6 +53,8 @@ extern "C" __global__ void gemm_bt_mxfp8_kernel(const unsigned char*, const unsi
 extern "C" __global__ void gemm_bt_mxfp8_256_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemm_bt_mxfp8_pipe_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m1(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_v3_m1(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_bf16_v3w_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m4(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m8(const ushort*, const ushort*, ushort*, int, int, int);
@@ -248,6 +250,16 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B) {
       kern<<<grid, dim3(256), 0, cur_stream()>>>(bf16_ptr(Ap), bf16_ptr(B),
                                                  bf16_mut(C), M, N, K);
     };
+    // M=1 wide-K rows: the loader/consumer LDS-DMA streaming engine
+    // measures +25% over v2 at K=14336 (5.9 vs 4.7 TB/s,
+    // profiles/r02_gemv_engine.txt); at K<=4096 its per-block prologue
+    // (x stage + 2-step ring fill) can't amortize over 4 steps and v2 wins
+    if (MM == 1 && K % 1024 == 0 && K > 4096 && K <= 14336 && N % 8 == 0) {
+      gemv_bt_bf16_v3w_m1<<<dim3(N / 8), dim3(256), 0, cur_stream()>>>(
+          bf16_ptr(Ap), bf16_ptr(B), bf16_mut(C), M, N, K);
+      HIP_CHECK_KERNEL();
+      return C;
+    }
     switch (MM) {
       case 1: launch(gemv_bt_bf16_v2_m1); break;
       case 2: launch(gemv_bt_bf16_v2_m2); break;
@@ -683,6 +695,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     HIP_CHECK_KERNEL();
     return C;
   }, "8-phase GEMM structural variants (A/B probe)");
+  m.def("gemv_bt_v3", [](torch::Tensor a, torch::Tensor b) {
+    check_bf16(a, "a"); check_bf16(b, "b");
+    const int M = a.size(0), K = a.size(1), N = b.size(0);
+    TORCH_CHECK(M == 1 && K % 512 == 0 && K <= 14336 && N % 8 == 0);
+    auto C = torch::empty({M, N}, a.options());
+    if (K <= 4096)
+      gemv_bt_bf16_v3_m1<<<dim3(N / 8), dim3(256), 0, cur_stream()>>>(
+          bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K);
+    else
+      gemv_bt_bf16_v3w_m1<<<dim3(N / 8), dim3(256), 0, cur_stream()>>>(
+          bf16_ptr(a), bf16_ptr(b), bf16_mut(C), M, N, K);
+    HIP_CHECK_KERNEL();
+    return C;
+  }, "loader/consumer LDS-DMA streaming GEMV (M=1)");
   m.def("gemm_bt_256x32", [](torch::Tensor a, torch::Tensor b) {
     check_bf16(a, "a"); check_bf16(b, "b");
     const int M = a.size(0), K = a.size(1), N = b.size(0);

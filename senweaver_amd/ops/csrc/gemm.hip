@@ -308,6 +308,245 @@ gemm_bt_bf16_256_kernel(const ushort* __restrict__ A, const ushort* __restrict__
     }                                                                         \
   }
 
+// ---------------------------------------------------------------------------
+// Decode GEMV v3 — loader/consumer LDS-DMA streaming engine (M = 1).
+// The v2 form (one wave per row, 2-deep nt register pipeline) measures ~85%
+// of the achievable HBM rate: its in-flight requests are single 1 KiB row
+// chunks scattered across many DRAM pages.  The microarch price list
+// (ldsdma-fill / nt-weights rows) measures 6.5-6.8 TB/s chip-wide when each
+// CU streams CONTIGUOUS 16 KiB fills via global_load_lds with nt.  Here a
+// block owns 16 consecutive output rows and walks K in 512-element chunks:
+// slot = 16 rows x 1 KiB, ring of 5 slots in LDS (<= 3 in flight keeps the
+// counted vmcnt under the 6-bit 63 cap), wave 3 is loader AND consumer, all
+// 4 waves split the 16 rows (16 lanes per row, 32 elems per lane per slot),
+// x is staged to LDS once up front.  One barrier per slot; counted
+// s_waitcnt vmcnt(16*inflight) before it (wait-then-barrier).
+// Constraints: M == 1, K % 512 == 0, N % 16 == 0.
+// ---------------------------------------------------------------------------
+#define GV3_KW 512         /* chunk elems per row per slot */
+
+template <int XCAP, int RD>
+static __device__ __forceinline__ void gemv3_body(
+    const ushort* __restrict__ A, const ushort* __restrict__ B,
+    ushort* __restrict__ C, int N, int K) {
+  // 8 rows per block, slot = 8 rows x 1 KiB = 8 KiB, ring of RD slots.
+  // K <= 4096: ring 7*8 + x 8 = 64 KiB -> TWO blocks/CU (one block's
+  // barrier/latency stalls hide under the other's stream).  K <= 14336
+  // (down-proj): x is 28 KiB -> one block/CU.
+  const int n0 = blockIdx.x * 8;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+
+  __shared__ __attribute__((aligned(16))) ushort gv3_smem[RD * 8 * GV3_KW + XCAP];
+  ushort* slots = gv3_smem;
+  ushort* x_lds = gv3_smem + RD * 8 * GV3_KW;
+
+  const int nslots = K / GV3_KW;
+
+  if (wid < 3) {
+    // stage x (1 KiB chunks, lane-linear)
+    for (int c = wid; c < nslots; c += 3)
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)
+              (A + (long long)c * GV3_KW + lane * 8),
+          (__attribute__((address_space(3))) unsigned int*)
+              (x_lds + (long long)c * GV3_KW + lane * 8),
+          16, 0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else {
+    // loader prologue: slots 0..RD-2 (8 nt glds each), then slot 0 landed
+    for (int t = 0; t < RD - 1 && t < nslots; ++t) {
+      const ushort* src0 = B + (long long)n0 * K + (long long)t * GV3_KW;
+      ushort* dst0 = slots + (t % RD) * 8 * GV3_KW;
+#pragma unroll
+      for (int r = 0; r < 8; ++r)
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)
+                (src0 + (long long)r * K + lane * 8),
+            (__attribute__((address_space(3))) unsigned int*)
+                (dst0 + r * GV3_KW + lane * 8),
+            16, 0, 2 /* nt */);
+    }
+    const int inflight0 = ((RD - 2) < (nslots - 1) ? (RD - 2) : (nslots - 1));
+    switch (inflight0 < 0 ? 0 : inflight0) {
+      case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+      case 1: asm volatile("s_waitcnt vmcnt(8)" ::: "memory"); break;
+      case 2: asm volatile("s_waitcnt vmcnt(16)" ::: "memory"); break;
+      case 3: asm volatile("s_waitcnt vmcnt(24)" ::: "memory"); break;
+      case 4: asm volatile("s_waitcnt vmcnt(32)" ::: "memory"); break;
+      default: asm volatile("s_waitcnt vmcnt(40)" ::: "memory"); break;
+    }
+  }
+  __builtin_amdgcn_s_barrier();
+
+  // consumer: wave w owns rows 2w, 2w+1; reads are lane-linear b128
+  // (conflict-free — the j*32 strided form measured a 16-way bank storm)
+  float acc0 = 0.f, acc1 = 0.f;
+  const int r0 = wid * 2;
+  for (int t = 0; t < nslots; ++t) {
+    const ushort* slot = slots + (t % RD) * 8 * GV3_KW;
+    bf16x8 xv = *reinterpret_cast<const bf16x8*>(
+        x_lds + (long long)t * GV3_KW + lane * 8);
+    bf16x8 w0 = *reinterpret_cast<const bf16x8*>(slot + r0 * GV3_KW + lane * 8);
+    bf16x8 w1 = *reinterpret_cast<const bf16x8*>(slot + (r0 + 1) * GV3_KW + lane * 8);
+    float xf[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) xf[e] = bf2f(xv.v[e]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc0 += bf2f(w0.v[e]) * xf[e];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc1 += bf2f(w1.v[e]) * xf[e];
+    if (wid == 3) {
+      const int nxt = t + RD - 1;
+      if (nxt < nslots) {
+        const ushort* src0 = B + (long long)n0 * K + (long long)nxt * GV3_KW;
+        ushort* dst0 = slots + (nxt % RD) * 8 * GV3_KW;
+#pragma unroll
+        for (int r = 0; r < 8; ++r)
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) unsigned int*)
+                  (src0 + (long long)r * K + lane * 8),
+              (__attribute__((address_space(3))) unsigned int*)
+                  (dst0 + r * GV3_KW + lane * 8),
+              16, 0, 2);
+      }
+      const int last = (nxt < nslots ? nxt : nslots - 1);
+      const int inflight = last - (t + 1);
+      switch (inflight < 0 ? 0 : inflight) {
+        case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+        case 1: asm volatile("s_waitcnt vmcnt(8)" ::: "memory"); break;
+        case 2: asm volatile("s_waitcnt vmcnt(16)" ::: "memory"); break;
+        case 3: asm volatile("s_waitcnt vmcnt(24)" ::: "memory"); break;
+        case 4: asm volatile("s_waitcnt vmcnt(32)" ::: "memory"); break;
+        default: asm volatile("s_waitcnt vmcnt(40)" ::: "memory"); break;
+      }
+    }
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+
+  acc0 = wave_reduce_sum(acc0);
+  acc1 = wave_reduce_sum(acc1);
+  if (lane == 0 && n0 + r0 < N) C[n0 + r0] = f2bf(acc0);
+  if (lane == 0 && n0 + r0 + 1 < N) C[n0 + r0 + 1] = f2bf(acc1);
+}
+
+// two slots per barrier round: the 1-slot cadence measured ~0.7 us/slot
+// against a 0.32 us fill (barrier+loader-wait dominated); pairing slots
+// halves the sync cost per byte.  Ring 6 (steps in flight: consuming,
+// landed, in-flight = 3 steps x 2 slots).  Needs nslots even (K % 1024).
+template <int XCAP>
+static __device__ __forceinline__ void gemv3_body2(
+    const ushort* __restrict__ A, const ushort* __restrict__ B,
+    ushort* __restrict__ C, int N, int K) {
+  const int n0 = blockIdx.x * 8;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+
+  __shared__ __attribute__((aligned(16))) ushort gv3_smem[6 * 8 * GV3_KW + XCAP];
+  ushort* slots = gv3_smem;
+  ushort* x_lds = gv3_smem + 6 * 8 * GV3_KW;
+
+  const int nsteps = K / (2 * GV3_KW);
+
+#define GV3_ISSUE_STEP(ST)                                                    \
+  do {                                                                        \
+    const ushort* s0_ = B + (long long)n0 * K + (long long)(ST) * 2 * GV3_KW; \
+    ushort* d0_ = slots + (((ST) * 2) % 6) * 8 * GV3_KW;                      \
+    _Pragma("unroll") for (int r = 0; r < 8; ++r) {                           \
+      __builtin_amdgcn_global_load_lds(                                       \
+          (const __attribute__((address_space(1))) unsigned int*)             \
+              (s0_ + (long long)r * K + lane * 8),                            \
+          (__attribute__((address_space(3))) unsigned int*)                   \
+              (d0_ + r * GV3_KW + lane * 8),                                  \
+          16, 0, 2);                                                          \
+      __builtin_amdgcn_global_load_lds(                                       \
+          (const __attribute__((address_space(1))) unsigned int*)             \
+              (s0_ + (long long)r * K + GV3_KW + lane * 8),                   \
+          (__attribute__((address_space(3))) unsigned int*)                   \
+              (d0_ + 8 * GV3_KW + r * GV3_KW + lane * 8),                     \
+          16, 0, 2);                                                          \
+    }                                                                         \
+  } while (0)
+
+  if (wid < 3) {
+    for (int c = wid; c < K / GV3_KW; c += 3)
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)
+              (A + (long long)c * GV3_KW + lane * 8),
+          (__attribute__((address_space(3))) unsigned int*)
+              (x_lds + (long long)c * GV3_KW + lane * 8),
+          16, 0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else {
+    GV3_ISSUE_STEP(0);
+    if (1 < nsteps) {
+      GV3_ISSUE_STEP(1);
+      asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+  }
+  __builtin_amdgcn_s_barrier();
+
+  float acc0 = 0.f, acc1 = 0.f;
+  const int r0 = wid * 2;
+  for (int st = 0; st < nsteps; ++st) {
+    const ushort* slot0 = slots + ((st * 2) % 6) * 8 * GV3_KW;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const ushort* slot = slot0 + h * 8 * GV3_KW;
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(
+          x_lds + ((long long)st * 2 + h) * GV3_KW + lane * 8);
+      bf16x8 w0 = *reinterpret_cast<const bf16x8*>(slot + r0 * GV3_KW + lane * 8);
+      bf16x8 w1 = *reinterpret_cast<const bf16x8*>(slot + (r0 + 1) * GV3_KW + lane * 8);
+      float xf[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) xf[e] = bf2f(xv.v[e]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc0 += bf2f(w0.v[e]) * xf[e];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc1 += bf2f(w1.v[e]) * xf[e];
+    }
+    if (wid == 3) {
+      const int nxt = st + 2;
+      if (nxt < nsteps) {
+        GV3_ISSUE_STEP(nxt);
+        // outstanding: step st+1 (16, oldest) + step st+2 (16)
+        asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+      } else if (st + 1 < nsteps) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+    }
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+#undef GV3_ISSUE_STEP
+
+  acc0 = wave_reduce_sum(acc0);
+  acc1 = wave_reduce_sum(acc1);
+  if (lane == 0 && n0 + r0 < N) C[n0 + r0] = f2bf(acc0);
+  if (lane == 0 && n0 + r0 + 1 < N) C[n0 + r0 + 1] = f2bf(acc1);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+gemv_bt_bf16_v3_m1(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                   ushort* __restrict__ C, int M, int N, int K) {
+  // 56 KiB LDS -> two blocks/CU; two slots per barrier round
+  gemv3_body2<4096>(A, B, C, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+gemv_bt_bf16_v3w_m1(const ushort* __restrict__ A, const ushort* __restrict__ B,
+                    ushort* __restrict__ C, int M, int N, int K) {
+  // wide-K (down-proj): 76 KiB LDS -> one block/CU
+  gemv3_body2<14336>(A, B, C, N, K);
+}
+
 GEMV2_INST(1)
 GEMV2_INST(2)
 GEMV2_INST(4)
