@@ -227,8 +227,12 @@ class LlamaModel:
         c = self.config
         if c.num_experts == 0:
             gateup = self._linear(h, L, "gateup")
-            act = ops.swiglu(gateup)
-            out = self._linear(act, L, "down")
+            if (self.quant == "bf16" and gateup.shape[0] == 1
+                    and gateup.is_cuda):
+                out = ops.swiglu_gemv(gateup, L["down"])
+            else:
+                act = ops.swiglu(gateup)
+                out = self._linear(act, L, "down")
             self.tp.all_reduce_(out)  # row-parallel down_proj partial sum
             return out
         return self._moe_ffn(h, L)

@@ -173,6 +173,23 @@ _GEMM_IMPL = os.environ.get("SENWEAVER_GEMM", "auto")  # auto | hip
 _ATTN_IMPL = os.environ.get("SENWEAVER_ATTN", "auto")  # auto | v2
 
 
+def swiglu_gemv(gateup: torch.Tensor, w_down: torch.Tensor) -> torch.Tensor:
+    """Fused silu(g)*u + down-projection GEMV (decode, M=1, wide K).
+
+    gateup [1, 2K]; w_down [N, K].  GPU: the loader/consumer LDS-DMA
+    streaming engine computes the activation into LDS on the way in —
+    one kernel replaces (swiglu + gemv).  Reference parity: reference
+    decoder MLP behavior; numerics vs the fp32 torch reference in
+    tests/test_kernels_gpu.py.
+    """
+    K = gateup.shape[-1] // 2
+    if (_on_gpu(gateup) and gateup.shape[0] == 1 and K % 1024 == 0
+            and 4096 < K <= 14336 and w_down.shape[0] % 8 == 0):
+        return hip_ext().swiglu_gemv_bt(gateup.contiguous(), w_down.contiguous())
+    act = swiglu(gateup)
+    return gemm_bt(act, w_down)
+
+
 def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """C[M,N] = A[M,K] @ B[N,K]^T.
 

@@ -55,6 +55,7 @@ extern "C" __global__ void gemm_bt_mxfp8_pipe_kernel(const unsigned char*, const
 extern "C" __global__ void gemv_bt_bf16_v2_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v3_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v3w_m1(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void swiglu_gemv_bt_bf16_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m4(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m8(const ushort*, const ushort*, ushort*, int, int, int);
@@ -709,6 +710,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     HIP_CHECK_KERNEL();
     return C;
   }, "loader/consumer LDS-DMA streaming GEMV (M=1)");
+  m.def("swiglu_gemv_bt", [](torch::Tensor gu, torch::Tensor b) {
+    check_bf16(gu, "gateup"); check_bf16(b, "w");
+    const int M = gu.size(0), K2 = gu.size(1), N = b.size(0);
+    const int K = K2 / 2;
+    TORCH_CHECK(M == 1 && K2 % 2 == 0 && K % 1024 == 0 && K > 4096 &&
+                K <= 14336 && N % 8 == 0 && b.size(1) == K);
+    auto C = torch::empty({M, N}, gu.options());
+    swiglu_gemv_bt_bf16_m1<<<dim3(N / 8), dim3(256), 0, cur_stream()>>>(
+        bf16_ptr(gu), bf16_ptr(b), bf16_mut(C), M, N, K);
+    HIP_CHECK_KERNEL();
+    return C;
+  }, "fused silu(g)*u + streaming GEMV (decode down-proj, M=1)");
   m.def("gemm_bt_256x32", [](torch::Tensor a, torch::Tensor b) {
     check_bf16(a, "a"); check_bf16(b, "b");
     const int M = a.size(0), K = a.size(1), N = b.size(0);

@@ -437,7 +437,7 @@ static __device__ __forceinline__ void gemv3_body(
 // against a 0.32 us fill (barrier+loader-wait dominated); pairing slots
 // halves the sync cost per byte.  Ring 6 (steps in flight: consuming,
 // landed, in-flight = 3 steps x 2 slots).  Needs nslots even (K % 1024).
-template <int XCAP>
+template <int XCAP, int FUSE_SWIGLU = 0>
 static __device__ __forceinline__ void gemv3_body2(
     const ushort* __restrict__ A, const ushort* __restrict__ B,
     ushort* __restrict__ C, int N, int K) {
@@ -473,14 +473,35 @@ static __device__ __forceinline__ void gemv3_body2(
   } while (0)
 
   if (wid < 3) {
-    for (int c = wid; c < K / GV3_KW; c += 3)
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)
-              (A + (long long)c * GV3_KW + lane * 8),
-          (__attribute__((address_space(3))) unsigned int*)
-              (x_lds + (long long)c * GV3_KW + lane * 8),
-          16, 0, 0);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (FUSE_SWIGLU) {
+      // x = silu(g) * u computed on the way in: A = gateup [1, 2K],
+      // g = A[0..K), u = A[K..2K) — replaces the standalone swiglu
+      // kernel (one ~5 us launch per decode layer)
+      for (int c = wid; c < K / GV3_KW; c += 3) {
+        bf16x8 g = *reinterpret_cast<const bf16x8*>(
+            A + (long long)c * GV3_KW + lane * 8);
+        bf16x8 u = *reinterpret_cast<const bf16x8*>(
+            A + (long long)K + (long long)c * GV3_KW + lane * 8);
+        bf16x8 out;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float gf = bf2f(g.v[e]);
+          out.v[e] = f2bf(gf / (1.f + __expf(-gf)) * bf2f(u.v[e]));
+        }
+        *reinterpret_cast<bf16x8*>(x_lds + (long long)c * GV3_KW + lane * 8) =
+            out;
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // writes visible
+    } else {
+      for (int c = wid; c < K / GV3_KW; c += 3)
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)
+                (A + (long long)c * GV3_KW + lane * 8),
+            (__attribute__((address_space(3))) unsigned int*)
+                (x_lds + (long long)c * GV3_KW + lane * 8),
+            16, 0, 0);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
   } else {
     GV3_ISSUE_STEP(0);
     if (1 < nsteps) {
@@ -545,6 +566,14 @@ gemv_bt_bf16_v3w_m1(const ushort* __restrict__ A, const ushort* __restrict__ B,
                     ushort* __restrict__ C, int M, int N, int K) {
   // wide-K (down-proj): 76 KiB LDS -> one block/CU
   gemv3_body2<14336>(A, B, C, N, K);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+swiglu_gemv_bt_bf16_m1(const ushort* __restrict__ GU,
+                       const ushort* __restrict__ B,
+                       ushort* __restrict__ C, int M, int N, int K) {
+  // fused silu(g)*u + down-proj GEMV (decode): GU = gateup [1, 2K]
+  gemv3_body2<14336, 1>(GU, B, C, N, K);
 }
 
 GEMV2_INST(1)

@@ -577,3 +577,16 @@ def test_gemv_v3_engine(dev):
     via_dispatch = ops.hip_ext().gemm_bt(x, w)
     torch.testing.assert_close(via_dispatch.float(), x.float() @ w.float().t(),
                                atol=2.0, rtol=2e-2)
+
+
+def test_swiglu_gemv_fused(dev):
+    """Fused silu(g)*u + down-proj GEMV vs the two-op fp32 reference."""
+    N, K = 4096, 14336
+    gu = torch.randn(1, 2 * K, dtype=torch.bfloat16, device=dev)
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+    got = ops.swiglu_gemv(gu, w)
+    g = gu[:, :K].float()
+    u = gu[:, K:].float()
+    act = g * torch.sigmoid(g) * u
+    ref_out = act @ w.float().t()
+    torch.testing.assert_close(got.float(), ref_out, atol=3.0, rtol=3e-2)
