@@ -14,6 +14,7 @@ is declared synthetic).
 from __future__ import annotations
 
 import math
+import os
 from typing import List, Optional
 
 import torch
@@ -228,7 +229,8 @@ class LlamaModel:
         if c.num_experts == 0:
             gateup = self._linear(h, L, "gateup")
             if (self.quant == "bf16" and gateup.shape[0] == 1
-                    and gateup.is_cuda):
+                    and gateup.is_cuda
+                    and os.environ.get("SENWEAVER_SWIGLU_FUSE", "0") == "1"):
                 out = ops.swiglu_gemv(gateup, L["down"])
             else:
                 act = ops.swiglu(gateup)
