@@ -42,8 +42,11 @@ class LlamaModel:
         # (BASELINE needs neither combined; the collectives would nest).
         self.ep = ep if ep is not None else EPContext()
         if self.ep.world > 1:
-            assert self.tp.world == 1, "EP and TP are not combined"
             assert config.num_experts % self.ep.world == 0
+            # EP composes with TP (parallel/grid.py): experts sharded over
+            # the EP group, each local expert TP-sharded like the dense FFN.
+            # The quantized grouped-GEMM path is EP-unaware (global counts).
+            assert quant == "bf16", "EP supports the bf16 expert path"
         c = config
         tpw = self.tp.world
         if tpw > 1:
@@ -100,22 +103,21 @@ class LlamaModel:
                 layer["router"] = W(c.num_experts, c.hidden_size)
                 w13_full = W(c.num_experts, 2 * c.intermediate_size, c.hidden_size)
                 w2_full = W(c.num_experts, c.hidden_size, c.intermediate_size)
-                if tpw > 1:
+                if self.ep.world > 1:  # EP first: slice this rank's experts
+                    elo, ehi = self.ep.local_experts(c.num_experts)
+                    w13_full = w13_full[elo:ehi]
+                    w2_full = w2_full[elo:ehi]
+                if tpw > 1:  # then TP-shard each (local) expert
                     layer["w13"] = torch.stack([
                         shard_gateup(w13_full[e], c.intermediate_size, self.tp)
-                        for e in range(c.num_experts)]).contiguous()
+                        for e in range(w13_full.shape[0])]).contiguous()
                     layer["w2"] = torch.stack([
                         shard_rows(w2_full[e], self.tp, dim=1)
-                        for e in range(c.num_experts)]).contiguous()
-                    del w13_full, w2_full
-                elif self.ep.world > 1:
-                    elo, ehi = self.ep.local_experts(c.num_experts)
-                    layer["w13"] = w13_full[elo:ehi].contiguous()
-                    layer["w2"] = w2_full[elo:ehi].contiguous()
+                        for e in range(w2_full.shape[0])]).contiguous()
                     del w13_full, w2_full
                 else:
-                    layer["w13"] = w13_full
-                    layer["w2"] = w2_full
+                    layer["w13"] = w13_full.contiguous()
+                    layer["w2"] = w2_full.contiguous()
             else:
                 gateup_full = W(2 * c.intermediate_size, c.hidden_size)
                 down_full = W(c.hidden_size, c.intermediate_size)
@@ -289,7 +291,9 @@ class LlamaModel:
                               device=h.device)
             res.index_add_(0, sorted_token,
                            down.float() * sorted_weight.unsqueeze(-1).float())
-            return res.to(h.dtype)
+            out = res.to(h.dtype)
+            self.tp.all_reduce_(out)  # row-parallel w2 partial sum (TP x EP)
+            return out
         if h.is_cuda:
             # sync-free routing: device-side cumulative ends feed the grouped
             # GEMM directly (the per-expert int(counts[e]) reads were 8 tiny

@@ -313,3 +313,36 @@ def test_ep2_matches_replicated():
     for rank, vals in results.items():
         assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
     assert results[0] == pytest.approx(results[1], abs=1e-5)
+
+
+def _payload_tp2_ep2_grid(rank, world):
+    import torch
+    from senweaver_amd.models import tiny_moe_tp
+    from senweaver_amd.models.llama import LlamaModel
+    from senweaver_amd.parallel import build_tp_ep_grid
+
+    tp, ep = build_tp_ep_grid(2, 2)
+    model = LlamaModel(tiny_moe_tp(), device="cpu", seed=13, tp=tp, ep=ep)
+    tokens = torch.randint(0, 512, (1, 48),
+                           generator=torch.Generator().manual_seed(6))
+    hidden = model.prefill(tokens)
+    return hidden.float().sum(-1).squeeze(0).tolist()[:8]
+
+
+def test_tp2_ep2_grid_matches_unsharded():
+    """World 4 as a TP2 x EP2 grid (subgroups): attention sharded over the
+    TP pair, experts sharded over the EP pair, result identical to the
+    unsharded single-rank forward."""
+    import torch
+    from senweaver_amd.models import tiny_moe_tp
+    from senweaver_amd.models.llama import LlamaModel
+
+    ref_model = LlamaModel(tiny_moe_tp(), device="cpu", seed=13)
+    tokens = torch.randint(0, 512, (1, 48),
+                           generator=torch.Generator().manual_seed(6))
+    ref = ref_model.prefill(tokens).float().sum(-1).squeeze(0).tolist()[:8]
+
+    results = _run_dist("_payload_tp2_ep2_grid", world=4)
+    for rank, vals in results.items():
+        assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
+    assert results[0] == pytest.approx(results[3], abs=1e-5)
