@@ -18,7 +18,7 @@ from senweaver_amd.ops import reference as ref  # noqa: E402
 
 dev = "cuda:0"
 ext = ops.hip_ext()
-VARS = [15, 16, 17, 18]
+VARS = [0, 1, 3, 5, 11, 12, 13, 14, 15]
 if len(sys.argv) > 1 and sys.argv[1] == "pmc":
     # counter run: a few dispatches of anchor + winner only, bench shape
     dev = "cuda:0"

@@ -591,6 +591,236 @@ static __device__ __forceinline__ void attn4_body2(
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// v6: persistent workgroups + cross-seam prefetch.  At S=2048 the launched
+// v15 pays ~4 us per 256-row block (prologue stage stall + Q load + tail)
+// and blocks run back-to-back on a CU at 1 block/CU — fully serialized.
+// Here gridDim.x (= #CUs' worth) blocks walk work items; during an item's
+// LAST tile the regular staging path parks the NEXT item's tile 0 in the
+// idle buffer, so the seam costs only a qf reload.  Work items decode with
+// qb heaviest-first, matching the launched mapping.
+// ---------------------------------------------------------------------------
+template <int DEFER, int DIET>
+static __device__ __forceinline__ void attn4_body_p(
+    const ushort* __restrict__ Q, const ushort* __restrict__ K,
+    const ushort* __restrict__ VT, ushort* __restrict__ OT, int B, int H,
+    int Hk, int S, float scale) {
+  const int D = 128;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int l31 = lane & 31;
+  const int lhi = lane >> 5;
+  const int nqb = (S + 255) / 256;
+  const int NW = nqb * H * B;
+
+  __shared__ __attribute__((aligned(16))) ushort smem[2][V4_KVBLK * 128 + 128 * V4_KVBLK];
+
+  const float scl = DIET ? scale * 1.44269504f : scale;
+  int buf = 0;
+  bool first = true;
+
+  for (int item = blockIdx.x; item < NW; item += gridDim.x) {
+    // decode work item (qb heaviest-first, like the launched grid)
+    const int x = item % nqb;
+    const int qb = nqb - 1 - x;
+    const int h = (item / nqb) % H;
+    const int b = item / (nqb * H);
+    const int kvh = h / (H / Hk);
+    const ushort* Qh = Q + (((long long)b * H + h) * S) * D;
+    const ushort* Kh = K + (((long long)b * Hk + kvh) * S) * D;
+    const ushort* VTh = VT + (((long long)b * Hk + kvh) * D) * S;
+    ushort* OTh = OT + (((long long)b * H + h) * D) * S;
+    const int q0 = qb * V4_QBLK + wid * 32;
+    const int q_lane = q0 + l31;
+    const bool live = q_lane < S;
+    const int kv_end = min(S, qb * V4_QBLK + V4_QBLK);
+
+    const bool more_items = item + gridDim.x < NW;
+
+    if (first) {  // only ever once: later items find tile 0 parked
+      v4_stage_glds(Kh, D, 16, smem[buf], V4_KVBLK * 16, tid, 512);
+      v4_stage_glds(VTh, S, 8, smem[buf] + V4_KVBLK * 128, 128 * 8, tid, 512);
+      first = false;
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+
+    short8 qf[8];
+    {
+      const long long qrow = (long long)(live ? q_lane : 0) * D;
+#pragma unroll
+      for (int st = 0; st < 8; ++st)
+        qf[st] = *reinterpret_cast<const short8*>(Qh + qrow + st * 16 + lhi * 8);
+    }
+
+    float m_run = -INFINITY, l_run = 0.f;
+    f32x16 o_acc[4];
+#pragma unroll
+    for (int db = 0; db < 4; ++db)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) o_acc[db][r] = 0.f;
+
+    for (int kv0 = 0; kv0 < kv_end; kv0 += V4_KVBLK) {
+      const bool more_tiles = kv0 + V4_KVBLK < kv_end;
+      const bool park = more_tiles || more_items;
+      const ushort* k_lds = smem[buf];
+      const ushort* vt_lds = smem[buf] + V4_KVBLK * 128;
+
+      __builtin_amdgcn_s_setprio(1);
+      f32x16 st[2];
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        f32x16 acc;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+#pragma unroll
+        for (int stp = 0; stp < 8; ++stp) {
+          short8 kf = v4_read(k_lds, sub * 32 + l31, stp * 2 + lhi, 16);
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp], acc, 0, 0, 0);
+        }
+        st[sub] = acc;
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // stage: next tile of THIS item, or tile 0 of the NEXT item
+      // (next-item bases decoded HERE so they are never live across the
+      // tile loop — computed at item top they spilled 388 B/lane)
+      short8 stg[4];
+      if (park) {
+        const ushort* Kn;
+        const ushort* VTn;
+        if (more_tiles) {
+          Kn = Kh + (long long)(kv0 + V4_KVBLK) * D;
+          VTn = VTh + kv0 + V4_KVBLK;
+        } else {
+          const int item2 = item + gridDim.x;
+          const int h2 = (item2 / nqb) % H;
+          const int b2 = item2 / (nqb * H);
+          const int kvh2 = h2 / (H / Hk);
+          Kn = K + (((long long)b2 * Hk + kvh2) * S) * D;
+          VTn = VT + (((long long)b2 * Hk + kvh2) * D) * S;
+        }
+        stg[0] = v4_gload(Kn, D, 16, tid);
+        stg[1] = v4_gload(Kn, D, 16, tid + 512);
+        stg[2] = v4_gload(VTn, S, 8, tid);
+        stg[3] = v4_gload(VTn, S, 8, tid + 512);
+      }
+
+      float vals[32];
+      float tile_max = -INFINITY;
+      const bool interior = DIET && (kv0 + V4_KVBLK - 1 <= q0);
+      if (interior) {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const float v = st[sub][r] * scl;
+            vals[sub * 16 + r] = v;
+            tile_max = v4_max(tile_max, v);
+          }
+      } else {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kv = kv0 + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+            float v = st[sub][r] * scl;
+            if (kv > q_lane) v = -INFINITY;
+            vals[sub * 16 + r] = v;
+            tile_max = DIET ? v4_max(tile_max, v) : fmaxf(tile_max, v);
+          }
+      }
+      tile_max = DIET ? v4_max(tile_max, __shfl_xor(tile_max, 32, 64))
+                      : fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+      float m_new, alpha;
+      bool defer = false;
+      if (DEFER) defer = __all((int)(tile_max - m_run <= (DIET ? 11.54f : 8.f)));
+      if (defer) {
+        m_new = m_run;
+        alpha = 1.f;
+      } else {
+        m_new = fmaxf(m_run, tile_max);
+        alpha = (m_run == -INFINITY)
+                    ? 0.f
+                    : (DIET ? v4_exp2(m_run - m_new) : __expf(m_run - m_new));
+      }
+      float rsum = 0.f;
+      m_run = m_new;
+      if (!defer) {
+#pragma unroll
+        for (int db = 0; db < 4; ++db)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
+      }
+
+      if (park) {
+        ushort* kd = smem[buf ^ 1];
+        ushort* vd = smem[buf ^ 1] + V4_KVBLK * 128;
+        *reinterpret_cast<short8*>(kd + (long long)tid * 8) = stg[0];
+        *reinterpret_cast<short8*>(kd + (long long)(tid + 512) * 8) = stg[1];
+        *reinterpret_cast<short8*>(vd + (long long)tid * 8) = stg[2];
+        *reinterpret_cast<short8*>(vd + (long long)(tid + 512) * 8) = stg[3];
+      }
+
+      // sm-split PV
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+          const int base = sub * 16 + g * 8;
+          float p[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            p[i] = DIET ? v4_exp2(vals[base + i] - m_new)
+                        : __expf(vals[base + i] - m_new);
+            rsum += p[i];
+          }
+          unsigned x0 = v4_cvt_pk_bf16(p[0], p[1]);
+          unsigned y0 = v4_cvt_pk_bf16(p[4], p[5]);
+          unsigned x1 = v4_cvt_pk_bf16(p[2], p[3]);
+          unsigned y1 = v4_cvt_pk_bf16(p[6], p[7]);
+          auto r0 = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
+          auto r1 = __builtin_amdgcn_permlane32_swap(x1, y1, false, false);
+          short8 pfrag;
+          unsigned* pw = reinterpret_cast<unsigned*>(&pfrag);
+          pw[0] = (unsigned)r0[0];
+          pw[1] = (unsigned)r1[0];
+          pw[2] = (unsigned)r0[1];
+          pw[3] = (unsigned)r1[1];
+          const int kvg = sub * 32 + g * 16;
+          __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+          for (int db = 0; db < 4; ++db) {
+            short8 vf = v4_read(vt_lds, db * 32 + l31, (kvg >> 3) + lhi, 8);
+            o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pfrag, o_acc[db], 0, 0, 0);
+          }
+          __builtin_amdgcn_s_setprio(0);
+        }
+      }
+      rsum += __shfl_xor(rsum, 32, 64);
+      l_run = l_run * alpha + rsum;
+
+      __syncthreads();
+      buf ^= 1;
+    }
+
+    // epilogue for this item (no LDS use: safe to overlap other waves'
+    // next-item QK against the parked buffer)
+    if (live) {
+      const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+      for (int db = 0; db < 4; ++db)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+          OTh[(long long)d * S + q_lane] = f2bf(o_acc[db][r] * inv_l);
+        }
+    }
+  }
+}
+
 #define V4_KERNEL(VAR, STAGE, DEFER)                                          \
   extern "C" __global__ void __launch_bounds__(512)                           \
       attn_fwd_v4_##VAR##_kernel(const ushort* __restrict__ Q,                \
@@ -950,4 +1180,14 @@ attn_fwd_v4_18_kernel(const ushort* __restrict__ Q,
                       float scale) {
   // 4-wave blocks + T21 epilogue
   attn4_body<2, 1, 1, 1, 256, 1>(Q, K, VT, OT, B, H, Hk, S, scale);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+attn_fwd_v4_19_kernel(const ushort* __restrict__ Q,
+                      const ushort* __restrict__ K,
+                      const ushort* __restrict__ VT,
+                      ushort* __restrict__ OT, int B, int H, int Hk, int S,
+                      float scale) {
+  // persistent + cross-seam prefetch + defer + diet (+ sm-split)
+  attn4_body_p<1, 1>(Q, K, VT, OT, B, H, Hk, S, scale);
 }

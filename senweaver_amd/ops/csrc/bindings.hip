@@ -512,7 +512,7 @@ torch::Tensor attn_fwd_v3(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
       int, float);
 DECL_V4(0) DECL_V4(1) DECL_V4(2) DECL_V4(3)
 DECL_V4(4) DECL_V4(5) DECL_V4(6) DECL_V4(7)
-DECL_V4(8) DECL_V4(9) DECL_V4(10) DECL_V4(11) DECL_V4(12) DECL_V4(13) DECL_V4(14) DECL_V4(15) DECL_V4(16) DECL_V4(17) DECL_V4(18)
+DECL_V4(8) DECL_V4(9) DECL_V4(10) DECL_V4(11) DECL_V4(12) DECL_V4(13) DECL_V4(14) DECL_V4(15) DECL_V4(16) DECL_V4(17) DECL_V4(18) DECL_V4(19)
 
 torch::Tensor attn_fwd_v4(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
                           double scale, long var) {
@@ -525,8 +525,12 @@ torch::Tensor attn_fwd_v4(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
   TORCH_CHECK(vt.size(1) == Hk && vt.size(2) == D && vt.size(3) == S);
   auto ot = torch::empty({B, H, D, S}, q.options());
   const bool four_wave = (var == 16 || var == 18);  // QBLK 128, 256 threads
-  const dim3 grid(four_wave ? (S + 127) / 128 : (S + 255) / 256, H, B),
-      blk(four_wave ? 256 : 512);
+  dim3 grid(four_wave ? (S + 127) / 128 : (S + 255) / 256, H, B);
+  if (var == 19) {  // persistent: one block per CU walks the work items
+    const int NW = ((S + 255) / 256) * H * B;
+    grid = dim3(NW < 256 ? NW : 256, 1, 1);
+  }
+  const dim3 blk(four_wave ? 256 : 512);
   auto s = cur_stream();
 #define LAUNCH_V4(VAR)                                                       \
   case VAR:                                                                  \
@@ -537,7 +541,7 @@ torch::Tensor attn_fwd_v4(torch::Tensor q, torch::Tensor k, torch::Tensor vt,
   switch (var) {
     LAUNCH_V4(0) LAUNCH_V4(1) LAUNCH_V4(2) LAUNCH_V4(3)
     LAUNCH_V4(4) LAUNCH_V4(5) LAUNCH_V4(6) LAUNCH_V4(7)
-    LAUNCH_V4(8) LAUNCH_V4(9) LAUNCH_V4(10) LAUNCH_V4(11) LAUNCH_V4(12) LAUNCH_V4(13) LAUNCH_V4(14) LAUNCH_V4(15) LAUNCH_V4(16) LAUNCH_V4(17) LAUNCH_V4(18)
+    LAUNCH_V4(8) LAUNCH_V4(9) LAUNCH_V4(10) LAUNCH_V4(11) LAUNCH_V4(12) LAUNCH_V4(13) LAUNCH_V4(14) LAUNCH_V4(15) LAUNCH_V4(16) LAUNCH_V4(17) LAUNCH_V4(18) LAUNCH_V4(19)
     default: TORCH_CHECK(false, "attn_fwd_v4: unknown variant ", var);
   }
 #undef LAUNCH_V4
