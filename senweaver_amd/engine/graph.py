@@ -73,3 +73,38 @@ class DecodeGraph:
         self.graph.replay()
         cache.seq_lens[seq] = pos + 1
         return self.out
+
+    def step_batch(self, token_ids, seqs) -> torch.Tensor:
+        """Lockstep decode of B sequences (len(seqs) == capture batch).
+
+        Serving-throughput path: one replay advances every sequence one
+        token; per-seq cache bookkeeping stays host-side exactly like
+        step().  Returns hidden [B, H].
+        """
+        cache = self.cache
+        assert len(seqs) == self.B and len(token_ids) == self.B
+        from .kvcache import PAGE_SIZE as _PS
+        poss, slots, cts = [], [], []
+        for sq in seqs:
+            pos = cache.seq_lens[sq]
+            cache._ensure_capacity(sq, pos + 1)
+            page = cache.block_tables[sq][pos // _PS]
+            poss.append(pos)
+            slots.append(page * _PS + pos % _PS)
+            cts.append(pos + 1)
+        dev = self.tokens.device
+        self.tokens.copy_(torch.tensor(token_ids, dtype=torch.long))
+        self.pos32.copy_(torch.tensor(poss, dtype=torch.int32))
+        self.slot.copy_(torch.tensor(slots, dtype=torch.int32))
+        self.bt.zero_()
+        for i, sq in enumerate(seqs):
+            pages = cache.block_tables[sq]
+            self.bt[i, : len(pages)] = torch.tensor(pages, dtype=torch.int32,
+                                                    device=dev)
+        self.ctx.copy_(torch.tensor(cts, dtype=torch.int32))
+        if self.graph is None:
+            self._capture()
+        self.graph.replay()
+        for sq in seqs:
+            cache.seq_lens[sq] += 1
+        return self.out

@@ -607,3 +607,51 @@ def test_decode_long_context_multipage(dev):
     out_eager = b_eager.generate(prompt, max_new_tokens=24)
     assert out_graph == out_eager
     assert len(out_graph) > 0
+
+
+def test_decode_step_batch_lockstep(dev):
+    """Batched graph decode (B=2 lockstep) must equal two independent B=1
+    decodes of the same prompts."""
+    from senweaver_amd.engine.graph import DecodeGraph
+    from senweaver_amd.engine.kvcache import PAGE_SIZE, PagedKVCache
+    from senweaver_amd.models import LlamaModel, tiny_debug
+
+    cfg = tiny_debug()
+    model = LlamaModel(cfg, device=dev, seed=5)
+    prompts = [torch.randint(0, 256, (1, 64), generator=torch.Generator().manual_seed(s0))
+               for s0 in (1, 2)]
+
+    def run_single(ptoks, steps=6):
+        cache = PagedKVCache(cfg, 16, torch.device(dev),
+                             num_kv_heads=model.local_kv_heads)
+        sq = cache.new_seq()
+        hidden = model.prefill(ptoks.to(dev), cache=cache, seqs=[sq],
+                               real_lens=[64])
+        graph = DecodeGraph(model, cache, 16, batch=1)
+        last = hidden[:, 63]
+        outs = []
+        nxt = int(ops.argmax_rows(model.logits(last))[0])
+        for _ in range(steps):
+            outs.append(nxt)
+            last = graph.step(nxt, sq)
+            nxt = int(ops.argmax_rows(model.logits(last))[0])
+        return outs
+
+    singles = [run_single(p) for p in prompts]
+
+    cache = PagedKVCache(cfg, 32, torch.device(dev),
+                         num_kv_heads=model.local_kv_heads)
+    seqs = [cache.new_seq() for _ in range(2)]
+    both = torch.cat(prompts, dim=0).to(dev)
+    hidden = model.prefill(both, cache=cache, seqs=seqs, real_lens=[64, 64])
+    graph = DecodeGraph(model, cache, 16, batch=2)
+    last = hidden[:, 63]
+    outs = [[], []]
+    nxt = [int(x) for x in ops.argmax_rows(model.logits(last)).cpu()]
+    for _ in range(6):
+        for i in range(2):
+            outs[i].append(nxt[i])
+        last = graph.step_batch(nxt, seqs)
+        nxt = [int(x) for x in ops.argmax_rows(model.logits(last)).cpu()]
+    assert outs[0] == singles[0]
+    assert outs[1] == singles[1]
