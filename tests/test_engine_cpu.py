@@ -170,3 +170,24 @@ def test_kvcache_page_reuse_and_exhaustion():
     s3 = cache.new_seq()
     fill(s3, 16 * 2)  # reuses freed pages
     assert cache.ctx_lens_tensor([s3]).tolist() == [32]
+
+
+def test_swiglu_gemv_cpu_fallback():
+    """ops.swiglu_gemv on CPU tensors runs the two-op reference path."""
+    import torch
+    from senweaver_amd import ops
+    gu = torch.randn(1, 2 * 8192, dtype=torch.bfloat16)
+    w = torch.randn(64, 8192, dtype=torch.bfloat16)
+    got = ops.swiglu_gemv(gu, w)
+    g = gu[:, :8192].float()
+    u = gu[:, 8192:].float()
+    ref = (g * torch.sigmoid(g) * u) @ w.float().t()
+    torch.testing.assert_close(got.float(), ref, atol=2.0, rtol=2e-2)
+
+
+def test_grid_world1():
+    """build_tp_ep_grid degrades to identity contexts without dist init."""
+    from senweaver_amd.parallel import build_tp_ep_grid
+    tp, ep = build_tp_ep_grid(1, 1)
+    assert tp.world == 1 and ep.world == 1
+    assert ep.local_experts(8) == (0, 8)
