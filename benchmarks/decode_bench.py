@@ -5,7 +5,9 @@ import os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__
 from senweaver_amd.engine.scorer import LlamaBackend
 
 def main():
-    backend = LlamaBackend("llama-3-8b", max_seq=1024)
+    quant = sys.argv[1] if len(sys.argv) > 1 else "bf16"
+    backend = LlamaBackend("llama-3-8b", max_seq=1024, quant=quant)
+    print("quant:", quant)
     # warmup + prefill
     t0 = time.perf_counter()
     out = backend.generate("warmup " * 50, max_new_tokens=4)

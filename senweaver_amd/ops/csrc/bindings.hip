@@ -54,6 +54,7 @@ extern "C" __global__ void gemm_bt_mxfp8_256_kernel(const unsigned char*, const 
 extern "C" __global__ void gemm_bt_mxfp8_pipe_kernel(const unsigned char*, const unsigned char*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v3_m1(const ushort*, const ushort*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_fp8w_m1(const ushort*, const unsigned char*, const float*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v3w_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void swiglu_gemv_bt_bf16_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
@@ -714,6 +715,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     HIP_CHECK_KERNEL();
     return C;
   }, "loader/consumer LDS-DMA streaming GEMV (M=1)");
+  m.def("gemv_bt_fp8w", [](torch::Tensor x, torch::Tensor bq, torch::Tensor bs) {
+    check_bf16(x, "x");
+    TORCH_CHECK(bq.scalar_type() == torch::kUInt8 && bs.scalar_type() == torch::kFloat32);
+    const int M = x.size(0), K = x.size(1), N = bq.size(0);
+    TORCH_CHECK(M == 1 && K % 1024 == 0 && bq.size(1) == K && N % 4 == 0);
+    auto C = torch::empty({M, N}, x.options());
+    gemv_bt_fp8w_m1<<<dim3(N / 4), dim3(256), 0, cur_stream()>>>(
+        bf16_ptr(x), bq.data_ptr<unsigned char>(), bs.data_ptr<float>(),
+        bf16_mut(C), M, N, K);
+    HIP_CHECK_KERNEL();
+    return C;
+  }, "fp8-weight x bf16-activation GEMV (decode, M=1)");
   m.def("swiglu_gemv_bt", [](torch::Tensor gu, torch::Tensor b) {
     check_bf16(gu, "gateup"); check_bf16(b, "w");
     const int M = gu.size(0), K2 = gu.size(1), N = b.size(0);

@@ -576,6 +576,54 @@ swiglu_gemv_bt_bf16_m1(const ushort* __restrict__ GU,
   gemv3_body2<14336, 1>(GU, B, C, N, K);
 }
 
+// ---------------------------------------------------------------------------
+// fp8-weight GEMV (decode, M = 1): W rows stored OCP e4m3 with a per-row
+// f32 scale (the model's existing fp8 weight format); x stays bf16 — no
+// per-step activation quant kernel and no M-pad through the 128-tile fp8
+// GEMM (which measured 101 tok/s vs 267 bf16 in decode).  Weight bytes
+// halve, and decode GEMV is weight-bandwidth-bound.
+// One wave per output row; 16 fp8 bytes per lane per iter via
+// v_cvt_pk_f32_fp8 pairs; K % 1024 == 0.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(2))) float f32x2_t;
+typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+
+extern "C" __global__ void __launch_bounds__(256)
+gemv_bt_fp8w_m1(const ushort* __restrict__ X, const unsigned char* __restrict__ Bq,
+                const float* __restrict__ Bs, ushort* __restrict__ C,
+                int M, int N, int K) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int n = blockIdx.x * 4 + wid;
+  if (n >= N) return;
+  const unsigned char* brow = Bq + (long long)n * K;
+  float acc = 0.f;
+  uint4_t cur = __builtin_nontemporal_load(
+      reinterpret_cast<const uint4_t*>(brow + lane * 16));
+  for (int k = lane * 16; k < K; k += 64 * 16) {
+    uint4_t nxt;
+    if (k + 64 * 16 < K)
+      nxt = __builtin_nontemporal_load(
+          reinterpret_cast<const uint4_t*>(brow + k + 64 * 16));
+    bf16x8 x0 = *reinterpret_cast<const bf16x8*>(X + k);
+    bf16x8 x1 = *reinterpret_cast<const bf16x8*>(X + k + 8);
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(cur[d], false);
+      f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(cur[d], true);
+      const bf16x8& xv = d < 2 ? x0 : x1;
+      const int e = (d & 1) * 4;
+      acc += lo[0] * bf2f(xv.v[e + 0]);
+      acc += lo[1] * bf2f(xv.v[e + 1]);
+      acc += hi[0] * bf2f(xv.v[e + 2]);
+      acc += hi[1] * bf2f(xv.v[e + 3]);
+    }
+    cur = nxt;
+  }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) C[n] = f2bf(acc * Bs[n]);
+}
+
 GEMV2_INST(1)
 GEMV2_INST(2)
 GEMV2_INST(4)

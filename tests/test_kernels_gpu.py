@@ -655,3 +655,15 @@ def test_decode_step_batch_lockstep(dev):
         nxt = [int(x) for x in ops.argmax_rows(model.logits(last)).cpu()]
     assert outs[0] == singles[0]
     assert outs[1] == singles[1]
+
+
+def test_gemv_fp8_weights(dev):
+    """fp8-weight x bf16-activation decode GEMV vs the dequant reference."""
+    for (N, K) in [(4096, 4096), (1024, 14336)]:
+        x = torch.randn(1, K, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+        wq, ws = ops.quant_fp8(w)
+        got = ops.gemv_fp8w(x, wq, ws)
+        wf = wq.view(torch.float8_e4m3fn).float() * ws.unsqueeze(1)
+        ref_out = x.float() @ wf.t()
+        torch.testing.assert_close(got.float(), ref_out, atol=1.5, rtol=2e-2)
