@@ -19,8 +19,10 @@ from senweaver_amd import ops  # noqa: E402
 
 
 def main():
+    quant = sys.argv[1] if len(sys.argv) > 1 else "bf16"
     cfg = get_config("llama-3-8b")
-    model = LlamaModel(cfg, device="cuda:0", seed=3)
+    model = LlamaModel(cfg, device="cuda:0", seed=3, quant=quant)
+    print("quant:", quant)
     n = 48
     prompt_len = 128
     for B in (1, 2, 4, 8):

@@ -157,7 +157,7 @@ class LlamaModel:
     def _linear(self, x: torch.Tensor, L: dict, name: str) -> torch.Tensor:
         """Projection through bf16 MFMA or the fp8 MFMA path."""
         if self.quant == "fp8":
-            if x.shape[0] == 1 and x.is_cuda and x.shape[-1] % 1024 == 0:
+            if x.shape[0] <= 8 and x.is_cuda and x.shape[-1] % 1024 == 0:
                 # decode: fp8-weight GEMV, activations stay bf16 (no
                 # per-step quant kernel, half the weight stream)
                 return ops.gemv_fp8w(x, L[name + "_q"], L[name + "_s"])

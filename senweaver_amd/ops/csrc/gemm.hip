@@ -588,16 +588,19 @@ swiglu_gemv_bt_bf16_m1(const ushort* __restrict__ GU,
 typedef __attribute__((ext_vector_type(2))) float f32x2_t;
 typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
 
-extern "C" __global__ void __launch_bounds__(256)
-gemv_bt_fp8w_m1(const ushort* __restrict__ X, const unsigned char* __restrict__ Bq,
-                const float* __restrict__ Bs, ushort* __restrict__ C,
-                int M, int N, int K) {
+template <int MM>
+static __device__ __forceinline__ void gemv_fp8w_body(
+    const ushort* __restrict__ X, const unsigned char* __restrict__ Bq,
+    const float* __restrict__ Bs, ushort* __restrict__ C, int M, int N,
+    int K) {
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int n = blockIdx.x * 4 + wid;
   if (n >= N) return;
   const unsigned char* brow = Bq + (long long)n * K;
-  float acc = 0.f;
+  float acc[MM];
+#pragma unroll
+  for (int m = 0; m < MM; ++m) acc[m] = 0.f;
   uint4_t cur = __builtin_nontemporal_load(
       reinterpret_cast<const uint4_t*>(brow + lane * 16));
   for (int k = lane * 16; k < K; k += 64 * 16) {
@@ -605,24 +608,45 @@ gemv_bt_fp8w_m1(const ushort* __restrict__ X, const unsigned char* __restrict__ 
     if (k + 64 * 16 < K)
       nxt = __builtin_nontemporal_load(
           reinterpret_cast<const uint4_t*>(brow + k + 64 * 16));
-    bf16x8 x0 = *reinterpret_cast<const bf16x8*>(X + k);
-    bf16x8 x1 = *reinterpret_cast<const bf16x8*>(X + k + 8);
+    float wf[16];
 #pragma unroll
     for (int d = 0; d < 4; ++d) {
       f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(cur[d], false);
       f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(cur[d], true);
-      const bf16x8& xv = d < 2 ? x0 : x1;
-      const int e = (d & 1) * 4;
-      acc += lo[0] * bf2f(xv.v[e + 0]);
-      acc += lo[1] * bf2f(xv.v[e + 1]);
-      acc += hi[0] * bf2f(xv.v[e + 2]);
-      acc += hi[1] * bf2f(xv.v[e + 3]);
+      wf[d * 4 + 0] = lo[0];
+      wf[d * 4 + 1] = lo[1];
+      wf[d * 4 + 2] = hi[0];
+      wf[d * 4 + 3] = hi[1];
+    }
+#pragma unroll
+    for (int m = 0; m < MM; ++m) {
+      bf16x8 x0 = *reinterpret_cast<const bf16x8*>(X + (long long)m * K + k);
+      bf16x8 x1 = *reinterpret_cast<const bf16x8*>(X + (long long)m * K + k + 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[m] += wf[e] * bf2f(x0.v[e]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[m] += wf[8 + e] * bf2f(x1.v[e]);
     }
     cur = nxt;
   }
-  acc = wave_reduce_sum(acc);
-  if (lane == 0) C[n] = f2bf(acc * Bs[n]);
+#pragma unroll
+  for (int m = 0; m < MM; ++m) {
+    const float v = wave_reduce_sum(acc[m]);
+    if (lane == 0 && m < M) C[(long long)m * N + n] = f2bf(v * Bs[n]);
+  }
 }
+
+#define GEMV_FP8W_INST(MM)                                                    \
+  extern "C" __global__ void __launch_bounds__(256)                           \
+  gemv_bt_fp8w_m##MM(const ushort* X, const unsigned char* Bq,                \
+                     const float* Bs, ushort* C, int M, int N, int K) {       \
+    gemv_fp8w_body<MM>(X, Bq, Bs, C, M, N, K);                                \
+  }
+
+GEMV_FP8W_INST(1)
+GEMV_FP8W_INST(2)
+GEMV_FP8W_INST(4)
+GEMV_FP8W_INST(8)
 
 GEMV2_INST(1)
 GEMV2_INST(2)

@@ -659,8 +659,9 @@ def test_decode_step_batch_lockstep(dev):
 
 def test_gemv_fp8_weights(dev):
     """fp8-weight x bf16-activation decode GEMV vs the dequant reference."""
-    for (N, K) in [(4096, 4096), (1024, 14336)]:
-        x = torch.randn(1, K, dtype=torch.bfloat16, device=dev)
+    for (N, K, M) in [(4096, 4096, 1), (1024, 14336, 1), (2048, 4096, 3),
+                      (1024, 4096, 8)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
         w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
         wq, ws = ops.quant_fp8(w)
         got = ops.gemv_fp8w(x, wq, ws)
