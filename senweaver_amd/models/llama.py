@@ -164,6 +164,8 @@ class LlamaModel:
             xq, xs = ops.quant_fp8(x)
             return ops.gemm_bt_fp8(xq, xs, L[name + "_q"], L[name + "_s"])
         if self.quant == "mxfp8":
+            if x.shape[0] <= 8 and x.is_cuda and x.shape[-1] % 1024 == 0:
+                return ops.gemv_mxfp8w(x, L[name + "_q"], L[name + "_s"])
             xq, xs = ops.quant_mxfp8(x)
             return ops.gemm_bt_mxfp8(xq, xs, L[name + "_q"], L[name + "_s"])
         return ops.gemm_bt(x, L[name])

@@ -207,6 +207,19 @@ def gemv_fp8w(x: torch.Tensor, w_q: torch.Tensor, w_s: torch.Tensor) -> torch.Te
     return (x.float() @ wf.t()).to(x.dtype)
 
 
+def gemv_mxfp8w(x: torch.Tensor, w_q: torch.Tensor, w_s: torch.Tensor) -> torch.Tensor:
+    """Decode GEMV with MX block-scaled fp8 weights, bf16 activations
+    (the mxfp8 analog of gemv_fp8w; e8m0 per-32-block scales)."""
+    K = x.shape[-1]
+    if _on_gpu(x) and x.shape[0] <= 8 and K % 1024 == 0 and w_q.shape[0] % 4 == 0:
+        return hip_ext().gemv_bt_mxfp8w(x.contiguous(), w_q.contiguous(),
+                                        w_s.contiguous())
+    rows = w_q.shape[0]
+    f = w_q.view(torch.float8_e4m3fn).float().view(rows, K // 32, 32)
+    wf = (f * torch.exp2(w_s.float() - 127).unsqueeze(-1)).reshape(rows, K)
+    return (x.float() @ wf.t()).to(x.dtype)
+
+
 def gemm_bt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """C[M,N] = A[M,K] @ B[N,K]^T.
 

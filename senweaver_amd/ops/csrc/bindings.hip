@@ -58,6 +58,10 @@ extern "C" __global__ void gemv_bt_fp8w_m1(const ushort*, const unsigned char*, 
 extern "C" __global__ void gemv_bt_fp8w_m2(const ushort*, const unsigned char*, const float*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_fp8w_m4(const ushort*, const unsigned char*, const float*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_fp8w_m8(const ushort*, const unsigned char*, const float*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_mxfp8w_m1(const ushort*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_mxfp8w_m2(const ushort*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_mxfp8w_m4(const ushort*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
+extern "C" __global__ void gemv_bt_mxfp8w_m8(const ushort*, const unsigned char*, const unsigned char*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v3w_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void swiglu_gemv_bt_bf16_m1(const ushort*, const ushort*, ushort*, int, int, int);
 extern "C" __global__ void gemv_bt_bf16_v2_m2(const ushort*, const ushort*, ushort*, int, int, int);
@@ -743,7 +747,34 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     }
     HIP_CHECK_KERNEL();
     return C;
-  }, "fp8-weight x bf16-activation GEMV (decode, M=1)");
+  }, "fp8-weight x bf16-activation GEMV (decode, M<=8)");
+  m.def("gemv_bt_mxfp8w", [](torch::Tensor x, torch::Tensor bq, torch::Tensor bs) {
+    check_bf16(x, "x");
+    TORCH_CHECK(bq.scalar_type() == torch::kUInt8 && bs.scalar_type() == torch::kUInt8);
+    const int M = x.size(0), K = x.size(1), N = bq.size(0);
+    TORCH_CHECK(M >= 1 && M <= 8 && K % 1024 == 0 && bq.size(1) == K &&
+                bs.size(1) == K / 32 && N % 4 == 0);
+    const int MM = M <= 1 ? 1 : M <= 2 ? 2 : M <= 4 ? 4 : 8;
+    torch::Tensor xp = x;
+    if (M < MM) {
+      xp = torch::zeros({MM, (long)K}, x.options());
+      xp.narrow(0, 0, M).copy_(x);
+    }
+    auto C = torch::empty({M, N}, x.options());
+    auto launch = [&](auto kern) {
+      kern<<<dim3(N / 4), dim3(256), 0, cur_stream()>>>(
+          bf16_ptr(xp), bq.data_ptr<unsigned char>(),
+          bs.data_ptr<unsigned char>(), bf16_mut(C), M, N, K);
+    };
+    switch (MM) {
+      case 1: launch(gemv_bt_mxfp8w_m1); break;
+      case 2: launch(gemv_bt_mxfp8w_m2); break;
+      case 4: launch(gemv_bt_mxfp8w_m4); break;
+      default: launch(gemv_bt_mxfp8w_m8); break;
+    }
+    HIP_CHECK_KERNEL();
+    return C;
+  }, "mxfp8-weight x bf16-activation GEMV (decode, M<=8)");
   m.def("swiglu_gemv_bt", [](torch::Tensor gu, torch::Tensor b) {
     check_bf16(gu, "gateup"); check_bf16(b, "w");
     const int M = gu.size(0), K2 = gu.size(1), N = b.size(0);

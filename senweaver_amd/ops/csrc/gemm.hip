@@ -636,6 +636,75 @@ static __device__ __forceinline__ void gemv_fp8w_body(
   }
 }
 
+// mxfp8-weight variant: e8m0 per-32-block scales.  A lane's 16 elems per
+// iter never straddle a 32-block (k % 16 == 0), so ONE scale byte per lane
+// per iter: wf *= 2^(s-127) via ldexpf on the per-iter partial: the scale is
+// folded per-iter partial sum (s is block-constant).
+template <int MM>
+static __device__ __forceinline__ void gemv_mxfp8w_body(
+    const ushort* __restrict__ X, const unsigned char* __restrict__ Bq,
+    const unsigned char* __restrict__ Bs, ushort* __restrict__ C, int M,
+    int N, int K) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int n = blockIdx.x * 4 + wid;
+  if (n >= N) return;
+  const unsigned char* brow = Bq + (long long)n * K;
+  const unsigned char* srow = Bs + (long long)n * (K / 32);
+  float acc[MM];
+#pragma unroll
+  for (int m = 0; m < MM; ++m) acc[m] = 0.f;
+  uint4_t cur = __builtin_nontemporal_load(
+      reinterpret_cast<const uint4_t*>(brow + lane * 16));
+  for (int k = lane * 16; k < K; k += 64 * 16) {
+    uint4_t nxt;
+    if (k + 64 * 16 < K)
+      nxt = __builtin_nontemporal_load(
+          reinterpret_cast<const uint4_t*>(brow + k + 64 * 16));
+    const float scale = exp2f((float)srow[k / 32] - 127.f);
+    float wf[16];
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(cur[d], false);
+      f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(cur[d], true);
+      wf[d * 4 + 0] = lo[0];
+      wf[d * 4 + 1] = lo[1];
+      wf[d * 4 + 2] = hi[0];
+      wf[d * 4 + 3] = hi[1];
+    }
+#pragma unroll
+    for (int m = 0; m < MM; ++m) {
+      bf16x8 x0 = *reinterpret_cast<const bf16x8*>(X + (long long)m * K + k);
+      bf16x8 x1 = *reinterpret_cast<const bf16x8*>(X + (long long)m * K + k + 8);
+      float part = 0.f;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) part += wf[e] * bf2f(x0.v[e]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) part += wf[8 + e] * bf2f(x1.v[e]);
+      acc[m] += part * scale;
+    }
+    cur = nxt;
+  }
+#pragma unroll
+  for (int m = 0; m < MM; ++m) {
+    const float v = wave_reduce_sum(acc[m]);
+    if (lane == 0 && m < M) C[(long long)m * N + n] = f2bf(v);
+  }
+}
+
+#define GEMV_MXFP8W_INST(MM)                                                  \
+  extern "C" __global__ void __launch_bounds__(256)                           \
+  gemv_bt_mxfp8w_m##MM(const ushort* X, const unsigned char* Bq,              \
+                       const unsigned char* Bs, ushort* C, int M, int N,      \
+                       int K) {                                               \
+    gemv_mxfp8w_body<MM>(X, Bq, Bs, C, M, N, K);                              \
+  }
+
+GEMV_MXFP8W_INST(1)
+GEMV_MXFP8W_INST(2)
+GEMV_MXFP8W_INST(4)
+GEMV_MXFP8W_INST(8)
+
 #define GEMV_FP8W_INST(MM)                                                    \
   extern "C" __global__ void __launch_bounds__(256)                           \
   gemv_bt_fp8w_m##MM(const ushort* X, const unsigned char* Bq,                \

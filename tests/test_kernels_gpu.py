@@ -668,3 +668,16 @@ def test_gemv_fp8_weights(dev):
         wf = wq.view(torch.float8_e4m3fn).float() * ws.unsqueeze(1)
         ref_out = x.float() @ wf.t()
         torch.testing.assert_close(got.float(), ref_out, atol=1.5, rtol=2e-2)
+
+
+def test_gemv_mxfp8_weights(dev):
+    """mxfp8-weight decode GEMV vs the block-dequant reference."""
+    for (N, K, M) in [(2048, 4096, 1), (1024, 4096, 4)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+        wq, ws = ops.quant_mxfp8(w)
+        got = ops.gemv_mxfp8w(x, wq, ws)
+        f = wq.view(torch.float8_e4m3fn).float().view(N, K // 32, 32)
+        wf = (f * torch.exp2(ws.float() - 127).unsqueeze(-1)).reshape(N, K)
+        ref_out = x.float() @ wf.t()
+        torch.testing.assert_close(got.float(), ref_out, atol=1.5, rtol=2e-2)
