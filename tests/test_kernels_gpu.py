@@ -681,3 +681,18 @@ def test_gemv_mxfp8_weights(dev):
         wf = (f * torch.exp2(ws.float() - 127).unsqueeze(-1)).reshape(N, K)
         ref_out = x.float() @ wf.t()
         torch.testing.assert_close(got.float(), ref_out, atol=1.5, rtol=2e-2)
+
+
+@pytest.mark.parametrize("quant", ["fp8", "mxfp8"])
+def test_decode_graph_quantized(dev, quant):
+    """Quantized-weight decode (fp8w/mxfp8w GEMV) through the captured
+    graph must match the eager path token-for-token."""
+    from senweaver_amd.engine.scorer import LlamaBackend
+    from senweaver_amd.models import tiny_debug
+    b_graph = LlamaBackend(tiny_debug(), device=dev, max_seq=256, quant=quant)
+    out_graph = b_graph.generate("quantized decode check", max_new_tokens=8)
+    b_eager = LlamaBackend(tiny_debug(), device=dev, max_seq=256, quant=quant)
+    b_eager._decode_state()
+    b_eager._decode_graph = None
+    out_eager = b_eager.generate("quantized decode check", max_new_tokens=8)
+    assert out_graph == out_eager
