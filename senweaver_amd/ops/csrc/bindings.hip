@@ -734,8 +734,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       xp.narrow(0, 0, M).copy_(x);
     }
     auto C = torch::empty({M, N}, x.options());
+    // MM==1 runs 1 row/wave (grid N/4); MM>1 runs 2 rows/wave (N/8)
+    const int nb = MM > 1 ? (N + 7) / 8 : (N + 3) / 4;
     auto launch = [&](auto kern) {
-      kern<<<dim3(N / 4), dim3(256), 0, cur_stream()>>>(
+      kern<<<dim3(nb), dim3(256), 0, cur_stream()>>>(
           bf16_ptr(xp), bq.data_ptr<unsigned char>(), bs.data_ptr<float>(),
           bf16_mut(C), M, N, K);
     };
