@@ -346,3 +346,33 @@ def test_tp2_ep2_grid_matches_unsharded():
     for rank, vals in results.items():
         assert vals == pytest.approx(ref, rel=0.05, abs=0.5)
     assert results[0] == pytest.approx(results[3], abs=1e-5)
+
+
+def _payload_ep_dispatch_empty_shard(rank, world):
+    import torch
+    from senweaver_amd.parallel.ep import EPContext
+
+    ep = EPContext.from_default_group()
+    E = 4  # rank0 owns experts 0-1, rank1 owns 2-3
+    # every token on every rank routes to experts 0/1 -> rank1's shard is EMPTY
+    counts = torch.tensor([4, 2, 0, 0])
+    x = (torch.arange(6 * 3, dtype=torch.float32).reshape(6, 3)
+         + 1000 * rank)
+    x_local, local_counts, meta = ep.dispatch(x, counts, E)
+    if rank == 0:
+        assert x_local.shape[0] == 12  # both ranks' 6 rows
+        assert local_counts.tolist() == [8, 4]
+    else:
+        assert x_local.shape[0] == 0
+        assert local_counts.tolist() == [0, 0]
+    y = ep.combine(x_local * 2, meta)
+    # roundtrip: every rank gets back exactly its own rows, doubled
+    return torch.allclose(y, x * 2), y.shape[0]
+
+
+def test_ep_dispatch_empty_shard():
+    """EP all-to-all with a rank that owns zero routed tokens: dispatch,
+    the empty-side grouped compute, and combine must all roundtrip."""
+    results = _run_dist("_payload_ep_dispatch_empty_shard")
+    for rank, (ok, rows) in results.items():
+        assert ok and rows == 6
