@@ -223,9 +223,17 @@ class LlamaBackend:
         """Incremental greedy decode for the transport layer: calls
         on_chunk(cumulative_text) per token, honors should_stop between
         tokens.  Same decode path as generate()."""
-        ids = [tok.BOS] + self.tokenizer.encode(prompt, max_tokens=self.max_seq - max_new_tokens - 8)
+        # context budget: a decode budget >= max_seq must not go negative
+        # (a negative python slice KEPT the whole prompt and overflowed the
+        # paged cache — examples/apo_demo.py with max_edit_tokens==max_seq)
+        ctx_budget = self.max_seq - max_new_tokens - 8
+        if ctx_budget < 32:
+            ctx_budget = max(32, self.max_seq // 4)
+        ids = [tok.BOS] + self.tokenizer.encode(prompt, max_tokens=ctx_budget)
         ids.append(tok.ROLE_ASSISTANT)
         real = len(ids)
+        # decode steps can never push past the cache (+2 slack pages)
+        max_new_tokens = max(1, min(max_new_tokens, self.max_seq - real))
         S = _pad64(real)
         cache, graph = self._decode_state()
         # full reset (also recovers pages leaked by an aborted prefill)

@@ -191,3 +191,13 @@ def test_grid_world1():
     tp, ep = build_tp_ep_grid(1, 1)
     assert tp.world == 1 and ep.world == 1
     assert ep.local_experts(8) == (0, 8)
+
+
+def test_generate_budget_clamp():
+    """A decode budget >= max_seq must not overflow the paged cache (the
+    negative context-budget slice kept the full prompt before)."""
+    from senweaver_amd.engine.scorer import LlamaBackend
+    from senweaver_amd.models import tiny_debug
+    b = LlamaBackend(tiny_debug(), device="cpu", max_seq=128)
+    out = b.generate("long prompt " * 200, max_new_tokens=128)
+    assert isinstance(out, str)
