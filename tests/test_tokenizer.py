@@ -77,3 +77,28 @@ def test_max_tokens_truncation():
 
 def test_for_vocab_prefers_bpe():
     assert isinstance(tok.for_vocab(128256), tok.BPETokenizer)
+
+
+def test_roundtrip_property():
+    """Property (hypothesis): encode->decode is lossless for printable
+    text on the full 32k tokenizer (byte-level BPE is invertible)."""
+    from hypothesis import given, settings, strategies as st
+    from senweaver_amd.engine.tokenizer import BPETokenizer
+
+    t = BPETokenizer(vocab_size=128256)
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.text(alphabet=st.characters(min_codepoint=32, max_codepoint=0x2FFF),
+                   min_size=0, max_size=120))
+    def check(s):
+        assert t.decode(t.encode(s)) == s
+
+    check()
+
+
+def test_roundtrip_multibyte_and_newlines():
+    from senweaver_amd.engine.tokenizer import BPETokenizer
+    t = BPETokenizer(vocab_size=128256)
+    for s in ["", "\n\n\t  mixed\nlines\n", "café über 中文 \U0001f600",
+              "def f(x):\n    return x * 2  # comment\n"]:
+        assert t.decode(t.encode(s)) == s
