@@ -208,3 +208,37 @@ def test_cli_singleton_start_status_stop(tmp_path, monkeypatch):
     (tmp_path / "d.lock").write_text("999999")
     assert cli.read_lock() is None
     assert not (tmp_path / "d.lock").exists()
+
+
+def test_daemon_slow_consumer_does_not_block_others(daemon):
+    """A client that sends a generation request and then never reads its
+    socket must not stall the daemon: events for it queue in its bounded
+    per-client outbuf (daemon.cpp kMaxClientOutbuf) while OTHER clients
+    keep getting full service."""
+    import json
+    import socket as socketlib
+
+    sock_path = daemon[0]
+    slow = socketlib.socket(socketlib.AF_UNIX, socketlib.SOCK_STREAM)
+    slow.connect(sock_path)
+    slow.sendall((json.dumps({
+        "method": "sendLLMMessage", "requestId": "slow-1",
+        "messages": [{"role": "user", "content": "slow consumer"}],
+        "maxNewTokens": 24}) + "\n").encode())
+    # deliberately never read from `slow`
+    try:
+        healthy = DaemonClient(sock_path)
+        assert healthy.ping(timeout=120)
+        texts = []
+        done = threading.Event()
+        healthy.send_llm_message(
+            [{"role": "user", "content": "healthy client"}],
+            on_text=lambda m: texts.append(m["fullText"]),
+            on_final=lambda m: (texts.append(m["fullText"]), done.set()),
+            on_error=lambda m: done.set(),
+            max_new_tokens=4)
+        assert done.wait(timeout=180), "healthy client starved by slow one"
+        assert texts and texts[-1]
+        healthy.close()
+    finally:
+        slow.close()
