@@ -201,3 +201,22 @@ def test_generate_budget_clamp():
     b = LlamaBackend(tiny_debug(), device="cpu", max_seq=128)
     out = b.generate("long prompt " * 200, max_new_tokens=128)
     assert isinstance(out, str)
+
+
+def test_gemv_quantized_cpu_fallbacks():
+    """gemv_fp8w / gemv_mxfp8w off-GPU run the dequant reference path."""
+    import torch
+    from senweaver_amd import ops
+    x = torch.randn(1, 2048, dtype=torch.bfloat16)
+    w = torch.randn(64, 2048, dtype=torch.bfloat16)
+    wq, ws = ops.quant_fp8(w)
+    got = ops.gemv_fp8w(x, wq, ws)
+    wf = wq.view(torch.float8_e4m3fn).float() * ws.unsqueeze(1)
+    torch.testing.assert_close(got.float(), x.float() @ wf.t(),
+                               atol=1.0, rtol=2e-2)
+    mq, ms = ops.quant_mxfp8(w)
+    got2 = ops.gemv_mxfp8w(x, mq, ms)
+    f = mq.view(torch.float8_e4m3fn).float().view(64, 2048 // 32, 32)
+    wf2 = (f * torch.exp2(ms.float() - 127).unsqueeze(-1)).reshape(64, 2048)
+    torch.testing.assert_close(got2.float(), x.float() @ wf2.t(),
+                               atol=1.0, rtol=2e-2)
