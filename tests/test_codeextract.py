@@ -91,3 +91,34 @@ def test_language_helpers():
     assert markdown_language_to_id("C++") == "cpp"
     assert markdown_language_to_id("") == "plaintext"
     assert markdown_language_to_id("python") == "python"
+
+
+def test_search_replace_fuzz_never_crashes_and_streams_monotonic():
+    """Property (hypothesis): for ANY text sliced at arbitrary chunk
+    boundaries, extract_search_replace_blocks never raises, and blocks
+    marked done in an earlier prefix keep identical orig/final in every
+    longer prefix (the monotonicity the streaming applier relies on)."""
+    from hypothesis import given, settings, strategies as st
+    from senweaver_amd.utils.codeextract import extract_search_replace_blocks
+
+    fragments = st.lists(st.sampled_from([
+        "<<<<<<< ORIGINAL\n", "=======\n", ">>>>>>> UPDATED\n",
+        "code line\n", "x = 1", "<<<<", ">>>", "====", "\n", "random text ",
+    ]), min_size=0, max_size=14)
+
+    @settings(max_examples=120, deadline=None)
+    @given(fragments, st.integers(min_value=1, max_value=9))
+    def check(frags, step):
+        s = "".join(frags)
+        done_seen = {}
+        for cut in range(0, len(s) + 1, step):
+            blocks = extract_search_replace_blocks(s[:cut])
+            for i, b in enumerate(blocks):
+                if b.state == "done":
+                    if i in done_seen:
+                        assert done_seen[i] == (b.orig, b.final)
+                    else:
+                        done_seen[i] = (b.orig, b.final)
+        extract_search_replace_blocks(s)  # full text never raises
+
+    check()
