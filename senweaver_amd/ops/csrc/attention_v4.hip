@@ -637,7 +637,7 @@ static __device__ __forceinline__ void attn4_body_p(
     const bool live = q_lane < S;
     const int kv_end = min(S, qb * V4_QBLK + V4_QBLK);
 
-    const bool more_items = item + gridDim.x < NW;
+    const bool more_items = item + (int)gridDim.x < NW;
 
     if (first) {  // only ever once: later items find tile 0 parked
       v4_stage_glds(Kh, D, 16, smem[buf], V4_KVBLK * 16, tid, 512);

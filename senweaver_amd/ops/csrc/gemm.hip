@@ -657,7 +657,7 @@ static __device__ __forceinline__ void gemv_fp8w_body(
       }
     }
     cur0 = nxt0;
-    cur1 = nxt1;
+    if (R > 1) cur1 = nxt1;
   }
 #pragma unroll
   for (int m = 0; m < MM; ++m) {
