@@ -219,10 +219,14 @@ class LlamaBackend:
 
     @torch.no_grad()
     def stream_generate(self, prompt: str, max_new_tokens: int,
-                        should_stop, on_chunk) -> str:
-        """Incremental greedy decode for the transport layer: calls
+                        should_stop, on_chunk, *, temperature: float = 0.0,
+                        top_p: float = 1.0, sample_seed: Optional[int] = None) -> str:
+        """Incremental decode for the transport layer: calls
         on_chunk(cumulative_text) per token, honors should_stop between
-        tokens.  Same decode path as generate()."""
+        tokens.  temperature <= 0 is greedy (the APO default — edits must
+        be deterministic); temperature > 0 samples from the
+        temperature-scaled softmax with nucleus (top-p) filtering, seeded
+        by ``sample_seed`` for reproducible sampling."""
         # context budget: a decode budget >= max_seq must not go negative
         # (a negative python slice KEPT the whole prompt and overflowed the
         # paged cache — examples/apo_demo.py with max_edit_tokens==max_seq)
@@ -247,11 +251,18 @@ class LlamaBackend:
         last_hidden = hidden[0, real - 1]
         out_ids = []
         text = ""
+        gen = None
+        if temperature > 0:
+            gen = torch.Generator(device=self.device)
+            gen.manual_seed(sample_seed if sample_seed is not None else 0)
         for step in range(max_new_tokens):
             if should_stop():
                 break
             logits = self.model.logits(last_hidden.reshape(1, -1))
-            nxt = int(ops.argmax_rows(logits)[0])
+            if temperature > 0:
+                nxt = _sample_token(logits, temperature, top_p, gen)
+            else:
+                nxt = int(ops.argmax_rows(logits)[0])
             if nxt == tok.EOS:
                 break
             out_ids.append(nxt)
@@ -261,7 +272,25 @@ class LlamaBackend:
         return text
 
     @torch.no_grad()
-    def generate(self, prompt: str, max_new_tokens: int = 256) -> str:
+    def generate(self, prompt: str, max_new_tokens: int = 256, *,
+                 temperature: float = 0.0, top_p: float = 1.0,
+                 sample_seed: Optional[int] = None) -> str:
         return self.stream_generate(prompt, max_new_tokens,
                                     should_stop=lambda: False,
-                                    on_chunk=lambda _t: None)
+                                    on_chunk=lambda _t: None,
+                                    temperature=temperature, top_p=top_p,
+                                    sample_seed=sample_seed)
+
+def _sample_token(logits: torch.Tensor, temperature: float, top_p: float,
+                  gen: torch.Generator) -> int:
+    """Nucleus sampling from [1, V] logits (f32 softmax for stability)."""
+    probs = torch.softmax(logits.float().squeeze(0) / temperature, dim=-1)
+    if top_p < 1.0:
+        sp, idx = torch.sort(probs, descending=True)
+        cum = torch.cumsum(sp, dim=-1)
+        keep = cum - sp < top_p  # always keeps the top token
+        sp = torch.where(keep, sp, torch.zeros_like(sp))
+        sp = sp / sp.sum()
+        pick = torch.multinomial(sp, 1, generator=gen)
+        return int(idx[pick])
+    return int(torch.multinomial(probs, 1, generator=gen))

@@ -80,6 +80,7 @@ class LLMMessageService:
         max_new_tokens: int = 512,
         think_tags: tuple = ("<think>", "</think>"),
         synchronous: bool = False,
+        model_options: Optional[dict] = None,
     ) -> str:
         """Start a streaming generation; returns the request id (abort token)."""
         request_id = str(uuid.uuid4())
@@ -103,8 +104,20 @@ class LLMMessageService:
         def run() -> None:
             try:
                 prompt = self.render_messages(messages)
-                final_raw = self._backend.stream_generate(
-                    prompt, max_new_tokens, req.abort_event.is_set, pump_chunk)
+                # modelSelectionOptions analog: temperature/top-p pass
+                # through to backends that sample (scripted test backends
+                # keep the positional-only signature)
+                opts = model_options or {}
+                if opts:
+                    final_raw = self._backend.stream_generate(
+                        prompt, max_new_tokens, req.abort_event.is_set,
+                        pump_chunk,
+                        temperature=float(opts.get("temperature", 0.0)),
+                        top_p=float(opts.get("topP", opts.get("top_p", 1.0))),
+                        sample_seed=opts.get("sampleSeed"))
+                else:
+                    final_raw = self._backend.stream_generate(
+                        prompt, max_new_tokens, req.abort_event.is_set, pump_chunk)
                 if req.abort_event.is_set():
                     if on_abort:
                         on_abort()

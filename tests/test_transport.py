@@ -88,3 +88,31 @@ def test_wait_joins_thread():
                                on_error=lambda **m: done.set())
     svc.wait(rid, timeout=5)
     assert done.is_set()
+
+
+def test_model_options_passthrough():
+    """model_options (temperature/topP) reach the backend's sampler."""
+    import threading
+    from senweaver_amd.transport.service import LLMMessageService
+
+    seen = {}
+
+    class SamplingBackend:
+        def stream_generate(self, prompt, max_new_tokens, should_stop,
+                            on_chunk, temperature=0.0, top_p=1.0,
+                            sample_seed=None):
+            seen.update(temperature=temperature, top_p=top_p,
+                        sample_seed=sample_seed)
+            on_chunk("ok")
+            return "ok"
+
+    svc = LLMMessageService(SamplingBackend())
+    done = threading.Event()
+    svc.send_llm_message([], on_text=lambda **k: None,
+                         on_final_message=lambda **k: done.set(),
+                         on_error=lambda **k: done.set(),
+                         model_options={"temperature": 0.7, "topP": 0.9,
+                                        "sampleSeed": 42},
+                         synchronous=True)
+    assert done.is_set()
+    assert seen == {"temperature": 0.7, "top_p": 0.9, "sample_seed": 42}
