@@ -76,16 +76,20 @@ class DaemonClient:
 
     def send_llm_message(self, messages: List[dict], on_text=None, on_final=None,
                          on_error=None, chat_mode: Optional[str] = None,
-                         max_new_tokens: int = 64) -> str:
+                         max_new_tokens: int = 64,
+                         model_options: Optional[dict] = None) -> str:
         rid = str(uuid.uuid4())
         with self._lock:
             self._hooks[rid] = {"onText": on_text or (lambda m: None),
                                 "onFinalMessage": on_final or (lambda m: None),
                                 "onError": on_error or (lambda m: None),
                                 "onAbort": lambda m: None}
-        self._send({"method": "sendLLMMessage", "requestId": rid,
-                    "messages": messages, "chatMode": chat_mode,
-                    "maxNewTokens": max_new_tokens})
+        req = {"method": "sendLLMMessage", "requestId": rid,
+               "messages": messages, "chatMode": chat_mode,
+               "maxNewTokens": max_new_tokens}
+        if model_options:
+            req["modelOptions"] = model_options
+        self._send(req)
         return rid
 
     def abort(self, request_id: str) -> None:

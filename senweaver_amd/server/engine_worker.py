@@ -73,7 +73,8 @@ def main() -> int:
             sid = service.send_llm_message(
                 messages, on_text, on_final, on_error,
                 chat_mode=req.get("chatMode"),
-                max_new_tokens=int(req.get("maxNewTokens", 256)))
+                max_new_tokens=int(req.get("maxNewTokens", 256)),
+                model_options=req.get("modelOptions"))
             request_map[rid] = sid
         elif method == "abort":
             sid = request_map.pop(rid, None)

@@ -242,3 +242,30 @@ def test_daemon_slow_consumer_does_not_block_others(daemon):
         healthy.close()
     finally:
         slow.close()
+
+
+def test_daemon_model_options_sampling(daemon):
+    """modelOptions (temperature, seed) flow client -> daemon -> worker ->
+    backend sampler: two identical seeded requests stream identical text;
+    a different seed explores differently (tiny random-init model)."""
+    c = DaemonClient(daemon[0])
+    assert c.ping(timeout=120)
+
+    def run(seed):
+        texts = []
+        done = threading.Event()
+        c.send_llm_message([{"role": "user", "content": "sample"}],
+                           on_final=lambda m: (texts.append(m["fullText"]),
+                                               done.set()),
+                           on_error=lambda m: done.set(),
+                           max_new_tokens=8,
+                           model_options={"temperature": 1.0,
+                                          "sampleSeed": seed})
+        assert done.wait(timeout=180)
+        return texts[0] if texts else ""
+
+    a1, a2 = run(7), run(7)
+    assert a1 == a2 and a1
+    outs = {run(sd) for sd in range(5)}
+    assert len(outs) > 1
+    c.close()
