@@ -220,7 +220,8 @@ class LlamaBackend:
     @torch.no_grad()
     def stream_generate(self, prompt: str, max_new_tokens: int,
                         should_stop, on_chunk, *, temperature: float = 0.0,
-                        top_p: float = 1.0, sample_seed: Optional[int] = None) -> str:
+                        top_p: float = 1.0, sample_seed: Optional[int] = None,
+                        stop: Optional[List[str]] = None) -> str:
         """Incremental decode for the transport layer: calls
         on_chunk(cumulative_text) per token, honors should_stop between
         tokens.  temperature <= 0 is greedy (the APO default — edits must
@@ -267,6 +268,13 @@ class LlamaBackend:
                 break
             out_ids.append(nxt)
             text = self.tokenizer.decode(out_ids)
+            if stop:
+                cut = min((text.find(sq) for sq in stop if sq in text),
+                          default=-1)
+                if cut >= 0:
+                    text = text[:cut]
+                    on_chunk(text)
+                    break
             on_chunk(text)
             last_hidden = self._decode_one(cache, graph, seq, nxt)
         return text

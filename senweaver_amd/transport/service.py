@@ -114,7 +114,8 @@ class LLMMessageService:
                         pump_chunk,
                         temperature=float(opts.get("temperature", 0.0)),
                         top_p=float(opts.get("topP", opts.get("top_p", 1.0))),
-                        sample_seed=opts.get("sampleSeed"))
+                        sample_seed=opts.get("sampleSeed"),
+                        stop=opts.get("stop"))
                 else:
                     final_raw = self._backend.stream_generate(
                         prompt, max_new_tokens, req.abort_event.is_set, pump_chunk)

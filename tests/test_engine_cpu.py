@@ -247,3 +247,17 @@ def test_sampling_decode():
                            torch.Generator().manual_seed(sd)) for sd in range(20)}
     assert picks <= {0, 1}
     del gen2
+
+
+def test_stop_sequences():
+    """Generation halts at the first stop string and truncates it away."""
+    from senweaver_amd.engine.scorer import LlamaBackend
+    b = LlamaBackend("tiny-debug", device="cpu", max_seq=128)
+    full = b.generate("halt test", max_new_tokens=8)
+    if len(full) < 4:
+        return  # degenerate tiny-model output; nothing to split on
+    stop_tok = full[2:4]
+    out = b.stream_generate("halt test", 8, lambda: False, lambda t: None,
+                            stop=[stop_tok])
+    assert stop_tok not in out
+    assert full.startswith(out)

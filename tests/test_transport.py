@@ -100,7 +100,7 @@ def test_model_options_passthrough():
     class SamplingBackend:
         def stream_generate(self, prompt, max_new_tokens, should_stop,
                             on_chunk, temperature=0.0, top_p=1.0,
-                            sample_seed=None):
+                            sample_seed=None, stop=None):
             seen.update(temperature=temperature, top_p=top_p,
                         sample_seed=sample_seed)
             on_chunk("ok")
