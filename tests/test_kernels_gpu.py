@@ -696,3 +696,16 @@ def test_decode_graph_quantized(dev, quant):
     b_eager._decode_graph = None
     out_eager = b_eager.generate("quantized decode check", max_new_tokens=8)
     assert out_graph == out_eager
+
+
+def test_checkpoint_roundtrip_gpu(dev, tmp_path):
+    """save/load on the device: restored weights reproduce logits."""
+    from senweaver_amd.models import LlamaModel, load_weights, save_weights, tiny_debug
+    a = LlamaModel(tiny_debug(), device=dev, seed=1)
+    b = LlamaModel(tiny_debug(), device=dev, seed=2)
+    toks = torch.randint(0, 256, (1, 64),
+                         generator=torch.Generator().manual_seed(3)).to(dev)
+    p = str(tmp_path / "w.safetensors")
+    save_weights(a, p)
+    load_weights(b, p)
+    torch.testing.assert_close(a.prefill(toks).float(), b.prefill(toks).float())
