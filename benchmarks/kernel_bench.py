@@ -114,6 +114,26 @@ def bench_attn():
         print(f"  torch sdpa:        {flops / tsdpa / 1e12:7.1f} TF/s")
 
 
+def bench_gemv():
+    """Decode GEMV family: bf16 v2/v3(engine) and fp8/mxfp8-weight forms."""
+    dev = torch.device("cuda:0")
+    ext = ops.hip_ext()
+    for (name, N, K) in [("qkv", 6144, 4096), ("gateup", 28672, 4096),
+                         ("down", 4096, 14336)]:
+        x = torch.randn(1, K, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+        wq, ws = ops.quant_fp8(w)
+        mq, ms = ops.quant_mxfp8(w)
+        t2 = timeit(lambda: ext.gemm_bt(x, w))
+        t3 = timeit(lambda: ext.gemv_bt_v3(x, w))
+        tf = timeit(lambda: ext.gemv_bt_fp8w(x, wq, ws))
+        tm = timeit(lambda: ext.gemv_bt_mxfp8w(x, mq, ms))
+        gb = N * K * 2
+        print(f"GEMV {name:7s} N{N} K{K}: v2 {gb/t2/1e12:5.2f}  v3 {gb/t3/1e12:5.2f}"
+              f"  fp8w {gb/2/tf/1e12:5.2f}  mxfp8w {gb/2/tm/1e12:5.2f} TB/s"
+              f"  (fp8w speedup vs v2 {t2/tf:4.2f}x)")
+
+
 def bench_memops():
     dev = torch.device("cuda:0")
     x = torch.randn(8192, 4096, dtype=torch.bfloat16, device=dev)
@@ -146,5 +166,7 @@ if __name__ == "__main__":
         bench_fp8()
     if which in ("all", "attn"):
         bench_attn()
+    if which in ("all", "gemv"):
+        bench_gemv()
     if which in ("all", "mem"):
         bench_memops()
