@@ -64,7 +64,11 @@ class LLMMessageService:
     def render_messages(messages: List[LLMChatMessage]) -> str:
         parts = []
         for m in messages:
-            parts.append(f"<|{m.role}|>\n{m.content}")
+            if isinstance(m, dict):  # duck-type the wire shape too
+                role, content = m.get("role", "user"), m.get("content", "")
+            else:
+                role, content = m.role, m.content
+            parts.append(f"<|{role}|>\n{content}")
         parts.append("<|assistant|>\n")
         return "\n".join(parts)
 

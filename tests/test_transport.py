@@ -116,3 +116,13 @@ def test_model_options_passthrough():
                          synchronous=True)
     assert done.is_set()
     assert seen == {"temperature": 0.7, "top_p": 0.9, "sample_seed": 42}
+
+
+def test_render_messages_accepts_dicts():
+    from senweaver_amd.transport.service import LLMChatMessage, LLMMessageService
+    a = LLMMessageService.render_messages(
+        [LLMChatMessage("user", "hi"), LLMChatMessage("assistant", "yo")])
+    b = LLMMessageService.render_messages(
+        [{"role": "user", "content": "hi"},
+         {"role": "assistant", "content": "yo"}])
+    assert a == b
