@@ -2,12 +2,17 @@
 reference's Rust `code-cli` (cli/src: launcher, singleton lock, RPC
 control plane; tunnels/update are N/A for a local engine).
 
-Commands (python -m senweaver_amd.server.cli <cmd>):
-  start   — build the daemon if stale, take the singleton lock, spawn it
-            detached with logs to the state dir, wait for the socket
-  stop    — graceful shutdown over the socket, SIGTERM fallback
-  status  — singleton/pid/socket/engine liveness (pings the worker)
-  logs    — tail the daemon's captured stderr
+Commands (python -m senweaver_amd.server.cli <cmd>, or the installed
+`senweaver-daemon` console script):
+  start [--model M]  — build the daemon if stale, take the singleton lock,
+                       spawn it detached with logs to the state dir, wait
+                       for the socket (model via SENWEAVER_MODEL)
+  stop               — graceful shutdown over the socket, SIGTERM fallback
+  status             — singleton/pid/socket/engine liveness (pings the worker)
+  logs               — tail the daemon's captured stderr
+  generate [opts] P  — one-shot: ensure the daemon is up, stream a
+                       generation for prompt P to stdout
+                       (--max-new N, --temperature T, --top-p P, --seed S)
 
 Singleton semantics match the reference launcher: an exclusive lock file
 with the owner pid; a stale lock (dead pid) is reclaimed; a second `start`
@@ -71,7 +76,9 @@ def _take_lock(pid: int) -> bool:
     return True
 
 
-def start(wait_s: float = 30.0) -> int:
+def start(wait_s: float = 30.0, model: Optional[str] = None) -> int:
+    if model:
+        os.environ["SENWEAVER_MODEL"] = model
     existing = read_lock()
     if existing is not None:
         print(f"daemon already running (pid {existing}, socket {SOCKET_PATH})")
@@ -165,17 +172,75 @@ def logs(lines: int = 40) -> int:
     return 0
 
 
+def generate(argv) -> int:
+    """One-shot streaming generation through the daemon (started if needed)."""
+    import argparse
+
+    ap = argparse.ArgumentParser(prog="senweaver-daemon generate")
+    ap.add_argument("prompt")
+    ap.add_argument("--max-new", type=int, default=64)
+    ap.add_argument("--temperature", type=float, default=0.0)
+    ap.add_argument("--top-p", type=float, default=1.0)
+    ap.add_argument("--seed", type=int, default=None)
+    ap.add_argument("--model", default=None)
+    args = ap.parse_args(argv)
+    if read_lock() is None and start(model=args.model) != 0:
+        return 1
+    import threading
+
+    done = threading.Event()
+    state = {"last": "", "err": None}
+
+    def on_text(m):
+        full = m.get("fullText", "")
+        sys.stdout.write(full[len(state["last"]):])
+        sys.stdout.flush()
+        state["last"] = full
+
+    def on_final(m):
+        on_text(m)
+        done.set()
+
+    def on_error(m):
+        state["err"] = m.get("message", "error")
+        done.set()
+
+    opts = {}
+    if args.temperature > 0:
+        opts = {"temperature": args.temperature, "topP": args.top_p}
+        if args.seed is not None:
+            opts["sampleSeed"] = args.seed
+    c = DaemonClient(SOCKET_PATH)
+    c.ping(timeout=180)  # engine warm-up
+    c.send_llm_message([{"role": "user", "content": args.prompt}],
+                       on_text=on_text, on_final=on_final, on_error=on_error,
+                       max_new_tokens=args.max_new,
+                       model_options=opts or None)
+    ok = done.wait(timeout=600)
+    c.close()
+    print()
+    if state["err"]:
+        print("error:", state["err"], file=sys.stderr)
+        return 1
+    return 0 if ok else 1
+
+
 def main(argv=None) -> int:
     argv = argv if argv is not None else sys.argv[1:]
     cmd = argv[0] if argv else "status"
     if cmd == "start":
-        return start()
+        model = None
+        if "--model" in argv:
+            model = argv[argv.index("--model") + 1]
+        return start(model=model)
     if cmd == "stop":
         return stop()
     if cmd == "status":
         return status()
     if cmd == "logs":
         return logs()
+    if cmd == "generate":
+        return generate(argv[1:])
     print(__doc__)
     return 2
 

@@ -269,3 +269,23 @@ def test_daemon_model_options_sampling(daemon):
     outs = {run(sd) for sd in range(5)}
     assert len(outs) > 1
     c.close()
+
+
+def test_cli_one_shot_generate(tmp_path):
+    """`senweaver-daemon generate` front door: auto-start, stream to
+    stdout, daemon reusable after, clean stop."""
+    import subprocess
+    import sys
+    env = dict(os.environ,
+               SENWEAVER_STATE_DIR=str(tmp_path),
+               SENWEAVER_SOCKET=str(tmp_path / "d.sock"),
+               SENWEAVER_MODEL="tiny-debug", SENWEAVER_MAX_SEQ="128")
+    r = subprocess.run([sys.executable, "-m", "senweaver_amd.server.cli",
+                        "generate", "--max-new", "6", "one shot"],
+                       capture_output=True, text=True, env=env, timeout=420)
+    assert r.returncode == 0, r.stderr[-500:]
+    assert r.stdout.strip()  # streamed something
+    r2 = subprocess.run([sys.executable, "-m", "senweaver_amd.server.cli",
+                         "stop"], capture_output=True, text=True, env=env,
+                        timeout=120)
+    assert r2.returncode == 0
