@@ -34,6 +34,7 @@
 #include <sys/socket.h>
 #include <sys/un.h>
 #include <sys/wait.h>
+#include <time.h>
 #include <unistd.h>
 
 #include <map>
@@ -107,8 +108,11 @@ class Daemon {
   std::map<int, Client> clients_;
   std::map<std::string, int> request_client_;  // requestId -> client fd
   std::set<std::string> aborted_;
+  time_t last_spawn_ = 0;
+  int fast_crashes_ = 0;  // crash-loop breaker (worker dead < 20 s after spawn)
 
   bool spawn_worker() {
+    last_spawn_ = time(nullptr);
     int in_pipe[2], out_pipe[2];
     if (pipe(in_pipe) || pipe(out_pipe)) return false;
     worker_pid_ = fork();
@@ -299,6 +303,19 @@ class Daemon {
                             "\",\"requestId\":\"" + rid +
                             "\",\"message\":\"engine worker crashed; request lost\"}";
           if (!send_to_client(cfd, msg)) drop_client(cfd);
+        }
+        // crash-loop breaker: a worker that dies within 20 s of spawn
+        // three times in a row will never come up (bad model, import
+        // failure, OOM) — exit instead of forking torch forever.  Clients
+        // see EOF; the launcher CLI reports the daemon down.
+        if (time(nullptr) - last_spawn_ < 20) {
+          if (++fast_crashes_ >= 3) {
+            fprintf(stderr, "[daemon] worker crashed %d times at startup; giving up\n",
+                    fast_crashes_);
+            break;
+          }
+        } else {
+          fast_crashes_ = 0;
         }
         if (!spawn_worker()) break;
         continue;

@@ -289,3 +289,22 @@ def test_cli_one_shot_generate(tmp_path):
                          "stop"], capture_output=True, text=True, env=env,
                         timeout=120)
     assert r2.returncode == 0
+
+
+def test_daemon_crash_loop_breaker(tmp_path):
+    """A worker that can never start (bogus model) must NOT respawn
+    forever: after 3 fast crashes the daemon exits, the socket is gone,
+    and the launcher reports it down."""
+    bin_path = build_daemon()
+    sock = str(tmp_path / "d.sock")
+    env = dict(os.environ, SENWEAVER_MODEL="no-such-model")
+    proc = subprocess.Popen([bin_path, "--socket", sock], env=env,
+                            stderr=subprocess.PIPE, text=True)
+    try:
+        proc.wait(timeout=240)  # 3 failed spawns, then give up
+    except subprocess.TimeoutExpired:
+        proc.kill()
+        pytest.fail("daemon kept respawning a dead worker")
+    err = proc.stderr.read()
+    assert "giving up" in err, err[-500:]
+    assert not os.path.exists(sock)
