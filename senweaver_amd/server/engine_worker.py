@@ -82,6 +82,8 @@ def main() -> int:
                 service.abort(sid)
         elif method == "list":
             emit({"event": "listResult", "models": service.list_models()})
+        elif method == "stats":
+            emit({"event": "statsResult", "usage": service.usage.stats()})
         elif method == "ping":
             emit({"event": "pong"})
         elif method == "shutdown":

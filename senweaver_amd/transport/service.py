@@ -57,6 +57,8 @@ class LLMMessageService:
         self._backend = backend
         self._requests: Dict[str, _Request] = {}
         self._lock = threading.Lock()
+        from ..utils.observability import TokenUsageTracker
+        self.usage = TokenUsageTracker()  # per-request in/out token log
 
     # -- prompt rendering (chat messages -> backbone prompt text) --
 
@@ -132,6 +134,7 @@ class LLMMessageService:
                 if not vis_text and not rsn and tool_call is None:
                     on_error(message="Response from model was empty.")
                     return
+                self._record_usage(request_id, prompt, final_raw)
                 on_final_message(full_text=vis_text, full_reasoning=rsn, tool_call=tool_call)
             except Exception as e:
                 try:
@@ -149,6 +152,17 @@ class LLMMessageService:
             req.thread = t
             t.start()
         return request_id
+
+    def _record_usage(self, request_id: str, prompt: str, output: str) -> None:
+        """Token accounting per request: exact counts when the backend has a
+        tokenizer, the reference's 4-chars/token estimate otherwise."""
+        tok = getattr(self._backend, "tokenizer", None)
+        if tok is not None:
+            tin, tout = len(tok.encode(prompt)), len(tok.encode(output))
+        else:
+            tin, tout = len(prompt) // 4, len(output) // 4
+        model = getattr(getattr(self._backend, "config", None), "name", "local")
+        self.usage.record(request_id, model, tin, tout)
 
     def abort(self, request_id: str) -> None:
         """Instant client-side abort (no round trip), like the reference."""

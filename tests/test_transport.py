@@ -126,3 +126,25 @@ def test_render_messages_accepts_dicts():
         [{"role": "user", "content": "hi"},
          {"role": "assistant", "content": "yo"}])
     assert a == b
+
+
+def test_usage_tracking_per_request():
+    from senweaver_amd.transport.service import LLMMessageService
+
+    class EchoBackend:
+        def stream_generate(self, prompt, max_new_tokens, should_stop,
+                            on_chunk, **kw):
+            on_chunk("four char out!")
+            return "four char out!"
+
+    svc = LLMMessageService(EchoBackend())
+    svc.send_llm_message([{"role": "user", "content": "x" * 40}],
+                         on_text=lambda **k: None,
+                         on_final_message=lambda **k: None,
+                         on_error=lambda **k: None, synchronous=True)
+    st = svc.usage.stats()
+    assert st["totalRequests"] == 1
+    # 4 c/t estimate path (no tokenizer on this backend): 40 chars -> 10
+    assert st["totalInputTokens"] == 10
+    assert st["totalOutputTokens"] == len("four char out!") // 4
+    assert st["byModel"]["local"]["requests"] == 1
