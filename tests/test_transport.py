@@ -144,7 +144,10 @@ def test_usage_tracking_per_request():
                          on_error=lambda **k: None, synchronous=True)
     st = svc.usage.stats()
     assert st["totalRequests"] == 1
-    # 4 c/t estimate path (no tokenizer on this backend): 40 chars -> 10
-    assert st["totalInputTokens"] == 10
+    # 4 c/t estimate over the RENDERED prompt (role wrappers included)
+    from senweaver_amd.transport.service import LLMChatMessage
+    rendered = LLMMessageService.render_messages(
+        [LLMChatMessage("user", "x" * 40)])
+    assert st["totalInputTokens"] == len(rendered) // 4
     assert st["totalOutputTokens"] == len("four char out!") // 4
     assert st["byModel"]["local"]["requests"] == 1
