@@ -81,7 +81,8 @@ def main() -> int:
             if sid:
                 service.abort(sid)
         elif method == "list":
-            emit({"event": "listResult", "models": service.list_models()})
+            emit({"event": "listResult", "models": service.list_models(),
+                  "detailed": service.list_models_detailed()})
         elif method == "stats":
             emit({"event": "statsResult", "usage": service.usage.stats()})
         elif method == "ping":

@@ -181,3 +181,19 @@ class LLMMessageService:
         """Local analog of ollamaList/openAICompatibleList."""
         name = getattr(getattr(self._backend, "config", None), "name", None)
         return [name] if name else []
+
+    def list_models_detailed(self) -> List[dict]:
+        """Model list with the capability record the reference's
+        RefreshModelService attaches (modelCapabilities.ts lookup)."""
+        from ..models.capabilities import get_model_capabilities
+        out = []
+        for name in self.list_models():
+            caps = get_model_capabilities(name)
+            out.append({
+                "name": name,
+                "contextWindow": caps.contextWindow,
+                "reservedOutputTokenSpace": caps.reservedOutputTokenSpace,
+                "supportsFIM": caps.supportsFIM,
+                "supportsSystemMessage": caps.supportsSystemMessage,
+            })
+        return out

@@ -151,3 +151,13 @@ def test_usage_tracking_per_request():
     assert st["totalInputTokens"] == len(rendered) // 4
     assert st["totalOutputTokens"] == len("four char out!") // 4
     assert st["byModel"]["local"]["requests"] == 1
+
+
+def test_list_models_detailed_capabilities():
+    from senweaver_amd.engine.scorer import LlamaBackend
+    from senweaver_amd.transport.service import LLMMessageService
+    svc = LLMMessageService(LlamaBackend("tiny-debug", device="cpu", max_seq=64))
+    det = svc.list_models_detailed()
+    assert det and det[0]["name"] == "tiny-debug"
+    assert det[0]["contextWindow"] == 2048  # capabilities registry record
+    assert det[0]["reservedOutputTokenSpace"] == 256
