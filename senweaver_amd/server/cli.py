@@ -153,6 +153,8 @@ def status() -> int:
         try:
             c = DaemonClient(SOCKET_PATH)
             info["engine"] = "ready" if c.ping(timeout=120) else "no pong"
+            if info["engine"] == "ready":
+                info["usage"] = c.stats(timeout=30)
             c.close()
         except OSError as e:
             info["engine"] = f"socket error: {e}"

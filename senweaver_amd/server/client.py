@@ -106,6 +106,18 @@ class DaemonClient:
             time.sleep(0.05)
         return False
 
+    def stats(self, timeout: float = 30.0):
+        """Engine token-usage stats (worker "stats" RPC), or None."""
+        n0 = len(self.events)
+        self._send({"method": "stats"})
+        t0 = time.time()
+        while time.time() - t0 < timeout:
+            for e in self.events[n0:]:
+                if e.get("event") == "statsResult":
+                    return e.get("usage")
+            time.sleep(0.05)
+        return None
+
     def shutdown(self) -> None:
         """Ask the daemon to shut down (worker + listener)."""
         self._send({"method": "shutdown"})

@@ -308,3 +308,17 @@ def test_daemon_crash_loop_breaker(tmp_path):
     err = proc.stderr.read()
     assert "giving up" in err, err[-500:]
     assert not os.path.exists(sock)
+
+
+def test_daemon_stats_rpc(daemon):
+    c = DaemonClient(daemon[0])
+    assert c.ping(timeout=120)
+    done = threading.Event()
+    c.send_llm_message([{"role": "user", "content": "count me"}],
+                       on_final=lambda m: done.set(),
+                       on_error=lambda m: done.set(), max_new_tokens=4)
+    assert done.wait(timeout=180)
+    st = c.stats()
+    assert st and st["totalRequests"] >= 1
+    assert st["totalInputTokens"] > 0
+    c.close()
