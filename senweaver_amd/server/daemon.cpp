@@ -208,7 +208,7 @@ class Daemon {
       request_client_.erase(rid);
       write_all(worker_in_, line + "\n");
       send_to_client(fd, "{\"event\":\"onAbort\",\"requestId\":\"" + rid + "\"}");
-    } else if (method == "list" || method == "ping") {
+    } else if (method == "list" || method == "ping" || method == "stats") {
       write_all(worker_in_, line + "\n");
       request_client_[method] = fd;  // single in-flight list/ping per method
     } else if (method == "shutdown") {
@@ -232,11 +232,15 @@ class Daemon {
       return;
     }
     std::string key = rid;
-    if (key.empty()) key = ev == "listResult" ? "list" : ev == "pong" ? "ping" : "";
+    if (key.empty())
+      key = ev == "listResult" ? "list"
+            : ev == "pong" ? "ping"
+            : ev == "statsResult" ? "stats" : "";
     auto it = request_client_.find(key);
     if (it == request_client_.end()) return;           // client gone
     int cfd = it->second;
-    if (terminal || ev == "listResult" || ev == "pong") request_client_.erase(it);
+    if (terminal || ev == "listResult" || ev == "pong" || ev == "statsResult")
+      request_client_.erase(it);
     if (!send_to_client(cfd, line)) drop_client(cfd);
   }
 
@@ -246,7 +250,7 @@ class Daemon {
     for (auto it = request_client_.begin(); it != request_client_.end();) {
       if (it->second == fd) {
         const std::string& rid = it->first;
-        if (rid != "list" && rid != "ping") {
+        if (rid != "list" && rid != "ping" && rid != "stats") {
           // stop the engine decoding for a dead client; tombstone the id so
           // late events are suppressed until the worker acks
           aborted_.insert(rid);
