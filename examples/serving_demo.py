@@ -56,10 +56,13 @@ def main():
         c.abort(rid)
         print(f"[4] abort returned in {1000 * (time.perf_counter() - t0):.1f} ms "
               "(client-side instant; the worker stops decoding asynchronously)")
+        st = c.stats()
+        print(f"[5] engine usage: {st['totalRequests']} requests, "
+              f"{st['totalInputTokens']} in / {st['totalOutputTokens']} out tokens")
         c.shutdown()
         c.close()
         proc.wait(timeout=30)
-        print("[5] daemon shut down cleanly; demo ok")
+        print("[6] daemon shut down cleanly; demo ok")
     finally:
         if proc.poll() is None:
             proc.terminate()
