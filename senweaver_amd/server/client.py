@@ -77,7 +77,8 @@ class DaemonClient:
     def send_llm_message(self, messages: List[dict], on_text=None, on_final=None,
                          on_error=None, chat_mode: Optional[str] = None,
                          max_new_tokens: int = 64,
-                         model_options: Optional[dict] = None) -> str:
+                         model_options: Optional[dict] = None,
+                         raw_prompt: Optional[str] = None) -> str:
         rid = str(uuid.uuid4())
         with self._lock:
             self._hooks[rid] = {"onText": on_text or (lambda m: None),
@@ -89,6 +90,8 @@ class DaemonClient:
                "maxNewTokens": max_new_tokens}
         if model_options:
             req["modelOptions"] = model_options
+        if raw_prompt is not None:
+            req["rawPrompt"] = raw_prompt
         self._send(req)
         return rid
 
@@ -128,3 +131,50 @@ class DaemonClient:
             self._sock.close()
         except OSError:
             pass
+
+class DaemonBackend:
+    """Backend adapter: the agent loop's LLMMessageService speaks to the
+    NATIVE daemon instead of an in-process model — the reference's actual
+    topology (renderer chat service -> main-process LLM channel).
+
+    stream_generate matches LlamaBackend's contract (cumulative on_chunk,
+    should_stop polled between events, abort forwarded to the worker).
+    """
+
+    def __init__(self, socket_path: str) -> None:
+        self._client = DaemonClient(socket_path)
+        if not self._client.ping(timeout=180):
+            raise RuntimeError("daemon engine did not answer ping")
+
+    def stream_generate(self, prompt: str, max_new_tokens: int, should_stop,
+                        on_chunk, temperature: float = 0.0, top_p: float = 1.0,
+                        sample_seed=None, stop=None) -> str:
+        done = threading.Event()
+        state = {"text": "", "aborted": False}
+
+        def on_text(m):
+            state["text"] = m.get("fullText", "")
+            on_chunk(state["text"])
+            if should_stop() and not state["aborted"]:
+                state["aborted"] = True
+                self._client.abort(rid)
+                done.set()
+
+        def fin(m):
+            state["text"] = m.get("fullText", state["text"])
+            done.set()
+
+        opts = None
+        if temperature > 0 or stop:
+            opts = {"temperature": temperature, "topP": top_p,
+                    "sampleSeed": sample_seed, "stop": stop}
+        rid = self._client.send_llm_message(
+            [], on_text=on_text, on_final=fin,
+            on_error=lambda m: done.set(),
+            max_new_tokens=max_new_tokens, model_options=opts,
+            raw_prompt=prompt)
+        done.wait(timeout=900)
+        return state["text"]
+
+    def close(self) -> None:
+        self._client.close()

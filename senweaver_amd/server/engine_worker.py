@@ -74,7 +74,8 @@ def main() -> int:
                 messages, on_text, on_final, on_error,
                 chat_mode=req.get("chatMode"),
                 max_new_tokens=int(req.get("maxNewTokens", 256)),
-                model_options=req.get("modelOptions"))
+                model_options=req.get("modelOptions"),
+                raw_prompt=req.get("rawPrompt"))
             request_map[rid] = sid
         elif method == "abort":
             sid = request_map.pop(rid, None)

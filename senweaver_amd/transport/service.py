@@ -87,6 +87,7 @@ class LLMMessageService:
         think_tags: tuple = ("<think>", "</think>"),
         synchronous: bool = False,
         model_options: Optional[dict] = None,
+        raw_prompt: Optional[str] = None,
     ) -> str:
         """Start a streaming generation; returns the request id (abort token)."""
         request_id = str(uuid.uuid4())
@@ -109,7 +110,8 @@ class LLMMessageService:
 
         def run() -> None:
             try:
-                prompt = self.render_messages(messages)
+                prompt = (raw_prompt if raw_prompt is not None
+                          else self.render_messages(messages))
                 # modelSelectionOptions analog: temperature/top-p pass
                 # through to backends that sample (scripted test backends
                 # keep the positional-only signature)

@@ -322,3 +322,31 @@ def test_daemon_stats_rpc(daemon):
     assert st and st["totalRequests"] >= 1
     assert st["totalInputTokens"] > 0
     c.close()
+
+
+def test_agent_loop_over_daemon(daemon, tmp_path):
+    """The full agent tool loop with its LLM calls served by the NATIVE
+    daemon (the reference's chat-service <-> native-channel topology):
+    DaemonBackend proxies stream_generate over the socket with rawPrompt
+    (no double role-rendering)."""
+    from senweaver_amd.chat import ChatThreadService, GlobalSettings
+    from senweaver_amd.server.client import DaemonBackend
+    from senweaver_amd.storage import MemoryStorage
+    from senweaver_amd.tools.service import ToolsService
+    from senweaver_amd.trace.collector import TraceCollector
+    from senweaver_amd.transport.service import LLMMessageService
+
+    backend = DaemonBackend(daemon[0])
+    try:
+        tc = TraceCollector(storage=MemoryStorage())
+        svc = ChatThreadService(LLMMessageService(backend),
+                                ToolsService(str(tmp_path)), tc,
+                                settings=GlobalSettings(auto_approve={}),
+                                sleep=lambda s: None)
+        thread = svc.open_thread()
+        svc.add_user_message_and_stream_response(thread.id, "say anything")
+        msgs = svc.get_thread(thread.id).messages
+        assert any(m.role == "assistant" for m in msgs)
+        assert tc.get_all_traces()  # turn traced end to end
+    finally:
+        backend.close()
