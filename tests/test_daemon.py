@@ -350,3 +350,34 @@ def test_agent_loop_over_daemon(daemon, tmp_path):
         assert tc.get_all_traces()  # turn traced end to end
     finally:
         backend.close()
+
+
+def test_daemon_many_concurrent_clients(daemon):
+    """8 clients streaming simultaneously: every request completes, every
+    stream stays cumulative, no cross-request text bleed."""
+    results = {}
+    lock = threading.Lock()
+
+    def one(i):
+        c = DaemonClient(daemon[0])
+        texts = []
+        done = threading.Event()
+        c.send_llm_message([{"role": "user", "content": f"client {i}"}],
+                           on_text=lambda m: texts.append(m["fullText"]),
+                           on_final=lambda m: (texts.append(m["fullText"]),
+                                               done.set()),
+                           on_error=lambda m: done.set(),
+                           max_new_tokens=5)
+        ok = done.wait(timeout=300)
+        for a, b in zip(texts, texts[1:]):
+            assert b.startswith(a)
+        with lock:
+            results[i] = ok and bool(texts)
+        c.close()
+
+    threads = [threading.Thread(target=one, args=(i,)) for i in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=360)
+    assert len(results) == 8 and all(results.values()), results
