@@ -57,6 +57,11 @@ class LLMMessageService:
         self._backend = backend
         self._requests: Dict[str, _Request] = {}
         self._lock = threading.Lock()
+        # one generation at a time per backend: stream_generate resets the
+        # shared paged cache (and on GPU replays one captured graph), so
+        # concurrent generations would corrupt each other — requests queue
+        # here and stream once started
+        self._gen_lock = threading.Lock()
         from ..utils.observability import TokenUsageTracker
         self.usage = TokenUsageTracker()  # per-request in/out token log
 
@@ -112,32 +117,12 @@ class LLMMessageService:
             try:
                 prompt = (raw_prompt if raw_prompt is not None
                           else self.render_messages(messages))
-                # modelSelectionOptions analog: temperature/top-p pass
-                # through to backends that sample (scripted test backends
-                # keep the positional-only signature)
-                opts = model_options or {}
-                if opts:
-                    final_raw = self._backend.stream_generate(
-                        prompt, max_new_tokens, req.abort_event.is_set,
-                        pump_chunk,
-                        temperature=float(opts.get("temperature", 0.0)),
-                        top_p=float(opts.get("topP", opts.get("top_p", 1.0))),
-                        sample_seed=opts.get("sampleSeed"),
-                        stop=opts.get("stop"))
-                else:
-                    final_raw = self._backend.stream_generate(
-                        prompt, max_new_tokens, req.abort_event.is_set, pump_chunk)
-                if req.abort_event.is_set():
-                    if on_abort:
-                        on_abort()
-                    return
-                text, rsn = reasoning.finalize(final_raw)
-                vis_text, tool_call = tools.finalize(text)
-                if not vis_text and not rsn and tool_call is None:
-                    on_error(message="Response from model was empty.")
-                    return
-                self._record_usage(request_id, prompt, final_raw)
-                on_final_message(full_text=vis_text, full_reasoning=rsn, tool_call=tool_call)
+                with self._gen_lock:
+                    if req.abort_event.is_set():
+                        if on_abort:
+                            on_abort()
+                        return
+                    return _generate(prompt)
             except Exception as e:
                 try:
                     on_error(message=str(e))
@@ -146,6 +131,34 @@ class LLMMessageService:
             finally:
                 with self._lock:
                     self._requests.pop(request_id, None)
+
+        def _generate(prompt: str) -> None:
+            # modelSelectionOptions analog: temperature/top-p pass through
+            # to backends that sample (scripted test backends keep the
+            # positional-only signature)
+            opts = model_options or {}
+            if opts:
+                final_raw = self._backend.stream_generate(
+                    prompt, max_new_tokens, req.abort_event.is_set,
+                    pump_chunk,
+                    temperature=float(opts.get("temperature", 0.0)),
+                    top_p=float(opts.get("topP", opts.get("top_p", 1.0))),
+                    sample_seed=opts.get("sampleSeed"),
+                    stop=opts.get("stop"))
+            else:
+                final_raw = self._backend.stream_generate(
+                    prompt, max_new_tokens, req.abort_event.is_set, pump_chunk)
+            if req.abort_event.is_set():
+                if on_abort:
+                    on_abort()
+                return
+            text, rsn = reasoning.finalize(final_raw)
+            vis_text, tool_call = tools.finalize(text)
+            if not vis_text and not rsn and tool_call is None:
+                on_error(message="Response from model was empty.")
+                return
+            self._record_usage(request_id, prompt, final_raw)
+            on_final_message(full_text=vis_text, full_reasoning=rsn, tool_call=tool_call)
 
         if synchronous:
             run()

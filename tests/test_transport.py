@@ -161,3 +161,37 @@ def test_list_models_detailed_capabilities():
     assert det and det[0]["name"] == "tiny-debug"
     assert det[0]["contextWindow"] == 2048  # capabilities registry record
     assert det[0]["reservedOutputTokenSpace"] == 256
+
+
+def test_generations_serialized_per_backend():
+    """Concurrent sends must not interleave stream_generate on the shared
+    backend (the paged cache/graph is single-stream state)."""
+    import threading
+    import time as _time
+    from senweaver_amd.transport.service import LLMMessageService
+
+    active = [0]
+    max_active = [0]
+    lock = threading.Lock()
+
+    class SlowBackend:
+        def stream_generate(self, prompt, max_new_tokens, should_stop,
+                            on_chunk, **kw):
+            with lock:
+                active[0] += 1
+                max_active[0] = max(max_active[0], active[0])
+            _time.sleep(0.05)
+            on_chunk("x")
+            with lock:
+                active[0] -= 1
+            return "x"
+
+    svc = LLMMessageService(SlowBackend())
+    done = [threading.Event() for _ in range(5)]
+    for i in range(5):
+        svc.send_llm_message([{"role": "user", "content": str(i)}],
+                             on_text=lambda **k: None,
+                             on_final_message=lambda i=i, **k: done[i].set(),
+                             on_error=lambda i=i, **k: done[i].set())
+    assert all(d.wait(timeout=60) for d in done)
+    assert max_active[0] == 1, f"generations overlapped: {max_active[0]}"
