@@ -285,6 +285,13 @@ def test_cli_one_shot_generate(tmp_path):
                        capture_output=True, text=True, env=env, timeout=420)
     assert r.returncode == 0, r.stderr[-500:]
     assert r.stdout.strip()  # streamed something
+    # seeded sampling through the CLI reuses the running daemon
+    r_s = subprocess.run([sys.executable, "-m", "senweaver_amd.server.cli",
+                          "generate", "--max-new", "6", "--temperature",
+                          "1.0", "--seed", "3", "sampled shot"],
+                         capture_output=True, text=True, env=env, timeout=420)
+    assert r_s.returncode == 0, r_s.stderr[-500:]
+    assert r_s.stdout.strip()
     r2 = subprocess.run([sys.executable, "-m", "senweaver_amd.server.cli",
                          "stop"], capture_output=True, text=True, env=env,
                         timeout=120)
