@@ -91,12 +91,26 @@ class PersistentTerminal:
         except Exception:
             pass
 
+    def _respawn(self) -> None:
+        self.proc = subprocess.Popen(
+            ["/bin/bash"], stdin=subprocess.PIPE, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT, cwd=self.cwd, text=True, bufsize=1)
+        self._reader = threading.Thread(target=self._read, daemon=True)
+        self._reader.start()
+
     def run(self, command: str, wait_s: float = MAX_TERMINAL_BG_COMMAND_TIME_S) -> str:
+        if self.proc.poll() is not None:  # shell exited (e.g. `exit`)
+            self._respawn()
         with self._lock:
             self._buffer.clear()
         marker = f"__DONE_{uuid.uuid4().hex[:8]}__"
-        self.proc.stdin.write(command + f"\necho {marker}\n")
-        self.proc.stdin.flush()
+        try:
+            self.proc.stdin.write(command + f"\necho {marker}\n")
+            self.proc.stdin.flush()
+        except (BrokenPipeError, OSError):
+            self._respawn()
+            self.proc.stdin.write(command + f"\necho {marker}\n")
+            self.proc.stdin.flush()
         deadline = time.time() + wait_s
         while time.time() < deadline:
             with self._lock:

@@ -349,3 +349,15 @@ def test_base_system_message_full_slots():
     # normal mode: no terminal section
     msg2 = base_system_message("normal", terminal_ids=["t1"])
     assert "t1" not in msg2
+
+
+def test_persistent_terminal_survives_shell_exit(tmp_path):
+    """`exit` (or a crashed shell) must not brick the persistent terminal:
+    the next run respawns bash in the same cwd."""
+    from senweaver_amd.tools.service import PersistentTerminal
+    t = PersistentTerminal(str(tmp_path))
+    assert "alive" in t.run("echo alive")
+    t.run("exit")
+    out = t.run("echo back from the dead")
+    assert "back from the dead" in out
+    t.kill()
