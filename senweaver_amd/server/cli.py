@@ -243,6 +243,9 @@ def main(argv=None) -> int:
         return logs()
     if cmd == "generate":
         return generate(argv[1:])
+    if cmd in ("--version", "version"):
+        print("senweaver-amd 0.2.0")
+        return 0
     print(__doc__)
     return 2
 
