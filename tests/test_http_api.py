@@ -1,0 +1,74 @@
+"""OpenAI-compatible HTTP surface (server/http_api.py) driven in-process."""
+
+import json
+
+import pytest
+
+from senweaver_amd.engine.scorer import LlamaBackend
+from senweaver_amd.server.http_api import create_app
+from senweaver_amd.transport.service import LLMMessageService
+
+
+@pytest.fixture(scope="module")
+def client():
+    from fastapi.testclient import TestClient
+    backend = LlamaBackend("tiny-debug", device="cpu", max_seq=128)
+    app = create_app(LLMMessageService(backend))
+    with TestClient(app) as c:
+        yield c
+
+
+def test_models_endpoint(client):
+    r = client.get("/v1/models")
+    assert r.status_code == 200
+    data = r.json()["data"]
+    assert data[0]["id"] == "tiny-debug"
+    assert data[0]["contextWindow"] == 2048
+
+
+def test_completions_non_stream(client):
+    r = client.post("/v1/completions",
+                    json={"prompt": "hello http", "max_tokens": 6})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "text.completion"
+    assert body["choices"][0]["finish_reason"] == "stop"
+    assert isinstance(body["choices"][0]["text"], str)
+
+
+def test_completions_stream_sse(client):
+    with client.stream("POST", "/v1/completions",
+                       json={"prompt": "stream http", "max_tokens": 6,
+                             "stream": True}) as r:
+        assert r.status_code == 200
+        assert r.headers["content-type"].startswith("text/event-stream")
+        deltas, done = [], False
+        for line in r.iter_lines():
+            if not line.startswith("data: "):
+                continue
+            payload = line[len("data: "):]
+            if payload == "[DONE]":
+                done = True
+                break
+            deltas.append(json.loads(payload)["choices"][0]["text"])
+    assert done
+    assert "".join(deltas)  # the concatenated deltas form the text
+
+
+def test_chat_completions_and_sampling(client):
+    r = client.post("/v1/chat/completions",
+                    json={"messages": [{"role": "user", "content": "chat"}],
+                          "max_tokens": 6, "temperature": 1.0, "seed": 5})
+    assert r.status_code == 200
+    msg = r.json()["choices"][0]["message"]
+    assert msg["role"] == "assistant"
+    # seeded sampling is reproducible over HTTP
+    r2 = client.post("/v1/chat/completions",
+                     json={"messages": [{"role": "user", "content": "chat"}],
+                           "max_tokens": 6, "temperature": 1.0, "seed": 5})
+    assert r2.json()["choices"][0]["message"]["content"] == msg["content"]
+
+
+def test_stats_endpoint(client):
+    st = client.get("/v1/stats").json()
+    assert st["totalRequests"] >= 1
