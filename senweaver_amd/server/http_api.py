@@ -157,16 +157,24 @@ def main(argv=None) -> int:
     ap.add_argument("--port", type=int, default=8008)
     ap.add_argument("--model", default=None)
     ap.add_argument("--max-seq", type=int, default=2048)
+    ap.add_argument("--daemon", default=None, metavar="SOCKET",
+                    help="front the native daemon at SOCKET instead of "
+                         "loading a model in-process")
     args = ap.parse_args(argv)
 
-    import torch
     import uvicorn
 
-    from ..engine.scorer import LlamaBackend
+    if args.daemon:
+        from .client import DaemonBackend
+        backend = DaemonBackend(args.daemon)
+    else:
+        import torch
 
-    model = args.model or ("llama-3-8b" if torch.cuda.is_available()
-                           else "tiny-debug")
-    backend = LlamaBackend(model, max_seq=args.max_seq)
+        from ..engine.scorer import LlamaBackend
+
+        model = args.model or ("llama-3-8b" if torch.cuda.is_available()
+                               else "tiny-debug")
+        backend = LlamaBackend(model, max_seq=args.max_seq)
     app = create_app(LLMMessageService(backend))
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
     return 0
