@@ -388,3 +388,23 @@ def test_daemon_many_concurrent_clients(daemon):
     for t in threads:
         t.join(timeout=360)
     assert len(results) == 8 and all(results.values()), results
+
+
+def test_http_over_daemon(daemon):
+    """The composed topology: OpenAI-compatible HTTP -> DaemonBackend ->
+    C++ daemon -> engine worker."""
+    from fastapi.testclient import TestClient
+    from senweaver_amd.server.client import DaemonBackend
+    from senweaver_amd.server.http_api import create_app
+    from senweaver_amd.transport.service import LLMMessageService
+
+    backend = DaemonBackend(daemon[0])
+    try:
+        app = create_app(LLMMessageService(backend))
+        with TestClient(app) as c:
+            r = c.post("/v1/completions",
+                       json={"prompt": "through every layer", "max_tokens": 5})
+            assert r.status_code == 200
+            assert isinstance(r.json()["choices"][0]["text"], str)
+    finally:
+        backend.close()
