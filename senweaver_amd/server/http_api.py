@@ -73,30 +73,41 @@ def create_app(service: LLMMessageService):
         stream = bool(body.get("stream", False))
 
         if not stream:
-            state = {}
-            done = threading.Event()
-            service.send_llm_message(
-                messages,
-                on_text=lambda **k: None,
-                on_final_message=lambda full_text="", **k: (
-                    state.update(text=full_text), done.set()),
-                on_error=lambda message="", **k: (
-                    state.update(error=message), done.set()),
-                max_new_tokens=max_tokens, model_options=_options(body),
-                raw_prompt=raw_prompt, synchronous=True)
-            done.wait(timeout=1)
-            if "error" in state:
-                return JSONResponse({"error": {"message": state["error"]}},
-                                    status_code=500)
-            text = state.get("text", "")
+            n = max(1, min(int(body.get("n", 1)), 16))
+            base_opts = _options(body)
+            choices = []
+            for i in range(n):
+                opts = dict(base_opts) if base_opts else None
+                if n > 1:
+                    # n samples need sampling; vary the seed per choice
+                    opts = opts or {"temperature": 0.8, "topP": 1.0}
+                    opts["sampleSeed"] = int(opts.get("sampleSeed", 0)) + i
+                state = {}
+                done = threading.Event()
+                service.send_llm_message(
+                    messages,
+                    on_text=lambda **k: None,
+                    on_final_message=lambda full_text="", **k: (
+                        state.update(text=full_text), done.set()),
+                    on_error=lambda message="", **k: (
+                        state.update(error=message), done.set()),
+                    max_new_tokens=max_tokens, model_options=opts,
+                    raw_prompt=raw_prompt, synchronous=True)
+                done.wait(timeout=1)
+                if "error" in state:
+                    return JSONResponse({"error": {"message": state["error"]}},
+                                        status_code=500)
+                text = state.get("text", "")
+                if kind == "chat":
+                    choices.append({"index": i, "finish_reason": "stop",
+                                    "message": {"role": "assistant",
+                                                "content": text}})
+                else:
+                    choices.append({"index": i, "finish_reason": "stop",
+                                    "text": text})
             usage = service.usage.stats()
-            if kind == "chat":
-                choice = {"index": 0, "finish_reason": "stop",
-                          "message": {"role": "assistant", "content": text}}
-            else:
-                choice = {"index": 0, "finish_reason": "stop", "text": text}
             return {"id": rid, "object": f"{kind}.completion",
-                    "created": created, "model": model, "choices": [choice],
+                    "created": created, "model": model, "choices": choices,
                     "usage": {"total_requests": usage["totalRequests"]}}
 
         q: "queue.Queue" = queue.Queue()

@@ -72,3 +72,13 @@ def test_chat_completions_and_sampling(client):
 def test_stats_endpoint(client):
     st = client.get("/v1/stats").json()
     assert st["totalRequests"] >= 1
+
+
+def test_completions_n_choices(client):
+    r = client.post("/v1/completions",
+                    json={"prompt": "pick", "max_tokens": 5, "n": 3,
+                          "temperature": 1.0, "seed": 11})
+    assert r.status_code == 200
+    ch = r.json()["choices"]
+    assert [c["index"] for c in ch] == [0, 1, 2]
+    assert len({c["text"] for c in ch}) >= 2  # per-choice seeds explore
