@@ -123,26 +123,31 @@ def create_app(service: LLMMessageService):
             on_text(full_text=full_text)
             q.put(None)
 
-        service.send_llm_message(
+        sid = service.send_llm_message(
             messages, on_text=on_text, on_final_message=on_final,
             on_error=lambda message="", **k: q.put(None),
             max_new_tokens=max_tokens, model_options=_options(body),
             raw_prompt=raw_prompt)
 
         def sse():
-            while True:
-                delta = q.get()
-                if delta is None:
-                    break
-                if kind == "chat":
-                    choice = {"index": 0, "delta": {"content": delta}}
-                else:
-                    choice = {"index": 0, "text": delta}
-                chunk = {"id": rid, "object": f"{kind}.completion.chunk",
-                         "created": created, "model": model,
-                         "choices": [choice]}
-                yield f"data: {json.dumps(chunk)}\n\n"
-            yield "data: [DONE]\n\n"
+            try:
+                while True:
+                    delta = q.get()
+                    if delta is None:
+                        break
+                    if kind == "chat":
+                        choice = {"index": 0, "delta": {"content": delta}}
+                    else:
+                        choice = {"index": 0, "text": delta}
+                    chunk = {"id": rid, "object": f"{kind}.completion.chunk",
+                             "created": created, "model": model,
+                             "choices": [choice]}
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                yield "data: [DONE]\n\n"
+            finally:
+                # client gone (GeneratorExit) or stream finished: stop the
+                # decode — a vanished consumer must not keep the GPU busy
+                service.abort(sid)
 
         return StreamingResponse(sse(), media_type="text/event-stream")
 

@@ -82,3 +82,18 @@ def test_completions_n_choices(client):
     ch = r.json()["choices"]
     assert [c["index"] for c in ch] == [0, 1, 2]
     assert len({c["text"] for c in ch}) >= 2  # per-choice seeds explore
+
+
+def test_stream_disconnect_aborts_generation(client):
+    """Closing the SSE stream early aborts the decode server-side."""
+    with client.stream("POST", "/v1/completions",
+                       json={"prompt": "long stream", "max_tokens": 64,
+                             "stream": True}) as r:
+        for line in r.iter_lines():
+            if line.startswith("data: ") and "[DONE]" not in line:
+                break  # bail after the first chunk — closes the stream
+    # the service must be free for the next request promptly (the aborted
+    # generation released the backend)
+    r2 = client.post("/v1/completions",
+                     json={"prompt": "after abort", "max_tokens": 4})
+    assert r2.status_code == 200
