@@ -145,6 +145,18 @@ class DaemonBackend:
         self._client = DaemonClient(socket_path)
         if not self._client.ping(timeout=180):
             raise RuntimeError("daemon engine did not answer ping")
+        self.model_names = self._fetch_models()
+
+    def _fetch_models(self, timeout: float = 30.0) -> list:
+        n0 = len(self._client.events)
+        self._client._send({"method": "list"})
+        t0 = time.time()
+        while time.time() - t0 < timeout:
+            for e in self._client.events[n0:]:
+                if e.get("event") == "listResult":
+                    return e.get("models", [])
+            time.sleep(0.05)
+        return []
 
     def stream_generate(self, prompt: str, max_new_tokens: int, should_stop,
                         on_chunk, temperature: float = 0.0, top_p: float = 1.0,

@@ -195,7 +195,10 @@ class LLMMessageService:
     def list_models(self) -> List[str]:
         """Local analog of ollamaList/openAICompatibleList."""
         name = getattr(getattr(self._backend, "config", None), "name", None)
-        return [name] if name else []
+        if name:
+            return [name]
+        # proxy backends (DaemonBackend) report the remote engine's models
+        return list(getattr(self._backend, "model_names", []) or [])
 
     def list_models_detailed(self) -> List[dict]:
         """Model list with the capability record the reference's

@@ -402,6 +402,8 @@ def test_http_over_daemon(daemon):
     try:
         app = create_app(LLMMessageService(backend))
         with TestClient(app) as c:
+            m = c.get("/v1/models").json()["data"]
+            assert m and m[0]["id"] == "tiny-debug"  # names proxied over UDS
             r = c.post("/v1/completions",
                        json={"prompt": "through every layer", "max_tokens": 5})
             assert r.status_code == 200
