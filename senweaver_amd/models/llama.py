@@ -495,10 +495,16 @@ class LlamaModel:
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
         """hidden [*, H] -> logits [*, vocab] through the lm_head GEMM."""
         flat = hidden.reshape(-1, self.config.hidden_size)
+        small = flat.shape[0] <= 8 and flat.is_cuda and \
+            flat.shape[-1] % 1024 == 0
         if self.quant == "fp8":
+            if small:  # decode: quantized-weight GEMV (no act quant, no M-pad)
+                return ops.gemv_fp8w(flat, self.lm_head_q, self.lm_head_s)
             xq, xs = ops.quant_fp8(flat)
             return ops.gemm_bt_fp8(xq, xs, self.lm_head_q, self.lm_head_s)
         if self.quant == "mxfp8":
+            if small:
+                return ops.gemv_mxfp8w(flat, self.lm_head_q, self.lm_head_s)
             xq, xs = ops.quant_mxfp8(flat)
             return ops.gemm_bt_mxfp8(xq, xs, self.lm_head_q, self.lm_head_s)
         return ops.gemm_bt(flat, self.lm_head)
