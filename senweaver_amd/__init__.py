@@ -12,10 +12,13 @@ A brand-new framework with the capabilities of SenWeaver-IDE's RL pipeline
   RoPE, paged/flash attention, sampling) for gfx950.
 - ``senweaver_amd.models``   — Llama / Mixtral backbones running on those kernels.
 - ``senweaver_amd.engine``   — KV cache, generation, log-prob scoring.
-- ``senweaver_amd.parallel`` — RCCL-over-xGMI candidate-parallel beam scoring.
+- ``senweaver_amd.parallel`` — RCCL-over-xGMI candidate-parallel beam scoring,
+  tensor parallelism, expert parallelism and the TP x EP subgroup grid.
+- ``senweaver_amd.server``   — native C++ daemon, engine worker, launcher CLI,
+  and the OpenAI-compatible HTTP serving surface.
 
 The reference's "model" is a remote LLM over HTTPS; here the critique generator and
 beam scorer run locally on MI355X GPUs.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
