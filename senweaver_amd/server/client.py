@@ -37,7 +37,10 @@ class DaemonClient:
         self._hooks: Dict[str, Dict[str, Callable]] = {}
         self._lock = threading.Lock()
         self._closed = False
+        # debug ring of recent events (bounded — a long-lived client must
+        # not grow without bound); control-plane replies use _waiters
         self.events: List[dict] = []
+        self._waiters: Dict[str, list] = {}  # event name -> [Event, payload]
         self._reader = threading.Thread(target=self._read_loop, daemon=True)
         self._reader.start()
 
@@ -59,11 +62,19 @@ class DaemonClient:
                     msg = json.loads(line)
                 except ValueError:
                     continue
-                self.events.append(msg)
+                ev = msg.get("event", "")
                 rid = msg.get("requestId", "")
                 with self._lock:
+                    self.events.append(msg)
+                    if len(self.events) > 2048:
+                        del self.events[:1024]
+                    w = self._waiters.pop(ev, None)
                     hooks = self._hooks.get(rid, {})
-                ev = msg.get("event", "")
+                    if ev in ("onFinalMessage", "onError", "onAbort"):
+                        self._hooks.pop(rid, None)  # terminal: free the entry
+                if w is not None:
+                    w[1] = msg
+                    w[0].set()
                 fn = hooks.get(ev)
                 if fn:
                     try:
@@ -98,28 +109,25 @@ class DaemonClient:
     def abort(self, request_id: str) -> None:
         self._send({"method": "abort", "requestId": request_id})
 
+    def _request_reply(self, method: str, event: str, timeout: float):
+        """One-shot control-plane roundtrip (pong/listResult/statsResult)."""
+        w = [threading.Event(), None]
+        with self._lock:
+            self._waiters[event] = w
+        self._send({"method": method})
+        if w[0].wait(timeout=timeout):
+            return w[1]
+        with self._lock:
+            self._waiters.pop(event, None)
+        return None
+
     def ping(self, timeout: float = 30.0) -> bool:
-        got = threading.Event()
-        n0 = len(self.events)
-        self._send({"method": "ping"})
-        t0 = time.time()
-        while time.time() - t0 < timeout:
-            if any(e.get("event") == "pong" for e in self.events[n0:]):
-                return True
-            time.sleep(0.05)
-        return False
+        return self._request_reply("ping", "pong", timeout) is not None
 
     def stats(self, timeout: float = 30.0):
         """Engine token-usage stats (worker "stats" RPC), or None."""
-        n0 = len(self.events)
-        self._send({"method": "stats"})
-        t0 = time.time()
-        while time.time() - t0 < timeout:
-            for e in self.events[n0:]:
-                if e.get("event") == "statsResult":
-                    return e.get("usage")
-            time.sleep(0.05)
-        return None
+        msg = self._request_reply("stats", "statsResult", timeout)
+        return msg.get("usage") if msg else None
 
     def shutdown(self) -> None:
         """Ask the daemon to shut down (worker + listener)."""
@@ -148,15 +156,8 @@ class DaemonBackend:
         self.model_names = self._fetch_models()
 
     def _fetch_models(self, timeout: float = 30.0) -> list:
-        n0 = len(self._client.events)
-        self._client._send({"method": "list"})
-        t0 = time.time()
-        while time.time() - t0 < timeout:
-            for e in self._client.events[n0:]:
-                if e.get("event") == "listResult":
-                    return e.get("models", [])
-            time.sleep(0.05)
-        return []
+        msg = self._client._request_reply("list", "listResult", timeout)
+        return msg.get("models", []) if msg else []
 
     def stream_generate(self, prompt: str, max_new_tokens: int, should_stop,
                         on_chunk, temperature: float = 0.0, top_p: float = 1.0,
