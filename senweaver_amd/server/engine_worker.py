@@ -63,11 +63,13 @@ def main() -> int:
                       "toolCall": tool_call.__dict__ if tool_call else None})
 
             def on_final(full_text="", full_reasoning="", tool_call=None, _rid=rid, **kw):
+                request_map.pop(_rid, None)  # terminal: free the entry
                 emit({"event": "onFinalMessage", "requestId": _rid, "fullText": full_text,
                       "fullReasoning": full_reasoning,
                       "toolCall": tool_call.__dict__ if tool_call else None})
 
             def on_error(message="", _rid=rid, **kw):
+                request_map.pop(_rid, None)
                 emit({"event": "onError", "requestId": _rid, "message": message})
 
             sid = service.send_llm_message(
