@@ -709,3 +709,12 @@ def test_checkpoint_roundtrip_gpu(dev, tmp_path):
     save_weights(a, p)
     load_weights(b, p)
     torch.testing.assert_close(a.prefill(toks).float(), b.prefill(toks).float())
+
+
+def test_add_bf16(dev):
+    """Residual-add utility op (paths that skip the fused add+norm)."""
+    a = torch.randn(333, 1024, dtype=torch.bfloat16, device=dev)
+    b = torch.randn(333, 1024, dtype=torch.bfloat16, device=dev)
+    got = ops.hip_ext().add_bf16(a, b)
+    torch.testing.assert_close(got.float(), a.float() + b.float(),
+                               atol=2e-2, rtol=2e-2)
