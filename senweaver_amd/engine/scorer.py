@@ -280,6 +280,23 @@ class LlamaBackend:
         return text
 
     @torch.no_grad()
+    def embed(self, texts: List[str], max_tokens: int = 512) -> List[List[float]]:
+        """Sequence embeddings: mean-pooled final hidden states (the
+        backbone has no trained embedding head; this is the standard
+        pooled-representation fallback, unit-normalized)."""
+        out = []
+        for text in texts:
+            ids = [tok.BOS] + self.tokenizer.encode(text, max_tokens=max_tokens)
+            S = _pad64(len(ids))
+            t = torch.zeros(1, S, dtype=torch.long)
+            t[0, : len(ids)] = torch.tensor(ids, dtype=torch.long)
+            hidden = self.model.prefill(t.to(self.device))  # [1, S, H]
+            v = hidden[0, : len(ids)].float().mean(dim=0)
+            v = v / (v.norm() + 1e-8)
+            out.append(v.cpu().tolist())
+        return out
+
+    @torch.no_grad()
     def generate(self, prompt: str, max_new_tokens: int = 256, *,
                  temperature: float = 0.0, top_p: float = 1.0,
                  sample_seed: Optional[int] = None) -> str:

@@ -151,6 +151,21 @@ def create_app(service: LLMMessageService):
 
         return StreamingResponse(sse(), media_type="text/event-stream")
 
+    @app.post("/v1/embeddings")
+    def embeddings(body: dict):
+        texts = body.get("input", [])
+        if isinstance(texts, str):
+            texts = [texts]
+        embed = getattr(service._backend, "embed", None)
+        if embed is None:
+            return JSONResponse(
+                {"error": {"message": "backend has no embedding support"}},
+                status_code=501)
+        vecs = embed([str(t) for t in texts])
+        return {"object": "list", "model": _model_name(),
+                "data": [{"object": "embedding", "index": i, "embedding": v}
+                         for i, v in enumerate(vecs)]}
+
     @app.post("/v1/completions")
     def completions(body: dict):
         return _generate(body, raw_prompt=str(body.get("prompt", "")),

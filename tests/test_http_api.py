@@ -97,3 +97,16 @@ def test_stream_disconnect_aborts_generation(client):
     r2 = client.post("/v1/completions",
                      json={"prompt": "after abort", "max_tokens": 4})
     assert r2.status_code == 200
+
+
+def test_embeddings_endpoint(client):
+    r = client.post("/v1/embeddings", json={"input": ["alpha text", "beta"]})
+    assert r.status_code == 200
+    data = r.json()["data"]
+    assert len(data) == 2 and len(data[0]["embedding"]) == 256  # tiny hidden
+    import math
+    n = math.sqrt(sum(x * x for x in data[0]["embedding"]))
+    assert abs(n - 1.0) < 1e-3  # unit-normalized
+    # deterministic for the same input
+    r2 = client.post("/v1/embeddings", json={"input": "alpha text"})
+    assert r2.json()["data"][0]["embedding"] == data[0]["embedding"]
