@@ -151,6 +151,23 @@ def create_app(service: LLMMessageService):
 
         return StreamingResponse(sse(), media_type="text/event-stream")
 
+    @app.post("/v1/tokenize")
+    def tokenize(body: dict):
+        tok = getattr(service._backend, "tokenizer", None)
+        if tok is None:
+            return JSONResponse(
+                {"error": {"message": "backend has no tokenizer"}}, 501)
+        ids = tok.encode(str(body.get("text", "")))
+        return {"tokens": ids, "count": len(ids)}
+
+    @app.post("/v1/detokenize")
+    def detokenize(body: dict):
+        tok = getattr(service._backend, "tokenizer", None)
+        if tok is None:
+            return JSONResponse(
+                {"error": {"message": "backend has no tokenizer"}}, 501)
+        return {"text": tok.decode([int(i) for i in body.get("tokens", [])])}
+
     @app.post("/v1/embeddings")
     def embeddings(body: dict):
         texts = body.get("input", [])
