@@ -10,6 +10,8 @@ the OpenAI API work against an MI355X node unchanged:
                                          DELTAS, then `data: [DONE]`
   POST /v1/chat/completions            — messages in (rendered through the
                                          engine's chat template)
+  POST /v1/embeddings                  — pooled-representation vectors
+  POST /v1/tokenize | /v1/detokenize   — budget math for clients
   GET  /v1/stats                       — per-request token usage
 
 Generation options map 1:1 onto the engine's sampler: temperature, top_p,
