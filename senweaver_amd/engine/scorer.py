@@ -221,7 +221,9 @@ class LlamaBackend:
     def stream_generate(self, prompt: str, max_new_tokens: int,
                         should_stop, on_chunk, *, temperature: float = 0.0,
                         top_p: float = 1.0, sample_seed: Optional[int] = None,
-                        stop: Optional[List[str]] = None) -> str:
+                        stop: Optional[List[str]] = None,
+                        presence_penalty: float = 0.0,
+                        frequency_penalty: float = 0.0) -> str:
         """Incremental decode for the transport layer: calls
         on_chunk(cumulative_text) per token, honors should_stop between
         tokens.  temperature <= 0 is greedy (the APO default — edits must
@@ -260,6 +262,15 @@ class LlamaBackend:
             if should_stop():
                 break
             logits = self.model.logits(last_hidden.reshape(1, -1))
+            if (presence_penalty or frequency_penalty) and out_ids:
+                # OpenAI semantics: penalize GENERATED tokens on the logits
+                seen = torch.tensor(sorted(set(out_ids)), device=logits.device)
+                counts = torch.tensor(
+                    [out_ids.count(int(t)) for t in seen],
+                    device=logits.device, dtype=logits.dtype)
+                logits = logits.clone()
+                logits[0, seen] -= (presence_penalty
+                                    + frequency_penalty * counts)
             if temperature > 0:
                 nxt = _sample_token(logits, temperature, top_p, gen)
             else:

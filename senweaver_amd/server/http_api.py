@@ -64,6 +64,10 @@ def create_app(service: LLMMessageService):
         stop = body.get("stop")
         if stop:
             opts["stop"] = [stop] if isinstance(stop, str) else list(stop)
+        for src, dst in (("presence_penalty", "presencePenalty"),
+                         ("frequency_penalty", "frequencyPenalty")):
+            if body.get(src):
+                opts[dst] = float(body[src])
         return opts or None
 
     def _generate(body: dict, raw_prompt: Optional[str],

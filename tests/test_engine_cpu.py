@@ -261,3 +261,19 @@ def test_stop_sequences():
                             stop=[stop_tok])
     assert stop_tok not in out
     assert full.startswith(out)
+
+
+def test_repetition_penalties_break_loops():
+    """Greedy decode of the tiny random-init model loops on a few tokens;
+    a frequency penalty must break the loop (OpenAI logit semantics)."""
+    from senweaver_amd.engine.scorer import LlamaBackend
+    b = LlamaBackend("tiny-debug", device="cpu", max_seq=128)
+    ids_plain, ids_pen = [], []
+    b.stream_generate("loop breaker", 10, lambda: False,
+                      lambda t: None)  # warm path
+    # capture token streams via the tokenizer-decoded text uniqueness
+    plain = b.generate("loop breaker", max_new_tokens=10)
+    pen = b.stream_generate("loop breaker", 10, lambda: False, lambda t: None,
+                            frequency_penalty=5.0)
+    uniq = lambda s: len(set(s.split()))
+    assert uniq(pen) > uniq(plain) or plain != pen
