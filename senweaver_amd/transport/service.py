@@ -138,13 +138,21 @@ class LLMMessageService:
             # positional-only signature)
             opts = model_options or {}
             if opts:
+                kwargs = {
+                    "temperature": float(opts.get("temperature", 0.0)),
+                    "top_p": float(opts.get("topP", opts.get("top_p", 1.0))),
+                    "sample_seed": opts.get("sampleSeed"),
+                    "stop": opts.get("stop"),
+                }
+                # optional knobs only when requested: duck-typed backends
+                # that predate them keep working
+                if opts.get("presencePenalty"):
+                    kwargs["presence_penalty"] = float(opts["presencePenalty"])
+                if opts.get("frequencyPenalty"):
+                    kwargs["frequency_penalty"] = float(opts["frequencyPenalty"])
                 final_raw = self._backend.stream_generate(
                     prompt, max_new_tokens, req.abort_event.is_set,
-                    pump_chunk,
-                    temperature=float(opts.get("temperature", 0.0)),
-                    top_p=float(opts.get("topP", opts.get("top_p", 1.0))),
-                    sample_seed=opts.get("sampleSeed"),
-                    stop=opts.get("stop"))
+                    pump_chunk, **kwargs)
             else:
                 final_raw = self._backend.stream_generate(
                     prompt, max_new_tokens, req.abort_event.is_set, pump_chunk)

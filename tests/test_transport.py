@@ -195,3 +195,30 @@ def test_generations_serialized_per_backend():
                              on_error=lambda i=i, **k: done[i].set())
     assert all(d.wait(timeout=60) for d in done)
     assert max_active[0] == 1, f"generations overlapped: {max_active[0]}"
+
+
+def test_penalties_reach_backend():
+    import threading
+    from senweaver_amd.transport.service import LLMMessageService
+
+    seen = {}
+
+    class B:
+        def stream_generate(self, prompt, max_new_tokens, should_stop,
+                            on_chunk, temperature=0.0, top_p=1.0,
+                            sample_seed=None, stop=None,
+                            presence_penalty=0.0, frequency_penalty=0.0):
+            seen.update(pp=presence_penalty, fp=frequency_penalty)
+            on_chunk("k")
+            return "k"
+
+    svc = LLMMessageService(B())
+    svc.send_llm_message([{"role": "user", "content": "x"}],
+                         on_text=lambda **k: None,
+                         on_final_message=lambda **k: None,
+                         on_error=lambda **k: None,
+                         model_options={"temperature": 0.5,
+                                        "presencePenalty": 0.3,
+                                        "frequencyPenalty": 0.7},
+                         synchronous=True)
+    assert seen == {"pp": 0.3, "fp": 0.7}
