@@ -223,7 +223,8 @@ class LlamaBackend:
                         top_p: float = 1.0, sample_seed: Optional[int] = None,
                         stop: Optional[List[str]] = None,
                         presence_penalty: float = 0.0,
-                        frequency_penalty: float = 0.0) -> str:
+                        frequency_penalty: float = 0.0,
+                        logit_bias: Optional[dict] = None) -> str:
         """Incremental decode for the transport layer: calls
         on_chunk(cumulative_text) per token, honors should_stop between
         tokens.  temperature <= 0 is greedy (the APO default — edits must
@@ -262,6 +263,10 @@ class LlamaBackend:
             if should_stop():
                 break
             logits = self.model.logits(last_hidden.reshape(1, -1))
+            if logit_bias:
+                logits = logits.clone()
+                for t, bias in logit_bias.items():
+                    logits[0, int(t)] += float(bias)
             if (presence_penalty or frequency_penalty) and out_ids:
                 # OpenAI semantics: penalize GENERATED tokens on the logits
                 seen = torch.tensor(sorted(set(out_ids)), device=logits.device)

@@ -68,6 +68,9 @@ def create_app(service: LLMMessageService):
                          ("frequency_penalty", "frequencyPenalty")):
             if body.get(src):
                 opts[dst] = float(body[src])
+        if body.get("logit_bias"):
+            opts["logitBias"] = {int(k): float(v)
+                                 for k, v in body["logit_bias"].items()}
         return opts or None
 
     def _generate(body: dict, raw_prompt: Optional[str],

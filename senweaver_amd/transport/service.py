@@ -150,6 +150,9 @@ class LLMMessageService:
                     kwargs["presence_penalty"] = float(opts["presencePenalty"])
                 if opts.get("frequencyPenalty"):
                     kwargs["frequency_penalty"] = float(opts["frequencyPenalty"])
+                if opts.get("logitBias"):
+                    kwargs["logit_bias"] = {int(k): float(v)
+                                            for k, v in opts["logitBias"].items()}
                 final_raw = self._backend.stream_generate(
                     prompt, max_new_tokens, req.abort_event.is_set,
                     pump_chunk, **kwargs)

@@ -277,3 +277,12 @@ def test_repetition_penalties_break_loops():
                             frequency_penalty=5.0)
     uniq = lambda s: len(set(s.split()))
     assert uniq(pen) > uniq(plain) or plain != pen
+
+
+def test_logit_bias_forces_token():
+    """A +1000 bias on one token makes greedy pick it every step."""
+    from senweaver_amd.engine.scorer import LlamaBackend
+    b = LlamaBackend("tiny-debug", device="cpu", max_seq=128)
+    out = b.stream_generate("bias me", 5, lambda: False, lambda t: None,
+                            logit_bias={300: 1000.0})
+    assert out.split() == ["tok300"] * 5
