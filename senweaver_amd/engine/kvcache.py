@@ -42,7 +42,10 @@ class PagedKVCache:
         bt = self.block_tables[seq]
         while len(bt) < need:
             if not self._free:
-                raise RuntimeError("KV cache out of pages")
+                raise RuntimeError(
+                f"KV cache out of pages: need {need - len(bt)} more for seq "
+                f"{seq} len {new_len} ({self.num_pages} total, "
+                f"{len(self._free)} free)")
             bt.append(self._free.pop())
 
     def free_seq(self, seq: int) -> None:
