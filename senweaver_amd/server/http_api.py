@@ -78,7 +78,8 @@ def create_app(service: LLMMessageService):
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
         created = int(time.time())
         model = _model_name()
-        max_tokens = int(body.get("max_tokens", 128))
+        max_tokens = int(body.get("max_tokens",
+                             body.get("max_completion_tokens", 128)))
         stream = bool(body.get("stream", False))
 
         if not stream:
